@@ -1,0 +1,74 @@
+"""Process orchestration: N actor processes + 1 replay-buffer process +
+learner in the main process (reference: /root/reference/train.py:20-44).
+
+The epsilon ladder is the reference's: eps_i = base_eps^(1 + i/(N-1) * alpha)
+(train.py:15-17).
+"""
+
+import random
+
+import numpy as np
+import torch
+import torch.multiprocessing as mp
+
+from . import config as cfg
+from .models.network import Network
+from .worker import Actor, Learner, ReplayBuffer
+
+
+def epsilon_ladder(num_actors=None, base_eps=None, alpha=None):
+    c = cfg.get()
+    n = num_actors or c.num_actors
+    base = base_eps or c.base_eps
+    a = alpha or c.alpha_eps
+    if n == 1:
+        return [base]
+    return [base ** (1 + i / (n - 1) * a) for i in range(n)]
+
+
+def _run_actor(epsilon, model, sample_queue, seed):
+    actor = Actor(epsilon, model, sample_queue, seed=seed)
+    actor.run()
+
+
+def _run_buffer(buffer: ReplayBuffer):
+    buffer.run()
+
+
+def train(seed: int = 0):
+    torch.manual_seed(seed)
+    np.random.seed(seed)
+    random.seed(seed)
+    torch.set_num_threads(1)
+
+    c = cfg.get()
+    model = Network(c.action_dim, c.obs_shape, c.hidden_dim, encoder=c.encoder,
+                    forward_steps=c.forward_steps, mlp_hidden=c.mlp_hidden)
+    model.share_memory()
+
+    sample_queues = [mp.Queue() for _ in range(c.num_actors)]
+    batch_queue = mp.Queue(c.batch_queue_size)
+    priority_queue = mp.Queue(c.batch_queue_size)
+
+    buffer = ReplayBuffer(sample_queues, batch_queue, priority_queue)
+    learner = Learner(batch_queue, priority_queue, model)
+
+    actor_procs = []
+    for i, eps in enumerate(epsilon_ladder()):
+        p = mp.Process(target=_run_actor,
+                       args=(eps, model, sample_queues[i], seed + 1 + i))
+        p.start()
+        actor_procs.append(p)
+
+    buffer_proc = mp.Process(target=_run_buffer, args=(buffer,))
+    buffer_proc.start()
+
+    learner.run()
+
+    buffer_proc.join()
+    for p in actor_procs:
+        p.terminate()
+
+
+if __name__ == "__main__":
+    train()

@@ -1,0 +1,672 @@
+"""Actor / Learner / ReplayBuffer — the core R2D2 runtime.
+
+API parity with the reference's worker.py (/root/reference/worker.py) at the
+class level: ``Block``, ``LocalBuffer``, ``ReplayBuffer``, ``Learner``,
+``Actor``, ``calculate_mixed_td_errors``, ``Learner.value_rescale``.  The
+implementation is new and MI355X-first:
+
+- the learner runs ONE online forward per update (``calculate_q_both``)
+  instead of the reference's two (worker.py:346,352), in bf16 on GPU;
+- batch assembly is vectorized numpy (no per-sample Python slice loop as in
+  reference worker.py:176-210);
+- the replay ingest thread blocks on a queue instead of busy-spinning
+  (reference worker.py:124-129 burns a core);
+- with ``config.gpu_replay`` the prioritized replay lives in GPU HBM with
+  on-device sampling/priority updates (replay/gpu_replay.py) and priorities
+  never leave the device;
+- stored recurrent states are aligned to each sequence's burn-in start
+  (the reference stores index ``i*learning_steps`` into the carried buffer,
+  which is only correct once the burn-in prefix is full — reference
+  worker.py:461 vs :186-188).
+
+Checkpoint format is the reference 4-tuple
+``(state_dict, num_updates, env_steps, minutes)`` (worker.py:380-381).
+"""
+
+import math
+import os
+import queue as queue_mod
+import threading
+import time
+from dataclasses import dataclass
+from typing import List, Optional, Tuple
+
+import numpy as np
+import torch
+import torch.nn as nn
+
+from . import config as cfg
+from .models.network import Network, AgentState
+from .ops import functional as Fn
+from .replay.priority_tree import PriorityTree
+
+
+############################## Data containers ##############################
+
+
+@dataclass
+class Block:
+    """Unit of actor->replay transfer: <=block_length transitions cut into
+    <=seq_per_block sequences of learning_steps each (reference worker.py:23-36)."""
+    obs: np.ndarray            # (burn_in0 + steps + 1, *obs_shape) uint8/float32
+    last_action: np.ndarray    # (burn_in0 + steps + 1, A) bool one-hot
+    last_reward: np.ndarray    # (burn_in0 + steps + 1,) float32
+    action: np.ndarray         # (steps,) uint8
+    n_step_reward: np.ndarray  # (steps,) float32
+    gamma: np.ndarray          # (steps,) float32 — per-step gamma^n, 0 at terminal
+    hidden: np.ndarray         # (num_sequences, 2, hidden_dim) float32
+    num_sequences: int
+    burn_in_steps: np.ndarray  # (num_sequences,) uint8
+    learning_steps: np.ndarray # (num_sequences,) uint8
+    forward_steps: np.ndarray  # (num_sequences,) uint8
+
+
+@dataclass
+class TrainingBatch:
+    """Device-ready training batch (the reference ships an anonymous 14-tuple,
+    worker.py:219-238; named fields here, same content)."""
+    obs: torch.Tensor            # (B, T, *obs_shape)
+    last_action: torch.Tensor    # (B, T, A)
+    last_reward: torch.Tensor    # (B, T)
+    hidden: torch.Tensor         # (2, B, H)
+    action: torch.Tensor         # (sum_learn, 1) long
+    n_step_reward: torch.Tensor  # (sum_learn,)
+    gamma: torch.Tensor          # (sum_learn,)
+    burn_in_steps: torch.Tensor  # (B,) CPU
+    learning_steps: torch.Tensor # (B,) CPU
+    forward_steps: torch.Tensor  # (B,) CPU
+    idxes: np.ndarray
+    is_weights: torch.Tensor     # (sum_learn,)
+    old_ptr: int
+    env_steps: int
+
+    def to(self, device, non_blocking=True):
+        self.obs = self.obs.to(device, non_blocking=non_blocking)
+        self.last_action = self.last_action.to(device, non_blocking=non_blocking)
+        self.last_reward = self.last_reward.to(device, non_blocking=non_blocking)
+        self.hidden = self.hidden.to(device, non_blocking=non_blocking)
+        self.action = self.action.to(device, non_blocking=non_blocking)
+        self.n_step_reward = self.n_step_reward.to(device, non_blocking=non_blocking)
+        self.gamma = self.gamma.to(device, non_blocking=non_blocking)
+        self.is_weights = self.is_weights.to(device, non_blocking=non_blocking)
+        return self
+
+    def pin(self):
+        for f in ("obs", "last_action", "last_reward", "hidden", "action",
+                  "n_step_reward", "gamma", "is_weights"):
+            setattr(self, f, getattr(self, f).pin_memory())
+        return self
+
+
+def calculate_mixed_td_errors(td_error: np.ndarray, learning_steps: np.ndarray,
+                              eta: float = 0.9) -> np.ndarray:
+    """Per-sequence priority = eta*max + (1-eta)*mean of |TD|
+    (reference worker.py:268-276)."""
+    return Fn.mixed_td_priority_np(td_error, learning_steps, eta)
+
+
+############################## LocalBuffer ##############################
+
+
+class LocalBuffer:
+    """Per-actor trajectory accumulator (reference worker.py:395-497).
+
+    Stores one episode's running transitions, cuts them into blocks of
+    <= block_length steps, computes on-actor: n-step returns, the per-step
+    gamma^n vector (0 at terminal -> no done flag stored), initial priorities
+    from the actor's own q-values, and carries the last burn_in+1 steps as
+    the next block's burn-in prefix.
+    """
+
+    def __init__(self, action_dim: int, forward_steps: Optional[int] = None,
+                 burn_in_steps: Optional[int] = None, learning_steps: Optional[int] = None,
+                 gamma: Optional[float] = None, hidden_dim: Optional[int] = None,
+                 block_length: Optional[int] = None):
+        c = cfg.get()
+        self.action_dim = action_dim
+        self.gamma = c.gamma if gamma is None else gamma
+        self.hidden_dim = c.hidden_dim if hidden_dim is None else hidden_dim
+        self.forward_steps = c.forward_steps if forward_steps is None else forward_steps
+        self.learning_steps = c.learning_steps if learning_steps is None else learning_steps
+        self.burn_in_steps = c.burn_in_steps if burn_in_steps is None else burn_in_steps
+        self.block_length = c.block_length if block_length is None else block_length
+        self.prio_eta = c.prio_eta
+        self.curr_burn_in_steps = 0
+        self.size = 0
+
+    def __len__(self):
+        return self.size
+
+    def reset(self, init_obs: np.ndarray):
+        self.obs_buffer: List[np.ndarray] = [init_obs]
+        first_action = np.zeros(self.action_dim, dtype=bool)
+        first_action[0] = True
+        self.last_action_buffer: List[np.ndarray] = [first_action]
+        self.last_reward_buffer: List[float] = [0.0]
+        self.hidden_buffer: List[np.ndarray] = [
+            np.zeros((2, self.hidden_dim), dtype=np.float32)]
+        self.action_buffer: List[int] = []
+        self.reward_buffer: List[float] = []
+        self.qval_buffer: List[np.ndarray] = []
+        self.curr_burn_in_steps = 0
+        self.size = 0
+        self.sum_reward = 0.0
+        self.done = False
+
+    def add(self, action: int, reward: float, next_obs: np.ndarray,
+            q_value: np.ndarray, hidden_state: np.ndarray):
+        """hidden_state: (2, H) — the LSTM state AFTER processing the current
+        obs, i.e. the correct initial state for a sequence starting at
+        next_obs."""
+        self.action_buffer.append(action)
+        self.reward_buffer.append(reward)
+        self.obs_buffer.append(next_obs)
+        onehot = np.zeros(self.action_dim, dtype=bool)
+        onehot[action] = True
+        self.last_action_buffer.append(onehot)
+        self.last_reward_buffer.append(reward)
+        self.hidden_buffer.append(hidden_state)
+        self.qval_buffer.append(q_value.reshape(-1))
+        self.sum_reward += reward
+        self.size += 1
+
+    def finish(self, last_qval: Optional[np.ndarray] = None):
+        assert 0 < self.size <= self.block_length
+        S, n, L = self.size, self.forward_steps, self.learning_steps
+        num_sequences = math.ceil(S / L)
+        mfs = min(S, n)
+
+        # per-step bootstrap discount (reference worker.py:443-455 semantics)
+        self.done = last_qval is None
+        gamma_vec = Fn.gamma_vector(S, n, self.gamma, self.done)
+        if self.done:
+            self.qval_buffer.append(np.zeros_like(self.qval_buffer[0]))
+        else:
+            self.qval_buffer.append(last_qval.reshape(-1))
+
+        obs = np.stack(self.obs_buffer)
+        last_action = np.stack(self.last_action_buffer)
+        last_reward = np.array(self.last_reward_buffer, dtype=np.float32)
+        actions = np.array(self.action_buffer, dtype=np.uint8)
+
+        # n-step return over this block's rewards (zero-padded tail)
+        n_step_reward = Fn.n_step_return(
+            np.array(self.reward_buffer, dtype=np.float32), n, self.gamma)
+
+        # per-sequence layout
+        burn_in = np.array([min(i * L + self.curr_burn_in_steps, self.burn_in_steps)
+                            for i in range(num_sequences)], dtype=np.uint8)
+        learning = np.array([min(L, S - i * L) for i in range(num_sequences)],
+                            dtype=np.uint8)
+        forward = np.array([min(n, S + 1 - int(np.sum(learning[:i + 1])))
+                            for i in range(num_sequences)], dtype=np.uint8)
+        assert forward[-1] == 1 and burn_in[0] == self.curr_burn_in_steps
+
+        # stored recurrent state, aligned to each sequence's burn-in start:
+        # learning start (buffer index) = curr_burn_in + i*L; burn-in start =
+        # that - burn_in[i].  (The reference stores index i*L, worker.py:461,
+        # which matches only when the carried prefix is full.)
+        h_idx = [self.curr_burn_in_steps + i * L - int(burn_in[i])
+                 for i in range(num_sequences)]
+        hiddens = np.stack([self.hidden_buffer[j] for j in h_idx])
+
+        # initial priorities from the actor's own q-values
+        qvals = np.stack(self.qval_buffer)          # (S+1, A); current block steps
+        max_q = qvals[mfs: S + 1].max(axis=1)
+        max_q = np.pad(max_q, (0, mfs - 1), mode="edge") if mfs > 1 else max_q
+        taken_q = qvals[np.arange(S), actions]
+        td_errors = np.abs(n_step_reward + gamma_vec * max_q - taken_q).astype(np.float32)
+        priorities = np.zeros(self.block_length // L, dtype=np.float32)
+        priorities[:num_sequences] = calculate_mixed_td_errors(
+            td_errors, learning, self.prio_eta)
+
+        block = Block(obs, last_action, last_reward, actions, n_step_reward,
+                      gamma_vec, hiddens, num_sequences, burn_in, learning, forward)
+
+        # carry burn-in prefix into the next block
+        keep = self.burn_in_steps + 1
+        self.obs_buffer = self.obs_buffer[-keep:]
+        self.last_action_buffer = self.last_action_buffer[-keep:]
+        self.last_reward_buffer = self.last_reward_buffer[-keep:]
+        self.hidden_buffer = self.hidden_buffer[-keep:]
+        self.action_buffer.clear()
+        self.reward_buffer.clear()
+        self.qval_buffer.clear()
+        self.curr_burn_in_steps = len(self.obs_buffer) - 1
+        self.size = 0
+
+        return [block, priorities, self.sum_reward if self.done else None]
+
+
+############################## ReplayBuffer ##############################
+
+
+class ReplayBuffer:
+    """Prioritized block ring buffer + batch assembler (host mode).
+
+    Runs in its own process (reference worker.py:38-261) with three worker
+    threads: ingest (actor queues -> ring), assemble (tree sample -> batches
+    -> batch_queue), and priority-update.  Also the console logger and the
+    run-termination condition.
+    """
+
+    def __init__(self, sample_queue_list, batch_queue, priority_queue,
+                 buffer_capacity: Optional[int] = None,
+                 alpha: Optional[float] = None, beta: Optional[float] = None,
+                 batch_size: Optional[int] = None, seed: Optional[int] = None):
+        c = cfg.get()
+        self.cfg = c
+        self.block_len = c.block_length
+        self.seq_len = c.learning_steps
+        self.buffer_capacity = buffer_capacity or c.buffer_capacity
+        self.num_sequences = self.buffer_capacity // self.seq_len
+        self.num_blocks = self.buffer_capacity // self.block_len
+        self.seq_per_block = self.block_len // self.seq_len
+        self.batch_size = batch_size or c.batch_size
+
+        self.priority_tree = PriorityTree(
+            self.num_sequences,
+            alpha if alpha is not None else c.prio_exponent,
+            beta if beta is not None else c.importance_sampling_exponent,
+            rng=np.random.default_rng(seed))
+
+        self.block_ptr = 0
+        self.size = 0
+        self.env_steps = 0
+        self.num_episodes = 0
+        self.episode_reward = 0.0
+        self.training_steps = 0
+        self.last_training_steps = 0
+        self.sum_loss = 0.0
+        self.last_size = 0
+        self.lock = threading.Lock()
+        self.buffer: List[Optional[Block]] = [None] * self.num_blocks
+        self.sample_queue_list = sample_queue_list
+        self.batch_queue = batch_queue
+        self.priority_queue = priority_queue
+        self.stop_flag = False
+
+    def __len__(self):
+        return self.size
+
+    # -- threads -----------------------------------------------------------
+
+    def run(self):
+        threads = [threading.Thread(target=f, daemon=True)
+                   for f in (self._ingest_loop, self._assemble_loop,
+                             self._priority_loop)]
+        for t in threads:
+            t.start()
+        log_interval = self.cfg.log_interval
+        while True:
+            self._log(log_interval)
+            if self.training_steps >= self.cfg.training_steps:
+                self.stop_flag = True
+                break
+            time.sleep(log_interval)
+
+    def _log(self, interval):
+        print(f"buffer size: {self.size}")
+        print(f"buffer update speed: {(self.size - self.last_size) / interval}/s")
+        self.last_size = self.size
+        print(f"number of environment steps: {self.env_steps}")
+        if self.num_episodes:
+            print(f"average episode return: {self.episode_reward / self.num_episodes:.4f}")
+            self.episode_reward = 0.0
+            self.num_episodes = 0
+        print(f"number of training steps: {self.training_steps}")
+        delta = self.training_steps - self.last_training_steps
+        print(f"training speed: {delta / interval}/s")
+        if delta:
+            print(f"loss: {self.sum_loss / delta:.4f}")
+            self.last_training_steps = self.training_steps
+            self.sum_loss = 0.0
+        print()
+
+    def _ingest_loop(self):
+        """Blocking multiplexed ingest (no busy-spin)."""
+        while not self.stop_flag:
+            got = False
+            for q in self.sample_queue_list:
+                try:
+                    data = q.get_nowait()
+                except queue_mod.Empty:
+                    continue
+                self.add(*data)
+                got = True
+            if not got:
+                time.sleep(0.005)
+
+    def _assemble_loop(self):
+        while self.size < self.cfg.learning_starts and not self.stop_flag:
+            time.sleep(0.5)
+        while not self.stop_flag:
+            if not self.batch_queue.full():
+                self.batch_queue.put(self.sample_batch())
+            else:
+                time.sleep(0.02)
+
+    def _priority_loop(self):
+        while not self.stop_flag:
+            try:
+                data = self.priority_queue.get(timeout=0.1)
+            except queue_mod.Empty:
+                continue
+            self.update_priorities(*data)
+
+    # -- operations ---------------------------------------------------------
+
+    def add(self, block: Block, priority: np.ndarray, episode_reward):
+        with self.lock:
+            idxes = np.arange(self.block_ptr * self.seq_per_block,
+                              (self.block_ptr + 1) * self.seq_per_block,
+                              dtype=np.int64)
+            self.priority_tree.update(idxes, priority)
+            old = self.buffer[self.block_ptr]
+            if old is not None:
+                self.size -= int(np.sum(old.learning_steps))
+            self.size += int(np.sum(block.learning_steps))
+            self.env_steps += int(np.sum(block.learning_steps))
+            self.buffer[self.block_ptr] = block
+            self.block_ptr = (self.block_ptr + 1) % self.num_blocks
+            if episode_reward is not None:
+                self.episode_reward += episode_reward
+                self.num_episodes += 1
+
+    def sample_batch(self) -> TrainingBatch:
+        """Vectorized batch assembly (replaces the reference's per-sample
+        Python slice loop, worker.py:176-210)."""
+        with self.lock:
+            idxes, is_weights = self.priority_tree.sample(self.batch_size)
+            block_idxes = idxes // self.seq_per_block
+            seq_idxes = idxes % self.seq_per_block
+
+            B = self.batch_size
+            burn = np.empty(B, dtype=np.int64)
+            learn = np.empty(B, dtype=np.int64)
+            fwd = np.empty(B, dtype=np.int64)
+            starts = np.empty(B, dtype=np.int64)
+            blocks = []
+            for i, (bi, si) in enumerate(zip(block_idxes, seq_idxes)):
+                blk = self.buffer[bi]
+                assert blk is not None and si < blk.num_sequences
+                blocks.append(blk)
+                burn[i] = blk.burn_in_steps[si]
+                learn[i] = blk.learning_steps[si]
+                fwd[i] = blk.forward_steps[si]
+                starts[i] = blk.burn_in_steps[0] + int(np.sum(blk.learning_steps[:si]))
+
+            T = int((burn + learn + fwd).max())
+            obs_shape = blocks[0].obs.shape[1:]
+            A = blocks[0].last_action.shape[1]
+            obs = np.zeros((B, T) + obs_shape, dtype=blocks[0].obs.dtype)
+            last_action = np.zeros((B, T, A), dtype=np.float32)
+            last_reward = np.zeros((B, T), dtype=np.float32)
+            hidden = np.empty((B, 2, blocks[0].hidden.shape[-1]), dtype=np.float32)
+            actions, rewards, gammas = [], [], []
+            for i, blk in enumerate(blocks):
+                si = seq_idxes[i]
+                s, L = starts[i], int(burn[i] + learn[i] + fwd[i])
+                obs[i, :L] = blk.obs[s - burn[i]: s + learn[i] + fwd[i]]
+                last_action[i, :L] = blk.last_action[s - burn[i]: s + learn[i] + fwd[i]]
+                last_reward[i, :L] = blk.last_reward[s - burn[i]: s + learn[i] + fwd[i]]
+                hidden[i] = blk.hidden[si]
+                ls = int(np.sum(blk.learning_steps[:si]))
+                le = ls + int(learn[i])
+                actions.append(blk.action[ls:le])
+                rewards.append(blk.n_step_reward[ls:le])
+                gammas.append(blk.gamma[ls:le])
+
+            is_rep = np.repeat(is_weights, learn).astype(np.float32)
+            batch = TrainingBatch(
+                obs=torch.from_numpy(obs),
+                last_action=torch.from_numpy(last_action),
+                last_reward=torch.from_numpy(last_reward),
+                hidden=torch.from_numpy(hidden).transpose(0, 1).contiguous(),
+                action=torch.from_numpy(np.concatenate(actions)).long().unsqueeze(1),
+                n_step_reward=torch.from_numpy(np.concatenate(rewards)),
+                gamma=torch.from_numpy(np.concatenate(gammas)),
+                burn_in_steps=torch.from_numpy(burn),
+                learning_steps=torch.from_numpy(learn),
+                forward_steps=torch.from_numpy(fwd),
+                idxes=idxes,
+                is_weights=torch.from_numpy(is_rep),
+                old_ptr=self.block_ptr,
+                env_steps=self.env_steps,
+            )
+        return batch
+
+    def update_priorities(self, idxes: np.ndarray, td_errors: np.ndarray,
+                          old_ptr: int, loss: float):
+        """Masks out indexes overwritten by the ring pointer since sampling
+        (reference worker.py:242-261 wraparound semantics), then tree update."""
+        with self.lock:
+            if self.block_ptr > old_ptr:
+                mask = ((idxes < old_ptr * self.seq_per_block)
+                        | (idxes >= self.block_ptr * self.seq_per_block))
+                idxes, td_errors = idxes[mask], td_errors[mask]
+            elif self.block_ptr < old_ptr:
+                mask = ((idxes < old_ptr * self.seq_per_block)
+                        & (idxes >= self.block_ptr * self.seq_per_block))
+                idxes, td_errors = idxes[mask], td_errors[mask]
+            if len(idxes):
+                self.priority_tree.update(idxes, td_errors)
+        self.training_steps += 1
+        self.sum_loss += loss
+
+
+############################## Learner ##############################
+
+
+class Learner:
+    """GPU learner: online + target nets, fused single-pass double-Q update,
+    Adam with grad-norm clip, priority feedback, weight publish, checkpoints
+    (reference worker.py:278-390)."""
+
+    def __init__(self, batch_queue, priority_queue, model: Network,
+                 grad_norm: Optional[float] = None, lr: Optional[float] = None,
+                 eps: Optional[float] = None, game_name: Optional[str] = None,
+                 target_net_update_interval: Optional[int] = None,
+                 save_interval: Optional[int] = None):
+        c = cfg.get()
+        self.cfg = c
+        self.device = torch.device(
+            c.device if torch.cuda.is_available() or c.device == "cpu" else "cpu")
+        self.online_net = Network(model.action_dim, model.obs_shape,
+                                  model.hidden_dim, encoder=c.encoder,
+                                  forward_steps=c.forward_steps,
+                                  mlp_hidden=c.mlp_hidden)
+        self.online_net.load_state_dict(model.state_dict())
+        self.online_net.to(self.device)
+        self.online_net.train()
+        self.target_net = Network(model.action_dim, model.obs_shape,
+                                  model.hidden_dim, encoder=c.encoder,
+                                  forward_steps=c.forward_steps,
+                                  mlp_hidden=c.mlp_hidden)
+        self.target_net.load_state_dict(model.state_dict())
+        self.target_net.to(self.device)
+        self.target_net.eval()
+        self.optimizer = torch.optim.Adam(self.online_net.parameters(),
+                                          lr=lr or c.lr, eps=eps or c.eps)
+        self.grad_norm = grad_norm or c.grad_norm
+        self.batch_queue = batch_queue
+        self.priority_queue = priority_queue
+        self.num_updates = 0
+        self.env_steps = 0
+        self.target_net_update_interval = (target_net_update_interval
+                                           or c.target_net_update_interval)
+        self.save_interval = save_interval or c.save_interval
+        self.game_name = game_name or c.game_name
+        self.shared_model = model
+        self.batched_data: List[TrainingBatch] = []
+        self.amp = c.amp and self.device.type == "cuda" and c.dtype == "bf16"
+
+    # -- weight publication -------------------------------------------------
+
+    def store_weights(self):
+        state = {k: v.cpu() for k, v in self.online_net.state_dict().items()}
+        self.shared_model.load_state_dict(state)
+
+    # -- data staging -------------------------------------------------------
+
+    def _prefetch_loop(self):
+        while True:
+            if len(self.batched_data) < 4:
+                try:
+                    data = self.batch_queue.get(timeout=0.1)
+                except queue_mod.Empty:
+                    continue
+                if self.device.type == "cuda":
+                    data.pin()
+                self.batched_data.append(data)
+            else:
+                time.sleep(0.01)
+
+    # -- the update step ----------------------------------------------------
+
+    def train_step(self, batch: TrainingBatch) -> Tuple[float, np.ndarray]:
+        """One update.  Returns (loss, per-sequence priorities)."""
+        c = self.cfg
+        batch.to(self.device)
+        h0 = (batch.hidden[:1], batch.hidden[1:])
+
+        ctx = torch.autocast("cuda", dtype=torch.bfloat16) if self.amp else _nullctx()
+        with ctx:
+            with torch.no_grad():
+                q_tgt_all = self.target_net.calculate_q_(
+                    batch.obs, batch.last_action, batch.last_reward, h0,
+                    batch.burn_in_steps, batch.learning_steps, batch.forward_steps)
+            q_learn, q_online_tgt = self.online_net.calculate_q_both(
+                batch.obs, batch.last_action, batch.last_reward, h0,
+                batch.burn_in_steps, batch.learning_steps, batch.forward_steps)
+
+        q_learn = q_learn.float()
+        with torch.no_grad():
+            target_q = Fn.double_q_target(
+                q_online_tgt.detach().float(), q_tgt_all.float(),
+                batch.n_step_reward, batch.gamma, c.rescale_eps)
+        batch_q = q_learn.gather(1, batch.action).squeeze(1)
+        loss = (batch.is_weights
+                * Fn.per_step_loss(batch_q, target_q, c.loss_fn, c.huber_kappa)).mean()
+
+        self.optimizer.zero_grad(set_to_none=True)
+        loss.backward()
+        nn.utils.clip_grad_norm_(self.online_net.parameters(), self.grad_norm)
+        self.optimizer.step()
+        self.num_updates += 1
+
+        td = (target_q - batch_q).detach().abs().cpu().numpy()
+        priorities = calculate_mixed_td_errors(
+            td, batch.learning_steps.numpy(), c.prio_eta)
+        return float(loss.item()), priorities
+
+    def run(self):
+        threading.Thread(target=self._prefetch_loop, daemon=True).start()
+        start_time = time.time()
+        os.makedirs("models", exist_ok=True)
+        while self.num_updates < self.cfg.training_steps:
+            while not self.batched_data:
+                time.sleep(0.05)
+            batch = self.batched_data.pop(0)
+            loss, priorities = self.train_step(batch)
+            self.priority_queue.put((batch.idxes, priorities, batch.old_ptr, loss))
+            self.env_steps = batch.env_steps
+            if self.num_updates % 4 == 0:
+                self.store_weights()
+            if self.num_updates % self.target_net_update_interval == 0:
+                self.target_net.load_state_dict(self.online_net.state_dict())
+            if self.num_updates % self.save_interval == 0:
+                self.save(start_time)
+
+    def save(self, start_time):
+        os.makedirs("models", exist_ok=True)
+        torch.save((self.online_net.state_dict(), self.num_updates,
+                    self.env_steps, (time.time() - start_time) / 60),
+                   os.path.join("models", f"{self.game_name}{self.num_updates}.pth"))
+
+    # reference-compat statics (worker.py:383-390)
+    value_rescale = staticmethod(Fn.value_rescale)
+    inverse_value_rescale = staticmethod(Fn.inverse_value_rescale)
+
+
+class _nullctx:
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        return False
+
+
+############################## Actor ##############################
+
+
+class Actor:
+    """CPU env loop: eps-greedy inference, LocalBuffer accumulation, periodic
+    weight pull from the shared model (reference worker.py:500-574)."""
+
+    def __init__(self, epsilon: float, model: Network, sample_queue,
+                 env_fn=None, max_episode_steps: Optional[int] = None,
+                 block_length: Optional[int] = None, seed: Optional[int] = None):
+        c = cfg.get()
+        from .envs import create_env
+        self.env = env_fn() if env_fn else create_env(seed=seed)
+        self.action_dim = self.env.action_dim
+        self.model = Network(self.action_dim, c.obs_shape, c.hidden_dim,
+                             encoder=c.encoder, forward_steps=c.forward_steps,
+                             mlp_hidden=c.mlp_hidden)
+        self.model.eval()
+        self.local_buffer = LocalBuffer(self.action_dim)
+        self.epsilon = epsilon
+        self.shared_model = model
+        self.sample_queue = sample_queue
+        self.max_episode_steps = max_episode_steps or c.max_episode_steps
+        self.block_length = block_length or c.block_length
+        self.update_interval = c.actor_update_interval
+        self.rng = np.random.default_rng(seed)
+        self.stop_after_steps = None  # test hook
+
+    def run(self):
+        actor_steps = 0
+        while True:
+            done = False
+            agent_state = self.reset()
+            episode_steps = 0
+            while not done and episode_steps < self.max_episode_steps:
+                with torch.no_grad():
+                    q_value, hidden = self.model(agent_state)
+                if self.rng.random() < self.epsilon:
+                    action = int(self.rng.integers(self.action_dim))
+                else:
+                    action = int(torch.argmax(q_value, 1).item())
+
+                next_obs, reward, done, _ = self.env.step(action)
+                agent_state.update(next_obs[None], action, [reward], hidden)
+                episode_steps += 1
+                actor_steps += 1
+                self.local_buffer.add(action, reward, next_obs,
+                                      q_value.numpy(),
+                                      torch.cat(hidden).squeeze(1).numpy())
+
+                if done:
+                    self.sample_queue.put(self.local_buffer.finish())
+                elif (len(self.local_buffer) == self.block_length
+                      or episode_steps == self.max_episode_steps):
+                    with torch.no_grad():
+                        q_value, _ = self.model(agent_state)
+                    data = self.local_buffer.finish(q_value.numpy())
+                    if self.epsilon > 0.01:
+                        data[2] = None  # only near-greedy actors report returns
+                    self.sample_queue.put(data)
+
+                if actor_steps % self.update_interval == 0:
+                    self.update_weights()
+                if self.stop_after_steps and actor_steps >= self.stop_after_steps:
+                    return actor_steps
+
+    def update_weights(self):
+        self.model.load_state_dict(self.shared_model.state_dict())
+
+    def reset(self):
+        obs = self.env.reset()
+        self.local_buffer.reset(obs)
+        return AgentState(torch.from_numpy(obs).unsqueeze(0), self.action_dim)

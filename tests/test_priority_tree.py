@@ -1,0 +1,68 @@
+import numpy as np
+
+from r2d2_amd.replay.priority_tree import PriorityTree
+
+
+def test_root_equals_leaf_sum_after_random_interleaving():
+    rng = np.random.default_rng(0)
+    tree = PriorityTree(1000, prio_exponent=0.9, is_exponent=0.6,
+                        rng=np.random.default_rng(1))
+    for _ in range(50):
+        n = rng.integers(1, 64)
+        idx = rng.integers(0, 1000, size=n)
+        td = rng.random(n).astype(np.float64) * 10
+        tree.update(idx, td)
+        if tree.total > 0:
+            tree.sample(16)  # sampling must not mutate the tree
+        assert abs(tree.total - tree.levels[-1].sum()) < 1e-9 * max(1, tree.total)
+    # every internal node consistent
+    for lvl in range(len(tree.levels) - 1):
+        child = tree.levels[lvl + 1]
+        np.testing.assert_allclose(tree.levels[lvl],
+                                   child[0::2] + child[1::2], rtol=1e-12)
+
+
+def test_sampled_frequency_proportional_to_priority():
+    tree = PriorityTree(8, prio_exponent=1.0, is_exponent=0.6,
+                        rng=np.random.default_rng(2))
+    prios = np.array([1, 1, 2, 4, 8, 0, 0, 16], dtype=np.float64)
+    tree.update(np.arange(8), prios)
+    counts = np.zeros(8)
+    n_draws = 2000
+    for _ in range(n_draws):
+        idx, _ = tree.sample(32)
+        np.add.at(counts, idx, 1)
+    freq = counts / counts.sum()
+    expect = prios / prios.sum()
+    assert counts[5] == 0 and counts[6] == 0
+    np.testing.assert_allclose(freq, expect, atol=0.01)
+
+
+def test_stratified_coverage():
+    """Stratified sampling: with uniform priorities every draw covers distinct
+    equal intervals -> all leaves hit with batch == capacity."""
+    tree = PriorityTree(64, 1.0, 0.6, rng=np.random.default_rng(3))
+    tree.update(np.arange(64), np.ones(64))
+    idx, w = tree.sample(64)
+    assert sorted(idx.tolist()) == list(range(64))
+    np.testing.assert_allclose(w, np.ones(64))
+
+
+def test_is_weights_formula():
+    tree = PriorityTree(4, prio_exponent=0.9, is_exponent=0.6,
+                        rng=np.random.default_rng(4))
+    td = np.array([1.0, 2.0, 3.0, 4.0])
+    tree.update(np.arange(4), td)
+    idx, w = tree.sample(256)
+    p = td[idx] ** 0.9
+    expect = (p / p.min()) ** -0.6
+    np.testing.assert_allclose(w, expect, rtol=1e-10)
+
+
+def test_update_overwrite():
+    tree = PriorityTree(16, 1.0, 0.6, rng=np.random.default_rng(5))
+    tree.update(np.arange(16), np.ones(16))
+    tree.update(np.array([3]), np.array([100.0]))
+    assert abs(tree.total - (15 + 100)) < 1e-9
+    idx, _ = tree.sample(64)
+    assert (idx == 3).mean() > 0.5

@@ -1,0 +1,104 @@
+import queue
+
+import numpy as np
+import pytest
+
+from r2d2_amd import config as cfg
+from r2d2_amd.worker import Block, LocalBuffer, ReplayBuffer
+
+
+def small_cfg():
+    return cfg.apply("cartpole", buffer_capacity=320, block_length=40,
+                     burn_in_steps=8, learning_steps=8, forward_steps=3,
+                     batch_size=4, learning_starts=40, hidden_dim=16)
+
+
+def make_block(value, steps=40, action_dim=2, hidden=16, burn=8, learn=8, n=3):
+    buf = LocalBuffer(action_dim, forward_steps=n, burn_in_steps=burn,
+                      learning_steps=learn, gamma=0.99, hidden_dim=hidden,
+                      block_length=steps)
+    buf.reset(np.full((4,), value, dtype=np.float32))
+    for t in range(steps):
+        buf.add(t % action_dim, float(value), np.full((4,), value, dtype=np.float32),
+                np.ones(action_dim, dtype=np.float32) * value,
+                np.full((2, hidden), value, dtype=np.float32))
+    block, prios, _ = buf.finish(np.zeros(action_dim, dtype=np.float32))
+    return block, prios
+
+
+def make_rb(seed=0):
+    return ReplayBuffer([queue.Queue()], queue.Queue(4), queue.Queue(4), seed=seed)
+
+
+def test_add_and_size_accounting():
+    small_cfg()
+    rb = make_rb()
+    blk, prios = make_block(1.0)
+    rb.add(blk, prios + 1.0, None)
+    assert len(rb) == 40
+    assert rb.env_steps == 40
+    # ring overwrite: fill all 8 slots then one more
+    for v in range(2, 10):
+        blk, prios = make_block(float(v))
+        rb.add(blk, prios + 1.0, None)
+    assert len(rb) == 320
+    assert rb.block_ptr == 1
+
+
+def test_sample_batch_contents():
+    small_cfg()
+    rb = make_rb(seed=3)
+    for v in range(1, 9):
+        blk, prios = make_block(float(v))
+        rb.add(blk, prios + 1.0, None)
+    batch = rb.sample_batch()
+    B = 4
+    assert batch.obs.shape[0] == B
+    assert batch.hidden.shape == (2, B, 16)
+    assert batch.action.shape[0] == int(batch.learning_steps.sum())
+    assert batch.is_weights.shape[0] == int(batch.learning_steps.sum())
+    # each sampled row's obs content equals the block value it came from
+    for i in range(B):
+        bi = batch.idxes[i] // rb.seq_per_block
+        v = float(bi + 1)
+        L = int(batch.burn_in_steps[i] + batch.learning_steps[i]
+                + batch.forward_steps[i])
+        assert np.allclose(batch.obs[i, :L].numpy(), v)
+        # fresh-episode blocks: seq 0 (burn 0) and seq 1 (burn-in start at
+        # buffer index 0) both start from the reset (zero) recurrent state
+        si = batch.idxes[i] % rb.seq_per_block
+        expect_h = 0.0 if si <= 1 else v
+        assert np.allclose(batch.hidden[:, i].numpy(), expect_h)
+
+
+def test_update_priorities_wraparound_mask():
+    small_cfg()
+    rb = make_rb()
+    for v in range(1, 9):
+        blk, prios = make_block(float(v))
+        rb.add(blk, prios + 1.0, None)
+    # pretend we sampled when ptr was 0 (old_ptr=0) and two blocks (0,1) have
+    # since been overwritten -> block_ptr = 2
+    for v in (10, 11):
+        blk, prios = make_block(float(v))
+        rb.add(blk, prios + 1.0, None)
+    assert rb.block_ptr == 2
+    idxes = np.array([0, 5, 10, 15])  # seqs in blocks 0,1,2,3
+    before = rb.priority_tree.levels[-1].copy()
+    rb.update_priorities(idxes, np.array([99.0, 99.0, 99.0, 99.0]), 0, 0.0)
+    after = rb.priority_tree.levels[-1]
+    # blocks 0,1 (leaves 0..9) were overwritten after sampling -> masked out
+    assert np.allclose(after[:10], before[:10])
+    assert after[10] != before[10] and after[15] != before[15]
+
+
+def test_priority_zero_blocks_never_sampled():
+    small_cfg()
+    rb = make_rb(seed=1)
+    blk, _ = make_block(1.0)
+    rb.add(blk, np.array([1.0, 1, 1, 1, 1]), None)
+    blk2, _ = make_block(2.0)
+    rb.add(blk2, np.zeros(5), None)
+    for _ in range(20):
+        idx, _ = rb.priority_tree.sample(8)
+        assert (idx < 5).all()
