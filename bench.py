@@ -1,0 +1,176 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: learner seq-samples/sec (bs=64, L=80, 84x84x4 frames).
+
+Measures the full R2D2 learner update — target-net forward, online forward,
+double-Q target with value rescaling, loss, backward, grad-norm clip, Adam,
+and per-sequence priority computation — on synthetic 84x84x4 uint8 frames
+with random-init weights (no datasets/checkpoints are downloadable in this
+environment).  This is BASELINE.json's metric on BASELINE.json's config;
+value is the WHOLE-JOB aggregate over all ranks.
+
+Single GPU:     python bench.py --gpus 1 --steps 30 --warmup 10
+Multi-GPU:      torchrun --nproc-per-node N bench.py --gpus N ...
+                (one rank per GPU over RCCL; reads RANK/WORLD_SIZE/MASTER_*)
+"""
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+
+
+def build_batch(c, device, seed, full_len=True):
+    """One synthetic device-resident training batch of the benchmark shape:
+    B sequences of burn_in+learning+forward steps of (C,84,84) uint8 frames."""
+    from r2d2_amd.worker import TrainingBatch
+
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    B = c.batch_size
+    burn = torch.full((B,), c.burn_in_steps, dtype=torch.int64)
+    learn = torch.full((B,), c.learning_steps, dtype=torch.int64)
+    fwd = torch.full((B,), c.forward_steps, dtype=torch.int64)
+    T = int((burn + learn + fwd).max())
+    A = c.action_dim
+    sum_learn = int(learn.sum())
+
+    obs = torch.randint(0, 256, (B, T) + tuple(c.obs_shape),
+                        dtype=torch.uint8, generator=g)
+    la = torch.zeros(B, T, A)
+    la[torch.arange(B)[:, None], torch.arange(T)[None, :],
+       torch.randint(0, A, (B, T), generator=g)] = 1.0
+    lr = torch.randn(B, T, generator=g) * 0.1
+    hidden = torch.randn(2, B, c.hidden_dim, generator=g) * 0.05
+    action = torch.randint(0, A, (sum_learn, 1), generator=g)
+    n_step_reward = torch.randn(sum_learn, generator=g).abs()
+    gamma = torch.full((sum_learn,), c.gamma ** c.forward_steps)
+    is_weights = torch.rand(sum_learn, generator=g) * 0.5 + 0.5
+
+    batch = TrainingBatch(
+        obs=obs, last_action=la, last_reward=lr, hidden=hidden,
+        action=action, n_step_reward=n_step_reward, gamma=gamma,
+        burn_in_steps=burn, learning_steps=learn, forward_steps=fwd,
+        idxes=np.arange(B), is_weights=is_weights, old_ptr=0, env_steps=0)
+    return batch.to(device, non_blocking=False)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--preset", type=str, default="mspacman")
+    ap.add_argument("--engine", type=str, default="auto",
+                    choices=["auto", "hip", "eager"],
+                    help="auto: HIP kernels on GPU when built, eager otherwise")
+    ap.add_argument("--batches", type=int, default=4,
+                    help="distinct synthetic batches rotated through")
+    args = ap.parse_args()
+
+    from r2d2_amd import config as cfg
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    have_cuda = torch.cuda.is_available()
+    device = torch.device(f"cuda:{local_rank}" if have_cuda else "cpu")
+    if have_cuda:
+        torch.cuda.set_device(device)
+
+    if world_size > 1:
+        import torch.distributed as dist
+        dist.init_process_group("nccl" if have_cuda else "gloo",
+                                rank=rank, world_size=world_size)
+
+    c = cfg.apply(args.preset, device="cuda" if have_cuda else "cpu")
+    use_hip = args.engine == "hip" or (args.engine == "auto" and have_cuda)
+    if not have_cuda:
+        c = cfg.apply(args.preset, device="cpu", dtype="fp32", amp=False,
+                      use_hip_kernels=False)
+
+    torch.manual_seed(1234)  # identical init on all ranks before broadcast
+    from r2d2_amd.models.network import Network
+    from r2d2_amd.worker import Learner
+
+    model = Network(c.action_dim, c.obs_shape, c.hidden_dim, encoder=c.encoder,
+                    forward_steps=c.forward_steps, mlp_hidden=c.mlp_hidden)
+    learner = Learner(batch_queue=None, priority_queue=None, model=model)
+    if use_hip and c.use_hip_kernels:
+        try:
+            learner.enable_hip_engine()
+        except AttributeError:
+            pass  # engine not built yet; eager path
+    batches = [build_batch(c, device, seed=1000 + rank * 100 + i)
+               for i in range(args.batches)]
+
+    def one_step(i):
+        loss, priorities = learner.train_step(batches[i % len(batches)])
+        return loss
+
+    def barrier():
+        if world_size > 1:
+            import torch.distributed as dist
+            dist.barrier()
+        if have_cuda:
+            torch.cuda.synchronize()
+
+    for i in range(args.warmup):
+        one_step(i)
+    barrier()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        one_step(args.warmup + i)
+    if have_cuda:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    barrier()
+
+    # max over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if world_size > 1 and have_cuda else "cpu")
+    if world_size > 1:
+        import torch.distributed as dist
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    n_gpus = world_size if world_size > 1 else args.gpus
+    value = c.batch_size * args.steps * n_gpus / elapsed
+    baseline = 363.0  # BASELINE.md derived learner seq-samples/sec
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "learner seq-samples/sec (bs=64, L=80, 84x84x4 frames)",
+            "value": round(value, 2),
+            "unit": "seq/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(value / baseline, 3),
+            "dtype": "bf16" if (have_cuda and c.dtype == "bf16") else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "r2d2-nature-cnn-lstm512-dueling",
+                "global_batch": c.batch_size * n_gpus,
+                "seq_len": c.burn_in_steps + c.learning_steps,
+                "burn_in": c.burn_in_steps,
+                "forward_steps": c.forward_steps,
+                "obs": list(c.obs_shape),
+                "parallelism": f"dp{n_gpus}",
+                "engine": "hip" if (use_hip and getattr(learner, "hip_engine", None)) else "eager",
+            },
+        }))
+
+    if world_size > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -501,6 +501,28 @@ class Learner:
         self.batched_data: List[TrainingBatch] = []
         self.amp = c.amp and self.device.type == "cuda" and c.dtype == "bf16"
 
+        # multi-learner data parallelism: bucketed grad all-reduce over
+        # RCCL/xGMI, overlapped with backward (parallel/ddp.py)
+        self.reducer = None
+        import torch.distributed as dist
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            from .parallel.ddp import GradAllReducer
+            self.reducer = GradAllReducer(list(self.online_net.parameters()))
+            self.reducer.broadcast_params(src=0)
+            self.target_net.load_state_dict(self.online_net.state_dict())
+
+    # -- HIP engine ---------------------------------------------------------
+
+    def enable_hip_engine(self):
+        """Switch the loss/priority path (and, as they land, the full forward/
+        backward) to the gfx950 HIP kernels.  Raises if the extension is not
+        built — on a GPU box the native path must be the one that runs."""
+        from .ops import hip_ops
+        hip_ops.ext(required=True)
+        self.hip_engine = True
+
+    hip_engine = False
+
     # -- weight publication -------------------------------------------------
 
     def store_weights(self):
@@ -541,23 +563,41 @@ class Learner:
                 batch.burn_in_steps, batch.learning_steps, batch.forward_steps)
 
         q_learn = q_learn.float()
-        with torch.no_grad():
-            target_q = Fn.double_q_target(
-                q_online_tgt.detach().float(), q_tgt_all.float(),
-                batch.n_step_reward, batch.gamma, c.rescale_eps)
-        batch_q = q_learn.gather(1, batch.action).squeeze(1)
-        loss = (batch.is_weights
-                * Fn.per_step_loss(batch_q, target_q, c.loss_fn, c.huber_kappa)).mean()
+        if self.hip_engine:
+            # fused double-Q target + TD + loss + priority, one kernel pair,
+            # dLoss/dQ computed in the same sweep (ops/hip/loss_kernels.hip)
+            from .ops import hip_ops
+            loss, prio_dev = hip_ops.fused_double_q_loss(
+                q_learn, q_online_tgt.detach().float(), q_tgt_all.float(),
+                batch.action, batch.n_step_reward, batch.gamma,
+                batch.is_weights, batch.learning_steps,
+                eps=c.rescale_eps, kappa=c.huber_kappa, loss_kind=c.loss_fn,
+                eta=c.prio_eta)
+        else:
+            with torch.no_grad():
+                target_q = Fn.double_q_target(
+                    q_online_tgt.detach().float(), q_tgt_all.float(),
+                    batch.n_step_reward, batch.gamma, c.rescale_eps)
+            batch_q = q_learn.gather(1, batch.action).squeeze(1)
+            loss = (batch.is_weights * Fn.per_step_loss(
+                batch_q, target_q, c.loss_fn, c.huber_kappa)).mean()
 
         self.optimizer.zero_grad(set_to_none=True)
+        if self.reducer is not None:
+            self.reducer.prepare()
         loss.backward()
+        if self.reducer is not None:
+            self.reducer.finish()
         nn.utils.clip_grad_norm_(self.online_net.parameters(), self.grad_norm)
         self.optimizer.step()
         self.num_updates += 1
 
-        td = (target_q - batch_q).detach().abs().cpu().numpy()
-        priorities = calculate_mixed_td_errors(
-            td, batch.learning_steps.numpy(), c.prio_eta)
+        if self.hip_engine:
+            priorities = prio_dev.cpu().numpy()
+        else:
+            td = (target_q - batch_q).detach().abs().cpu().numpy()
+            priorities = calculate_mixed_td_errors(
+                td, batch.learning_steps.numpy(), c.prio_eta)
         return float(loss.item()), priorities
 
     def run(self):
