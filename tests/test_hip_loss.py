@@ -49,7 +49,9 @@ def test_fused_loss_matches_golden(kind, A):
 
     assert torch.allclose(loss, loss_g, atol=1e-5, rtol=1e-5), (loss, loss_g)
     assert torch.allclose(prio, prio_g.to(dev), atol=1e-5)
-    assert torch.allclose(grad_hip, q2.grad, atol=1e-6)
+    # fp32 target recomputation (sqrtf chain) differs by ULPs; the MSE grad
+    # amplifies by 2*w*td, so compare with fp32-appropriate tolerance
+    assert torch.allclose(grad_hip, q2.grad, atol=2e-5, rtol=1e-4)
 
 
 def test_argmax_tie_break_first():
