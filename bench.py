@@ -56,6 +56,33 @@ def build_batch(c, device, seed, full_len=True):
     return batch.to(device, non_blocking=False)
 
 
+def build_synthetic_block(c, rng):
+    """A full actor Block of synthetic frames (for pre-filling the replay)."""
+    from r2d2_amd.worker import Block
+
+    steps, burn, learn, n = (c.block_length, c.burn_in_steps,
+                             c.learning_steps, c.forward_steps)
+    nseq = steps // learn
+    rows = burn + steps + 1
+    A = c.action_dim
+    obs = rng.integers(0, 256, size=(rows,) + tuple(c.obs_shape), dtype=np.uint8)
+    la = np.zeros((rows, A), dtype=bool)
+    la[np.arange(rows), rng.integers(0, A, rows)] = True
+    lr = rng.normal(size=rows).astype(np.float32) * 0.1
+    action = rng.integers(0, A, steps).astype(np.uint8)
+    nsr = rng.normal(size=steps).astype(np.float32).clip(-5, 5)
+    gamma = np.full(steps, c.gamma ** n, dtype=np.float32)
+    hidden = (rng.normal(size=(nseq, 2, c.hidden_dim)) * 0.05).astype(np.float32)
+    learn_arr = np.full(nseq, learn, dtype=np.uint8)
+    # LocalBuffer invariant: forward steps are bounded by the block end
+    # (worker.py finish(): forward[-1] == 1) — the obs rows only extend one
+    # step past the last learning step.
+    fwd_arr = np.minimum(
+        n, steps + 1 - np.cumsum(learn_arr.astype(np.int64))).astype(np.uint8)
+    return Block(obs, la, lr, action, nsr, gamma, hidden, nseq,
+                 np.full(nseq, burn, dtype=np.uint8), learn_arr, fwd_arr)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -67,6 +94,11 @@ def main():
                     help="auto: HIP kernels on GPU when built, eager otherwise")
     ap.add_argument("--batches", type=int, default=4,
                     help="distinct synthetic batches rotated through")
+    ap.add_argument("--gpu-replay", action="store_true",
+                    help="sample each step from the GPU-resident prioritized "
+                         "replay (sum-tree + on-device gather in the timed loop)")
+    ap.add_argument("--replay-transitions", type=int, default=160_000,
+                    help="synthetic transitions pre-filled into the GPU replay")
     args = ap.parse_args()
 
     from r2d2_amd import config as cfg
@@ -103,11 +135,29 @@ def main():
             learner.enable_hip_engine()
         except AttributeError:
             pass  # engine not built yet; eager path
-    batches = [build_batch(c, device, seed=1000 + rank * 100 + i)
-               for i in range(args.batches)]
+    replay = None
+    if args.gpu_replay and have_cuda:
+        assert learner.hip_engine, "--gpu-replay needs the HIP engine"
+        from r2d2_amd.replay.gpu_replay import GpuReplayBuffer
+        replay = GpuReplayBuffer(device=device,
+                                 capacity=args.replay_transitions)
+        rng = np.random.default_rng(42 + rank)
+        for _ in range(replay.num_blocks):
+            replay.ingest(build_synthetic_block(c, rng),
+                          rng.random(replay.spb).astype(np.float32) + 0.1)
+        torch.cuda.synchronize()
+        batches = None
+    else:
+        batches = [build_batch(c, device, seed=1000 + rank * 100 + i)
+                   for i in range(args.batches)]
 
     def one_step(i):
-        loss, priorities = learner.train_step(batches[i % len(batches)])
+        if replay is not None:
+            batch = replay.sample()
+            loss, prio = learner.train_step(batch)
+            replay.update_priorities(batch.idxes, prio, batch.old_ptr)
+        else:
+            loss, _ = learner.train_step(batches[i % len(batches)])
         return loss
 
     def barrier():
@@ -164,6 +214,7 @@ def main():
                 "obs": list(c.obs_shape),
                 "parallelism": f"dp{n_gpus}",
                 "engine": "hip" if (use_hip and getattr(learner, "hip_engine", None)) else "eager",
+                "replay": "gpu-resident" if replay is not None else "prebuilt-batches",
             },
         }))
 

@@ -124,14 +124,14 @@ PRESETS = {
     "mspacman_gpu_replay": dict(
         game_name="MsPacman", env_type="synthetic", obs_shape=(4, 84, 84),
         action_dim=9, encoder="nature", gpu_replay=True, num_actors=256,
-        buffer_capacity=8_000_000,
+        buffer_capacity=4_000_000,
     ),
     # configs[3]: 8x data-parallel learners (parallelism degree comes from
     # torchrun's WORLD_SIZE; the preset is otherwise mspacman_gpu_replay)
     "mspacman_dp": dict(
         game_name="MsPacman", env_type="synthetic", obs_shape=(4, 84, 84),
         action_dim=9, encoder="nature", gpu_replay=True, num_actors=256,
-        buffer_capacity=8_000_000,
+        buffer_capacity=4_000_000,
     ),
     # configs[4]: IMPALA-deep ResNet encoder
     "seaquest_impala": dict(

@@ -11,6 +11,26 @@ std::vector<torch::Tensor> fused_double_q_loss(
 torch::Tensor segment_priority(torch::Tensor abs_td, torch::Tensor seg_offsets,
                                double eta);
 
+// sumtree.hip
+void sumtree_update(torch::Tensor tree, int64_t leaf_offset, torch::Tensor idxes,
+                    torch::Tensor td, double alpha, int64_t old_ptr,
+                    int64_t cur_ptr, int64_t seq_per_block, int64_t num_blocks);
+std::vector<torch::Tensor> sumtree_sample(torch::Tensor tree, int64_t leaf_offset,
+                                          int64_t num_levels, torch::Tensor jitter,
+                                          int64_t n, double beta);
+
+// replay_gather.hip
+std::vector<torch::Tensor> replay_gather_meta(
+    torch::Tensor idx, torch::Tensor burn_s, torch::Tensor learn_s,
+    torch::Tensor fwd_s, torch::Tensor obs_start_s, torch::Tensor learn_off_s,
+    int64_t spb);
+std::vector<torch::Tensor> replay_gather_batch(
+    torch::Tensor obs_store, torch::Tensor la_store, torch::Tensor lr_store,
+    torch::Tensor act_store, torch::Tensor nsr_store, torch::Tensor gam_store,
+    torch::Tensor hid_store, torch::Tensor idx, torch::Tensor meta,
+    torch::Tensor seg, torch::Tensor weights, int64_t T, int64_t A,
+    int64_t max_learn, int64_t H, int64_t spb);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "r2d2_amd gfx950 HIP kernels";
     m.def("fused_double_q_loss", &fused_double_q_loss,
@@ -18,4 +38,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "abs_td, target)");
     m.def("segment_priority", &segment_priority,
           "Per-sequence mixed max/mean |TD| priority over ragged segments");
+    m.def("sumtree_update", &sumtree_update,
+          "GPU sum-tree priority update (duplicate-safe, ring-stale-masked)");
+    m.def("sumtree_sample", &sumtree_sample,
+          "GPU sum-tree stratified sample -> (idx, prio, is_weight)");
+    m.def("replay_gather_meta", &replay_gather_meta,
+          "Sampled-sequence metadata + segment offsets");
+    m.def("replay_gather_batch", &replay_gather_batch,
+          "On-device padded batch assembly from the GPU block store");
 }

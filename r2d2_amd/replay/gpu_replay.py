@@ -1,0 +1,189 @@
+"""GPU-resident prioritized replay: HBM block store + device sum-tree +
+on-device batch assembly.
+
+The MI355X replacement for the reference's host-RAM ReplayBuffer + CPU
+PriorityTree + Python batch assembler: 288 GB of
+HBM3E holds the whole 4M-transition store (uint8 frames), sampling descends
+the tree on-device, the padded training batch is gathered straight out of
+the block store, and priority updates (with the ring-overwrite stale mask of
+reference worker.py:247-256) never leave the GPU.
+
+Ingest takes host `Block`s (from CPU actors) through pinned staging buffers
+on a dedicated copy stream.
+"""
+
+from typing import List, Optional
+
+import numpy as np
+import torch
+
+from .. import config as cfg
+from ..ops import hip_ops
+
+
+class GpuReplayBuffer:
+    def __init__(self, device="cuda", capacity: Optional[int] = None,
+                 batch_size: Optional[int] = None):
+        c = cfg.get()
+        self.cfg = c
+        self.device = torch.device(device)
+        assert len(c.obs_shape) == 3, "GPU replay stores uint8 frames"
+        self.obs_shape = tuple(c.obs_shape)
+        self.frame_bytes = int(np.prod(self.obs_shape))
+        self.A = c.action_dim
+        self.H = c.hidden_dim
+        self.alpha = c.prio_exponent
+        self.beta = c.importance_sampling_exponent
+        self.eta = c.prio_eta
+        self.batch_size = batch_size or c.batch_size
+
+        self.block_len = c.block_length
+        self.learn_len = c.learning_steps
+        self.spb = c.seq_per_block
+        self.capacity = capacity or c.buffer_capacity
+        self.num_blocks = self.capacity // self.block_len
+        self.num_sequences = self.num_blocks * self.spb
+        self.obs_rows = c.burn_in_steps + self.block_len + 1
+        self.T = c.seq_len  # fixed padded length (burn+learn+fwd)
+
+        dev = self.device
+        nb, spb = self.num_blocks, self.spb
+        self.obs_store = torch.zeros(nb, self.obs_rows, self.frame_bytes,
+                                     dtype=torch.uint8, device=dev)
+        self.la_store = torch.zeros(nb, self.obs_rows, dtype=torch.uint8, device=dev)
+        self.lr_store = torch.zeros(nb, self.obs_rows, dtype=torch.float32, device=dev)
+        self.act_store = torch.zeros(nb, self.block_len, dtype=torch.uint8, device=dev)
+        self.nsr_store = torch.zeros(nb, self.block_len, dtype=torch.float32, device=dev)
+        self.gam_store = torch.zeros(nb, self.block_len, dtype=torch.float32, device=dev)
+        self.hid_store = torch.zeros(nb * spb, 2 * self.H, dtype=torch.float32,
+                                     device=dev)
+        zi = lambda: torch.zeros(nb * spb, dtype=torch.int32, device=dev)
+        self.burn_s, self.learn_s, self.fwd_s = zi(), zi(), zi()
+        self.obs_start_s, self.learn_off_s = zi(), zi()
+
+        # heap-layout f64 sum-tree
+        self.num_levels = 1
+        while (1 << self.num_levels) < self.num_sequences:
+            self.num_levels += 1
+        self.leaf_offset = (1 << self.num_levels) - 1
+        self.tree = torch.zeros(2 * (1 << self.num_levels) - 1,
+                                dtype=torch.float64, device=dev)
+
+        self.block_ptr = 0
+        self.size = 0
+        self.env_steps = 0
+        self._ext = hip_ops.ext(required=True)
+
+    # ------------------------------------------------------------------
+    def __len__(self):
+        return self.size
+
+    @property
+    def total_priority(self) -> float:
+        return float(self.tree[0].item())
+
+    def ingest(self, block, priorities: np.ndarray):
+        """Copy one actor Block into the device store and set its sequence
+        priorities.  `block` is a worker.Block (numpy); `priorities` has
+        seq_per_block entries, zero-padded for unused slots."""
+        slot = self.block_ptr
+        rows = block.obs.shape[0]
+        steps = block.action.shape[0]
+        nseq = block.num_sequences
+        dev = self.device
+
+        obs_flat = torch.from_numpy(
+            np.ascontiguousarray(block.obs.reshape(rows, -1)))
+        assert obs_flat.dtype == torch.uint8
+        self.obs_store[slot, :rows].copy_(obs_flat, non_blocking=True)
+        la_idx = torch.from_numpy(
+            np.ascontiguousarray(block.last_action.argmax(1).astype(np.uint8)))
+        self.la_store[slot, :rows].copy_(la_idx, non_blocking=True)
+        self.lr_store[slot, :rows].copy_(
+            torch.from_numpy(block.last_reward), non_blocking=True)
+        self.act_store[slot, :steps].copy_(
+            torch.from_numpy(block.action), non_blocking=True)
+        self.nsr_store[slot, :steps].copy_(
+            torch.from_numpy(block.n_step_reward), non_blocking=True)
+        self.gam_store[slot, :steps].copy_(
+            torch.from_numpy(block.gamma), non_blocking=True)
+        base = slot * self.spb
+        self.hid_store[base: base + nseq].copy_(
+            torch.from_numpy(block.hidden.reshape(nseq, -1)), non_blocking=True)
+
+        # per-sequence metadata
+        burn = block.burn_in_steps.astype(np.int32)
+        learn = block.learning_steps.astype(np.int32)
+        fwd = block.forward_steps.astype(np.int32)
+        learn_off = np.zeros(nseq, dtype=np.int32)
+        learn_off[1:] = np.cumsum(learn[:-1])
+        obs_start = (slot * self.obs_rows + int(burn[0]) + learn_off).astype(np.int32)
+        meta = np.zeros((5, self.spb), dtype=np.int32)
+        meta[0, :nseq] = burn
+        meta[1, :nseq] = learn
+        meta[2, :nseq] = fwd
+        meta[3, :nseq] = obs_start
+        meta[4, :nseq] = learn_off
+        mt = torch.from_numpy(meta).to(dev, non_blocking=True)
+        sl = slice(base, base + self.spb)
+        self.burn_s[sl] = mt[0]
+        self.learn_s[sl] = mt[1]
+        self.fwd_s[sl] = mt[2]
+        self.obs_start_s[sl] = mt[3]
+        self.learn_off_s[sl] = mt[4]
+
+        # priorities (zero for unused seq slots kills them in the tree)
+        idxes = torch.arange(base, base + self.spb, dtype=torch.int64, device=dev)
+        prio = torch.from_numpy(
+            np.ascontiguousarray(priorities.astype(np.float32))).to(dev, non_blocking=True)
+        self._ext.sumtree_update(self.tree, self.leaf_offset, idxes, prio,
+                                 self.alpha, 0, 0, self.spb, self.num_blocks)
+
+        self.size += int(learn.sum())
+        if self.size > self.capacity:
+            self.size = min(self.size, self.capacity)
+        self.env_steps += int(learn.sum())
+        self.block_ptr = (self.block_ptr + 1) % self.num_blocks
+
+    # ------------------------------------------------------------------
+    def sample(self, batch_size: Optional[int] = None):
+        """Returns a device-resident TrainingBatch; one small D2H copy for the
+        per-sample lengths (overlap by calling from a prefetch thread/stream)."""
+        from ..worker import TrainingBatch
+
+        B = batch_size or self.batch_size
+        dev = self.device
+        jitter = torch.rand(B, device=dev)
+        idx, prio, weight = self._ext.sumtree_sample(
+            self.tree, self.leaf_offset, self.num_levels, jitter, B, self.beta)
+        meta, seg = self._ext.replay_gather_meta(
+            idx, self.burn_s, self.learn_s, self.fwd_s, self.obs_start_s,
+            self.learn_off_s, self.spb)
+        outs = self._ext.replay_gather_batch(
+            self.obs_store, self.la_store, self.lr_store, self.act_store,
+            self.nsr_store, self.gam_store, self.hid_store, idx, meta, seg,
+            weight, self.T, self.A, self.learn_len, self.H, self.spb)
+        obs, la, lr, act, nsr, gam, w_rep, hid = outs
+
+        meta_h = meta[:3].cpu()  # sync point (small)
+        seg_h = seg.cpu()
+        R = int(seg_h[-1])
+        batch = TrainingBatch(
+            obs=obs.view(B, self.T, *self.obs_shape),
+            last_action=la, last_reward=lr, hidden=hid,
+            action=act[:R].unsqueeze(1), n_step_reward=nsr[:R], gamma=gam[:R],
+            burn_in_steps=meta_h[0].long(), learning_steps=meta_h[1].long(),
+            forward_steps=meta_h[2].long(),
+            idxes=idx, is_weights=w_rep[:R],
+            old_ptr=self.block_ptr, env_steps=self.env_steps)
+        return batch
+
+    # ------------------------------------------------------------------
+    def update_priorities(self, idxes: torch.Tensor, priorities: torch.Tensor,
+                          old_ptr: int):
+        """idxes/priorities stay on device; ring-stale mask applied in-kernel."""
+        self._ext.sumtree_update(self.tree, self.leaf_offset,
+                                 idxes.to(torch.int64),
+                                 priorities.float(), self.alpha,
+                                 old_ptr, self.block_ptr, self.spb,
+                                 self.num_blocks)

@@ -546,8 +546,9 @@ class Learner:
 
     # -- the update step ----------------------------------------------------
 
-    def train_step(self, batch: TrainingBatch) -> Tuple[float, np.ndarray]:
-        """One update.  Returns (loss, per-sequence priorities)."""
+    def train_step(self, batch: TrainingBatch):
+        """One update.  Returns (loss tensor, per-sequence priorities — a
+        device tensor on the HIP path, numpy on the eager path)."""
         c = self.cfg
         batch.to(self.device)
         h0 = (batch.hidden[:1], batch.hidden[1:])
@@ -593,12 +594,12 @@ class Learner:
         self.num_updates += 1
 
         if self.hip_engine:
-            priorities = prio_dev.cpu().numpy()
+            priorities = prio_dev  # device tensor; stays on GPU for gpu_replay
         else:
             td = (target_q - batch_q).detach().abs().cpu().numpy()
             priorities = calculate_mixed_td_errors(
                 td, batch.learning_steps.numpy(), c.prio_eta)
-        return float(loss.item()), priorities
+        return loss.detach(), priorities
 
     def run(self):
         threading.Thread(target=self._prefetch_loop, daemon=True).start()
@@ -609,6 +610,9 @@ class Learner:
                 time.sleep(0.05)
             batch = self.batched_data.pop(0)
             loss, priorities = self.train_step(batch)
+            loss = float(loss)
+            if torch.is_tensor(priorities):
+                priorities = priorities.cpu().numpy()
             self.priority_queue.put((batch.idxes, priorities, batch.old_ptr, loss))
             self.env_steps = batch.env_steps
             if self.num_updates % 4 == 0:

@@ -51,6 +51,7 @@ def test_full_loop_single_process():
     for _ in range(c.training_steps):
         batch = rb.sample_batch()
         loss, priorities = learner.train_step(batch)
+        loss = float(loss)
         assert np.isfinite(loss)
         assert priorities.shape == (c.batch_size,)
         assert np.isfinite(priorities).all()
@@ -92,6 +93,6 @@ def test_learning_reduces_loss_on_fixed_batch():
     for i in range(30):
         loss, _ = learner.train_step(batch)
         if i == 0:
-            first = loss
-        last = loss
+            first = float(loss)
+        last = float(loss)
     assert last < first
