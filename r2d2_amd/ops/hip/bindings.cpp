@@ -31,6 +31,30 @@ std::vector<torch::Tensor> replay_gather_batch(
     torch::Tensor seg, torch::Tensor weights, int64_t T, int64_t A,
     int64_t max_learn, int64_t H, int64_t spb);
 
+// gemm_kernels.hip
+torch::Tensor gemm_bias_act(torch::Tensor A, torch::Tensor Wt,
+                            torch::Tensor bias, int64_t act, bool out_f32);
+torch::Tensor gemm_dgrad(torch::Tensor dY, torch::Tensor act_out,
+                         torch::Tensor W, bool relu_mask);
+std::vector<torch::Tensor> gemm_wgrad(torch::Tensor dY, torch::Tensor act_out,
+                                      torch::Tensor A, bool relu_mask,
+                                      bool want_bias);
+
+// conv_kernels.hip
+torch::Tensor conv_fwd(torch::Tensor in, torch::Tensor Wt, torch::Tensor bias,
+                       int64_t conv_id, int64_t N, int64_t INH, int64_t INW,
+                       int64_t OH, int64_t OW, bool relu);
+torch::Tensor conv_dgrad(torch::Tensor dYp, torch::Tensor Wd, torch::Tensor taps,
+                         int64_t N, int64_t PH, int64_t PW, int64_t COUT,
+                         int64_t XH, int64_t XW, int64_t CIN,
+                         int64_t y0, int64_t x0, int64_t S, int64_t pad,
+                         torch::Tensor dX);
+std::vector<torch::Tensor> conv_wgrad(torch::Tensor dY, torch::Tensor act,
+                                      torch::Tensor in, int64_t conv_id,
+                                      int64_t N, int64_t INH, int64_t INW,
+                                      int64_t OH, int64_t OW, int64_t COUT,
+                                      int64_t K);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "r2d2_amd gfx950 HIP kernels";
     m.def("fused_double_q_loss", &fused_double_q_loss,
@@ -46,4 +70,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "Sampled-sequence metadata + segment offsets");
     m.def("replay_gather_batch", &replay_gather_batch,
           "On-device padded batch assembly from the GPU block store");
+    m.def("gemm_bias_act", &gemm_bias_act, "MFMA GEMM + bias + activation");
+    m.def("gemm_dgrad", &gemm_dgrad, "MFMA GEMM backward-data (+ fused ReLU mask)");
+    m.def("gemm_wgrad", &gemm_wgrad, "MFMA GEMM backward-weight (+ bias grad)");
+    m.def("conv_fwd", &conv_fwd, "Implicit-GEMM MFMA conv forward (NHWC)");
+    m.def("conv_dgrad", &conv_dgrad, "MFMA conv backward-data (tap classes)");
+    m.def("conv_wgrad", &conv_wgrad, "MFMA conv backward-weight");
 }

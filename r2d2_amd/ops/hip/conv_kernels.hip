@@ -1,0 +1,451 @@
+// Implicit-GEMM MFMA convolution kernels for the Nature-CNN encoder (gfx950).
+//
+// Replaces MIOpen's conv paths, which on this workload fall back to naive
+// f64-accumulation kernels + 261k per-image im2col launches (see
+// profiles/r01_eager_step_kernel_stats.csv: 65% of all GPU time).
+//
+// Layout: NHWC, bf16 activations (uint8 input for conv1 with fused /255
+// dequant).  K-order is (ky, kx, c), so a lane's 8 consecutive k-elements
+// are 8 CONTIGUOUS input bytes/bf16 (x-then-c is memory-contiguous in NHWC
+// and KW*CIN % 8 == 0 for all three convs) — A fragments load straight from
+// global, no im2col materialization.  Weights are prepacked (COUT, K) for
+// forward / wgrad and (CIN, TAPS*COUT) per tap-class for dgrad.
+//
+// Forward:  out(M=N*OH*OW, COUT) = act(patch(M,K) @ Wt^T + bias)
+// Dgrad:    dX from a zero-padded dY via uniform tap tables (stride-2 convs
+//           split into parity classes so every row in a tile has the same
+//           taps; no divergence).
+// Wgrad:    dWt(COUT,K) += dY^T @ patch with LDS-staged 32-row tiles and
+//           f32 atomics; fused ReLU mask + bias grad.
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+union cbf8u {
+    bf16x8 v;
+    uint4 u;
+    __bf16 e[8];
+    unsigned short s[8];
+};
+
+__device__ __forceinline__ bf16x8 cload_bf16x8(const __hip_bfloat16* p) {
+    cbf8u r;
+    r.u = *reinterpret_cast<const uint4*>(p);
+    return r.v;
+}
+
+__device__ __forceinline__ bf16x8 czero() {
+    cbf8u r;
+    r.u = uint4{0, 0, 0, 0};
+    return r.v;
+}
+
+// load 8 uint8 bytes and dequantize to bf16/255
+__device__ __forceinline__ bf16x8 load_dequant8(const unsigned char* p) {
+    uint2 raw = *reinterpret_cast<const uint2*>(p);
+    cbf8u r;
+    const float inv = 1.f / 255.f;
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+        r.e[i] = (__bf16)(((raw.x >> (8 * i)) & 0xff) * inv);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+        r.e[4 + i] = (__bf16)(((raw.y >> (8 * i)) & 0xff) * inv);
+    return r.v;
+}
+
+// ---------------------------------------------------------------------------
+// Forward
+// ---------------------------------------------------------------------------
+template <bool IN_U8, int KH, int KW, int CIN, int S, bool RELU>
+__global__ __launch_bounds__(256) void conv_fwd_kernel(
+    const void* __restrict__ in,            // (N, INH, INW, CIN)
+    const __hip_bfloat16* __restrict__ Wt,  // (COUT, K)
+    const float* __restrict__ bias,         // (COUT,)
+    __hip_bfloat16* __restrict__ out,       // (M, COUT)
+    int M, int INH, int INW, int OH, int OW, int COUT) {
+    constexpr int K = KH * KW * CIN;
+    constexpr int KWC = KW * CIN;
+    int wave = threadIdx.x / WAVE;
+    int lane = threadIdx.x & (WAVE - 1);
+    int wr = wave >> 1, wc = wave & 1;
+    long row0 = (long)blockIdx.x * 64 + wr * 32;
+    long col0 = (long)blockIdx.y * 64 + wc * 32;
+    int frow = lane & 15;
+    int kseg = (lane >> 4) * 8;
+
+    // per-lane patch base addresses for its two A rows
+    long abase[2];
+    bool avalid[2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+        long r = row0 + i * 16 + frow;
+        avalid[i] = r < M;
+        if (avalid[i]) {
+            long n = r / (OH * OW);
+            int p = (int)(r % (OH * OW));
+            int oy = p / OW, ox = p % OW;
+            abase[i] = ((n * INH + (long)oy * S) * INW + (long)ox * S) * CIN;
+        } else {
+            abase[i] = 0;
+        }
+    }
+
+    f32x4 acc[2][2] = {};
+    for (int k0 = 0; k0 < K; k0 += 32) {
+        int k = k0 + kseg;
+        int dy = k / KWC;
+        int rem = k % KWC;
+        long off = (long)dy * INW * CIN + rem;
+        bf16x8 a[2], b[2];
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            if (avalid[i]) {
+                if (IN_U8)
+                    a[i] = load_dequant8(
+                        reinterpret_cast<const unsigned char*>(in) + abase[i] + off);
+                else
+                    a[i] = cload_bf16x8(
+                        reinterpret_cast<const __hip_bfloat16*>(in) + abase[i] + off);
+            } else {
+                a[i] = czero();
+            }
+            long c = col0 + i * 16 + frow;
+            b[i] = (c < COUT) ? cload_bf16x8(Wt + c * K + k0 + kseg) : czero();
+        }
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+            for (int j = 0; j < 2; ++j)
+                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+
+    int ccol = lane & 15;
+    int crow = (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                long rr = row0 + i * 16 + crow + r;
+                long cc = col0 + j * 16 + ccol;
+                if (rr < M && cc < COUT) {
+                    float v = acc[i][j][r] + bias[cc];
+                    if (RELU) v = fmaxf(v, 0.f);
+                    out[rr * COUT + cc] = f2bf(v);
+                }
+            }
+}
+
+// ---------------------------------------------------------------------------
+// Dgrad: dX[n, y, x, ci] = sum_t sum_co dYp[n, (y-dy_t)/S + pad, ...] *
+//                          Wd[ci][t*COUT + co]
+// Launched per tap-class; rows enumerate the class's (n, yy, xx) grid with
+// y = y0 + yy*S.  dYp is the zero-PADDED upstream gradient (pre-masked by
+// the ReLU of this conv's output).
+// ---------------------------------------------------------------------------
+template <int TAPS>
+__global__ __launch_bounds__(256) void conv_dgrad_kernel(
+    const __hip_bfloat16* __restrict__ dYp,  // (N, PH, PW, COUT)
+    const __hip_bfloat16* __restrict__ Wd,   // (CIN, TAPS*COUT)
+    __hip_bfloat16* __restrict__ dX,         // (N, XH, XW, CIN)
+    const int* __restrict__ taps,            // (TAPS, 2): dy, dx
+    int Mc, int YY, int XX, int y0, int x0, int S, int pad,
+    int PH, int PW, int COUT, int XH, int XW, int CIN) {
+    const int K = TAPS * COUT;
+    int wave = threadIdx.x / WAVE;
+    int lane = threadIdx.x & (WAVE - 1);
+    int wr = wave >> 1, wc = wave & 1;
+    long row0 = (long)blockIdx.x * 64 + wr * 32;
+    long col0 = (long)blockIdx.y * 64 + wc * 32;
+    int frow = lane & 15;
+    int kseg = (lane >> 4) * 8;
+
+    // per-lane (n, y, x) for its two rows
+    long nbase[2];
+    int yv[2], xv[2];
+    bool avalid[2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+        long r = row0 + i * 16 + frow;
+        avalid[i] = r < Mc;
+        long n = avalid[i] ? r / (YY * XX) : 0;
+        int p = avalid[i] ? (int)(r % (YY * XX)) : 0;
+        yv[i] = y0 + (p / XX) * S;
+        xv[i] = x0 + (p % XX) * S;
+        nbase[i] = n * PH * PW;
+    }
+
+    f32x4 acc[2][2] = {};
+    for (int k0 = 0; k0 < K; k0 += 32) {
+        int k = k0 + kseg;
+        int t = k / COUT;
+        int co = k % COUT;
+        int dy = taps[2 * t], dx = taps[2 * t + 1];
+        bf16x8 a[2], b[2];
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            if (avalid[i]) {
+                int oy = (yv[i] - dy) / S + pad;
+                int ox = (xv[i] - dx) / S + pad;
+                a[i] = cload_bf16x8(dYp + (nbase[i] + (long)oy * PW + ox) * COUT + co);
+            } else {
+                a[i] = czero();
+            }
+            long c = col0 + i * 16 + frow;
+            b[i] = (c < CIN) ? cload_bf16x8(Wd + c * K + k0 + kseg) : czero();
+        }
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+            for (int j = 0; j < 2; ++j)
+                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+
+    int ccol = lane & 15;
+    int crow = (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                long rr = row0 + i * 16 + crow + r;
+                long cc = col0 + j * 16 + ccol;
+                if (rr < Mc && cc < CIN) {
+                    long n = rr / (YY * XX);
+                    int p = (int)(rr % (YY * XX));
+                    int y = y0 + (p / XX) * S, x = x0 + (p % XX) * S;
+                    dX[((n * XH + y) * XW + x) * CIN + cc] = f2bf(acc[i][j][r]);
+                }
+            }
+}
+
+// ---------------------------------------------------------------------------
+// Wgrad: dWt(COUT, K) += sum_rows relu_mask(dY)[row][co] * patch[row][k]
+// LDS-staged 32-row tiles; f32 atomics; db fused.
+// ---------------------------------------------------------------------------
+template <bool IN_U8, int KH, int KW, int CIN, int S, bool RELU>
+__global__ __launch_bounds__(256) void conv_wgrad_kernel(
+    const __hip_bfloat16* __restrict__ dY,   // (M, COUT)
+    const __hip_bfloat16* __restrict__ act,  // (M, COUT) forward output
+    const void* __restrict__ in,             // (N, INH, INW, CIN)
+    float* __restrict__ dWt,                 // (COUT, K) f32
+    float* __restrict__ db,                  // (COUT,) f32
+    int M, int INH, int INW, int OH, int OW, int COUT, int rows_per_chunk) {
+    constexpr int K = KH * KW * CIN;
+    constexpr int KWC = KW * CIN;
+    __shared__ __hip_bfloat16 s_dy[32][64 + 8];
+    __shared__ __hip_bfloat16 s_a[32][64 + 8];
+    int wave = threadIdx.x / WAVE;
+    int lane = threadIdx.x & (WAVE - 1);
+    int wr = wave >> 1, wc = wave & 1;
+    long mstart = (long)blockIdx.x * rows_per_chunk;
+    long mend = min((long)M, mstart + rows_per_chunk);
+    long cocol0 = (long)blockIdx.y * 64;   // co tile
+    long kcol0 = (long)blockIdx.z * 64;    // k tile
+    int frow = lane & 15;
+    int mseg = (lane >> 4) * 8;
+
+    f32x4 acc[2][2] = {};
+    float bias_acc = 0.f;
+
+    for (long m0 = mstart; m0 < mend; m0 += 32) {
+        __syncthreads();
+        {
+            int t = threadIdx.x;
+            int mrow = t / 8;
+            int col = (t % 8) * 8;
+            long gm = m0 + mrow;
+            cbf8u v;
+            v.v = czero();
+            if (gm < mend) {
+#pragma unroll
+                for (int e = 0; e < 8; ++e) {
+                    long c = cocol0 + col + e;
+                    if (c < COUT) {
+                        float g = bf2f(dY[gm * COUT + c]);
+                        if (RELU && !(bf2f(act[gm * COUT + c]) > 0.f)) g = 0.f;
+                        v.e[e] = (__bf16)g;
+                    }
+                }
+            }
+            *reinterpret_cast<bf16x8*>(&s_dy[mrow][col]) = v.v;
+
+            cbf8u w;
+            w.v = czero();
+            if (gm < mend && kcol0 + col < K) {
+                long n = gm / (OH * OW);
+                int p = (int)(gm % (OH * OW));
+                int oy = p / OW, ox = p % OW;
+                long base = ((n * INH + (long)oy * S) * INW + (long)ox * S) * CIN;
+                int k = (int)kcol0 + col;
+                int dy = k / KWC, rem = k % KWC;
+                long off = base + (long)dy * INW * CIN + rem;
+                // 8 contiguous k (never crosses a dy boundary: KWC % 8 == 0)
+                if (IN_U8)
+                    w.v = load_dequant8(
+                        reinterpret_cast<const unsigned char*>(in) + off);
+                else
+                    w.v = cload_bf16x8(
+                        reinterpret_cast<const __hip_bfloat16*>(in) + off);
+                if (kcol0 + col + 8 > K) {
+#pragma unroll
+                    for (int e = 0; e < 8; ++e)
+                        if (kcol0 + col + e >= K) w.e[e] = (__bf16)0.f;
+                }
+            }
+            *reinterpret_cast<bf16x8*>(&s_a[mrow][col]) = w.v;
+        }
+        __syncthreads();
+
+        bf16x8 fa[2], fb[2];
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            cbf8u va, vb;
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                va.e[e] = s_dy[mseg + e][wr * 32 + i * 16 + frow];
+                vb.e[e] = s_a[mseg + e][wc * 32 + i * 16 + frow];
+            }
+            fa[i] = va.v;
+            fb[i] = vb.v;
+        }
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+            for (int j = 0; j < 2; ++j)
+                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    fa[i], fb[j], acc[i][j], 0, 0, 0);
+
+        if (blockIdx.z == 0 && threadIdx.x < 64) {
+            int c = threadIdx.x;
+            for (int mr = 0; mr < 32; ++mr) bias_acc += bf2f(s_dy[mr][c]);
+        }
+    }
+
+    int ccol = lane & 15;
+    int crow = (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                long co = cocol0 + wr * 32 + i * 16 + crow + r;
+                long kk = kcol0 + wc * 32 + j * 16 + ccol;
+                if (co < COUT && kk < K)
+                    atomicAdd(&dWt[co * K + kk], acc[i][j][r]);
+            }
+    if (blockIdx.z == 0 && threadIdx.x < 64) {
+        long c = cocol0 + threadIdx.x;
+        if (c < COUT) atomicAdd(&db[c], bias_acc);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Host wrappers.  Conv geometry is dispatched over the three Nature-CNN
+// layers (and reusable for any conv matching a template instance).
+// ---------------------------------------------------------------------------
+
+static inline int ccdiv(long a, long b) { return (int)((a + b - 1) / b); }
+
+// conv_id: 1 = 8x8 s4 CIN variable(4) u8-in, 2 = 4x4 s2 CIN 32, 3 = 3x3 s1 CIN 64
+torch::Tensor conv_fwd(torch::Tensor in, torch::Tensor Wt, torch::Tensor bias,
+                       int64_t conv_id, int64_t N, int64_t INH, int64_t INW,
+                       int64_t OH, int64_t OW, bool relu) {
+    long M = N * OH * OW;
+    long COUT = Wt.size(0);
+    auto out = torch::empty({M, COUT},
+                            in.options().dtype(torch::kBFloat16));
+    dim3 grid(ccdiv(M, 64), ccdiv(COUT, 64));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    auto* w = reinterpret_cast<const __hip_bfloat16*>(Wt.data_ptr());
+    auto* o = reinterpret_cast<__hip_bfloat16*>(out.data_ptr());
+    const float* b = bias.data_ptr<float>();
+    const void* x = in.data_ptr();
+#define CLAUNCH(U8, KH_, KW_, CIN_, S_)                                        \
+    hipLaunchKernelGGL((conv_fwd_kernel<U8, KH_, KW_, CIN_, S_, true>), grid,  \
+                       dim3(256), 0, stream.stream(), x, w, b, o, (int)M,      \
+                       (int)INH, (int)INW, (int)OH, (int)OW, (int)COUT)
+    if (conv_id == 1) {
+        TORCH_CHECK(in.dtype() == torch::kUInt8);
+        CLAUNCH(true, 8, 8, 4, 4);
+    } else if (conv_id == 2) {
+        TORCH_CHECK(in.dtype() == torch::kBFloat16);
+        CLAUNCH(false, 4, 4, 32, 2);
+    } else if (conv_id == 3) {
+        CLAUNCH(false, 3, 3, 64, 1);
+    } else {
+        TORCH_CHECK(false, "unknown conv_id");
+    }
+#undef CLAUNCH
+    return out;
+}
+
+torch::Tensor conv_dgrad(torch::Tensor dYp, torch::Tensor Wd, torch::Tensor taps,
+                         int64_t N, int64_t PH, int64_t PW, int64_t COUT,
+                         int64_t XH, int64_t XW, int64_t CIN,
+                         int64_t y0, int64_t x0, int64_t S, int64_t pad,
+                         torch::Tensor dX) {
+    long YY = (XH - 1 - y0) / S + 1;
+    long XX = (XW - 1 - x0) / S + 1;
+    long Mc = N * YY * XX;
+    long TAPS = taps.size(0);
+    dim3 grid(ccdiv(Mc, 64), ccdiv(CIN, 64));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    auto* dy = reinterpret_cast<const __hip_bfloat16*>(dYp.data_ptr());
+    auto* w = reinterpret_cast<const __hip_bfloat16*>(Wd.data_ptr());
+    auto* dx = reinterpret_cast<__hip_bfloat16*>(dX.data_ptr());
+    const int* tp = taps.data_ptr<int>();
+#define DLAUNCH(T)                                                             \
+    hipLaunchKernelGGL((conv_dgrad_kernel<T>), grid, dim3(256), 0,             \
+                       stream.stream(), dy, w, dx, tp, (int)Mc, (int)YY,       \
+                       (int)XX, (int)y0, (int)x0, (int)S, (int)pad, (int)PH,   \
+                       (int)PW, (int)COUT, (int)XH, (int)XW, (int)CIN)
+    if (TAPS == 4) DLAUNCH(4);
+    else if (TAPS == 9) DLAUNCH(9);
+    else TORCH_CHECK(false, "unsupported tap count");
+#undef DLAUNCH
+    return dX;
+}
+
+std::vector<torch::Tensor> conv_wgrad(torch::Tensor dY, torch::Tensor act,
+                                      torch::Tensor in, int64_t conv_id,
+                                      int64_t N, int64_t INH, int64_t INW,
+                                      int64_t OH, int64_t OW, int64_t COUT,
+                                      int64_t K) {
+    long M = N * OH * OW;
+    auto dWt = torch::zeros({COUT, K}, dY.options().dtype(torch::kFloat32));
+    auto db = torch::zeros({COUT}, dY.options().dtype(torch::kFloat32));
+    long tiles = (long)ccdiv(COUT, 64) * ccdiv(K, 64);
+    long target_chunks = std::max(1L, 2048L / std::max(1L, tiles));
+    long rows_per_chunk = std::max(32L, (M + target_chunks - 1) / target_chunks);
+    rows_per_chunk = ((rows_per_chunk + 31) / 32) * 32;
+    dim3 grid(ccdiv(M, rows_per_chunk), ccdiv(COUT, 64), ccdiv(K, 64));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    auto* dy = reinterpret_cast<const __hip_bfloat16*>(dY.data_ptr());
+    auto* ac = reinterpret_cast<const __hip_bfloat16*>(act.data_ptr());
+    const void* x = in.data_ptr();
+#define WLAUNCH(U8, KH_, KW_, CIN_, S_)                                        \
+    hipLaunchKernelGGL((conv_wgrad_kernel<U8, KH_, KW_, CIN_, S_, true>),      \
+                       grid, dim3(256), 0, stream.stream(), dy, ac, x,         \
+                       dWt.data_ptr<float>(), db.data_ptr<float>(), (int)M,    \
+                       (int)INH, (int)INW, (int)OH, (int)OW, (int)COUT,        \
+                       (int)rows_per_chunk)
+    if (conv_id == 1) WLAUNCH(true, 8, 8, 4, 4);
+    else if (conv_id == 2) WLAUNCH(false, 4, 4, 32, 2);
+    else if (conv_id == 3) WLAUNCH(false, 3, 3, 64, 1);
+    else TORCH_CHECK(false, "unknown conv_id");
+#undef WLAUNCH
+    return {dWt, db};
+}

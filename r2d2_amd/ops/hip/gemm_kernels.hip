@@ -1,0 +1,397 @@
+// MFMA GEMM kernels for the R2D2 network's dense layers (gfx950).
+//
+// C(M,N) = A(M,K) @ W^T + bias, optional ReLU.  Weights are prepacked
+// row-major as Wt(N,K) so both A and B fragments load 8 CONTIGUOUS bf16
+// (16 B) per lane — the mfma_f32_16x16x32_bf16 operand layout puts 8
+// consecutive k-elements in each lane (A: row = lane&15; B: col = lane&15;
+// k = (lane>>4)*8 + j).  Replaces hipBLASLt/eager Linear for the encoder FC
+// (3136->512), the dueling heads, and the LSTM gate GEMMs.
+//
+// Structure: 4-wave workgroups, 64x64 block tile, each wave a 32x32 tile
+// (2x2 fragments, A/B frags shared across the wave's row/col pair), K-major
+// loop.  Weights are L2-resident (<= 3.2 MB); correctness-first, tuned via
+// rocprof (see profiles/).
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+union bf8u {
+    bf16x8 v;
+    uint4 u;
+    __bf16 e[8];
+};
+
+__device__ __forceinline__ bf16x8 load_bf16x8(const __hip_bfloat16* p) {
+    bf8u r;
+    r.u = *reinterpret_cast<const uint4*>(p);
+    return r.v;
+}
+
+__device__ __forceinline__ bf16x8 zero_bf16x8() {
+    bf8u r;
+    r.u = uint4{0, 0, 0, 0};
+    return r.v;
+}
+
+// ---------------------------------------------------------------------------
+// gemm_bias_act: out(M,N) = act(A(M,K) @ Wt(N,K)^T + bias)
+//   OUT_F32: write f32 (head outputs feeding the loss kernel) else bf16.
+//   ACT: 0 none, 1 relu.
+// ---------------------------------------------------------------------------
+template <int ACT, bool HAS_BIAS, bool OUT_F32>
+__global__ __launch_bounds__(256) void gemm_bias_act_kernel(
+    const __hip_bfloat16* __restrict__ A,   // (M, K)
+    const __hip_bfloat16* __restrict__ Wt,  // (N, K)
+    const float* __restrict__ bias,         // (N,)
+    void* __restrict__ out,                 // (M, N) bf16 or f32
+    int M, int N, int K) {
+    // block tile 64x64: wave w covers rows [wr*32, +32), cols [wc*32, +32)
+    int wave = threadIdx.x / WAVE;
+    int lane = threadIdx.x & (WAVE - 1);
+    int wr = wave >> 1, wc = wave & 1;
+    long row0 = (long)blockIdx.x * 64 + wr * 32;
+    long col0 = (long)blockIdx.y * 64 + wc * 32;
+
+    int frow = lane & 15;            // fragment row (A) / col (B)
+    int kseg = (lane >> 4) * 8;      // k offset of this lane's 8 elements
+
+    f32x4 acc[2][2] = {};
+    for (int k0 = 0; k0 < K; k0 += 32) {
+        bf16x8 a[2], b[2];
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            long r = row0 + i * 16 + frow;
+            a[i] = (r < M) ? load_bf16x8(A + r * K + k0 + kseg) : zero_bf16x8();
+            long c = col0 + i * 16 + frow;
+            b[i] = (c < N) ? load_bf16x8(Wt + c * K + k0 + kseg) : zero_bf16x8();
+        }
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+            for (int j = 0; j < 2; ++j)
+                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+
+    // C/D layout: col = lane&15, row = (lane>>4)*4 + r
+    int ccol = lane & 15;
+    int crow = (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                long rr = row0 + i * 16 + crow + r;
+                long cc = col0 + j * 16 + ccol;
+                if (rr < M && cc < N) {
+                    float v = acc[i][j][r];
+                    if (HAS_BIAS) v += bias[cc];
+                    if (ACT == 1) v = fmaxf(v, 0.f);
+                    if (OUT_F32)
+                        reinterpret_cast<float*>(out)[rr * N + cc] = v;
+                    else
+                        reinterpret_cast<__hip_bfloat16*>(out)[rr * N + cc] =
+                            f2bf(v);
+                }
+            }
+}
+
+// ---------------------------------------------------------------------------
+// gemm_dgrad: dA(M,K) = dY(M,N) @ W(K,N)^T-with-W-stored-(K,N)... i.e.
+//   dA[m][k] = sum_n dY[m][n] * W[k][n], W prepacked row-major (K, N).
+//   RELU_MASK: multiply dY by (act_out > 0) on load (fused ReLU backward,
+//   act_out is the forward output of this layer, same shape as dY).
+// ---------------------------------------------------------------------------
+template <bool RELU_MASK>
+__global__ __launch_bounds__(256) void gemm_dgrad_kernel(
+    const __hip_bfloat16* __restrict__ dY,      // (M, N)
+    const __hip_bfloat16* __restrict__ act,     // (M, N) or null
+    const __hip_bfloat16* __restrict__ W,       // (K, N) row-major
+    __hip_bfloat16* __restrict__ dA,            // (M, K)
+    int M, int N, int K) {
+    int wave = threadIdx.x / WAVE;
+    int lane = threadIdx.x & (WAVE - 1);
+    int wr = wave >> 1, wc = wave & 1;
+    long row0 = (long)blockIdx.x * 64 + wr * 32;   // m
+    long col0 = (long)blockIdx.y * 64 + wc * 32;   // k (output feature dim)
+    int frow = lane & 15;
+    int nseg = (lane >> 4) * 8;
+
+    f32x4 acc[2][2] = {};
+    for (int n0 = 0; n0 < N; n0 += 32) {
+        bf16x8 a[2], b[2];
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            long r = row0 + i * 16 + frow;
+            if (r < M) {
+                a[i] = load_bf16x8(dY + r * N + n0 + nseg);
+                if (RELU_MASK) {
+                    bf8u av, mv;
+                    av.v = a[i];
+                    mv.v = load_bf16x8(act + r * N + n0 + nseg);
+#pragma unroll
+                    for (int e = 0; e < 8; ++e)
+                        if (!((float)mv.e[e] > 0.f)) av.e[e] = (__bf16)0.f;
+                    a[i] = av.v;
+                }
+            } else {
+                a[i] = zero_bf16x8();
+            }
+            long c = col0 + i * 16 + frow;
+            b[i] = (c < K) ? load_bf16x8(W + c * N + n0 + nseg) : zero_bf16x8();
+        }
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+            for (int j = 0; j < 2; ++j)
+                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+    int ccol = lane & 15;
+    int crow = (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                long rr = row0 + i * 16 + crow + r;
+                long cc = col0 + j * 16 + ccol;
+                if (rr < M && cc < K)
+                    dA[rr * K + cc] = f2bf(acc[i][j][r]);
+            }
+}
+
+// ---------------------------------------------------------------------------
+// gemm_wgrad: dWt(N,K) += dY(M,N)^T @ A(M,K), reduction over M with
+// LDS-staged 32-row tiles; row-chunked grid with f32 atomics into dW.
+// Also accumulates db(N) = sum_m dY[m][n] when HAS_BIAS.
+//   RELU_MASK as in dgrad (applied to dY).
+// ---------------------------------------------------------------------------
+template <bool RELU_MASK, bool HAS_BIAS>
+__global__ __launch_bounds__(256) void gemm_wgrad_kernel(
+    const __hip_bfloat16* __restrict__ dY,   // (M, N)
+    const __hip_bfloat16* __restrict__ act,  // (M, N) or null
+    const __hip_bfloat16* __restrict__ A,    // (M, K)
+    float* __restrict__ dWt,                 // (N, K) f32 accumulate
+    float* __restrict__ db,                  // (N,) f32 accumulate
+    int M, int N, int K, int rows_per_chunk) {
+    // grid.x: row chunks; grid.y: N tiles of 64; grid.z: K tiles of 64
+    __shared__ __hip_bfloat16 s_dy[32][64 + 4];  // [m][n]
+    __shared__ __hip_bfloat16 s_a[32][64 + 4];   // [m][k]
+    int wave = threadIdx.x / WAVE;
+    int lane = threadIdx.x & (WAVE - 1);
+    int wr = wave >> 1, wc = wave & 1;
+    long mstart = (long)blockIdx.x * rows_per_chunk;
+    long mend = min((long)M, mstart + rows_per_chunk);
+    long ncol0 = (long)blockIdx.y * 64;
+    long kcol0 = (long)blockIdx.z * 64;
+
+    int frow = lane & 15;
+    int mseg = (lane >> 4) * 8;
+
+    f32x4 acc[2][2] = {};
+    float bias_acc = 0.f;   // lane-partial db for column ncol0 + (tid%64)
+
+    for (long m0 = mstart; m0 < mend; m0 += 32) {
+        // stage 32 rows x 64 cols of dY and A into LDS (coalesced)
+        __syncthreads();
+        // 256 threads, each loads 8 elements: 32*64/8 = 256 loads per tile
+        {
+            int t = threadIdx.x;
+            int mrow = t / 8;           // 0..31
+            int ncol = (t % 8) * 8;     // 0..56
+            long gm = m0 + mrow;
+            bf8u v;
+            if (gm < mend) {
+                if (ncol0 + ncol + 8 <= N)
+                    v.v = load_bf16x8(dY + gm * N + ncol0 + ncol);
+                else {
+#pragma unroll
+                    for (int e = 0; e < 8; ++e) {
+                        long c = ncol0 + ncol + e;
+                        v.e[e] = (c < N) ? ((const __bf16*)dY)[gm * N + c]
+                                         : (__bf16)0.f;
+                    }
+                }
+                if (RELU_MASK) {
+#pragma unroll
+                    for (int e = 0; e < 8; ++e) {
+                        long c = ncol0 + ncol + e;
+                        float m_ = (c < N) ? (float)((const __bf16*)act)[gm * N + c]
+                                           : 0.f;
+                        if (!(m_ > 0.f)) v.e[e] = (__bf16)0.f;
+                    }
+                }
+            } else {
+                v.v = zero_bf16x8();
+            }
+            *reinterpret_cast<bf16x8*>(&s_dy[mrow][ncol]) = v.v;
+            bf8u w;
+            if (gm < mend) {
+                if (kcol0 + ncol + 8 <= K)
+                    w.v = load_bf16x8(A + gm * K + kcol0 + ncol);
+                else {
+#pragma unroll
+                    for (int e = 0; e < 8; ++e) {
+                        long c = kcol0 + ncol + e;
+                        w.e[e] = (c < K) ? ((const __bf16*)A)[gm * K + c]
+                                         : (__bf16)0.f;
+                    }
+                }
+            } else {
+                w.v = zero_bf16x8();
+            }
+            *reinterpret_cast<bf16x8*>(&s_a[mrow][ncol]) = w.v;
+        }
+        __syncthreads();
+
+        // MFMA: out tile (n, k); gemm-M = n-dim, gemm-N = k-dim, gemm-K = m
+        // A-frag: s_dy column (n = row0 + frow), 8 m values
+        // B-frag: s_a column (k), 8 m values — both are LDS column reads
+#pragma unroll
+        for (int ks = 0; ks < 32; ks += 32) {
+            bf16x8 fa[2], fb[2];
+#pragma unroll
+            for (int i = 0; i < 2; ++i) {
+                bf8u va, vb;
+#pragma unroll
+                for (int e = 0; e < 8; ++e) {
+                    va.e[e] = s_dy[mseg + e][wr * 32 + i * 16 + frow];
+                    vb.e[e] = s_a[mseg + e][wc * 32 + i * 16 + frow];
+                }
+                fa[i] = va.v;
+                fb[i] = vb.v;
+            }
+#pragma unroll
+            for (int i = 0; i < 2; ++i)
+#pragma unroll
+                for (int j = 0; j < 2; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        fa[i], fb[j], acc[i][j], 0, 0, 0);
+        }
+        if (HAS_BIAS) {
+            // db: threads 0..63 own column tid%64; sum the 32 rows staged
+            int c = threadIdx.x % 64;
+            if (threadIdx.x < 64) {
+                for (int mr = 0; mr < 32; ++mr)
+                    bias_acc += (float)s_dy[mr][c];
+            }
+        }
+    }
+
+    int ccol = lane & 15;
+    int crow = (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                long nn = ncol0 + wr * 32 + i * 16 + crow + r;
+                long kk = kcol0 + wc * 32 + j * 16 + ccol;
+                if (nn < N && kk < K)
+                    atomicAdd(&dWt[nn * K + kk], acc[i][j][r]);
+            }
+    if (HAS_BIAS && threadIdx.x < 64 && blockIdx.z == 0) {
+        long c = ncol0 + threadIdx.x;
+        if (c < N) atomicAdd(&db[c], bias_acc);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Host wrappers
+// ---------------------------------------------------------------------------
+
+static inline int cdiv(long a, long b) { return (int)((a + b - 1) / b); }
+
+torch::Tensor gemm_bias_act(torch::Tensor A, torch::Tensor Wt,
+                            torch::Tensor bias, int64_t act, bool out_f32) {
+    TORCH_CHECK(A.is_cuda() && A.dtype() == torch::kBFloat16 && A.is_contiguous());
+    TORCH_CHECK(Wt.dtype() == torch::kBFloat16 && Wt.is_contiguous());
+    long M = A.size(0), K = A.size(1), N = Wt.size(0);
+    TORCH_CHECK(Wt.size(1) == K && K % 32 == 0, "K must be a multiple of 32");
+    bool has_bias = bias.defined() && bias.numel() > 0;
+    auto out = torch::empty({M, N}, A.options().dtype(
+        out_f32 ? torch::kFloat32 : torch::kBFloat16));
+    dim3 grid(cdiv(M, 64), cdiv(N, 64));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    const float* bptr = has_bias ? bias.data_ptr<float>() : nullptr;
+    auto* a = reinterpret_cast<const __hip_bfloat16*>(A.data_ptr());
+    auto* w = reinterpret_cast<const __hip_bfloat16*>(Wt.data_ptr());
+
+#define LAUNCH(ACT, HB, OF)                                                    \
+    hipLaunchKernelGGL((gemm_bias_act_kernel<ACT, HB, OF>), grid, dim3(256),   \
+                       0, stream.stream(), a, w, bptr, out.data_ptr(),         \
+                       (int)M, (int)N, (int)K)
+    if (has_bias) {
+        if (act == 1) { if (out_f32) LAUNCH(1, true, true); else LAUNCH(1, true, false); }
+        else          { if (out_f32) LAUNCH(0, true, true); else LAUNCH(0, true, false); }
+    } else {
+        if (act == 1) { if (out_f32) LAUNCH(1, false, true); else LAUNCH(1, false, false); }
+        else          { if (out_f32) LAUNCH(0, false, true); else LAUNCH(0, false, false); }
+    }
+#undef LAUNCH
+    return out;
+}
+
+torch::Tensor gemm_dgrad(torch::Tensor dY, torch::Tensor act_out,
+                         torch::Tensor W, bool relu_mask) {
+    TORCH_CHECK(dY.is_cuda() && dY.dtype() == torch::kBFloat16 && dY.is_contiguous());
+    TORCH_CHECK(W.dtype() == torch::kBFloat16 && W.is_contiguous());
+    long M = dY.size(0), N = dY.size(1), K = W.size(0);
+    TORCH_CHECK(W.size(1) == N && N % 32 == 0, "N must be a multiple of 32");
+    auto dA = torch::empty({M, K}, dY.options());
+    dim3 grid(cdiv(M, 64), cdiv(K, 64));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    auto* dy = reinterpret_cast<const __hip_bfloat16*>(dY.data_ptr());
+    auto* ac = relu_mask
+        ? reinterpret_cast<const __hip_bfloat16*>(act_out.data_ptr()) : nullptr;
+    auto* w = reinterpret_cast<const __hip_bfloat16*>(W.data_ptr());
+    auto* da = reinterpret_cast<__hip_bfloat16*>(dA.data_ptr());
+    if (relu_mask)
+        hipLaunchKernelGGL((gemm_dgrad_kernel<true>), grid, dim3(256), 0,
+                           stream.stream(), dy, ac, w, da, (int)M, (int)N, (int)K);
+    else
+        hipLaunchKernelGGL((gemm_dgrad_kernel<false>), grid, dim3(256), 0,
+                           stream.stream(), dy, ac, w, da, (int)M, (int)N, (int)K);
+    return dA;
+}
+
+std::vector<torch::Tensor> gemm_wgrad(torch::Tensor dY, torch::Tensor act_out,
+                                      torch::Tensor A, bool relu_mask,
+                                      bool want_bias) {
+    TORCH_CHECK(dY.is_cuda() && dY.dtype() == torch::kBFloat16 && dY.is_contiguous());
+    TORCH_CHECK(A.dtype() == torch::kBFloat16 && A.is_contiguous());
+    long M = dY.size(0), N = dY.size(1), K = A.size(1);
+    auto dWt = torch::zeros({N, K}, dY.options().dtype(torch::kFloat32));
+    auto db = torch::zeros({want_bias ? N : 1},
+                           dY.options().dtype(torch::kFloat32));
+    // chunk rows so ~1024 blocks exist for the z=0 plane
+    long tiles = (long)cdiv(N, 64) * cdiv(K, 64);
+    long target_chunks = std::max(1L, 1024L / std::max(1L, tiles));
+    long rows_per_chunk = std::max(32L, (M + target_chunks - 1) / target_chunks);
+    rows_per_chunk = ((rows_per_chunk + 31) / 32) * 32;
+    dim3 grid(cdiv(M, rows_per_chunk), cdiv(N, 64), cdiv(K, 64));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    auto* dy = reinterpret_cast<const __hip_bfloat16*>(dY.data_ptr());
+    auto* ac = relu_mask
+        ? reinterpret_cast<const __hip_bfloat16*>(act_out.data_ptr()) : nullptr;
+    auto* a = reinterpret_cast<const __hip_bfloat16*>(A.data_ptr());
+#define LAUNCHW(RM, HB)                                                        \
+    hipLaunchKernelGGL((gemm_wgrad_kernel<RM, HB>), grid, dim3(256), 0,        \
+                       stream.stream(), dy, ac, a, dWt.data_ptr<float>(),      \
+                       db.data_ptr<float>(), (int)M, (int)N, (int)K,           \
+                       (int)rows_per_chunk)
+    if (relu_mask) { if (want_bias) LAUNCHW(true, true); else LAUNCHW(true, false); }
+    else           { if (want_bias) LAUNCHW(false, true); else LAUNCHW(false, false); }
+#undef LAUNCHW
+    return {dWt, db};
+}
