@@ -55,6 +55,16 @@ std::vector<torch::Tensor> conv_wgrad(torch::Tensor dY, torch::Tensor act,
                                       int64_t OH, int64_t OW, int64_t COUT,
                                       int64_t K);
 
+// lstm_kernels.hip
+std::vector<torch::Tensor> lstm_fwd(
+    torch::Tensor X0, torch::Tensor X1, torch::Tensor Whh0, torch::Tensor Whh1,
+    torch::Tensor init0, torch::Tensor init1, torch::Tensor lens,
+    torch::Tensor barrier_ws, bool want_stash);
+torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
+                       torch::Tensor Hout, torch::Tensor dHext,
+                       torch::Tensor Whh_bwd, torch::Tensor lens,
+                       torch::Tensor barrier_ws);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "r2d2_amd gfx950 HIP kernels";
     m.def("fused_double_q_loss", &fused_double_q_loss,
@@ -76,4 +86,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv_fwd", &conv_fwd, "Implicit-GEMM MFMA conv forward (NHWC)");
     m.def("conv_dgrad", &conv_dgrad, "MFMA conv backward-data (tap classes)");
     m.def("conv_wgrad", &conv_wgrad, "MFMA conv backward-weight");
+    m.def("lstm_fwd", &lstm_fwd,
+          "Persistent fused LSTM forward (dual-network, length-masked)");
+    m.def("lstm_bwd", &lstm_bwd, "Persistent fused LSTM BPTT backward");
 }

@@ -184,8 +184,8 @@ __global__ __launch_bounds__(256) void gemm_wgrad_kernel(
     float* __restrict__ db,                  // (N,) f32 accumulate
     int M, int N, int K, int rows_per_chunk) {
     // grid.x: row chunks; grid.y: N tiles of 64; grid.z: K tiles of 64
-    __shared__ __hip_bfloat16 s_dy[32][64 + 4];  // [m][n]
-    __shared__ __hip_bfloat16 s_a[32][64 + 4];   // [m][k]
+    __shared__ __hip_bfloat16 s_dy[32][64 + 8];  // [m][n]
+    __shared__ __hip_bfloat16 s_a[32][64 + 8];   // [m][k]
     int wave = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
     int wr = wave >> 1, wc = wave & 1;

@@ -68,21 +68,22 @@ def test_gemm_dgrad_wgrad_match_autograd():
     out = M_.gemm_bias_act(A, W, b, 1, False)
     dY = torch.randn(M, N, device="cuda").bfloat16()
 
-    # reference graph in fp32 on the same rounded values
-    A32 = A.float().requires_grad_(True)
-    W32 = W.float().requires_grad_(True)
-    b32 = b.clone().requires_grad_(True)
-    out32 = F.relu(A32 @ W32.t() + b32)
-    out32.backward(dY.float())
+    # reference in fp32 using the KERNEL's activation mask (the kernel masks
+    # on its bf16 forward; an f32 forward flips boundary elements)
+    mask = (out.float() > 0).float()
+    dout = dY.float() * mask
+    ref_dA = dout @ W.float()
+    ref_dW = dout.t() @ A.float()
+    ref_db = dout.sum(0)
 
     # dgrad: dA = (dY*mask) @ W  — kernel takes W stored (K, N)
     W_kn = W.t().contiguous()
     dA = M_.gemm_dgrad(dY, out, W_kn, True)
-    close(dA, A32.grad, rtol=5e-2, name="dgrad")
+    close(dA, ref_dA, rtol=5e-2, name="dgrad")
     # wgrad
     dWt, db = M_.gemm_wgrad(dY, out, A, True, True)
-    close(dWt, W32.grad, rtol=5e-2, atol=0.5, name="wgrad")
-    close(db, b32.grad, rtol=5e-2, atol=0.5, name="bgrad")
+    close(dWt, ref_dW, rtol=5e-2, atol=0.5, name="wgrad")
+    close(db, ref_db, rtol=5e-2, atol=0.5, name="bgrad")
 
 
 # -------------------------------------------------------------------------
@@ -198,7 +199,8 @@ def test_conv_wgrad(conv_id, cin, cout, k, s, inhw):
         x_nchw = torch.randint(0, 256, (N, cin, inhw, inhw),
                                dtype=torch.uint8, device="cuda")
         x_in = x_nchw.permute(0, 2, 3, 1).contiguous()
-        ref_in = (x_nchw.float() / 255.0).requires_grad_(False)
+        # the kernel dequantizes to bf16; round the reference the same way
+        ref_in = (x_nchw.float() / 255.0).bfloat16().float()
     else:
         xb = torch.randn(N, cin, inhw, inhw, device="cuda").bfloat16()
         x_in = xb.permute(0, 2, 3, 1).contiguous()
