@@ -1,0 +1,372 @@
+"""The MI355X training engine: runs the full R2D2 learner update through the
+hand-written gfx950 HIP kernels with a manual (graph-free) backward pass.
+
+Forward per update (fused): ONE online pass over the full
+sequence (serving both learning- and target-position Q, replacing the
+reference's two online passes, worker.py:346+352) and one target-net pass —
+conv1..3 (implicit-GEMM MFMA, NHWC, fused u8 dequant) -> FC -> input GEMM ->
+persistent fused LSTM (both networks in one launch) -> gathered dueling
+heads -> fused double-Q/rescale/Huber loss + on-device priorities.
+
+Backward: fused loss emits dQ; dueling-combine bwd; head GEMM d/wgrads;
+scatter-add into dH_ext; persistent BPTT kernel -> dgates; LSTM weight grads
+as bulk GEMMs over B*T rows; FC + conv d/wgrads (tap-class dgrad).  All
+gradients land in the nn.Module parameters' .grad (f32), so the existing
+clip + Adam + DDP all-reduce path applies unchanged.
+
+Weights are prepacked bf16 (transposed/permuted per kernel contract) after
+every optimizer step (`refresh_online`), and on target-net sync
+(`refresh_target`).
+"""
+
+from typing import Optional
+
+import numpy as np
+import torch
+
+from .. import config as cfg
+from . import hip_ops
+
+PAD_HEAD = 32   # padded output width for the A-dim and V-dim head GEMMs
+KPAD = 32       # LSTM input features padded to a multiple of 32
+
+
+def _round_up(x, m):
+    return (x + m - 1) // m * m
+
+
+class _NetPack:
+    """Prepacked bf16 weights for one network (online or target)."""
+
+    def __init__(self, net, device, A, with_bwd: bool):
+        self.net = net
+        self.device = device
+        self.A = A
+        self.with_bwd = with_bwd
+        self.refresh()
+
+    def refresh(self):
+        net = self.net
+        dev = self.device
+        A = self.A
+        enc = net.encoder
+        to_bf = lambda t: t.detach().to(dev).bfloat16().contiguous()
+        f32 = lambda t: t.detach().to(dev).float().contiguous()
+
+        # convs: (COUT, CIN, KH, KW) -> (COUT, KH*KW*CIN)
+        def pack_conv(conv):
+            w = conv.weight.detach()
+            wt = w.permute(0, 2, 3, 1).reshape(w.shape[0], -1)
+            return to_bf(wt), f32(conv.bias)
+
+        self.w1t, self.b1 = pack_conv(enc.conv1)
+        self.w2t, self.b2 = pack_conv(enc.conv2)
+        self.w3t, self.b3 = pack_conv(enc.conv3)
+
+        # FC: torch flattens NCHW (64,7,7); our conv3 output flattens HWC
+        wf = enc.fc.weight.detach()          # (512, 3136) over (c,h,w)
+        wf_hwc = wf.reshape(512, 64, 7, 7).permute(0, 2, 3, 1).reshape(512, 3136)
+        self.wft = to_bf(wf_hwc)
+        self.bf = f32(enc.fc.bias)
+
+        # LSTM: weight_ih (4H, 512+A+1) padded to KPAD multiple
+        wih = net.recurrent.weight_ih_l0.detach()
+        H4, kin = wih.shape
+        self.kin = kin
+        self.kin_pad = _round_up(kin, KPAD)
+        wih_pad = torch.zeros(H4, self.kin_pad, device=dev)
+        wih_pad[:, :kin] = wih.to(dev)
+        self.wih_t = wih_pad.bfloat16().contiguous()
+        self.whh_t = to_bf(net.recurrent.weight_hh_l0)        # (4H, H)
+        self.lstm_bias = f32(net.recurrent.bias_ih_l0
+                             + net.recurrent.bias_hh_l0)      # (4H,)
+
+        # heads (padded to PAD_HEAD output cols, zero rows beyond real dims)
+        def pack_head_out(lin, rows):
+            w = torch.zeros(PAD_HEAD, lin.weight.shape[1], device=dev)
+            w[:rows] = lin.weight.detach().to(dev)
+            b = torch.zeros(PAD_HEAD, device=dev)
+            b[:rows] = lin.bias.detach().to(dev)
+            return w.bfloat16().contiguous(), b.contiguous()
+
+        adv1, adv2 = net.advantage[0], net.advantage[2]
+        val1, val2 = net.value[0], net.value[2]
+        self.wa1t, self.ba1 = to_bf(adv1.weight), f32(adv1.bias)
+        self.wa2t, self.ba2 = pack_head_out(adv2, A)
+        self.wv1t, self.bv1 = to_bf(val1.weight), f32(val1.bias)
+        self.wv2t, self.bv2 = pack_head_out(val2, 1)
+
+        if self.with_bwd:
+            # dgrad prepacks (W stored (K, N))
+            self.wf_kn = self.wft.t().contiguous()             # (3136, 512)
+            self.wih_kn = self.wih_t.t().contiguous()          # (kin_pad, 4H)
+            self.whh_bwd = self.whh_t.t().contiguous()         # (H, 4H)
+            self.wa1_kn = self.wa1t.t().contiguous()
+            self.wa2_kn = self.wa2t.t().contiguous()           # (512, PAD)
+            self.wv1_kn = self.wv1t.t().contiguous()
+            self.wv2_kn = self.wv2t.t().contiguous()
+            # conv dgrad prepacks
+            w3 = enc.conv3.weight.detach().to(dev)             # (64,64,3,3)
+            w3_nhwc = w3.permute(0, 2, 3, 1)                   # (co,dy,dx,ci)
+            self.w3d = (w3_nhwc.permute(3, 1, 2, 0)            # (ci,dy,dx,co)
+                        .reshape(64, 9 * 64).bfloat16().contiguous())
+            self.taps3 = torch.tensor(
+                [[dy, dx] for dy in range(3) for dx in range(3)],
+                dtype=torch.int32, device=dev)
+            w2 = enc.conv2.weight.detach().to(dev)             # (64,32,4,4)
+            w2_nhwc = w2.permute(0, 2, 3, 1)                   # (co,dy,dx,ci)
+            self.w2d = {}
+            self.taps2 = {}
+            for py in range(2):
+                for px in range(2):
+                    tap_list = [(dy, dx) for dy in range(py, 4, 2)
+                                for dx in range(px, 4, 2)]
+                    wd = torch.stack([w2_nhwc[:, d, x, :] for d, x in tap_list],
+                                     dim=0)                    # (t, co, ci)
+                    self.w2d[(py, px)] = (wd.permute(2, 0, 1)
+                                          .reshape(32, 4 * 64)
+                                          .bfloat16().contiguous())
+                    self.taps2[(py, px)] = torch.tensor(
+                        tap_list, dtype=torch.int32, device=dev)
+
+
+class HipNetworkEngine:
+    def __init__(self, online_net, target_net, device, config=None):
+        c = config or cfg.get()
+        self.cfg = c
+        assert c.encoder == "nature" and c.hidden_dim == 512, \
+            "HIP engine supports the flagship nature/512 config"
+        assert tuple(c.obs_shape[1:]) == (84, 84)
+        self.C = c.obs_shape[0]
+        assert self.C == 4, "conv1 kernel instantiated for 4 input channels"
+        self.A = c.action_dim
+        self.H = 512
+        self.device = torch.device(device)
+        self.m = hip_ops.ext(required=True)
+        self.online = _NetPack(online_net, self.device, self.A, with_bwd=True)
+        self.target = _NetPack(target_net, self.device, self.A, with_bwd=False)
+        self.online_net = online_net
+        self.target_net = target_net
+        self.bar = torch.zeros(64, dtype=torch.int32, device=self.device)
+        self._empty = torch.Tensor()
+
+    def refresh_online(self):
+        self.online.refresh()
+
+    def refresh_target(self):
+        self.target.refresh()
+
+    # ------------------------------------------------------------------
+    def _encoder_fwd(self, pack, obs_hwc_u8):
+        """obs: (M, 84, 84, C) uint8 -> latent (M, 512) bf16 + stashes."""
+        m = self.m
+        M = obs_hwc_u8.shape[0]
+        a1 = m.conv_fwd(obs_hwc_u8, pack.w1t, pack.b1, 1, M, 84, 84, 20, 20, True)
+        a2 = m.conv_fwd(a1, pack.w2t, pack.b2, 2, M, 20, 20, 9, 9, True)
+        a3 = m.conv_fwd(a2, pack.w3t, pack.b3, 3, M, 9, 9, 7, 7, True)
+        flat = a3.view(M, 3136)
+        latent = m.gemm_bias_act(flat, pack.wft, pack.bf, 1, False)
+        return latent, (a1, a2, a3, flat)
+
+    def _lstm_input(self, pack, latent, last_action, last_reward):
+        """Build padded rin and the input-GEMM X (Mrows, 4H)."""
+        M = latent.shape[0]
+        rin = torch.zeros(M, pack.kin_pad, device=self.device,
+                          dtype=torch.bfloat16)
+        rin[:, :512] = latent
+        rin[:, 512:512 + self.A] = last_action.reshape(M, self.A).bfloat16()
+        rin[:, 512 + self.A] = last_reward.reshape(M).bfloat16()
+        X = self.m.gemm_bias_act(rin, pack.wih_t, pack.lstm_bias, 0, False)
+        return rin, X
+
+    def _heads_fwd(self, pack, h):
+        """h: (R, 512) bf16 -> q (R, A) f32 + head stashes."""
+        m = self.m
+        adv1 = m.gemm_bias_act(h, pack.wa1t, pack.ba1, 1, False)
+        adv2 = m.gemm_bias_act(adv1, pack.wa2t, pack.ba2, 0, True)
+        val1 = m.gemm_bias_act(h, pack.wv1t, pack.bv1, 1, False)
+        val2 = m.gemm_bias_act(val1, pack.wv2t, pack.bv2, 0, True)
+        q = m.dueling_combine(adv2, val2, self.A)
+        return q, (adv1, val1)
+
+    # ------------------------------------------------------------------
+    def _positions(self, burn, learn, fwd, T):
+        """Host-side gather indices.  Returns (learn_pos, tgt_pos) flat int64
+        arrays indexing (b * (T+1) + t + 1) rows of Hout, plus per-row b,t
+        (for the dHext scatter)."""
+        n = self.cfg.forward_steps
+        lp, tp, lbt = [], [], []
+        for b, (bu, le, fw) in enumerate(zip(burn.tolist(), learn.tolist(),
+                                             fwd.tolist())):
+            t_learn = np.arange(le) + bu
+            t_tgt = np.minimum(t_learn + n, bu + le + fw - 1)
+            lp.append(b * (T + 1) + t_learn + 1)
+            tp.append(b * (T + 1) + t_tgt + 1)
+            lbt.append(b * T + t_learn)
+        return (np.concatenate(lp), np.concatenate(tp), np.concatenate(lbt))
+
+    # ------------------------------------------------------------------
+    def train_step(self, batch):
+        """Full fused update.  Fills .grad on the online nn.Module params and
+        returns (loss tensor, per-sequence priority tensor, both on device)."""
+        m = self.m
+        c = self.cfg
+        dev = self.device
+        B, T = batch.obs.shape[:2]
+        A, H = self.A, self.H
+
+        # ---- inputs ----------------------------------------------------
+        obs = batch.obs
+        if obs.shape[2] == self.C:                 # (B,T,C,84,84) -> NHWC
+            obs_hwc = obs.permute(0, 1, 3, 4, 2).reshape(B * T, 84, 84, self.C)
+            obs_hwc = obs_hwc.contiguous()
+        else:
+            obs_hwc = obs.reshape(B * T, 84, 84, self.C).contiguous()
+        la = batch.last_action.to(dev)
+        lr = batch.last_reward.to(dev)
+        lens = (batch.burn_in_steps + batch.learning_steps
+                + batch.forward_steps).to(torch.int32).to(dev)
+        init = batch.hidden.float().contiguous()   # (2, B, H)
+
+        # ---- forward ---------------------------------------------------
+        lat_o, enc_stash = self._encoder_fwd(self.online, obs_hwc)
+        lat_t, _ = self._encoder_fwd(self.target, obs_hwc)
+        rin_o, X_o = self._lstm_input(self.online, lat_o, la, lr)
+        _, X_t = self._lstm_input(self.target, lat_t, la, lr)
+        Xo = X_o.view(B, T, 4 * H)
+        Xt = X_t.view(B, T, 4 * H)
+        Ho, Co, Ht, Ct, stash = m.lstm_fwd(
+            Xo, Xt, self.online.whh_t, self.target.whh_t,
+            init, init, lens, self.bar, True)
+
+        lp, tp, lbt = self._positions(batch.burn_in_steps,
+                                      batch.learning_steps,
+                                      batch.forward_steps, T)
+        lp_t = torch.from_numpy(lp).to(dev)
+        tp_t = torch.from_numpy(tp).to(dev)
+        R = lp.shape[0]
+
+        Ho_flat = Ho.view(-1, H)
+        h_learn = Ho_flat.index_select(0, lp_t)
+        h_tgt_o = Ho_flat.index_select(0, tp_t)
+        h_tgt_t = Ht.view(-1, H).index_select(0, tp_t)
+
+        # online heads on [learn; tgt] rows in one pass
+        h_cat = torch.cat([h_learn, h_tgt_o], 0)
+        q_cat, (adv1_o, val1_o) = self._heads_fwd(self.online, h_cat)
+        q_learn, q_online_tgt = q_cat[:R], q_cat[R:]
+        q_tgt, _ = self._heads_fwd(self.target, h_tgt_t)
+
+        # ---- fused loss + priorities ----------------------------------
+        seg = torch.zeros(B + 1, dtype=torch.int32)
+        seg[1:] = torch.cumsum(batch.learning_steps.to(torch.int32), 0)
+        seg = seg.to(dev)
+        loss, dq, abs_td, _ = m.fused_double_q_loss(
+            q_learn.contiguous(), q_online_tgt.contiguous(), q_tgt.contiguous(),
+            batch.action.view(-1).long(), batch.n_step_reward, batch.gamma,
+            batch.is_weights, c.rescale_eps, c.huber_kappa,
+            0 if c.loss_fn == "mse" else 1)
+        prio = m.segment_priority(abs_td, seg, c.prio_eta)
+
+        # ---- backward --------------------------------------------------
+        ON = self.online
+        dq_pad = torch.zeros(2 * R, A, device=dev)  # learn rows only get grad
+        dq_pad[:R] = dq
+        dadv2, dval2 = m.dueling_combine_bwd(dq_pad.contiguous(),
+                                             PAD_HEAD, PAD_HEAD)
+        # adv path
+        dadv1 = m.gemm_dgrad(dadv2, self._empty, ON.wa2_kn, False)
+        dh_a = m.gemm_dgrad(dadv1, adv1_o, ON.wa1_kn, True)
+        dWa2, dba2 = m.gemm_wgrad(dadv2, self._empty, adv1_o, False, True)
+        dWa1, dba1 = m.gemm_wgrad(dadv1, adv1_o, h_cat, True, True)
+        # value path
+        dval1 = m.gemm_dgrad(dval2, self._empty, ON.wv2_kn, False)
+        dh_v = m.gemm_dgrad(dval1, val1_o, ON.wv1_kn, True)
+        dWv2, dbv2 = m.gemm_wgrad(dval2, self._empty, val1_o, False, True)
+        dWv1, dbv1 = m.gemm_wgrad(dval1, val1_o, h_cat, True, True)
+
+        dh_rows = (dh_a.float() + dh_v.float())     # (2R, 512)
+        # scatter-add into dHext (B, T, H): learn rows at t, tgt rows at t_tgt
+        dHext = torch.zeros(B * T, H, device=dev)
+        dHext.index_add_(0, torch.from_numpy(lbt).to(dev), dh_rows[:R])
+        tgt_bt = tp_t - (tp_t // (T + 1)) - 1       # b*(T+1)+t+1 -> b*T + t
+        dHext.index_add_(0, tgt_bt, dh_rows[R:])
+        dHext = dHext.view(B, T, H)
+
+        dgates = m.lstm_bwd(stash, Co, Ho, dHext.contiguous(), ON.whh_bwd,
+                            lens, self.bar)
+        dgates_flat = dgates.view(B * T, 4 * H)
+
+        h_prev = Ho[:, :T].reshape(B * T, H).contiguous()
+        dWhh, _ = m.gemm_wgrad(dgates_flat, self._empty, h_prev, False, False)
+        dWih_pad, db_lstm = m.gemm_wgrad(dgates_flat, self._empty, rin_o,
+                                         False, True)
+        drin = m.gemm_dgrad(dgates_flat, self._empty, ON.wih_kn, False)
+        dlat = drin[:, :512].contiguous()
+
+        a1, a2, a3, flat = enc_stash
+        lat_bf = lat_o  # forward output (relu mask source)
+        dflat = m.gemm_dgrad(dlat, lat_bf, ON.wf_kn, True)
+        dWf, dbf = m.gemm_wgrad(dlat, lat_bf, flat, True, True)
+
+        M = B * T
+        # conv3 backward
+        dW3, db3 = m.conv_wgrad(dflat.view(M * 49, 64), a3, a2, 3,
+                                M, 9, 9, 7, 7, 64, 9 * 64)
+        d3m = (dflat.view(M * 49, 64)
+               * (a3.view(M * 49, 64) > 0).bfloat16()).view(M, 7, 7, 64)
+        dyp3 = torch.zeros(M, 11, 11, 64, device=dev, dtype=torch.bfloat16)
+        dyp3[:, 2:9, 2:9] = d3m
+        d_a2 = torch.empty(M, 9, 9, 64, device=dev, dtype=torch.bfloat16)
+        m.conv_dgrad(dyp3.contiguous(), ON.w3d, ON.taps3, M, 11, 11, 64,
+                     9, 9, 64, 0, 0, 1, 2, d_a2)
+        # conv2 backward
+        dW2, db2 = m.conv_wgrad(d_a2.view(M * 81, 64), a2, a1, 2,
+                                M, 20, 20, 9, 9, 64, 4 * 4 * 32)
+        d2m = (d_a2.view(M * 81, 64)
+               * (a2.view(M * 81, 64) > 0).bfloat16()).view(M, 9, 9, 64)
+        dyp2 = torch.zeros(M, 11, 11, 64, device=dev, dtype=torch.bfloat16)
+        dyp2[:, 1:10, 1:10] = d2m
+        d_a1 = torch.empty(M, 20, 20, 32, device=dev, dtype=torch.bfloat16)
+        dyp2c = dyp2.contiguous()
+        for py in range(2):
+            for px in range(2):
+                m.conv_dgrad(dyp2c, ON.w2d[(py, px)], ON.taps2[(py, px)],
+                             M, 11, 11, 64, 20, 20, 32, py, px, 2, 1, d_a1)
+        # conv1 wgrad (no dgrad: input is data)
+        dW1, db1 = m.conv_wgrad(d_a1.view(M * 400, 32), a1, obs_hwc, 1,
+                                M, 84, 84, 20, 20, 32, 8 * 8 * self.C)
+
+        # ---- write grads into the nn.Module (f32) ----------------------
+        net = self.online_net
+        enc = net.encoder
+
+        def setg(p, g):
+            p.grad = g.to(p.dtype)
+
+        def conv_grad(dwt, cout, kh, kw, cin):
+            return dwt.view(cout, kh, kw, cin).permute(0, 3, 1, 2).contiguous()
+
+        setg(enc.conv1.weight, conv_grad(dW1, 32, 8, 8, self.C))
+        setg(enc.conv1.bias, db1)
+        setg(enc.conv2.weight, conv_grad(dW2, 64, 4, 4, 32))
+        setg(enc.conv2.bias, db2)
+        setg(enc.conv3.weight, conv_grad(dW3, 64, 3, 3, 64))
+        setg(enc.conv3.bias, db3)
+        setg(enc.fc.weight, dWf.view(512, 7, 7, 64).permute(0, 3, 1, 2)
+             .reshape(512, 3136).contiguous())
+        setg(enc.fc.bias, dbf)
+        setg(net.recurrent.weight_ih_l0, dWih_pad[:, :ON.kin].contiguous())
+        setg(net.recurrent.weight_hh_l0, dWhh)
+        setg(net.recurrent.bias_ih_l0, db_lstm)
+        setg(net.recurrent.bias_hh_l0, db_lstm.clone())
+        setg(net.advantage[0].weight, dWa1)
+        setg(net.advantage[0].bias, dba1)
+        setg(net.advantage[2].weight, dWa2[:A].contiguous())
+        setg(net.advantage[2].bias, dba2[:A].contiguous())
+        setg(net.value[0].weight, dWv1)
+        setg(net.value[0].bias, dbv1)
+        setg(net.value[2].weight, dWv2[:1].contiguous())
+        setg(net.value[2].bias, dbv2[:1].contiguous())
+
+        return loss.squeeze(0), prio

@@ -60,6 +60,9 @@ std::vector<torch::Tensor> lstm_fwd(
     torch::Tensor X0, torch::Tensor X1, torch::Tensor Whh0, torch::Tensor Whh1,
     torch::Tensor init0, torch::Tensor init1, torch::Tensor lens,
     torch::Tensor barrier_ws, bool want_stash);
+torch::Tensor dueling_combine(torch::Tensor adv, torch::Tensor val, int64_t A);
+std::vector<torch::Tensor> dueling_combine_bwd(torch::Tensor dq, int64_t PADA,
+                                               int64_t PADV);
 torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
                        torch::Tensor Hout, torch::Tensor dHext,
                        torch::Tensor Whh_bwd, torch::Tensor lens,
@@ -89,4 +92,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("lstm_fwd", &lstm_fwd,
           "Persistent fused LSTM forward (dual-network, length-masked)");
     m.def("lstm_bwd", &lstm_bwd, "Persistent fused LSTM BPTT backward");
+    m.def("dueling_combine", &dueling_combine, "q = V + A - mean(A)");
+    m.def("dueling_combine_bwd", &dueling_combine_bwd, "dueling combine backward");
 }

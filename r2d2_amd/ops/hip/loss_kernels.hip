@@ -169,3 +169,68 @@ torch::Tensor segment_priority(torch::Tensor abs_td, torch::Tensor seg_offsets,
                        B, (float)eta);
     return prio;
 }
+
+// ---------------------------------------------------------------------------
+// Dueling head combine: q[r][a] = v[r] + adv[r][a] - mean_{a<A}(adv[r][:A])
+// adv/val come from padded 32-col head GEMMs (cols >= A are zero-weighted).
+// Backward: dadv[a] = dq[a] - sum(dq)/A (a < A, else 0); dval = sum(dq).
+// ---------------------------------------------------------------------------
+__global__ void dueling_combine_kernel(
+    const float* __restrict__ adv,   // (R, PADA)
+    const float* __restrict__ val,   // (R, PADV) — col 0 is V
+    float* __restrict__ q,           // (R, A)
+    int R, int A, int PADA, int PADV) {
+    int r = blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+    int lane = threadIdx.x & (WAVE - 1);
+    if (r >= R) return;
+    float a = (lane < A) ? adv[(long)r * PADA + lane] : 0.f;
+    float s = wave_allreduce_sum(a) / A;
+    if (lane < A) q[(long)r * A + lane] = val[(long)r * PADV] + a - s;
+}
+
+__global__ void dueling_combine_bwd_kernel(
+    const float* __restrict__ dq,    // (R, A)
+    __hip_bfloat16* __restrict__ dadv,  // (R, PADA) bf16 (feeds gemm bwd)
+    __hip_bfloat16* __restrict__ dval,  // (R, PADV)
+    int R, int A, int PADA, int PADV) {
+    int r = blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+    int lane = threadIdx.x & (WAVE - 1);
+    if (r >= R) return;
+    float d = (lane < A) ? dq[(long)r * A + lane] : 0.f;
+    float s = wave_allreduce_sum(d);
+    if (lane < PADA)
+        dadv[(long)r * PADA + lane] =
+            f2bf((lane < A) ? (d - s / A) : 0.f);
+    if (lane < PADV)
+        dval[(long)r * PADV + lane] = f2bf((lane == 0) ? s : 0.f);
+}
+
+torch::Tensor dueling_combine(torch::Tensor adv, torch::Tensor val, int64_t A) {
+    int R = adv.size(0), PADA = adv.size(1), PADV = val.size(1);
+    auto q = torch::empty({R, A}, adv.options());
+    const int wpb = 4;
+    dim3 grid((R + wpb - 1) / wpb);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(dueling_combine_kernel, grid, dim3(WAVE * wpb), 0,
+                       stream.stream(), adv.data_ptr<float>(),
+                       val.data_ptr<float>(), q.data_ptr<float>(), R, (int)A,
+                       PADA, PADV);
+    return q;
+}
+
+std::vector<torch::Tensor> dueling_combine_bwd(torch::Tensor dq, int64_t PADA,
+                                               int64_t PADV) {
+    int R = dq.size(0), A = dq.size(1);
+    auto opts = dq.options().dtype(torch::kBFloat16);
+    auto dadv = torch::empty({R, PADA}, opts);
+    auto dval = torch::empty({R, PADV}, opts);
+    const int wpb = 4;
+    dim3 grid((R + wpb - 1) / wpb);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(dueling_combine_bwd_kernel, grid, dim3(WAVE * wpb), 0,
+                       stream.stream(), dq.data_ptr<float>(),
+                       reinterpret_cast<__hip_bfloat16*>(dadv.data_ptr()),
+                       reinterpret_cast<__hip_bfloat16*>(dval.data_ptr()),
+                       R, A, (int)PADA, (int)PADV);
+    return {dadv, dval};
+}

@@ -103,6 +103,28 @@ class GradAllReducer:
                 for p, g in zip(bucket, outs):
                     p.grad.copy_(g)
 
+    def reduce_all(self):
+        """Synchronous bucket all-reduce of existing .grads (for the manual-
+        backward HIP engine path, where no autograd hooks fire)."""
+        if not self.enabled:
+            return
+        works = []
+        flats = []
+        for bucket in self.buckets:
+            for p in bucket:
+                if p.grad is None:
+                    p.grad = torch.zeros_like(p)
+            flat = torch._utils._flatten_dense_tensors([p.grad for p in bucket])
+            flat.div_(self.world_size)
+            works.append(dist.all_reduce(flat, group=self.group, async_op=True))
+            flats.append(flat)
+        for bucket, work, flat in zip(self.buckets, works, flats):
+            work.wait()
+            outs = torch._utils._unflatten_dense_tensors(
+                flat, [p.grad for p in bucket])
+            for p, g in zip(bucket, outs):
+                p.grad.copy_(g)
+
     def broadcast_params(self, src: int = 0):
         if not self.enabled:
             return

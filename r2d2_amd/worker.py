@@ -514,14 +514,24 @@ class Learner:
     # -- HIP engine ---------------------------------------------------------
 
     def enable_hip_engine(self):
-        """Switch the loss/priority path (and, as they land, the full forward/
-        backward) to the gfx950 HIP kernels.  Raises if the extension is not
-        built — on a GPU box the native path must be the one that runs."""
+        """Switch to the gfx950 HIP kernels.  For the flagship nature/512
+        config the FULL forward/backward runs through the hand-written engine
+        (ops/engine.py); otherwise the fused loss/priority kernels are used
+        under the eager forward.  Raises if the extension is not built — on a
+        GPU box the native path must be the one that runs."""
         from .ops import hip_ops
         hip_ops.ext(required=True)
         self.hip_engine = True
+        c = self.cfg
+        if (c.encoder == "nature" and c.hidden_dim == 512
+                and len(c.obs_shape) == 3 and c.obs_shape[0] == 4
+                and self.device.type == "cuda"):
+            from .ops.engine import HipNetworkEngine
+            self.engine = HipNetworkEngine(self.online_net, self.target_net,
+                                           self.device, c)
 
     hip_engine = False
+    engine = None
 
     # -- weight publication -------------------------------------------------
 
@@ -551,6 +561,19 @@ class Learner:
         device tensor on the HIP path, numpy on the eager path)."""
         c = self.cfg
         batch.to(self.device)
+
+        if self.engine is not None:
+            # full HIP path: manual backward fills .grads directly
+            loss, prio = self.engine.train_step(batch)
+            if self.reducer is not None:
+                self.reducer.reduce_all()
+            nn.utils.clip_grad_norm_(self.online_net.parameters(), self.grad_norm)
+            self.optimizer.step()
+            self.optimizer.zero_grad(set_to_none=True)
+            self.engine.refresh_online()
+            self.num_updates += 1
+            return loss, prio
+
         h0 = (batch.hidden[:1], batch.hidden[1:])
 
         ctx = torch.autocast("cuda", dtype=torch.bfloat16) if self.amp else _nullctx()
@@ -619,6 +642,8 @@ class Learner:
                 self.store_weights()
             if self.num_updates % self.target_net_update_interval == 0:
                 self.target_net.load_state_dict(self.online_net.state_dict())
+                if self.engine is not None:
+                    self.engine.refresh_target()
             if self.num_updates % self.save_interval == 0:
                 self.save(start_time)
 
