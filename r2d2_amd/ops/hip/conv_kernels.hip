@@ -49,15 +49,15 @@ __device__ __forceinline__ bf16x8 czero() {
 // load 8 uint8 bytes and dequantize to bf16/255
 __device__ __forceinline__ bf16x8 load_dequant8(const unsigned char* p) {
     uint2 raw = *reinterpret_cast<const uint2*>(p);
-    cbf8u r;
+    bf16x8 r;
     const float inv = 1.f / 255.f;
 #pragma unroll
     for (int i = 0; i < 4; ++i)
-        r.e[i] = (__bf16)(((raw.x >> (8 * i)) & 0xff) * inv);
+        r[i] = (__bf16)(((raw.x >> (8 * i)) & 0xff) * inv);
 #pragma unroll
     for (int i = 0; i < 4; ++i)
-        r.e[4 + i] = (__bf16)(((raw.y >> (8 * i)) & 0xff) * inv);
-    return r.v;
+        r[4 + i] = (__bf16)(((raw.y >> (8 * i)) & 0xff) * inv);
+    return r;
 }
 
 // ---------------------------------------------------------------------------
@@ -266,8 +266,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
             int mrow = t / 8;
             int col = (t % 8) * 8;
             long gm = m0 + mrow;
-            cbf8u v;
-            v.v = czero();
+            bf16x8 v = czero();
             if (gm < mend) {
 #pragma unroll
                 for (int e = 0; e < 8; ++e) {
@@ -275,14 +274,13 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
                     if (c < COUT) {
                         float g = bf2f(dY[gm * COUT + c]);
                         if (RELU && !(bf2f(act[gm * COUT + c]) > 0.f)) g = 0.f;
-                        v.e[e] = (__bf16)g;
+                        v[e] = (__bf16)g;
                     }
                 }
             }
-            *reinterpret_cast<bf16x8*>(&s_dy[mrow][col]) = v.v;
+            *reinterpret_cast<bf16x8*>(&s_dy[mrow][col]) = v;
 
-            cbf8u w;
-            w.v = czero();
+            bf16x8 w = czero();
             if (gm < mend && kcol0 + col < K) {
                 long n = gm / (OH * OW);
                 int p = (int)(gm % (OH * OW));
@@ -293,32 +291,29 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
                 long off = base + (long)dy * INW * CIN + rem;
                 // 8 contiguous k (never crosses a dy boundary: KWC % 8 == 0)
                 if (IN_U8)
-                    w.v = load_dequant8(
+                    w = load_dequant8(
                         reinterpret_cast<const unsigned char*>(in) + off);
                 else
-                    w.v = cload_bf16x8(
+                    w = cload_bf16x8(
                         reinterpret_cast<const __hip_bfloat16*>(in) + off);
                 if (kcol0 + col + 8 > K) {
 #pragma unroll
                     for (int e = 0; e < 8; ++e)
-                        if (kcol0 + col + e >= K) w.e[e] = (__bf16)0.f;
+                        if (kcol0 + col + e >= K) w[e] = (__bf16)0.f;
                 }
             }
-            *reinterpret_cast<bf16x8*>(&s_a[mrow][col]) = w.v;
+            *reinterpret_cast<bf16x8*>(&s_a[mrow][col]) = w;
         }
         __syncthreads();
 
         bf16x8 fa[2], fb[2];
 #pragma unroll
         for (int i = 0; i < 2; ++i) {
-            cbf8u va, vb;
 #pragma unroll
             for (int e = 0; e < 8; ++e) {
-                va.e[e] = s_dy[mseg + e][wr * 32 + i * 16 + frow];
-                vb.e[e] = s_a[mseg + e][wc * 32 + i * 16 + frow];
+                fa[i][e] = *(const __bf16*)&s_dy[mseg + e][wr * 32 + i * 16 + frow];
+                fb[i][e] = *(const __bf16*)&s_a[mseg + e][wc * 32 + i * 16 + frow];
             }
-            fa[i] = va.v;
-            fb[i] = vb.v;
         }
 #pragma unroll
         for (int i = 0; i < 2; ++i)

@@ -133,13 +133,10 @@ __global__ __launch_bounds__(256) void gemm_dgrad_kernel(
             if (r < M) {
                 a[i] = load_bf16x8(dY + r * N + n0 + nseg);
                 if (RELU_MASK) {
-                    bf8u av, mv;
-                    av.v = a[i];
-                    mv.v = load_bf16x8(act + r * N + n0 + nseg);
+                    bf16x8 mv = load_bf16x8(act + r * N + n0 + nseg);
 #pragma unroll
                     for (int e = 0; e < 8; ++e)
-                        if (!((float)mv.e[e] > 0.f)) av.e[e] = (__bf16)0.f;
-                    a[i] = av.v;
+                        if (!((float)mv[e] > 0.f)) a[i][e] = (__bf16)0.f;
                 }
             } else {
                 a[i] = zero_bf16x8();
@@ -209,16 +206,15 @@ __global__ __launch_bounds__(256) void gemm_wgrad_kernel(
             int mrow = t / 8;           // 0..31
             int ncol = (t % 8) * 8;     // 0..56
             long gm = m0 + mrow;
-            bf8u v;
+            bf16x8 v = zero_bf16x8();
             if (gm < mend) {
                 if (ncol0 + ncol + 8 <= N)
-                    v.v = load_bf16x8(dY + gm * N + ncol0 + ncol);
+                    v = load_bf16x8(dY + gm * N + ncol0 + ncol);
                 else {
 #pragma unroll
                     for (int e = 0; e < 8; ++e) {
                         long c = ncol0 + ncol + e;
-                        v.e[e] = (c < N) ? ((const __bf16*)dY)[gm * N + c]
-                                         : (__bf16)0.f;
+                        if (c < N) v[e] = ((const __bf16*)dY)[gm * N + c];
                     }
                 }
                 if (RELU_MASK) {
@@ -227,29 +223,24 @@ __global__ __launch_bounds__(256) void gemm_wgrad_kernel(
                         long c = ncol0 + ncol + e;
                         float m_ = (c < N) ? (float)((const __bf16*)act)[gm * N + c]
                                            : 0.f;
-                        if (!(m_ > 0.f)) v.e[e] = (__bf16)0.f;
+                        if (!(m_ > 0.f)) v[e] = (__bf16)0.f;
                     }
                 }
-            } else {
-                v.v = zero_bf16x8();
             }
-            *reinterpret_cast<bf16x8*>(&s_dy[mrow][ncol]) = v.v;
-            bf8u w;
+            *reinterpret_cast<bf16x8*>(&s_dy[mrow][ncol]) = v;
+            bf16x8 w = zero_bf16x8();
             if (gm < mend) {
                 if (kcol0 + ncol + 8 <= K)
-                    w.v = load_bf16x8(A + gm * K + kcol0 + ncol);
+                    w = load_bf16x8(A + gm * K + kcol0 + ncol);
                 else {
 #pragma unroll
                     for (int e = 0; e < 8; ++e) {
                         long c = kcol0 + ncol + e;
-                        w.e[e] = (c < K) ? ((const __bf16*)A)[gm * K + c]
-                                         : (__bf16)0.f;
+                        if (c < K) w[e] = ((const __bf16*)A)[gm * K + c];
                     }
                 }
-            } else {
-                w.v = zero_bf16x8();
             }
-            *reinterpret_cast<bf16x8*>(&s_a[mrow][ncol]) = w.v;
+            *reinterpret_cast<bf16x8*>(&s_a[mrow][ncol]) = w;
         }
         __syncthreads();
 
@@ -261,14 +252,11 @@ __global__ __launch_bounds__(256) void gemm_wgrad_kernel(
             bf16x8 fa[2], fb[2];
 #pragma unroll
             for (int i = 0; i < 2; ++i) {
-                bf8u va, vb;
 #pragma unroll
                 for (int e = 0; e < 8; ++e) {
-                    va.e[e] = s_dy[mseg + e][wr * 32 + i * 16 + frow];
-                    vb.e[e] = s_a[mseg + e][wc * 32 + i * 16 + frow];
+                    fa[i][e] = *(const __bf16*)&s_dy[mseg + e][wr * 32 + i * 16 + frow];
+                    fb[i][e] = *(const __bf16*)&s_a[mseg + e][wc * 32 + i * 16 + frow];
                 }
-                fa[i] = va.v;
-                fb[i] = vb.v;
             }
 #pragma unroll
             for (int i = 0; i < 2; ++i)
