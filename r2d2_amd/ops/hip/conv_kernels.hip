@@ -271,11 +271,12 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
 #pragma unroll
                 for (int e = 0; e < 8; ++e) {
                     long c = cocol0 + col + e;
-                    if (c < COUT) {
-                        float g = bf2f(dY[gm * COUT + c]);
-                        if (RELU && !(bf2f(act[gm * COUT + c]) > 0.f)) g = 0.f;
-                        v[e] = (__bf16)g;
+                    float g = (c < COUT) ? bf2f(dY[gm * COUT + c]) : 0.f;
+                    if (RELU) {
+                        float m_ = (c < COUT) ? bf2f(act[gm * COUT + c]) : 0.f;
+                        g = (m_ > 0.f) ? g : 0.f;
                     }
+                    v[e] = (__bf16)g;
                 }
             }
             *reinterpret_cast<bf16x8*>(&s_dy[mrow][col]) = v;
@@ -298,8 +299,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
                         reinterpret_cast<const __hip_bfloat16*>(in) + off);
                 if (kcol0 + col + 8 > K) {
 #pragma unroll
-                    for (int e = 0; e < 8; ++e)
-                        if (kcol0 + col + e >= K) w[e] = (__bf16)0.f;
+                    for (int e = 0; e < 8; ++e) {
+                        float vv = (float)w[e];
+                        w[e] = (__bf16)((kcol0 + col + e < K) ? vv : 0.f);
+                    }
                 }
             }
             *reinterpret_cast<bf16x8*>(&s_a[mrow][col]) = w;

@@ -135,8 +135,10 @@ __global__ __launch_bounds__(256) void gemm_dgrad_kernel(
                 if (RELU_MASK) {
                     bf16x8 mv = load_bf16x8(act + r * N + n0 + nseg);
 #pragma unroll
-                    for (int e = 0; e < 8; ++e)
-                        if (!((float)mv[e] > 0.f)) a[i][e] = (__bf16)0.f;
+                    for (int e = 0; e < 8; ++e) {
+                        float vv = (float)a[i][e];
+                        a[i][e] = (__bf16)(((float)mv[e] > 0.f) ? vv : 0.f);
+                    }
                 }
             } else {
                 a[i] = zero_bf16x8();
@@ -208,22 +210,28 @@ __global__ __launch_bounds__(256) void gemm_wgrad_kernel(
             long gm = m0 + mrow;
             bf16x8 v = zero_bf16x8();
             if (gm < mend) {
-                if (ncol0 + ncol + 8 <= N)
+                if (ncol0 + ncol + 8 <= N) {
                     v = load_bf16x8(dY + gm * N + ncol0 + ncol);
-                else {
+                    if (RELU_MASK) {
+                        bf16x8 mv = load_bf16x8(act + gm * N + ncol0 + ncol);
 #pragma unroll
-                    for (int e = 0; e < 8; ++e) {
-                        long c = ncol0 + ncol + e;
-                        if (c < N) v[e] = ((const __bf16*)dY)[gm * N + c];
+                        for (int e = 0; e < 8; ++e) {
+                            float vv = (float)v[e];
+                            v[e] = (__bf16)(((float)mv[e] > 0.f) ? vv : 0.f);
+                        }
                     }
-                }
-                if (RELU_MASK) {
+                } else {
 #pragma unroll
                     for (int e = 0; e < 8; ++e) {
                         long c = ncol0 + ncol + e;
-                        float m_ = (c < N) ? (float)((const __bf16*)act)[gm * N + c]
+                        float vv = (c < N) ? (float)((const __bf16*)dY)[gm * N + c]
                                            : 0.f;
-                        if (!(m_ > 0.f)) v[e] = (__bf16)0.f;
+                        if (RELU_MASK) {
+                            float m_ = (c < N)
+                                ? (float)((const __bf16*)act)[gm * N + c] : 0.f;
+                            vv = (m_ > 0.f) ? vv : 0.f;
+                        }
+                        v[e] = (__bf16)vv;
                     }
                 }
             }
