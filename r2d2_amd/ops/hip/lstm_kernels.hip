@@ -1,26 +1,32 @@
 // Fused persistent LSTM kernels (gfx950) — SURVEY.md §2.3 K6, the hard one.
 //
 // Replaces nn.LSTM + pack_padded_sequence (reference model.py:95-100,136-141;
-// 87k hipBLASLt launches per profile) with ONE kernel per direction:
+// 87k hipBLASLt launches per profile) with ONE kernel per direction.
 //
-// - The input GEMM X = rin @ W_ih^T + b is precomputed for all (B, T) steps
-//   by the MFMA GEMM (gemm_kernels.hip) — it has no sequential dependence.
-// - The recurrent part runs as a single persistent launch: each workgroup
-//   owns a slice of hidden units, keeps its W_hh slice resident in LDS for
-//   all T steps, computes its gate columns with MFMA from the previous
-//   step's h (read through L2), applies the gate nonlinearities, and
-//   advances h/c.  Steps are separated by an XCD-sharded global barrier
-//   (agent-scope release/acquire, relaxed polling + s_sleep, cumulative
-//   epoch counters zeroed by hipMemsetAsync before every launch — HIP guide
-//   §6 G16).  Per-sample length masks replace pack_padded semantics: masked
-//   steps copy h/c through unchanged.
-// - Two networks (online + target) ride in one launch (disjoint workgroup
-//   ranges), sharing the per-step barrier.
-// - Backward (online only) runs the reverse-time recurrence: per step one
-//   MFMA GEMM dh += dgates_{t+1} @ W_hh with the wg's W_hh^T slice LDS-
-//   resident, elementwise gate backward, dgates written to a global stash.
-//   The big weight gradients (dW_hh, dW_ih, db) and dX are then plain
-//   GEMMs outside the kernel (gemm_wgrad / gemm_dgrad over B*T rows).
+// Structure (forward):
+// - X = rin @ W_ih^T + b precomputed for all (B, T) steps by the MFMA GEMM.
+// - Persistent launch, one workgroup per 8 hidden units (x2 networks in one
+//   launch).  W_hh slice and the cell state c stay LDS-resident for all T
+//   steps; each step bulk-stages the previous h (B x H) into LDS with
+//   independent coalesced 16 B loads, runs the gate GEMM entirely out of
+//   LDS, and advances h/c with vectorized stores.
+// - NO global barrier: the time index makes the h exchange naturally
+//   double-buffered, so each step is synchronized by PRODUCER FLAGS only
+//   (HIP guide §6 G16: plain h stores -> per-wave vmcnt drain ->
+//   __syncthreads -> release fence -> relaxed per-slice flag; consumers
+//   poll 64 flags lane-parallel, one acquire fence, then plain loads).
+//   Flag words are zeroed by hipMemsetAsync before every launch; max skew
+//   between workgroups is self-limiting (each wg is both producer and
+//   consumer).  Spins are bounded; a poison word aborts the launch.
+// - Per-sample length masks replace pack_padded semantics: masked steps
+//   copy h/c through unchanged.
+//
+// Backward (online net): reverse recurrence with the same flag protocol on
+// the dgates stream; dgates_{t+1} (B x 4H) is streamed through LDS in
+// 256-column pieces (cooperative staging, then all-LDS MFMA); per-step gate
+// math reads the stashed activations via vectorized row staging.  The bulk
+// weight gradients (dW_hh, dW_ih, db) and dX are plain GEMMs outside the
+// kernel (gemm_wgrad / gemm_dgrad over B*T rows).
 //
 // Gate order matches torch.nn.LSTM: [i, f, g, o] chunks of H.
 
@@ -36,13 +42,18 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 union lbf8u {
     bf16x8 v;
     uint4 u;
-    __bf16 e[8];
 };
 
 __device__ __forceinline__ bf16x8 lload8(const __hip_bfloat16* p) {
     lbf8u r;
     r.u = *reinterpret_cast<const uint4*>(p);
     return r.v;
+}
+
+__device__ __forceinline__ void lstore8(__hip_bfloat16* p, bf16x8 v) {
+    lbf8u r;
+    r.v = v;
+    *reinterpret_cast<uint4*>(p) = r.u;
 }
 
 __device__ __forceinline__ bf16x8 lzero8() {
@@ -56,25 +67,79 @@ __device__ __forceinline__ float sigmoidf_(float x) {
 }
 
 // ---------------------------------------------------------------------------
-// XCD-sharded grid barrier (plain launch; grid <= 256 blocks => resident).
-// Cumulative counters: epoch e complete when bucket counts reach e*per_bucket
-// and top count reaches e*8.  One release fence before arrival, one acquire
-// after the generation flip; relaxed polling with s_sleep.
+// Producer-flag hand-off.  flags[slice] counts published time indices of that
+// producer slice.  publish: called by ALL threads after the slice stores.
+// await: wave 0 polls `nflags` flag words lane-parallel (relaxed), one
+// acquire fence, __syncthreads.  Returns false on bounded-spin timeout.
 // ---------------------------------------------------------------------------
+#define LSTM_MAX_FLAGS 128
+
+__device__ __forceinline__ void publish_slice(unsigned* flag, unsigned value) {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // every storing wave
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __hip_atomic_store(flag, value, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+    }
+}
+
+__device__ __forceinline__ bool await_slices(unsigned* flags, int nflags,
+                                             unsigned value, unsigned* poison) {
+    __shared__ unsigned ok_s;
+    if (threadIdx.x < WAVE) {
+        int lane = threadIdx.x;
+        unsigned ok = 1;
+        long spins = 0;
+        for (;;) {
+            unsigned f1 = (lane < nflags)
+                ? __hip_atomic_load(&flags[lane], __ATOMIC_RELAXED,
+                                    __HIP_MEMORY_SCOPE_AGENT) : ~0u;
+            unsigned f2 = (lane + WAVE < nflags)
+                ? __hip_atomic_load(&flags[lane + WAVE], __ATOMIC_RELAXED,
+                                    __HIP_MEMORY_SCOPE_AGENT) : ~0u;
+            if (__all(f1 >= value && f2 >= value)) break;
+            __builtin_amdgcn_s_sleep(1);
+            if (++spins > (long)2e8) {
+                if (lane == 0)
+                    __hip_atomic_store(poison, 1u, __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+                ok = 0;
+                break;
+            }
+            if (__hip_atomic_load(poison, __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT)) {
+                ok = 0;
+                break;
+            }
+        }
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+        if (lane == 0) ok_s = ok;
+    }
+    __syncthreads();
+    return ok_s != 0;
+}
+
+// layout of the workspace (int32 words): [0..127] flags, [128] poison,
+// [129..137] legacy barrier words (barrier_bench)
 struct GridBar {
+    unsigned flags[LSTM_MAX_FLAGS];
+    unsigned poison;
     unsigned bucket[8];
     unsigned top;
     unsigned gen;
-    unsigned poison;
+    unsigned gen_b[8];
 };
 
+// legacy counter barrier, kept for the microbench / comparison
 __device__ __forceinline__ bool grid_barrier(GridBar* bar, unsigned epoch,
                                              int nblocks) {
     __syncthreads();
     __shared__ unsigned ok_s;
     if (threadIdx.x == 0) {
         int b = blockIdx.x & 7;
-        int per = (nblocks + 7 - b) >> 3;  // blocks with id%8 == b
+        int per = (nblocks + 7 - b) >> 3;
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
         unsigned prev = __hip_atomic_fetch_add(&bar->bucket[b], 1u,
@@ -92,18 +157,8 @@ __device__ __forceinline__ bool grid_barrier(GridBar* bar, unsigned epoch,
         long spins = 0;
         while (__hip_atomic_load(&bar->gen, __ATOMIC_RELAXED,
                                  __HIP_MEMORY_SCOPE_AGENT) < epoch) {
-            __builtin_amdgcn_s_sleep(8);
-            if (++spins > (long)2e8) {  // bounded spin: poison and bail
-                __hip_atomic_store(&bar->poison, 1u, __ATOMIC_RELAXED,
-                                   __HIP_MEMORY_SCOPE_AGENT);
-                ok = 0;
-                break;
-            }
-            if (__hip_atomic_load(&bar->poison, __ATOMIC_RELAXED,
-                                  __HIP_MEMORY_SCOPE_AGENT)) {
-                ok = 0;
-                break;
-            }
+            __builtin_amdgcn_s_sleep(2);
+            if (++spins > (long)2e8) { ok = 0; break; }
         }
         __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
         ok_s = ok;
@@ -113,12 +168,10 @@ __device__ __forceinline__ bool grid_barrier(GridBar* bar, unsigned epoch,
 }
 
 // ---------------------------------------------------------------------------
-// Forward.  Grid: wgs_per_net = H/8 blocks per network (net1 optional).
-// Each wg: 8 hidden units -> 32 gate columns, W_hh slice (32 x H) in LDS.
-// Per step: gates(B,32) = h_prev(B,H) @ Whh_slice^T + X[t] -> c,h update.
+// Forward
 // ---------------------------------------------------------------------------
 template <int H>
-__global__ __launch_bounds__(256) void lstm_fwd_kernel(
+__global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
     const __hip_bfloat16* __restrict__ X0,    // (B, T, 4H)
     const __hip_bfloat16* __restrict__ X1,    // or null
     const __hip_bfloat16* __restrict__ Whh0,  // (4H, H) row-major
@@ -143,119 +196,161 @@ __global__ __launch_bounds__(256) void lstm_fwd_kernel(
     __hip_bfloat16* Hout = net ? Hout1 : Hout0;
     float* Cout = net ? Cout1 : Cout0;
     __hip_bfloat16* stash = net ? nullptr : stash0;
+    unsigned* flags = bar->flags + net * WGS_PER_NET;
+    unsigned* myflag = &flags[wid];
 
     __shared__ __hip_bfloat16 s_whh[32][H + 8];
+    __shared__ __hip_bfloat16 s_h[64][H + 8];
     __shared__ float s_gates[64][32 + 4];
+    __shared__ float s_c[64][8];
+    __shared__ __hip_bfloat16 s_hrow[64][8];
 
-    // stage the wg's 32 W_hh rows (gate g, unit u0+j -> row g*H + u0 + j;
-    // local col c = g*8 + j)
     for (int e = threadIdx.x * 8; e < 32 * H; e += blockDim.x * 8) {
         int c = e / H;
         int k = e % H;
         int g = c / 8, j = c % 8;
-        *reinterpret_cast<bf16x8*>(&s_whh[c][k]) =
-            lload8(Whh + (long)(g * H + u0 + j) * H + k);
+        lstore8(&s_whh[c][k], lload8(Whh + (long)(g * H + u0 + j) * H + k));
     }
-    // write h0/c0 into the output buffers (the wg's 8 units)
+    // h0 -> Hout[:,0] (this wg's slice); c0 -> LDS-resident cell state
     for (int p = threadIdx.x; p < B * 8; p += blockDim.x) {
         int b = p / 8, j = p % 8;
         int u = u0 + j;
         Hout[((long)b * (T + 1)) * H + u] = f2bf(init[(long)b * H + u]);
         Cout[((long)b * (T + 1)) * H + u] = init[((long)B + b) * H + u];
+        s_c[b][j] = init[((long)B + b) * H + u];
     }
-    if (!grid_barrier(bar, 1, nblocks)) return;
+    publish_slice(myflag, 1u);
 
     const int wave = threadIdx.x / WAVE;
     const int lane = threadIdx.x & (WAVE - 1);
-    const int wrow0 = (wave >> 1) * 32;        // rows 0..63 (batch)
-    const int wcol0 = (wave & 1) * 16;         // cols 0..31 (gate local)
+    const int wrow0 = (wave >> 1) * 32;        // rows (batch)
+    const int wcol0 = (wave & 1) * 16;         // cols (gate local)
     const int frow = lane & 15;
     const int kseg = (lane >> 4) * 8;
 
+    constexpr int CHUNKS = (64 * H) / 8;
+
     for (int t = 0; t < T; ++t) {
-        // gates = h_prev @ Whh_slice^T
+        if (!await_slices(flags, WGS_PER_NET, (unsigned)(t + 1), &bar->poison))
+            return;
+        // bulk-stage h_prev (B x H) into LDS
+        {
+            const long base = (long)t * H;
+#pragma unroll
+            for (int e = threadIdx.x; e < CHUNKS; e += 256) {
+                int row = e / (H / 8);
+                int k8 = (e % (H / 8)) * 8;
+                bf16x8 v = (row < B)
+                    ? lload8(Hout + ((long)row * (T + 1)) * H + base + k8)
+                    : lzero8();
+                lstore8(&s_h[row][k8], v);
+            }
+        }
+        __syncthreads();
+
+        // gates = h_prev @ Whh_slice^T  (all-LDS MFMA)
         f32x4 acc[2] = {};
-#pragma unroll 4
+#pragma unroll
         for (int k0 = 0; k0 < H; k0 += 32) {
             bf16x8 bfr = lload8(&s_whh[wcol0 + frow][k0 + kseg]);
 #pragma unroll
             for (int i = 0; i < 2; ++i) {
-                int row = wrow0 + i * 16 + frow;
-                bf16x8 afr = (row < B)
-                    ? lload8(Hout + ((long)row * (T + 1) + t) * H + k0 + kseg)
-                    : lzero8();
+                bf16x8 afr = lload8(&s_h[wrow0 + i * 16 + frow][k0 + kseg]);
                 acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                     afr, bfr, acc[i], 0, 0, 0);
             }
         }
-        // + X[t], stage to LDS
         {
             int ccol = lane & 15;
             int crow = (lane >> 4) * 4;
 #pragma unroll
             for (int i = 0; i < 2; ++i)
 #pragma unroll
-                for (int r = 0; r < 4; ++r) {
-                    int row = wrow0 + i * 16 + crow + r;
-                    int c = wcol0 + ccol;
-                    if (row < B) {
-                        int g = c / 8, j = c % 8;
-                        float x = bf2f(X[((long)row * T + t) * 4 * H
-                                         + g * H + u0 + j]);
-                        s_gates[row][c] = acc[i][r] + x;
-                    }
-                }
+                for (int r = 0; r < 4; ++r)
+                    s_gates[wrow0 + i * 16 + crow + r][wcol0 + ccol] = acc[i][r];
         }
         __syncthreads();
-        // gate nonlinearities + state advance for (b, j) pairs
+        // + X[t] (vectorized), then nonlinearities + state advance
+        {
+            int row = threadIdx.x / 4;
+            int g = threadIdx.x % 4;
+            if (row < B) {
+                bf16x8 x8 = lload8(X + ((long)row * T + t) * 4 * H + g * H + u0);
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    s_gates[row][g * 8 + j] += (float)x8[j];
+            }
+        }
+        __syncthreads();
         for (int p = threadIdx.x; p < B * 8; p += blockDim.x) {
             int b = p / 8, j = p % 8;
-            int u = u0 + j;
-            long prev_off = ((long)b * (T + 1) + t) * H + u;
-            long cur_off = prev_off + H;
-            float c_prev = Cout[prev_off];
             bool active = t < lens[b];
-            float i_ = 0.f, f_ = 0.f, g_ = 0.f, o_ = 0.f, c, h;
+            float i_ = 0.f, f_ = 0.f, g_ = 0.f, o_ = 0.f;
+            float c = s_c[b][j], h;
             if (active) {
                 i_ = sigmoidf_(s_gates[b][0 + j]);
                 f_ = sigmoidf_(s_gates[b][8 + j]);
                 g_ = tanhf(s_gates[b][16 + j]);
                 o_ = sigmoidf_(s_gates[b][24 + j]);
-                c = f_ * c_prev + i_ * g_;
+                c = f_ * c + i_ * g_;
                 h = o_ * tanhf(c);
             } else {
-                c = c_prev;
-                h = bf2f(Hout[prev_off]);
+                h = (float)*(const __bf16*)&s_h[b][u0 + j];
             }
-            Cout[cur_off] = c;
-            Hout[cur_off] = f2bf(h);
-            if (stash) {
-                long so = ((long)b * T + t) * 4 * H + u;
-                stash[so] = f2bf(i_);
-                stash[so + H] = f2bf(f_);
-                stash[so + 2 * H] = f2bf(g_);
-                stash[so + 3 * H] = f2bf(o_);
+            s_c[b][j] = c;
+            *(__bf16*)&s_hrow[b][j] = (__bf16)h;
+            s_gates[b][0 + j] = i_;
+            s_gates[b][8 + j] = f_;
+            s_gates[b][16 + j] = g_;
+            s_gates[b][24 + j] = o_;
+        }
+        __syncthreads();
+        // vectorized writers: h slice (16 B/row), c (32 B/row), gate stash
+        {
+            int tid = threadIdx.x;
+            if (tid < 64 && tid < B) {
+                long off = ((long)tid * (T + 1) + t + 1) * H + u0;
+                lstore8(Hout + off, *reinterpret_cast<bf16x8*>(&s_hrow[tid][0]));
+            } else if (tid >= 64 && tid < 192) {
+                int b = (tid - 64) / 2, half = (tid - 64) & 1;
+                if (b < B) {
+                    long off = ((long)b * (T + 1) + t + 1) * H + u0 + half * 4;
+                    *reinterpret_cast<float4*>(Cout + off) =
+                        *reinterpret_cast<float4*>(&s_c[b][half * 4]);
+                }
             }
         }
-        if (!grid_barrier(bar, (unsigned)(t + 2), nblocks)) return;
+        if (stash) {
+            int row = threadIdx.x / 4;
+            int g = threadIdx.x % 4;
+            if (row < B) {
+                bf16x8 v;
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    v[j] = (__bf16)s_gates[row][g * 8 + j];
+                lstore8(stash + ((long)row * T + t) * 4 * H + g * H + u0, v);
+            }
+        }
+        publish_slice(myflag, (unsigned)(t + 2));
     }
 }
 
 // ---------------------------------------------------------------------------
-// Backward (online net).  Grid: H/16 wgs, each 16 hidden units; W_hh^T
-// slice (16 x 4H) in LDS.  Reverse loop; dgates -> global stash (zeros at
-// masked steps), dh/dc kept per-wg in LDS.
+// Backward (online net).  H/16 wgs, each 16 hidden units; W_hh^T slice in
+// LDS; dgates_{t+1} streamed through LDS in 256-col pieces; vectorized row
+// staging of stash/C/dHext; flag hand-off on the dgates stream.
 // ---------------------------------------------------------------------------
 template <int H>
-__global__ __launch_bounds__(256) void lstm_bwd_kernel(
+__global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
     const __hip_bfloat16* __restrict__ stash,  // (B, T, 4H) i,f,g,o
     const float* __restrict__ Cout,            // (B, T+1, H)
-    const __hip_bfloat16* __restrict__ Hout,   // (B, T+1, H) (unused, kept)
+    const __hip_bfloat16* __restrict__ Hout,   // (unused)
     const float* __restrict__ dHext,           // (B, T, H) upstream
     const __hip_bfloat16* __restrict__ Whh_bwd,// (H, 4H): W_hh^T row-major
     const int* __restrict__ lens,
     __hip_bfloat16* __restrict__ dgates,       // (B, T, 4H) out
     GridBar* bar, int B, int T, int nblocks) {
+    constexpr int WGS = H / 16;
     const int wid = blockIdx.x;
     const int u0 = wid * 16;
 
@@ -263,37 +358,65 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
     __shared__ float s_dh[64][16 + 1];
     __shared__ float s_dc[64][16 + 1];
     __shared__ float s_rec[64][16 + 1];
+    __shared__ __hip_bfloat16 s_dgout[64][64 + 8];  // this wg's dgates cols
+
+    unsigned* flags = bar->flags;
+    unsigned* myflag = &flags[wid];
 
     for (int e = threadIdx.x * 8; e < 16 * 4 * H; e += blockDim.x * 8) {
         int c = e / (4 * H);
         int k = e % (4 * H);
-        *reinterpret_cast<bf16x8*>(&s_wb[c][k]) =
-            lload8(Whh_bwd + (long)(u0 + c) * 4 * H + k);
+        lstore8(&s_wb[c][k], lload8(Whh_bwd + (long)(u0 + c) * 4 * H + k));
     }
     for (int p = threadIdx.x; p < B * 16; p += blockDim.x) {
         s_dh[p / 16][p % 16] = 0.f;
         s_dc[p / 16][p % 16] = 0.f;
     }
     __syncthreads();
+    if (threadIdx.x == 0)
+        __hip_atomic_store(myflag, 1u, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
 
     const int wave = threadIdx.x / WAVE;
     const int lane = threadIdx.x & (WAVE - 1);
-    const int wrow0 = wave * 16;   // 4 waves x 16 rows = 64 batch rows
+    const int wrow0 = wave * 16;
     const int frow = lane & 15;
     const int kseg = (lane >> 4) * 8;
 
     for (int t = T - 1; t >= 0; --t) {
-        // recurrent contribution: rec(B,16) = dgates_{t+1}(B,4H) @ s_wb^T
+        unsigned need = (unsigned)(T - t);   // pieces published for t+1
+        if (!await_slices(flags, WGS, need, &bar->poison)) return;
+        // rec(B,16) = dgates_{t+1}(B,4H) @ s_wb^T with a 4-deep register
+        // prefetch ring on the dgates stream (loads stay in flight across
+        // MFMAs; no per-piece barriers)
         if (t < T - 1) {
+            const int arow = wrow0 + frow;
+            const __hip_bfloat16* dgrow =
+                dgates + ((long)arow * T + t + 1) * 4 * H;
+            const bool rvalid = arow < B;
             f32x4 acc = {};
-            for (int k0 = 0; k0 < 4 * H; k0 += 32) {
-                bf16x8 bfr = lload8(&s_wb[frow][k0 + kseg]);
-                int row = wrow0 + frow;
-                bf16x8 afr = (row < B)
-                    ? lload8(dgates + ((long)row * T + t + 1) * 4 * H + k0 + kseg)
-                    : lzero8();
-                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr, bfr, acc,
-                                                              0, 0, 0);
+            bf16x8 a0 = rvalid ? lload8(dgrow + 0 * 32 + kseg) : lzero8();
+            bf16x8 a1 = rvalid ? lload8(dgrow + 1 * 32 + kseg) : lzero8();
+            bf16x8 a2 = rvalid ? lload8(dgrow + 2 * 32 + kseg) : lzero8();
+            bf16x8 a3 = rvalid ? lload8(dgrow + 3 * 32 + kseg) : lzero8();
+#pragma unroll 4
+            for (int kc = 0; kc < 4 * H / 32; kc += 4) {
+                bf16x8 b0 = lload8(&s_wb[frow][(kc + 0) * 32 + kseg]);
+                bf16x8 b1 = lload8(&s_wb[frow][(kc + 1) * 32 + kseg]);
+                bf16x8 b2 = lload8(&s_wb[frow][(kc + 2) * 32 + kseg]);
+                bf16x8 b3 = lload8(&s_wb[frow][(kc + 3) * 32 + kseg]);
+                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc, 0, 0, 0);
+                if (kc + 4 < 4 * H / 32)
+                    a0 = rvalid ? lload8(dgrow + (kc + 4) * 32 + kseg) : lzero8();
+                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc, 0, 0, 0);
+                if (kc + 5 < 4 * H / 32)
+                    a1 = rvalid ? lload8(dgrow + (kc + 5) * 32 + kseg) : lzero8();
+                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b2, acc, 0, 0, 0);
+                if (kc + 6 < 4 * H / 32)
+                    a2 = rvalid ? lload8(dgrow + (kc + 6) * 32 + kseg) : lzero8();
+                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b3, acc, 0, 0, 0);
+                if (kc + 7 < 4 * H / 32)
+                    a3 = rvalid ? lload8(dgrow + (kc + 7) * 32 + kseg) : lzero8();
             }
             int ccol = lane & 15;
             int crow = (lane >> 4) * 4;
@@ -304,7 +427,6 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
             }
         }
         __syncthreads();
-
         for (int p = threadIdx.x; p < B * 16; p += blockDim.x) {
             int b = p / 16, jl = p % 16;
             int u = u0 + jl;
@@ -326,6 +448,7 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
             }
             bool active = t < lens[b];
             long so = ((long)b * T + t) * 4 * H + u;
+            float di = 0.f, df = 0.f, dg = 0.f, do_ = 0.f;
             if (active) {
                 float i_ = bf2f(stash[so]);
                 float f_ = bf2f(stash[so + H]);
@@ -334,28 +457,89 @@ __global__ __launch_bounds__(256) void lstm_bwd_kernel(
                 float tc = tanhf(Cout[((long)b * (T + 1) + t + 1) * H + u]);
                 float c_prev = Cout[((long)b * (T + 1) + t) * H + u];
                 float dc = dc_in + dh * o_ * (1.f - tc * tc);
-                dgates[so] = f2bf(dc * g_ * i_ * (1.f - i_));
-                dgates[so + H] = f2bf(dc * c_prev * f_ * (1.f - f_));
-                dgates[so + 2 * H] = f2bf(dc * i_ * (1.f - g_ * g_));
-                dgates[so + 3 * H] = f2bf(dh * tc * o_ * (1.f - o_));
+                di = dc * g_ * i_ * (1.f - i_);
+                df = dc * c_prev * f_ * (1.f - f_);
+                dg = dc * i_ * (1.f - g_ * g_);
+                do_ = dh * tc * o_ * (1.f - o_);
                 s_dh[b][jl] = dh;
                 s_dc[b][jl] = dc;
             } else {
-                dgates[so] = (__hip_bfloat16)0.f;
-                dgates[so + H] = (__hip_bfloat16)0.f;
-                dgates[so + 2 * H] = (__hip_bfloat16)0.f;
-                dgates[so + 3 * H] = (__hip_bfloat16)0.f;
                 s_dh[b][jl] = dh;
                 s_dc[b][jl] = dc_in;
             }
+            *(__bf16*)&s_dgout[b][0 + jl] = (__bf16)di;
+            *(__bf16*)&s_dgout[b][16 + jl] = (__bf16)df;
+            *(__bf16*)&s_dgout[b][32 + jl] = (__bf16)dg;
+            *(__bf16*)&s_dgout[b][48 + jl] = (__bf16)do_;
         }
-        if (!grid_barrier(bar, (unsigned)(T - t), nblocks)) return;
+        __syncthreads();
+        // vectorized dgates writes: thread (b, g, half) -> 16 B
+        {
+            int tid = threadIdx.x;
+            int b = tid / 4;
+            int gh = tid % 4;
+            if (b < B) {
+#pragma unroll
+                for (int rep = 0; rep < 2; ++rep) {
+                    int g = gh;
+                    int half = rep;
+                    long off = ((long)b * T + t) * 4 * H + g * H + u0 + half * 8;
+                    lstore8(dgates + off,
+                            *reinterpret_cast<bf16x8*>(&s_dgout[b][g * 16 + half * 8]));
+                }
+            }
+        }
+        publish_slice(myflag, need + 1);
+    }
+}
+
+// barrier-only microbench
+__global__ void barrier_bench_kernel(GridBar* bar, int steps, int nblocks) {
+    for (int t = 0; t < steps; ++t)
+        if (!grid_barrier(bar, (unsigned)(t + 1), nblocks)) return;
+}
+
+// handoff-only microbench: prices one flag publish/await round at `nblocks`
+__global__ void handoff_bench_kernel(GridBar* bar, int steps, int nblocks) {
+    unsigned* flags = bar->flags;
+    unsigned* myflag = &flags[blockIdx.x];
+    if (threadIdx.x == 0)
+        __hip_atomic_store(myflag, 1u, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+    for (int t = 0; t < steps; ++t) {
+        if (!await_slices(flags, nblocks, (unsigned)(t + 1), &bar->poison))
+            return;
+        publish_slice(myflag, (unsigned)(t + 2));
     }
 }
 
 // ---------------------------------------------------------------------------
 // Host wrappers
 // ---------------------------------------------------------------------------
+
+static void zero_ws(torch::Tensor& ws, hipStream_t stream) {
+    TORCH_CHECK(ws.numel() * ws.element_size() >= (long)sizeof(GridBar),
+                "barrier workspace too small (need >= 256 int32)");
+    hipMemsetAsync(ws.data_ptr(), 0, sizeof(GridBar), stream);
+}
+
+void barrier_bench(torch::Tensor barrier_ws, int64_t steps, int64_t nblocks) {
+    auto stream = at::cuda::getCurrentCUDAStream();
+    zero_ws(barrier_ws, stream.stream());
+    hipLaunchKernelGGL(barrier_bench_kernel, dim3((int)nblocks), dim3(256), 0,
+                       stream.stream(),
+                       reinterpret_cast<GridBar*>(barrier_ws.data_ptr()),
+                       (int)steps, (int)nblocks);
+}
+
+void handoff_bench(torch::Tensor barrier_ws, int64_t steps, int64_t nblocks) {
+    auto stream = at::cuda::getCurrentCUDAStream();
+    zero_ws(barrier_ws, stream.stream());
+    hipLaunchKernelGGL(handoff_bench_kernel, dim3((int)nblocks), dim3(256), 0,
+                       stream.stream(),
+                       reinterpret_cast<GridBar*>(barrier_ws.data_ptr()),
+                       (int)steps, (int)nblocks);
+}
 
 std::vector<torch::Tensor> lstm_fwd(
     torch::Tensor X0, torch::Tensor X1, torch::Tensor Whh0, torch::Tensor Whh1,
@@ -375,15 +559,12 @@ std::vector<torch::Tensor> lstm_fwd(
     auto C0 = torch::empty({B, T + 1, H}, f32);
     auto H1 = two ? torch::empty({B, T + 1, H}, bf) : torch::Tensor();
     auto C1 = two ? torch::empty({B, T + 1, H}, f32) : torch::Tensor();
-    auto stash = want_stash ? torch::empty({B, T, H4}, bf)
-                            : torch::Tensor();
+    auto stash = want_stash ? torch::empty({B, T, H4}, bf) : torch::Tensor();
 
     int wgs = (int)H / 8;
     int nblocks = wgs * (two ? 2 : 1);
     auto stream = at::cuda::getCurrentCUDAStream();
-    TORCH_CHECK(barrier_ws.numel() * barrier_ws.element_size()
-                >= (long)sizeof(GridBar));
-    hipMemsetAsync(barrier_ws.data_ptr(), 0, sizeof(GridBar), stream.stream());
+    zero_ws(barrier_ws, stream.stream());
 
     auto bp = [](torch::Tensor& t) {
         return t.defined()
@@ -420,7 +601,7 @@ torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
     auto dgates = torch::empty({B, T, H4}, stash.options());
     int nblocks = (int)H / 16;
     auto stream = at::cuda::getCurrentCUDAStream();
-    hipMemsetAsync(barrier_ws.data_ptr(), 0, sizeof(GridBar), stream.stream());
+    zero_ws(barrier_ws, stream.stream());
     hipLaunchKernelGGL((lstm_bwd_kernel<512>), dim3(nblocks), dim3(256), 0,
         stream.stream(),
         reinterpret_cast<const __hip_bfloat16*>(stash.data_ptr()),
