@@ -217,11 +217,11 @@ class HipNetworkEngine:
 
         # ---- inputs ----------------------------------------------------
         obs = batch.obs
-        if obs.shape[2] == self.C:                 # (B,T,C,84,84) -> NHWC
+        if obs.shape[-1] == self.C and obs.shape[2] == 84:   # already HWC
+            obs_hwc = obs.reshape(B * T, 84, 84, self.C)
+        else:                                      # (B,T,C,84,84) -> NHWC
             obs_hwc = obs.permute(0, 1, 3, 4, 2).reshape(B * T, 84, 84, self.C)
             obs_hwc = obs_hwc.contiguous()
-        else:
-            obs_hwc = obs.reshape(B * T, 84, 84, self.C).contiguous()
         la = batch.last_action.to(dev)
         lr = batch.last_reward.to(dev)
         lens = (batch.burn_in_steps + batch.learning_steps

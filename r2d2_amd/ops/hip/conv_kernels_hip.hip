@@ -232,7 +232,10 @@ __global__ __launch_bounds__(256) void conv_dgrad_kernel(
 
 // ---------------------------------------------------------------------------
 // Wgrad: dWt(COUT, K) += sum_rows relu_mask(dY)[row][co] * patch[row][k]
-// LDS-staged 32-row tiles; f32 atomics; db fused.
+// The FULL K extent (<= 576) and full COUT (<= 64) are staged per 32-row
+// tile, so dY and the patches are each read exactly once; per-wave
+// accumulators cover all k-tiles and are flushed with f32 atomics once per
+// chunk.  Fused ReLU mask + bias grad.
 // ---------------------------------------------------------------------------
 template <bool IN_U8, int KH, int KW, int CIN, int S, bool RELU>
 __global__ __launch_bounds__(256) void conv_wgrad_kernel(
@@ -244,88 +247,92 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     int M, int INH, int INW, int OH, int OW, int COUT, int rows_per_chunk) {
     constexpr int K = KH * KW * CIN;
     constexpr int KWC = KW * CIN;
+    constexpr int KHALF = ((K / 2 + 31) / 32) * 32;   // per-wave k extent
+    constexpr int KFRAG = KHALF / 16;                 // 16-col frags per wave
     __shared__ __hip_bfloat16 s_dy[32][64 + 8];
-    __shared__ __hip_bfloat16 s_a[32][64 + 8];
+    __shared__ __hip_bfloat16 s_a[32][K + 8];
     int wave = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
-    int wr = wave >> 1, wc = wave & 1;
+    int wr = wave >> 1, wc = wave & 1;            // co half, k half
     long mstart = (long)blockIdx.x * rows_per_chunk;
     long mend = min((long)M, mstart + rows_per_chunk);
-    long cocol0 = (long)blockIdx.y * 64;   // co tile
-    long kcol0 = (long)blockIdx.z * 64;    // k tile
     int frow = lane & 15;
     int mseg = (lane >> 4) * 8;
 
-    f32x4 acc[2][2] = {};
+    f32x4 acc[2][KFRAG] = {};
     float bias_acc = 0.f;
 
     for (long m0 = mstart; m0 < mend; m0 += 32) {
         __syncthreads();
         {
             int t = threadIdx.x;
-            int mrow = t / 8;
-            int col = (t % 8) * 8;
-            long gm = m0 + mrow;
-            bf16x8 v = czero();
-            if (gm < mend) {
-#pragma unroll
-                for (int e = 0; e < 8; ++e) {
-                    long c = cocol0 + col + e;
-                    float g = (c < COUT) ? bf2f(dY[gm * COUT + c]) : 0.f;
-                    if (RELU) {
-                        float m_ = (c < COUT) ? bf2f(act[gm * COUT + c]) : 0.f;
-                        g = (m_ > 0.f) ? g : 0.f;
-                    }
-                    v[e] = (__bf16)g;
-                }
-            }
-            *reinterpret_cast<bf16x8*>(&s_dy[mrow][col]) = v;
-
-            bf16x8 w = czero();
-            if (gm < mend && kcol0 + col < K) {
-                long n = gm / (OH * OW);
-                int p = (int)(gm % (OH * OW));
-                int oy = p / OW, ox = p % OW;
-                long base = ((n * INH + (long)oy * S) * INW + (long)ox * S) * CIN;
-                int k = (int)kcol0 + col;
-                int dy = k / KWC, rem = k % KWC;
-                long off = base + (long)dy * INW * CIN + rem;
-                // 8 contiguous k (never crosses a dy boundary: KWC % 8 == 0)
-                if (IN_U8)
-                    w = load_dequant8(
-                        reinterpret_cast<const unsigned char*>(in) + off);
-                else
-                    w = cload_bf16x8(
-                        reinterpret_cast<const __hip_bfloat16*>(in) + off);
-                if (kcol0 + col + 8 > K) {
+            {
+                int mrow = t / 8;
+                int col = (t % 8) * 8;
+                long gm = m0 + mrow;
+                bf16x8 v = czero();
+                if (gm < mend) {
 #pragma unroll
                     for (int e = 0; e < 8; ++e) {
-                        float vv = (float)w[e];
-                        w[e] = (__bf16)((kcol0 + col + e < K) ? vv : 0.f);
+                        long c = col + e;
+                        float g = (c < COUT) ? bf2f(dY[gm * COUT + c]) : 0.f;
+                        if (RELU) {
+                            float m_ = (c < COUT) ? bf2f(act[gm * COUT + c]) : 0.f;
+                            g = (m_ > 0.f) ? g : 0.f;
+                        }
+                        v[e] = (__bf16)g;
                     }
                 }
+                *reinterpret_cast<bf16x8*>(&s_dy[mrow][col]) = v;
             }
-            *reinterpret_cast<bf16x8*>(&s_a[mrow][col]) = w;
+            for (int e8 = t; e8 < 32 * (K / 8); e8 += 256) {
+                int mrow = e8 / (K / 8);
+                int k = (e8 % (K / 8)) * 8;
+                long gm = m0 + mrow;
+                bf16x8 w = czero();
+                if (gm < mend) {
+                    long n = gm / (OH * OW);
+                    int p = (int)(gm % (OH * OW));
+                    int oy = p / OW, ox = p % OW;
+                    long base = ((n * INH + (long)oy * S) * INW
+                                 + (long)ox * S) * CIN;
+                    int dy_ = k / KWC, rem = k % KWC;
+                    long off = base + (long)dy_ * INW * CIN + rem;
+                    if (IN_U8)
+                        w = load_dequant8(
+                            reinterpret_cast<const unsigned char*>(in) + off);
+                    else
+                        w = cload_bf16x8(
+                            reinterpret_cast<const __hip_bfloat16*>(in) + off);
+                }
+                *reinterpret_cast<bf16x8*>(&s_a[mrow][k]) = w;
+            }
         }
         __syncthreads();
 
-        bf16x8 fa[2], fb[2];
+        bf16x8 fa[2];
 #pragma unroll
         for (int i = 0; i < 2; ++i) {
 #pragma unroll
-            for (int e = 0; e < 8; ++e) {
+            for (int e = 0; e < 8; ++e)
                 fa[i][e] = *(const __bf16*)&s_dy[mseg + e][wr * 32 + i * 16 + frow];
-                fb[i][e] = *(const __bf16*)&s_a[mseg + e][wc * 32 + i * 16 + frow];
-            }
         }
 #pragma unroll
-        for (int i = 0; i < 2; ++i)
+        for (int kf = 0; kf < KFRAG; ++kf) {
+            int kcol = wc * KHALF + kf * 16 + frow;
+            bf16x8 fb = czero();
+            if (kcol < K) {
 #pragma unroll
-            for (int j = 0; j < 2; ++j)
-                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    fa[i], fb[j], acc[i][j], 0, 0, 0);
+                for (int e = 0; e < 8; ++e)
+                    fb[e] = *(const __bf16*)&s_a[mseg + e][kcol];
+            }
+#pragma unroll
+            for (int i = 0; i < 2; ++i)
+                acc[i][kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    fa[i], fb, acc[i][kf], 0, 0, 0);
+        }
 
-        if (blockIdx.z == 0 && threadIdx.x < 64) {
+        if (threadIdx.x < 64) {
             int c = threadIdx.x;
             for (int mr = 0; mr < 32; ++mr) bias_acc += bf2f(s_dy[mr][c]);
         }
@@ -336,16 +343,16 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
 #pragma unroll
     for (int i = 0; i < 2; ++i)
 #pragma unroll
-        for (int j = 0; j < 2; ++j)
+        for (int kf = 0; kf < KFRAG; ++kf)
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                long co = cocol0 + wr * 32 + i * 16 + crow + r;
-                long kk = kcol0 + wc * 32 + j * 16 + ccol;
+                long co = wr * 32 + i * 16 + crow + r;
+                long kk = wc * KHALF + kf * 16 + ccol;
                 if (co < COUT && kk < K)
-                    atomicAdd(&dWt[co * K + kk], acc[i][j][r]);
+                    atomicAdd(&dWt[co * K + kk], acc[i][kf][r]);
             }
-    if (blockIdx.z == 0 && threadIdx.x < 64) {
-        long c = cocol0 + threadIdx.x;
+    if (threadIdx.x < 64) {
+        long c = threadIdx.x;
         if (c < COUT) atomicAdd(&db[c], bias_acc);
     }
 }
@@ -425,11 +432,10 @@ std::vector<torch::Tensor> conv_wgrad(torch::Tensor dY, torch::Tensor act,
     long M = N * OH * OW;
     auto dWt = torch::zeros({COUT, K}, dY.options().dtype(torch::kFloat32));
     auto db = torch::zeros({COUT}, dY.options().dtype(torch::kFloat32));
-    long tiles = (long)ccdiv(COUT, 64) * ccdiv(K, 64);
-    long target_chunks = std::max(1L, 2048L / std::max(1L, tiles));
+    long target_chunks = 1024;
     long rows_per_chunk = std::max(32L, (M + target_chunks - 1) / target_chunks);
     rows_per_chunk = ((rows_per_chunk + 31) / 32) * 32;
-    dim3 grid(ccdiv(M, rows_per_chunk), ccdiv(COUT, 64), ccdiv(K, 64));
+    dim3 grid(ccdiv(M, rows_per_chunk));
     auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
     auto* dy = reinterpret_cast<const __hip_bfloat16*>(dY.data_ptr());
     auto* ac = reinterpret_cast<const __hip_bfloat16*>(act.data_ptr());

@@ -92,8 +92,12 @@ class GpuReplayBuffer:
         nseq = block.num_sequences
         dev = self.device
 
-        obs_flat = torch.from_numpy(
-            np.ascontiguousarray(block.obs.reshape(rows, -1)))
+        # store frames HWC (channels innermost) — the conv kernels are NHWC,
+        # so gathered batches feed conv1 with no device-side permute
+        obs_np = block.obs
+        if obs_np.ndim == 4:
+            obs_np = np.ascontiguousarray(obs_np.transpose(0, 2, 3, 1))
+        obs_flat = torch.from_numpy(obs_np.reshape(rows, -1))
         assert obs_flat.dtype == torch.uint8
         self.obs_store[slot, :rows].copy_(obs_flat, non_blocking=True)
         la_idx = torch.from_numpy(
@@ -168,8 +172,9 @@ class GpuReplayBuffer:
         meta_h = meta[:3].cpu()  # sync point (small)
         seg_h = seg.cpu()
         R = int(seg_h[-1])
+        c_, h_, w_ = self.obs_shape
         batch = TrainingBatch(
-            obs=obs.view(B, self.T, *self.obs_shape),
+            obs=obs.view(B, self.T, h_, w_, c_),   # HWC layout
             last_action=la, last_reward=lr, hidden=hid,
             action=act[:R].unsqueeze(1), n_step_reward=nsr[:R], gamma=gam[:R],
             burn_in_steps=meta_h[0].long(), learning_steps=meta_h[1].long(),

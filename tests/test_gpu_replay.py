@@ -121,7 +121,7 @@ def test_gpu_replay_roundtrip():
     batch = replay.sample(16)
     torch.cuda.synchronize()
     B, T = 16, c.seq_len
-    assert batch.obs.shape == (B, T, 4, 84, 84)
+    assert batch.obs.shape == (B, T, 84, 84, 4)  # HWC store
     assert batch.hidden.shape == (2, B, c.hidden_dim)
     R = int(batch.learning_steps.sum())
     assert batch.action.shape == (R, 1)
