@@ -21,6 +21,8 @@ every optimizer step (`refresh_online`), and on target-net sync
 
 from typing import Optional
 
+import os
+
 import numpy as np
 import torch
 
@@ -150,8 +152,52 @@ class HipNetworkEngine:
         self.bar = torch.zeros(256, dtype=torch.int32, device=self.device)
         self._empty = torch.Tensor()
 
+        # flat parameter/grad/Adam-moment buffers: module params become views
+        # so the whole network is clipped + stepped by TWO kernels
+        # (ops/hip/optim_kernels.hip) and DP all-reduce is ONE collective.
+        params = [p for p in online_net.parameters()]
+        n_total = sum(p.numel() for p in params)
+        dev = self.device
+        self.flat_param = torch.empty(n_total, dtype=torch.float32, device=dev)
+        self.flat_grad = torch.zeros(n_total, dtype=torch.float32, device=dev)
+        self.exp_avg = torch.zeros(n_total, dtype=torch.float32, device=dev)
+        self.exp_avg_sq = torch.zeros(n_total, dtype=torch.float32, device=dev)
+        self.norm_buf = torch.zeros(1, dtype=torch.float32, device=dev)
+        self.adam_t = 0
+        ofs = 0
+        for p in params:
+            n = p.numel()
+            self.flat_param[ofs:ofs + n].copy_(p.data.to(dev).view(-1))
+            p.data = self.flat_param[ofs:ofs + n].view(p.shape)
+            p.grad = self.flat_grad[ofs:ofs + n].view(p.shape)
+            ofs += n
+        self.timing = bool(os.environ.get("R2D2_ENGINE_TIMING"))
+        self._events = []
+
+    def _mark(self, name):
+        if self.timing:
+            ev = torch.cuda.Event(enable_timing=True)
+            ev.record()
+            self._events.append((name, ev))
+
+    def timing_report(self):
+        torch.cuda.synchronize()
+        out = []
+        for (n0, e0), (n1, e1) in zip(self._events, self._events[1:]):
+            out.append((n1, e0.elapsed_time(e1)))
+        self._events.clear()
+        return out
+
     def refresh_online(self):
         self.online.refresh()
+
+    def optimizer_step(self, lr, eps, max_norm, betas=(0.9, 0.999)):
+        """Fused grad-norm clip + Adam over the flat buffers (K12/K13)."""
+        self.adam_t += 1
+        self.m.grad_sumsq(self.flat_grad, self.norm_buf)
+        self.m.adam_step(self.flat_param, self.flat_grad, self.exp_avg,
+                         self.exp_avg_sq, self.norm_buf, max_norm, lr,
+                         betas[0], betas[1], eps, self.adam_t)
 
     def refresh_target(self):
         self.target.refresh()
@@ -228,16 +274,21 @@ class HipNetworkEngine:
                 + batch.forward_steps).to(torch.int32).to(dev)
         init = batch.hidden.float().contiguous()   # (2, B, H)
 
+        self._mark("start")
         # ---- forward ---------------------------------------------------
         lat_o, enc_stash = self._encoder_fwd(self.online, obs_hwc)
+        self._mark("enc_online")
         lat_t, _ = self._encoder_fwd(self.target, obs_hwc)
+        self._mark("enc_target")
         rin_o, X_o = self._lstm_input(self.online, lat_o, la, lr)
         _, X_t = self._lstm_input(self.target, lat_t, la, lr)
+        self._mark("lstm_input")
         Xo = X_o.view(B, T, 4 * H)
         Xt = X_t.view(B, T, 4 * H)
         Ho, Co, Ht, Ct, stash = m.lstm_fwd(
             Xo, Xt, self.online.whh_t, self.target.whh_t,
             init, init, lens, self.bar, True)
+        self._mark("lstm_fwd")
 
         lp, tp, lbt = self._positions(batch.burn_in_steps,
                                       batch.learning_steps,
@@ -256,6 +307,7 @@ class HipNetworkEngine:
         q_cat, (adv1_o, val1_o) = self._heads_fwd(self.online, h_cat)
         q_learn, q_online_tgt = q_cat[:R], q_cat[R:]
         q_tgt, _ = self._heads_fwd(self.target, h_tgt_t)
+        self._mark("heads")
 
         # ---- fused loss + priorities ----------------------------------
         seg = torch.zeros(B + 1, dtype=torch.int32)
@@ -267,6 +319,7 @@ class HipNetworkEngine:
             batch.is_weights, c.rescale_eps, c.huber_kappa,
             0 if c.loss_fn == "mse" else 1)
         prio = m.segment_priority(abs_td, seg, c.prio_eta)
+        self._mark("loss")
 
         # ---- backward --------------------------------------------------
         ON = self.online
@@ -292,9 +345,11 @@ class HipNetworkEngine:
         tgt_bt = tp_t - (tp_t // (T + 1)) - 1       # b*(T+1)+t+1 -> b*T + t
         dHext.index_add_(0, tgt_bt, dh_rows[R:])
         dHext = dHext.view(B, T, H)
+        self._mark("head_bwd")
 
         dgates = m.lstm_bwd(stash, Co, Ho, dHext.contiguous(), ON.whh_bwd,
                             lens, self.bar)
+        self._mark("lstm_bwd")
         dgates_flat = dgates.view(B * T, 4 * H)
 
         h_prev = Ho[:, :T].reshape(B * T, H).contiguous()
@@ -303,6 +358,7 @@ class HipNetworkEngine:
                                          False, True)
         drin = m.gemm_dgrad(dgates_flat, self._empty, ON.wih_kn, False)
         dlat = drin[:, :512].contiguous()
+        self._mark("lstm_wgrads")
 
         a1, a2, a3, flat = enc_stash
         lat_bf = lat_o  # forward output (relu mask source)
@@ -311,6 +367,7 @@ class HipNetworkEngine:
 
         M = B * T
         # conv3 backward
+        self._mark("fc_bwd")
         dW3, db3 = m.conv_wgrad(dflat.view(M * 49, 64), a3, a2, 3,
                                 M, 9, 9, 7, 7, 64, 9 * 64)
         d3m = (dflat.view(M * 49, 64)
@@ -336,13 +393,14 @@ class HipNetworkEngine:
         # conv1 wgrad (no dgrad: input is data)
         dW1, db1 = m.conv_wgrad(d_a1.view(M * 400, 32), a1, obs_hwc, 1,
                                 M, 84, 84, 20, 20, 32, 8 * 8 * self.C)
+        self._mark("conv_bwd")
 
         # ---- write grads into the nn.Module (f32) ----------------------
         net = self.online_net
         enc = net.encoder
 
         def setg(p, g):
-            p.grad = g.to(p.dtype)
+            p.grad.copy_(g.reshape(p.shape))
 
         def conv_grad(dwt, cout, kh, kw, cin):
             return dwt.view(cout, kh, kw, cin).permute(0, 3, 1, 2).contiguous()
@@ -369,4 +427,5 @@ class HipNetworkEngine:
         setg(net.value[2].weight, dWv2[:1].contiguous())
         setg(net.value[2].bias, dbv2[:1].contiguous())
 
+        self._mark("grad_write")
         return loss.squeeze(0), prio

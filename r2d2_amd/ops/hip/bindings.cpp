@@ -70,6 +70,13 @@ torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
                        torch::Tensor Whh_bwd, torch::Tensor lens,
                        torch::Tensor barrier_ws);
 
+// optim_kernels.hip
+torch::Tensor grad_sumsq(torch::Tensor grad, torch::Tensor norm_buf);
+void adam_step(torch::Tensor param, torch::Tensor grad, torch::Tensor m,
+               torch::Tensor v, torch::Tensor norm_buf, double max_norm,
+               double lr, double beta1, double beta2, double eps,
+               int64_t step);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "r2d2_amd gfx950 HIP kernels";
     m.def("fused_double_q_loss", &fused_double_q_loss,
@@ -98,4 +105,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("handoff_bench", &handoff_bench, "producer-flag handoff microbench");
     m.def("dueling_combine", &dueling_combine, "q = V + A - mean(A)");
     m.def("dueling_combine_bwd", &dueling_combine_bwd, "dueling combine backward");
+    m.def("grad_sumsq", &grad_sumsq, "flat gradient squared-norm reduction");
+    m.def("adam_step", &adam_step,
+          "fused multi-tensor clip + Adam on the flat parameter buffer");
 }

@@ -563,13 +563,16 @@ class Learner:
         batch.to(self.device)
 
         if self.engine is not None:
-            # full HIP path: manual backward fills .grads directly
+            # full HIP path: manual backward fills the flat grad buffer
             loss, prio = self.engine.train_step(batch)
             if self.reducer is not None:
-                self.reducer.reduce_all()
-            nn.utils.clip_grad_norm_(self.online_net.parameters(), self.grad_norm)
-            self.optimizer.step()
-            self.optimizer.zero_grad(set_to_none=True)
+                import torch.distributed as dist
+                self.engine.flat_grad.div_(self.reducer.world_size)
+                dist.all_reduce(self.engine.flat_grad)
+            self.engine.optimizer_step(
+                lr=self.optimizer.param_groups[0]["lr"],
+                eps=self.optimizer.param_groups[0]["eps"],
+                max_norm=self.grad_norm)
             self.engine.refresh_online()
             self.num_updates += 1
             return loss, prio
@@ -649,7 +652,9 @@ class Learner:
 
     def save(self, start_time):
         os.makedirs("models", exist_ok=True)
-        torch.save((self.online_net.state_dict(), self.num_updates,
+        state = {k: v.detach().clone() for k, v in
+                 self.online_net.state_dict().items()}
+        torch.save((state, self.num_updates,
                     self.env_steps, (time.time() - start_time) / 60),
                    os.path.join("models", f"{self.game_name}{self.num_updates}.pth"))
 
