@@ -63,7 +63,7 @@ __device__ __forceinline__ bf16x8 load_dequant8(const unsigned char* p) {
 // ---------------------------------------------------------------------------
 // Forward
 // ---------------------------------------------------------------------------
-template <bool IN_U8, int KH, int KW, int CIN, int S, bool RELU>
+template <bool IN_U8, int KH, int KW, int CIN, int S, int NCOL, bool RELU>
 __global__ __launch_bounds__(256) void conv_fwd_kernel(
     const void* __restrict__ in,            // (N, INH, INW, CIN)
     const __hip_bfloat16* __restrict__ Wt,  // (COUT, K)
@@ -74,8 +74,11 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
     constexpr int KWC = KW * CIN;
     int wave = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
-    int wr = wave >> 1, wc = wave & 1;
-    long row0 = (long)blockIdx.x * 64 + wr * 32;
+    // NCOL == 32 (e.g. conv1 COUT=32): 4 waves stack on the M dim so no
+    // wave computes guarded-away columns; else classic 2x2
+    int wr = (NCOL == 32) ? wave : (wave >> 1);
+    int wc = (NCOL == 32) ? 0 : (wave & 1);
+    long row0 = (long)blockIdx.x * (NCOL == 32 ? 128 : 64) + wr * 32;
     long col0 = (long)blockIdx.y * 64 + wc * 32;
     int frow = lane & 15;
     int kseg = (lane >> 4) * 8;
@@ -152,7 +155,7 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
 // y = y0 + yy*S.  dYp is the zero-PADDED upstream gradient (pre-masked by
 // the ReLU of this conv's output).
 // ---------------------------------------------------------------------------
-template <int TAPS>
+template <int TAPS, int NCOL>
 __global__ __launch_bounds__(256) void conv_dgrad_kernel(
     const __hip_bfloat16* __restrict__ dYp,  // (N, PH, PW, COUT)
     const __hip_bfloat16* __restrict__ Wd,   // (CIN, TAPS*COUT)
@@ -163,8 +166,9 @@ __global__ __launch_bounds__(256) void conv_dgrad_kernel(
     const int K = TAPS * COUT;
     int wave = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
-    int wr = wave >> 1, wc = wave & 1;
-    long row0 = (long)blockIdx.x * 64 + wr * 32;
+    int wr = (NCOL == 32) ? wave : (wave >> 1);
+    int wc = (NCOL == 32) ? 0 : (wave & 1);
+    long row0 = (long)blockIdx.x * (NCOL == 32 ? 128 : 64) + wr * 32;
     long col0 = (long)blockIdx.y * 64 + wc * 32;
     int frow = lane & 15;
     int kseg = (lane >> 4) * 8;
@@ -237,7 +241,7 @@ __global__ __launch_bounds__(256) void conv_dgrad_kernel(
 // accumulators cover all k-tiles and are flushed with f32 atomics once per
 // chunk.  Fused ReLU mask + bias grad.
 // ---------------------------------------------------------------------------
-template <bool IN_U8, int KH, int KW, int CIN, int S, bool RELU>
+template <bool IN_U8, int KH, int KW, int CIN, int S, int NCOT, bool RELU>
 __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     const __hip_bfloat16* __restrict__ dY,   // (M, COUT)
     const __hip_bfloat16* __restrict__ act,  // (M, COUT) forward output
@@ -247,18 +251,23 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     int M, int INH, int INW, int OH, int OW, int COUT, int rows_per_chunk) {
     constexpr int K = KH * KW * CIN;
     constexpr int KWC = KW * CIN;
-    constexpr int KHALF = ((K / 2 + 31) / 32) * 32;   // per-wave k extent
+    // NCOT = co tiles of 32 (1 when COUT<=32): waves split K 4/NCOT ways so
+    // no wave computes guarded-away co columns
+    constexpr int NKW = 4 / NCOT;                     // k splits
+    constexpr int KHALF = ((K / NKW + 31) / 32) * 32; // per-wave k extent
     constexpr int KFRAG = KHALF / 16;                 // 16-col frags per wave
     __shared__ __hip_bfloat16 s_dy[32][64 + 8];
     __shared__ __hip_bfloat16 s_a[32][K + 8];
     int wave = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
-    int wr = wave >> 1, wc = wave & 1;            // co half, k half
+    int wr = (NCOT == 1) ? 0 : (wave >> 1);           // co tile
+    int wc = (NCOT == 1) ? wave : (wave & 1);         // k split
     long mstart = (long)blockIdx.x * rows_per_chunk;
     long mend = min((long)M, mstart + rows_per_chunk);
     int frow = lane & 15;
     int mseg = (lane >> 4) * 8;
 
+    // co extent per wave is always 32 (2 fragments of 16)
     f32x4 acc[2][KFRAG] = {};
     float bias_acc = 0.f;
 
@@ -372,19 +381,22 @@ torch::Tensor conv_fwd(torch::Tensor in, torch::Tensor Wt, torch::Tensor bias,
     long COUT = Wt.size(0);
     auto out = torch::empty({M, COUT},
                             in.options().dtype(torch::kBFloat16));
-    dim3 grid(ccdiv(M, 64), ccdiv(COUT, 64));
+    dim3 grid(ccdiv(M, COUT <= 32 ? 128 : 64), ccdiv(COUT, 64));
     auto stream = at::cuda::getCurrentCUDAStream();
     auto* w = reinterpret_cast<const __hip_bfloat16*>(Wt.data_ptr());
     auto* o = reinterpret_cast<__hip_bfloat16*>(out.data_ptr());
     const float* b = bias.data_ptr<float>();
     const void* x = in.data_ptr();
 #define CLAUNCH(U8, KH_, KW_, CIN_, S_)                                        \
-    hipLaunchKernelGGL((conv_fwd_kernel<U8, KH_, KW_, CIN_, S_, true>), grid,  \
+    hipLaunchKernelGGL((conv_fwd_kernel<U8, KH_, KW_, CIN_, S_, 64, true>),    \
+                       grid,                                                   \
                        dim3(256), 0, stream.stream(), x, w, b, o, (int)M,      \
                        (int)INH, (int)INW, (int)OH, (int)OW, (int)COUT)
     if (conv_id == 1) {
         TORCH_CHECK(in.dtype() == torch::kUInt8);
-        CLAUNCH(true, 8, 8, 4, 4);
+        hipLaunchKernelGGL((conv_fwd_kernel<true, 8, 8, 4, 4, 32, true>), grid,
+                           dim3(256), 0, stream.stream(), x, w, b, o, (int)M,
+                           (int)INH, (int)INW, (int)OH, (int)OW, (int)COUT);
     } else if (conv_id == 2) {
         TORCH_CHECK(in.dtype() == torch::kBFloat16);
         CLAUNCH(false, 4, 4, 32, 2);
@@ -406,14 +418,20 @@ torch::Tensor conv_dgrad(torch::Tensor dYp, torch::Tensor Wd, torch::Tensor taps
     long XX = (XW - 1 - x0) / S + 1;
     long Mc = N * YY * XX;
     long TAPS = taps.size(0);
-    dim3 grid(ccdiv(Mc, 64), ccdiv(CIN, 64));
+    dim3 grid(ccdiv(Mc, CIN <= 32 ? 128 : 64), ccdiv(CIN, 64));
     auto stream = at::cuda::getCurrentCUDAStream();
     auto* dy = reinterpret_cast<const __hip_bfloat16*>(dYp.data_ptr());
     auto* w = reinterpret_cast<const __hip_bfloat16*>(Wd.data_ptr());
     auto* dx = reinterpret_cast<__hip_bfloat16*>(dX.data_ptr());
     const int* tp = taps.data_ptr<int>();
 #define DLAUNCH(T)                                                             \
-    hipLaunchKernelGGL((conv_dgrad_kernel<T>), grid, dim3(256), 0,             \
+    if (CIN <= 32)                                                             \
+        hipLaunchKernelGGL((conv_dgrad_kernel<T, 32>), grid, dim3(256), 0,     \
+                       stream.stream(), dy, w, dx, tp, (int)Mc, (int)YY,       \
+                       (int)XX, (int)y0, (int)x0, (int)S, (int)pad, (int)PH,   \
+                       (int)PW, (int)COUT, (int)XH, (int)XW, (int)CIN);        \
+    else                                                                       \
+        hipLaunchKernelGGL((conv_dgrad_kernel<T, 64>), grid, dim3(256), 0,     \
                        stream.stream(), dy, w, dx, tp, (int)Mc, (int)YY,       \
                        (int)XX, (int)y0, (int)x0, (int)S, (int)pad, (int)PH,   \
                        (int)PW, (int)COUT, (int)XH, (int)XW, (int)CIN)
@@ -440,15 +458,16 @@ std::vector<torch::Tensor> conv_wgrad(torch::Tensor dY, torch::Tensor act,
     auto* dy = reinterpret_cast<const __hip_bfloat16*>(dY.data_ptr());
     auto* ac = reinterpret_cast<const __hip_bfloat16*>(act.data_ptr());
     const void* x = in.data_ptr();
-#define WLAUNCH(U8, KH_, KW_, CIN_, S_)                                        \
-    hipLaunchKernelGGL((conv_wgrad_kernel<U8, KH_, KW_, CIN_, S_, true>),      \
+#define WLAUNCH(U8, KH_, KW_, CIN_, S_, NCOT_)                                 \
+    hipLaunchKernelGGL((conv_wgrad_kernel<U8, KH_, KW_, CIN_, S_, NCOT_,       \
+                                          true>),                              \
                        grid, dim3(256), 0, stream.stream(), dy, ac, x,         \
                        dWt.data_ptr<float>(), db.data_ptr<float>(), (int)M,    \
                        (int)INH, (int)INW, (int)OH, (int)OW, (int)COUT,        \
                        (int)rows_per_chunk)
-    if (conv_id == 1) WLAUNCH(true, 8, 8, 4, 4);
-    else if (conv_id == 2) WLAUNCH(false, 4, 4, 32, 2);
-    else if (conv_id == 3) WLAUNCH(false, 3, 3, 64, 1);
+    if (conv_id == 1) WLAUNCH(true, 8, 8, 4, 4, 1);
+    else if (conv_id == 2) WLAUNCH(false, 4, 4, 32, 2, 2);
+    else if (conv_id == 3) WLAUNCH(false, 3, 3, 64, 1, 2);
     else TORCH_CHECK(false, "unknown conv_id");
 #undef WLAUNCH
     return {dWt, db};
