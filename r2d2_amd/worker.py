@@ -467,7 +467,8 @@ class Learner:
                  grad_norm: Optional[float] = None, lr: Optional[float] = None,
                  eps: Optional[float] = None, game_name: Optional[str] = None,
                  target_net_update_interval: Optional[int] = None,
-                 save_interval: Optional[int] = None):
+                 save_interval: Optional[int] = None,
+                 model_dir: str = "models"):
         c = cfg.get()
         self.cfg = c
         self.device = torch.device(
@@ -497,6 +498,7 @@ class Learner:
                                            or c.target_net_update_interval)
         self.save_interval = save_interval or c.save_interval
         self.game_name = game_name or c.game_name
+        self.model_dir = model_dir
         self.shared_model = model
         self.batched_data: List[TrainingBatch] = []
         self.amp = c.amp and self.device.type == "cuda" and c.dtype == "bf16"
@@ -629,8 +631,8 @@ class Learner:
 
     def run(self):
         threading.Thread(target=self._prefetch_loop, daemon=True).start()
-        start_time = time.time()
-        os.makedirs("models", exist_ok=True)
+        start_time = time.time() - self._resumed_minutes * 60.0
+        os.makedirs(self.model_dir, exist_ok=True)
         while self.num_updates < self.cfg.training_steps:
             while not self.batched_data:
                 time.sleep(0.05)
@@ -651,12 +653,64 @@ class Learner:
                 self.save(start_time)
 
     def save(self, start_time):
-        os.makedirs("models", exist_ok=True)
-        state = {k: v.detach().clone() for k, v in
+        """Write the reference 4-tuple checkpoint (worker.py:380-381 /
+        test.py:27) plus a ``.train.pth`` sidecar (optimizer + target net)
+        that enables exact resume — the reference has no resume path
+        (SURVEY §5); the sidecar adds one without changing the 4-tuple
+        contract the eval harness consumes."""
+        os.makedirs(self.model_dir, exist_ok=True)
+        state = {k: v.detach().cpu().clone() for k, v in
                  self.online_net.state_dict().items()}
-        torch.save((state, self.num_updates,
-                    self.env_steps, (time.time() - start_time) / 60),
-                   os.path.join("models", f"{self.game_name}{self.num_updates}.pth"))
+        minutes = (time.time() - start_time) / 60
+        stem = os.path.join(self.model_dir, f"{self.game_name}{self.num_updates}")
+        torch.save((state, self.num_updates, self.env_steps, minutes),
+                   stem + ".pth")
+        extra = {
+            "optimizer": self.optimizer.state_dict(),
+            "target_net": {k: v.detach().cpu().clone() for k, v in
+                           self.target_net.state_dict().items()},
+            "num_updates": self.num_updates,
+            "env_steps": self.env_steps,
+            "minutes": minutes,
+        }
+        if self.engine is not None:
+            # HIP path: Adam runs in the engine's fused flat-buffer kernels,
+            # not the torch optimizer — persist those moments.
+            extra["engine_adam"] = {
+                "exp_avg": self.engine.exp_avg.cpu().clone(),
+                "exp_avg_sq": self.engine.exp_avg_sq.cpu().clone(),
+                "adam_t": self.engine.adam_t,
+            }
+        torch.save(extra, stem + ".train.pth")
+
+    _resumed_minutes = 0.0
+
+    def load_checkpoint(self, path: str):
+        """Resume from a 4-tuple checkpoint; restores optimizer/target state
+        from the ``.train.pth`` sidecar when present (else target := online,
+        fresh optimizer)."""
+        state, num_updates, env_steps, minutes = torch.load(
+            path, map_location="cpu", weights_only=False)
+        self.online_net.load_state_dict(state)
+        self.num_updates = int(num_updates)
+        self.env_steps = int(env_steps)
+        self._resumed_minutes = float(minutes)
+        sidecar = path[:-len(".pth")] + ".train.pth"
+        if os.path.exists(sidecar):
+            extra = torch.load(sidecar, map_location="cpu", weights_only=False)
+            self.optimizer.load_state_dict(extra["optimizer"])
+            self.target_net.load_state_dict(extra["target_net"])
+            if self.engine is not None and "engine_adam" in extra:
+                ea = extra["engine_adam"]
+                self.engine.exp_avg.copy_(ea["exp_avg"].to(self.device))
+                self.engine.exp_avg_sq.copy_(ea["exp_avg_sq"].to(self.device))
+                self.engine.adam_t = int(ea["adam_t"])
+        else:
+            self.target_net.load_state_dict(state)
+        if self.engine is not None:
+            self.engine.refresh_online()
+            self.engine.refresh_target()
+        self.store_weights()
 
     # reference-compat statics (worker.py:383-390)
     value_rescale = staticmethod(Fn.value_rescale)

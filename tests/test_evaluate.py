@@ -1,0 +1,102 @@
+"""Evaluation harness + checkpoint/resume tests (reference test.py:14-88 and
+the 4-tuple checkpoint contract at reference worker.py:380-381)."""
+
+import os
+import queue
+
+import numpy as np
+import pytest
+import torch
+
+from r2d2_amd import config as cfg
+from r2d2_amd import evaluate
+from r2d2_amd.models.network import Network
+from r2d2_amd.worker import Learner
+
+
+def setup_cartpole(**kw):
+    base = dict(buffer_capacity=640, block_length=40, burn_in_steps=8,
+                learning_steps=8, forward_steps=3, batch_size=8,
+                learning_starts=80, hidden_dim=32, mlp_hidden=32,
+                training_steps=5, num_actors=1, max_episode_steps=50,
+                save_interval=2)
+    base.update(kw)
+    return cfg.apply("cartpole", **base)
+
+
+def make_learner(tmp_path, seed=0):
+    c = setup_cartpole()
+    torch.manual_seed(seed)
+    model = Network(c.action_dim, c.obs_shape, c.hidden_dim, encoder="mlp",
+                    forward_steps=c.forward_steps, mlp_hidden=c.mlp_hidden)
+    learner = Learner(queue.Queue(), queue.Queue(), model,
+                      model_dir=str(tmp_path))
+    return c, learner
+
+
+def test_checkpoint_4tuple_and_sidecar(tmp_path):
+    c, learner = make_learner(tmp_path)
+    learner.num_updates = 2
+    learner.env_steps = 123
+    import time
+    learner.save(time.time() - 90)
+    path = tmp_path / f"{c.game_name}2.pth"
+    assert path.exists()
+    state, n, steps, minutes = torch.load(path, weights_only=False)
+    assert n == 2 and steps == 123
+    assert 1.0 <= minutes <= 2.0
+    # loadable by a fresh Network (the reference test.py:27,30 contract)
+    net = Network(c.action_dim, c.obs_shape, c.hidden_dim, encoder="mlp",
+                  forward_steps=c.forward_steps, mlp_hidden=c.mlp_hidden)
+    net.load_state_dict(state)
+    assert (tmp_path / f"{c.game_name}2.train.pth").exists()
+
+
+def test_resume_roundtrip(tmp_path):
+    from bench import build_batch
+    c, learner = make_learner(tmp_path)
+    dev = torch.device("cpu")
+    batch = build_batch(c, dev, seed=3)
+    for _ in range(3):
+        learner.train_step(batch)
+    learner.num_updates = 2
+    learner.env_steps = 50
+    import time
+    learner.save(time.time())
+
+    # fresh learner resumes: same params, same optimizer moments
+    _, learner2 = make_learner(tmp_path, seed=99)
+    learner2.load_checkpoint(str(tmp_path / f"{c.game_name}2.pth"))
+    assert learner2.num_updates == 2 and learner2.env_steps == 50
+    for p1, p2 in zip(learner.online_net.parameters(),
+                      learner2.online_net.parameters()):
+        assert torch.equal(p1, p2)
+    s1 = learner.optimizer.state_dict()["state"]
+    s2 = learner2.optimizer.state_dict()["state"]
+    assert set(s1) == set(s2)
+    for k in s1:
+        assert torch.allclose(s1[k]["exp_avg"], s2[k]["exp_avg"])
+    # training continues identically from the restored state
+    l1, _ = learner.train_step(batch)
+    l2, _ = learner2.train_step(batch)
+    assert torch.allclose(torch.as_tensor(l1), torch.as_tensor(l2))
+
+
+@pytest.mark.timeout(300)
+def test_eval_harness(tmp_path):
+    c, learner = make_learner(tmp_path)
+    import time
+    for n in (2, 4):
+        learner.num_updates = n
+        learner.env_steps = 100 * n
+        learner.save(time.time() - 60)
+    results = evaluate.test(model_dir=str(tmp_path), num_episodes=2,
+                            pool_size=2, out_dir=str(tmp_path))
+    assert len(results) == 2
+    assert [r["num_updates"] for r in results] == [2, 4]
+    for r in results:
+        assert r["env_frames"] == r["env_steps"] * c.frame_skip
+        assert len(r["rewards"]) == 2
+        assert np.isfinite(r["mean_reward"])
+    assert (tmp_path / f"{c.game_name}_eval.csv").exists()
+    assert (tmp_path / f"{c.game_name}_eval.jsonl").exists()
