@@ -206,7 +206,7 @@ def main():
             "dtype": "bf16" if (have_cuda and c.dtype == "bf16") else "fp32",
             "data": "synthetic",
             "config": {
-                "model": "r2d2-nature-cnn-lstm512-dueling",
+                "model": f"r2d2-{c.encoder}-cnn-lstm{c.hidden_dim}-dueling",
                 "global_batch": c.batch_size * n_gpus,
                 "seq_len": c.burn_in_steps + c.learning_steps,
                 "burn_in": c.burn_in_steps,

@@ -70,6 +70,7 @@ class Config:
     use_hip_kernels: bool = True                 # HIP path on GPU; eager is the CPU/golden path
     gpu_replay: bool = True                      # GPU-resident block store + sum-tree
     log_interval: int = 10                       # seconds; reference: config.py (log_interval)
+    metrics_path: Optional[str] = None           # JSONL metrics emit (None = console only)
     batch_queue_size: int = 8
     amp: bool = True
 

@@ -253,7 +253,8 @@ class ReplayBuffer:
     def __init__(self, sample_queue_list, batch_queue, priority_queue,
                  buffer_capacity: Optional[int] = None,
                  alpha: Optional[float] = None, beta: Optional[float] = None,
-                 batch_size: Optional[int] = None, seed: Optional[int] = None):
+                 batch_size: Optional[int] = None, seed: Optional[int] = None,
+                 metrics_path: Optional[str] = None):
         c = cfg.get()
         self.cfg = c
         self.block_len = c.block_length
@@ -285,6 +286,10 @@ class ReplayBuffer:
         self.batch_queue = batch_queue
         self.priority_queue = priority_queue
         self.stop_flag = False
+        # observability: each _log() tick also appends one JSONL record
+        # (SURVEY §5 — the reference only print()s, worker.py:89-111)
+        self.metrics_path = metrics_path or getattr(c, "metrics_path", None)
+        self._t0 = time.time()
 
     def __len__(self):
         return self.size
@@ -306,22 +311,37 @@ class ReplayBuffer:
             time.sleep(log_interval)
 
     def _log(self, interval):
+        rec = {
+            "wall_s": round(time.time() - self._t0, 1),
+            "buffer_size": self.size,
+            "buffer_fill_rate": (self.size - self.last_size) / interval,
+            "env_steps": self.env_steps,
+            "training_steps": self.training_steps,
+        }
         print(f"buffer size: {self.size}")
-        print(f"buffer update speed: {(self.size - self.last_size) / interval}/s")
+        print(f"buffer update speed: {rec['buffer_fill_rate']}/s")
         self.last_size = self.size
         print(f"number of environment steps: {self.env_steps}")
         if self.num_episodes:
-            print(f"average episode return: {self.episode_reward / self.num_episodes:.4f}")
+            rec["avg_episode_return"] = self.episode_reward / self.num_episodes
+            print(f"average episode return: {rec['avg_episode_return']:.4f}")
             self.episode_reward = 0.0
             self.num_episodes = 0
         print(f"number of training steps: {self.training_steps}")
         delta = self.training_steps - self.last_training_steps
+        rec["training_rate"] = delta / interval
+        rec["seq_samples_per_s"] = rec["training_rate"] * self.batch_size
         print(f"training speed: {delta / interval}/s")
         if delta:
-            print(f"loss: {self.sum_loss / delta:.4f}")
+            rec["loss"] = self.sum_loss / delta
+            print(f"loss: {rec['loss']:.4f}")
             self.last_training_steps = self.training_steps
             self.sum_loss = 0.0
         print()
+        if self.metrics_path:
+            import json
+            with open(self.metrics_path, "a") as f:
+                f.write(json.dumps(rec) + "\n")
 
     def _ingest_loop(self):
         """Blocking multiplexed ingest (no busy-spin)."""
@@ -798,3 +818,148 @@ class Actor:
         obs = self.env.reset()
         self.local_buffer.reset(obs)
         return AgentState(torch.from_numpy(obs).unsqueeze(0), self.action_dim)
+
+
+############################## VectorActor ##############################
+
+
+class VectorActor:
+    """Batched-inference actor driver: E environments stepped in lockstep
+    with ONE batched network forward per tick (SURVEY §2.4 — scaling the
+    reference's 8 single-env CPU actor processes, worker.py:500-574, to
+    hundreds of envs by moving inference to a single GPU context).
+
+    Each env keeps its own LocalBuffer, epsilon (the reference ladder,
+    train.py:15-17), and episode; the network forward runs once per tick
+    over the (E, ...) batch — on cuda this is the K15 T=1 fast path and a
+    single H2D frame copy per tick instead of E per-process forwards.
+
+    Block cuts are finished one tick late: the reference finishes a full
+    block with the q-values of the *post-step* state (worker.py:552-554);
+    here that q arrives with the next tick's batched forward, so full
+    buffers are flagged and finished at the top of the next tick with
+    exactly the same values.
+    """
+
+    def __init__(self, epsilons: List[float], model: Network, sample_queues,
+                 num_envs: Optional[int] = None, device: str = "cpu",
+                 seed: Optional[int] = None):
+        c = cfg.get()
+        self.cfg = c
+        E = num_envs or len(epsilons)
+        assert len(epsilons) == E and len(sample_queues) in (1, E)
+        self.E = E
+        self.epsilons = np.asarray(epsilons, dtype=np.float64)
+        self.device = torch.device(device if torch.cuda.is_available()
+                                   or device == "cpu" else "cpu")
+        from .envs import create_env
+        self.envs = [create_env(seed=None if seed is None else seed + i)
+                     for i in range(E)]
+        self.action_dim = self.envs[0].action_dim
+        self.model = Network(self.action_dim, c.obs_shape, c.hidden_dim,
+                             encoder=c.encoder, forward_steps=c.forward_steps,
+                             mlp_hidden=c.mlp_hidden)
+        self.model.load_state_dict(model.state_dict())
+        self.model.to(self.device).eval()
+        self.shared_model = model
+        self.queues = sample_queues
+        self.buffers = [LocalBuffer(self.action_dim) for _ in range(E)]
+        self.rng = np.random.default_rng(seed)
+        self.update_interval = max(1, c.actor_update_interval // E)
+        self.max_episode_steps = c.max_episode_steps
+        self.block_length = c.block_length
+        H = c.hidden_dim
+        # batched device-side inference state
+        self.obs_t = torch.zeros((E,) + tuple(c.obs_shape), dtype=torch.uint8,
+                                 device=self.device)
+        self.la_t = torch.zeros(E, self.action_dim, device=self.device)
+        self.la_t[:, 0] = 1.0
+        self.lr_t = torch.zeros(E, 1, device=self.device)
+        self.h_t = torch.zeros(1, E, H, device=self.device)
+        self.c_t = torch.zeros(1, E, H, device=self.device)
+        self._obs_host = np.zeros((E,) + tuple(c.obs_shape), dtype=np.uint8)
+        self.episode_steps = np.zeros(E, dtype=np.int64)
+        self._pending_finish: List[int] = []
+
+    def _queue(self, i):
+        return self.queues[i % len(self.queues)]
+
+    def _reset_env(self, i):
+        obs = self.envs[i].reset()
+        self.buffers[i].reset(obs)
+        self._obs_host[i] = obs
+        self.obs_t[i] = torch.from_numpy(np.ascontiguousarray(obs)).to(
+            self.device, non_blocking=True)
+        self.la_t[i].zero_(); self.la_t[i, 0] = 1.0
+        self.lr_t[i].zero_()
+        self.h_t[0, i].zero_(); self.c_t[0, i].zero_()
+        self.episode_steps[i] = 0
+
+    def run(self, stop_after_steps: Optional[int] = None):
+        """Drive all envs until stop_after_steps total env steps (None =
+        forever).  Returns total env steps taken."""
+        for i in range(self.E):
+            self._reset_env(i)
+        total_steps = 0
+        tick = 0
+        use_amp = (self.device.type == "cuda" and self.cfg.dtype == "bf16")
+        state = AgentState.__new__(AgentState)  # batched raw-tensor state
+        while stop_after_steps is None or total_steps < stop_after_steps:
+            state.obs = self.obs_t
+            state.last_action = self.la_t
+            state.last_reward = self.lr_t
+            state.hidden_state = (self.h_t, self.c_t)
+            with torch.no_grad():
+                if use_amp:
+                    with torch.autocast("cuda", dtype=torch.bfloat16):
+                        q, (h, c) = self.model(state)
+                    q = q.float()
+                else:
+                    q, (h, c) = self.model(state)
+            q_cpu = q.cpu().numpy()                       # (E, A)
+            hid_cpu = torch.stack((h[0], c[0]), dim=1).float().cpu().numpy()
+
+            for i in self._pending_finish:                # late block cuts
+                self._queue(i).put(self.buffers[i].finish(q_cpu[i]))
+            self._pending_finish.clear()
+
+            greedy = q_cpu.argmax(axis=1)
+            explore = self.rng.random(self.E) < self.epsilons
+            rand_a = self.rng.integers(0, self.action_dim, self.E)
+            actions = np.where(explore, rand_a, greedy)
+
+            rewards = np.zeros(self.E, dtype=np.float32)
+            dones = np.zeros(self.E, dtype=bool)
+            for i, env in enumerate(self.envs):
+                a = int(actions[i])
+                next_obs, r, done, _ = env.step(a)
+                self.buffers[i].add(a, float(r), next_obs, q_cpu[i], hid_cpu[i])
+                self._obs_host[i] = next_obs
+                rewards[i] = r
+                dones[i] = done
+            total_steps += self.E
+            self.episode_steps += 1
+            tick += 1
+
+            # advance batched state (one H2D copy for all frames)
+            self.obs_t.copy_(torch.from_numpy(self._obs_host),
+                             non_blocking=True)
+            self.la_t.zero_()
+            self.la_t[np.arange(self.E), actions] = 1.0
+            self.lr_t.copy_(torch.from_numpy(rewards).unsqueeze(1))
+            self.h_t, self.c_t = h.detach(), c.detach()
+
+            for i in range(self.E):
+                if dones[i]:
+                    data = self.buffers[i].finish()
+                    if self.epsilons[i] > 0.01:
+                        data[2] = None   # only near-greedy actors report
+                    self._queue(i).put(data)
+                    self._reset_env(i)
+                elif (len(self.buffers[i]) == self.block_length
+                      or self.episode_steps[i] >= self.max_episode_steps):
+                    self._pending_finish.append(i)
+
+            if tick % self.update_interval == 0:
+                self.model.load_state_dict(self.shared_model.state_dict())
+        return total_steps
