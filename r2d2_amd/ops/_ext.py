@@ -25,6 +25,7 @@ SOURCES = [
     os.path.join(SRC_DIR, "lstm_kernels.hip"),
     os.path.join(SRC_DIR, "gemm_kernels.hip"),
     os.path.join(SRC_DIR, "conv_kernels.hip"),
+    os.path.join(SRC_DIR, "impala_kernels.hip"),
     os.path.join(SRC_DIR, "optim_kernels.hip"),
 ]
 

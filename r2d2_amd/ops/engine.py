@@ -27,7 +27,9 @@ import numpy as np
 import torch
 
 from .. import config as cfg
+from ..models.encoders import ImpalaCNN
 from . import hip_ops
+from . import impala as impala_ops
 
 PAD_HEAD = 32   # padded output width for the A-dim and V-dim head GEMMs
 KPAD = 32       # LSTM input features padded to a multiple of 32
@@ -55,21 +57,29 @@ class _NetPack:
         to_bf = lambda t: t.detach().to(dev).bfloat16().contiguous()
         f32 = lambda t: t.detach().to(dev).float().contiguous()
 
-        # convs: (COUT, CIN, KH, KW) -> (COUT, KH*KW*CIN)
-        def pack_conv(conv):
-            w = conv.weight.detach()
-            wt = w.permute(0, 2, 3, 1).reshape(w.shape[0], -1)
-            return to_bf(wt), f32(conv.bias)
+        self.impala = isinstance(enc, ImpalaCNN)
+        if self.impala:
+            if hasattr(self, "imp"):
+                self.imp.refresh()
+            else:
+                self.imp = impala_ops.ImpalaPack(enc, dev, self.with_bwd)
+        else:
+            # convs: (COUT, CIN, KH, KW) -> (COUT, KH*KW*CIN)
+            def pack_conv(conv):
+                w = conv.weight.detach()
+                wt = w.permute(0, 2, 3, 1).reshape(w.shape[0], -1)
+                return to_bf(wt), f32(conv.bias)
 
-        self.w1t, self.b1 = pack_conv(enc.conv1)
-        self.w2t, self.b2 = pack_conv(enc.conv2)
-        self.w3t, self.b3 = pack_conv(enc.conv3)
+            self.w1t, self.b1 = pack_conv(enc.conv1)
+            self.w2t, self.b2 = pack_conv(enc.conv2)
+            self.w3t, self.b3 = pack_conv(enc.conv3)
 
-        # FC: torch flattens NCHW (64,7,7); our conv3 output flattens HWC
-        wf = enc.fc.weight.detach()          # (512, 3136) over (c,h,w)
-        wf_hwc = wf.reshape(512, 64, 7, 7).permute(0, 2, 3, 1).reshape(512, 3136)
-        self.wft = to_bf(wf_hwc)
-        self.bf = f32(enc.fc.bias)
+            # FC: torch flattens NCHW (64,7,7); conv3 output flattens HWC
+            wf = enc.fc.weight.detach()          # (512, 3136) over (c,h,w)
+            wf_hwc = (wf.reshape(512, 64, 7, 7).permute(0, 2, 3, 1)
+                      .reshape(512, 3136))
+            self.wft = to_bf(wf_hwc)
+            self.bf = f32(enc.fc.bias)
 
         # LSTM: weight_ih (4H, 512+A+1) padded to KPAD multiple
         wih = net.recurrent.weight_ih_l0.detach()
@@ -100,14 +110,17 @@ class _NetPack:
 
         if self.with_bwd:
             # dgrad prepacks (W stored (K, N))
-            self.wf_kn = self.wft.t().contiguous()             # (3136, 512)
+            if not self.impala:
+                self.wf_kn = self.wft.t().contiguous()         # (3136, 512)
             self.wih_kn = self.wih_t.t().contiguous()          # (kin_pad, 4H)
             self.whh_bwd = self.whh_t.t().contiguous()         # (H, 4H)
             self.wa1_kn = self.wa1t.t().contiguous()
             self.wa2_kn = self.wa2t.t().contiguous()           # (512, PAD)
             self.wv1_kn = self.wv1t.t().contiguous()
             self.wv2_kn = self.wv2t.t().contiguous()
-            # conv dgrad prepacks
+            if self.impala:
+                return
+            # nature conv dgrad prepacks
             w3 = enc.conv3.weight.detach().to(dev)             # (64,64,3,3)
             w3_nhwc = w3.permute(0, 2, 3, 1)                   # (co,dy,dx,ci)
             self.w3d = (w3_nhwc.permute(3, 1, 2, 0)            # (ci,dy,dx,co)
@@ -136,11 +149,12 @@ class HipNetworkEngine:
     def __init__(self, online_net, target_net, device, config=None):
         c = config or cfg.get()
         self.cfg = c
-        assert c.encoder == "nature" and c.hidden_dim == 512, \
-            "HIP engine supports the flagship nature/512 config"
+        assert c.encoder in ("nature", "impala") and c.hidden_dim == 512, \
+            "HIP engine supports the nature/impala 512-hidden configs"
         assert tuple(c.obs_shape[1:]) == (84, 84)
         self.C = c.obs_shape[0]
         assert self.C == 4, "conv1 kernel instantiated for 4 input channels"
+        self.impala = c.encoder == "impala"
         self.A = c.action_dim
         self.H = 512
         self.device = torch.device(device)
@@ -203,9 +217,11 @@ class HipNetworkEngine:
         self.target.refresh()
 
     # ------------------------------------------------------------------
-    def _encoder_fwd(self, pack, obs_hwc_u8):
+    def _encoder_fwd(self, pack, obs_hwc_u8, want_stash=True):
         """obs: (M, 84, 84, C) uint8 -> latent (M, 512) bf16 + stashes."""
         m = self.m
+        if self.impala:
+            return impala_ops.encoder_fwd(m, pack.imp, obs_hwc_u8, want_stash)
         M = obs_hwc_u8.shape[0]
         a1 = m.conv_fwd(obs_hwc_u8, pack.w1t, pack.b1, 1, M, 84, 84, 20, 20, True)
         a2 = m.conv_fwd(a1, pack.w2t, pack.b2, 2, M, 20, 20, 9, 9, True)
@@ -278,7 +294,7 @@ class HipNetworkEngine:
         # ---- forward ---------------------------------------------------
         lat_o, enc_stash = self._encoder_fwd(self.online, obs_hwc)
         self._mark("enc_online")
-        lat_t, _ = self._encoder_fwd(self.target, obs_hwc)
+        lat_t, _ = self._encoder_fwd(self.target, obs_hwc, want_stash=False)
         self._mark("enc_target")
         rin_o, X_o = self._lstm_input(self.online, lat_o, la, lr)
         _, X_t = self._lstm_input(self.target, lat_t, la, lr)
@@ -360,12 +376,20 @@ class HipNetworkEngine:
         dlat = drin[:, :512].contiguous()
         self._mark("lstm_wgrads")
 
+        M = B * T
+        if self.impala:
+            imp_grads = impala_ops.encoder_bwd(m, ON.imp, enc_stash,
+                                               dlat, lat_o)
+            self._mark("conv_bwd")
+            return self._write_grads_common(
+                batch, imp_grads, dWih_pad, dWhh, db_lstm,
+                dWa1, dba1, dWa2, dba2, dWv1, dbv1, dWv2, dbv2, loss, prio)
+
         a1, a2, a3, flat = enc_stash
         lat_bf = lat_o  # forward output (relu mask source)
         dflat = m.gemm_dgrad(dlat, lat_bf, ON.wf_kn, True)
         dWf, dbf = m.gemm_wgrad(dlat, lat_bf, flat, True, True)
 
-        M = B * T
         # conv3 backward
         self._mark("fc_bwd")
         dW3, db3 = m.conv_wgrad(dflat.view(M * 49, 64), a3, a2, 3,
@@ -427,5 +451,34 @@ class HipNetworkEngine:
         setg(net.value[2].weight, dWv2[:1].contiguous())
         setg(net.value[2].bias, dbv2[:1].contiguous())
 
+        self._mark("grad_write")
+        return loss.squeeze(0), prio
+
+    def _write_grads_common(self, batch, imp_grads, dWih_pad, dWhh, db_lstm,
+                            dWa1, dba1, dWa2, dba2, dWv1, dbv1, dWv2, dbv2,
+                            loss, prio):
+        """IMPALA-path grad write: encoder grads via impala_ops, LSTM/head
+        grads identical to the nature path."""
+        net = self.online_net
+        ON = self.online
+        A = self.A
+
+        impala_ops.write_grads(net.encoder, imp_grads)
+
+        def setg(p, g):
+            p.grad.copy_(g.reshape(p.shape))
+
+        setg(net.recurrent.weight_ih_l0, dWih_pad[:, :ON.kin].contiguous())
+        setg(net.recurrent.weight_hh_l0, dWhh)
+        setg(net.recurrent.bias_ih_l0, db_lstm)
+        setg(net.recurrent.bias_hh_l0, db_lstm.clone())
+        setg(net.advantage[0].weight, dWa1)
+        setg(net.advantage[0].bias, dba1)
+        setg(net.advantage[2].weight, dWa2[:A].contiguous())
+        setg(net.advantage[2].bias, dba2[:A].contiguous())
+        setg(net.value[0].weight, dWv1)
+        setg(net.value[0].bias, dbv1)
+        setg(net.value[2].weight, dWv2[:1].contiguous())
+        setg(net.value[2].bias, dbv2[:1].contiguous())
         self._mark("grad_write")
         return loss.squeeze(0), prio

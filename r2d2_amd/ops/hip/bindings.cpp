@@ -70,6 +70,24 @@ torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
                        torch::Tensor Whh_bwd, torch::Tensor lens,
                        torch::Tensor barrier_ws);
 
+// impala_kernels.hip
+void conv3p(torch::Tensor in, torch::Tensor Wt, torch::Tensor bias,
+            torch::Tensor res, torch::Tensor mask, torch::Tensor out,
+            int64_t N, int64_t H, int64_t W, bool relu_in, bool has_bias,
+            int64_t epi);
+std::vector<torch::Tensor> conv3p_wgrad(torch::Tensor dY, torch::Tensor in,
+                                        int64_t N, int64_t H, int64_t W,
+                                        bool relu_in);
+std::vector<torch::Tensor> maxpool3s2_fwd(torch::Tensor in, int64_t N,
+                                          int64_t H, int64_t W);
+void maxpool3s2_bwd(torch::Tensor dOut, torch::Tensor arg, torch::Tensor dIn,
+                    int64_t N, int64_t H, int64_t W, int64_t OH, int64_t OW);
+torch::Tensor pack_frames(torch::Tensor frames, int64_t H, int64_t W);
+torch::Tensor pad2dense(torch::Tensor in, int64_t N, int64_t H, int64_t W,
+                        bool relu);
+void dense2pad_mask(torch::Tensor dflat, torch::Tensor act_pad,
+                    torch::Tensor out, int64_t N, int64_t H, int64_t W);
+
 // optim_kernels.hip
 torch::Tensor grad_sumsq(torch::Tensor grad, torch::Tensor norm_buf);
 void adam_step(torch::Tensor param, torch::Tensor grad, torch::Tensor m,
@@ -105,6 +123,20 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("handoff_bench", &handoff_bench, "producer-flag handoff microbench");
     m.def("dueling_combine", &dueling_combine, "q = V + A - mean(A)");
     m.def("dueling_combine_bwd", &dueling_combine_bwd, "dueling combine backward");
+    m.def("conv3p", &conv3p,
+          "IMPALA 3x3 s1 p1 conv on halo-padded NHWC (fwd & dgrad-as-conv, "
+          "fused relu-in / residual / mask epilogues)");
+    m.def("conv3p_wgrad", &conv3p_wgrad,
+          "IMPALA 3x3 conv backward-weight (padded dY, relu-in patches)");
+    m.def("maxpool3s2_fwd", &maxpool3s2_fwd,
+          "maxpool 3x3 s2 p1 forward + tap argmax");
+    m.def("maxpool3s2_bwd", &maxpool3s2_bwd,
+          "maxpool backward (atomic-free input-side gather)");
+    m.def("pack_frames", &pack_frames,
+          "u8 HWC frames -> halo-padded 8-channel u8");
+    m.def("pad2dense", &pad2dense, "padded NHWC -> dense rows (+relu)");
+    m.def("dense2pad_mask", &dense2pad_mask,
+          "dense grad -> padded, masked by act>0 (relu backward)");
     m.def("grad_sumsq", &grad_sumsq, "flat gradient squared-norm reduction");
     m.def("adam_step", &adam_step,
           "fused multi-tensor clip + Adam on the flat parameter buffer");

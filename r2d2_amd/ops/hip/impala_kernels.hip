@@ -1,0 +1,653 @@
+// IMPALA-deep ResNet encoder kernels (gfx950) — BASELINE.json configs[4]
+// (reference encoder semantics: Espeholt et al. 2018; eager golden in
+// r2d2_amd/models/encoders.py ImpalaCNN).
+//
+// Every 3x3 stride-1 pad-1 conv runs as a VALID conv over activations kept
+// in HBM with a one-pixel zero halo: tensor (N, H+2, W+2, C) NHWC bf16 with
+// real data at [1..H, 1..W].  That removes all boundary divergence from the
+// MFMA inner loop — the same implicit-GEMM scheme as conv_kernels.hip:
+// K-order (ky, kx, c) so each lane's 8 k-elements are contiguous (3*C % 8
+// == 0 for C in {8, 16, 32}; the 4-channel frame input is zero-padded to 8
+// channels by pack_frames).  Dgrad of a 3x3 s1 p1 conv is the SAME kernel
+// over the halo-padded upstream gradient with spatially-flipped transposed
+// prepacked weights, so forward and backward-data share one template.
+//
+// Epilogues fuse the IMPALA pre-activation residual arithmetic:
+//   EPI 0: out = acc (+bias)
+//   EPI 1: out = residual + acc          (second conv of a residual block)
+//   EPI 2: out = (mask > 0) * acc        (ReLU backward through conv input)
+//   EPI 3: out = residual + (mask>0)*acc (residual-block input gradient)
+// RELU_IN applies ReLU on patch load (res-block convs consume relu(x)
+// without materializing it).
+//
+// maxpool 3x3 s2 p1 fwd records a per-output tap argmax (u8) and treats
+// out-of-image taps as -inf (torch semantics); backward is atomic-free:
+// each input pixel gathers from the <=4 windows that can claim it.
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+union ibf8u {
+    bf16x8 v;
+    uint4 u;
+    __bf16 e[8];
+};
+
+__device__ __forceinline__ bf16x8 iload8(const __hip_bfloat16* p) {
+    ibf8u r;
+    r.u = *reinterpret_cast<const uint4*>(p);
+    return r.v;
+}
+
+__device__ __forceinline__ bf16x8 izero() {
+    ibf8u r;
+    r.u = uint4{0, 0, 0, 0};
+    return r.v;
+}
+
+__device__ __forceinline__ bf16x8 irelu8(bf16x8 a) {
+    ibf8u r;
+    r.v = a;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+        float f = bf2f(r.e[i]);
+        r.e[i] = (__bf16)fmaxf(f, 0.f);
+    }
+    return r.v;
+}
+
+__device__ __forceinline__ bf16x8 idequant8(const unsigned char* p) {
+    uint2 raw = *reinterpret_cast<const uint2*>(p);
+    bf16x8 r;
+    const float inv = 1.f / 255.f;
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+        r[i] = (__bf16)(((raw.x >> (8 * i)) & 0xff) * inv);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+        r[4 + i] = (__bf16)(((raw.y >> (8 * i)) & 0xff) * inv);
+    return r;
+}
+
+// ---------------------------------------------------------------------------
+// conv3p: 3x3 valid conv over halo-padded NHWC input, halo-padded output.
+//   in  (N, H+2, W+2, CIN)  bf16 (or u8 for the frame conv)
+//   Wt  (COUT, K=9*CIN)     k-order (ky, kx, c)
+//   out (N, H+2, W+2, COUT) written at the +1 halo offset
+// NCOL 16: each wave 32 rows x 16 cols (1 B frag); NCOL 32: 32x32 (2 B
+// frags).  4 waves always stack the M dim -> 128 rows per workgroup.
+// ---------------------------------------------------------------------------
+template <bool IN_U8, int CIN, int NCOL, bool RELU_IN, bool HAS_BIAS, int EPI>
+__global__ __launch_bounds__(256) void conv3p_kernel(
+    const void* __restrict__ in,
+    const __hip_bfloat16* __restrict__ Wt,
+    const float* __restrict__ bias,
+    const __hip_bfloat16* __restrict__ res,   // padded, EPI 1/3
+    const __hip_bfloat16* __restrict__ mask,  // padded, EPI 2/3
+    __hip_bfloat16* __restrict__ out,
+    int M, int H, int W, int COUT) {
+    constexpr int K = 9 * CIN;
+    constexpr int KROW = 3 * CIN;  // contiguous k per ky
+    const int PW = W + 2;
+    const int PH = H + 2;
+    int wave = threadIdx.x / WAVE;
+    int lane = threadIdx.x & (WAVE - 1);
+    long row0 = (long)blockIdx.x * 128 + wave * 32;
+    int frow = lane & 15;
+    int kseg = (lane >> 4) * 8;
+
+    long abase[2];
+    bool avalid[2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+        long r = row0 + i * 16 + frow;
+        avalid[i] = r < M;
+        if (avalid[i]) {
+            long n = r / (H * W);
+            int p = (int)(r % (H * W));
+            int oy = p / W, ox = p % W;
+            // window top-left in padded coords = (oy, ox)
+            abase[i] = ((n * PH + oy) * PW + ox) * CIN;
+        } else {
+            abase[i] = 0;
+        }
+    }
+
+    constexpr int NB = (NCOL == 32) ? 2 : 1;
+    f32x4 acc[2][NB] = {};
+    for (int k0 = 0; k0 < K; k0 += 32) {
+        int k = k0 + kseg;
+        bool kval = k < K;
+        int dy = kval ? k / KROW : 0;
+        int rem = kval ? k % KROW : 0;
+        long off = (long)dy * PW * CIN + rem;
+        bf16x8 a[2], b[NB];
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            if (avalid[i] && kval) {
+                if (IN_U8)
+                    a[i] = idequant8(
+                        reinterpret_cast<const unsigned char*>(in) + abase[i] + off);
+                else {
+                    a[i] = iload8(
+                        reinterpret_cast<const __hip_bfloat16*>(in) + abase[i] + off);
+                    if (RELU_IN) a[i] = irelu8(a[i]);
+                }
+            } else {
+                a[i] = izero();
+            }
+        }
+#pragma unroll
+        for (int j = 0; j < NB; ++j) {
+            int c = j * 16 + frow;
+            b[j] = (c < COUT && kval) ? iload8(Wt + (long)c * K + k) : izero();
+        }
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+            for (int j = 0; j < NB; ++j)
+                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+
+    int ccol = lane & 15;
+    int crow = (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < NB; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                long rr = row0 + i * 16 + crow + r;
+                int cc = j * 16 + ccol;
+                if (rr < M && cc < COUT) {
+                    long n = rr / (H * W);
+                    int p = (int)(rr % (H * W));
+                    int oy = p / W, ox = p % W;
+                    long oidx = ((n * PH + oy + 1) * PW + ox + 1) * COUT + cc;
+                    float v = acc[i][j][r];
+                    if (HAS_BIAS) v += bias[cc];
+                    if (EPI == 2 || EPI == 3)
+                        v = (bf2f(mask[oidx]) > 0.f) ? v : 0.f;
+                    if (EPI == 1 || EPI == 3) v += bf2f(res[oidx]);
+                    out[oidx] = f2bf(v);
+                }
+            }
+}
+
+// ---------------------------------------------------------------------------
+// conv3p_wgrad: dWt(COUT, K=9*CIN) += dY^T @ patches, dY and input both in
+// the halo-padded layout; RELU_IN applies relu on patch load.  Same LDS
+// full-K staging scheme as conv_kernels.hip conv_wgrad (dY and patches are
+// each read exactly once per 32-row tile).
+// ---------------------------------------------------------------------------
+template <bool IN_U8, int CIN, bool RELU_IN>
+__global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
+    const __hip_bfloat16* __restrict__ dY,   // (N, H+2, W+2, COUT) padded
+    const void* __restrict__ in,             // (N, H+2, W+2, CIN) padded
+    float* __restrict__ dWt,                 // (COUT, K) f32
+    float* __restrict__ db,                  // (COUT,) f32
+    int M, int H, int W, int COUT, int rows_per_chunk) {
+    constexpr int K = 9 * CIN;
+    constexpr int KROW = 3 * CIN;
+    constexpr int KHALF = ((K / 4 + 31) / 32) * 32;
+    constexpr int KFRAG = KHALF / 16;
+    const int PW = W + 2;
+    const int PH = H + 2;
+    __shared__ __hip_bfloat16 s_dy[32][32 + 8];
+    __shared__ __hip_bfloat16 s_a[32][K + 8];
+    int wave = threadIdx.x / WAVE;
+    int lane = threadIdx.x & (WAVE - 1);
+    long mstart = (long)blockIdx.x * rows_per_chunk;
+    long mend = min((long)M, mstart + rows_per_chunk);
+    int frow = lane & 15;
+    int mseg = (lane >> 4) * 8;
+
+    f32x4 acc[2][KFRAG] = {};
+    float bias_acc = 0.f;
+
+    for (long m0 = mstart; m0 < mend; m0 += 32) {
+        __syncthreads();
+        {
+            int t = threadIdx.x;
+            if (t < 32 * 4) {  // 32 rows x 32 cols of dY, 8 cols per thread
+                int mrow = t / 4;
+                int col = (t % 4) * 8;
+                long gm = m0 + mrow;
+                bf16x8 v = izero();
+                if (gm < mend) {
+                    long n = gm / (H * W);
+                    int p = (int)(gm % (H * W));
+                    int oy = p / W, ox = p % W;
+                    long base = ((n * PH + oy + 1) * PW + ox + 1) * COUT;
+                    ibf8u u;
+#pragma unroll
+                    for (int e = 0; e < 8; ++e)
+                        u.e[e] = (col + e < COUT)
+                                     ? (__bf16)bf2f(dY[base + col + e])
+                                     : (__bf16)0.f;
+                    v = u.v;
+                }
+                *reinterpret_cast<bf16x8*>(&s_dy[mrow][col]) = v;
+            }
+            for (int e8 = t; e8 < 32 * (K / 8); e8 += 256) {
+                int mrow = e8 / (K / 8);
+                int k = (e8 % (K / 8)) * 8;
+                long gm = m0 + mrow;
+                bf16x8 w = izero();
+                if (gm < mend) {
+                    long n = gm / (H * W);
+                    int p = (int)(gm % (H * W));
+                    int oy = p / W, ox = p % W;
+                    long base = ((n * PH + oy) * PW + ox) * CIN;
+                    int dy_ = k / KROW, rem = k % KROW;
+                    long off = base + (long)dy_ * PW * CIN + rem;
+                    if (IN_U8)
+                        w = idequant8(
+                            reinterpret_cast<const unsigned char*>(in) + off);
+                    else {
+                        w = iload8(
+                            reinterpret_cast<const __hip_bfloat16*>(in) + off);
+                        if (RELU_IN) w = irelu8(w);
+                    }
+                }
+                *reinterpret_cast<bf16x8*>(&s_a[mrow][k]) = w;
+            }
+        }
+        __syncthreads();
+
+        bf16x8 fa;  // 16 cout cols only (COUT <= 32 -> 2 frag rows)
+        bf16x8 fa2;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+            fa[e] = *(const __bf16*)&s_dy[mseg + e][frow];
+            fa2[e] = *(const __bf16*)&s_dy[mseg + e][16 + frow];
+        }
+#pragma unroll
+        for (int kf = 0; kf < KFRAG; ++kf) {
+            int kcol = wave * KHALF + kf * 16 + frow;
+            bf16x8 fb = izero();
+            if (kcol < K) {
+#pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    fb[e] = *(const __bf16*)&s_a[mseg + e][kcol];
+            }
+            acc[0][kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                fa, fb, acc[0][kf], 0, 0, 0);
+            acc[1][kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                fa2, fb, acc[1][kf], 0, 0, 0);
+        }
+
+        if (threadIdx.x < 32) {
+            int c = threadIdx.x;
+            for (int mr = 0; mr < 32; ++mr) bias_acc += bf2f(s_dy[mr][c]);
+        }
+    }
+
+    int ccol = lane & 15;
+    int crow = (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int kf = 0; kf < KFRAG; ++kf)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                long co = i * 16 + crow + r;
+                long kk = wave * KHALF + kf * 16 + ccol;
+                if (co < COUT && kk < K)
+                    atomicAdd(&dWt[co * K + kk], acc[i][kf][r]);
+            }
+    if (threadIdx.x < 32) {
+        long c = threadIdx.x;
+        if (c < COUT) atomicAdd(&db[c], bias_acc);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// maxpool 3x3 stride 2 pad 1: padded in (N,H+2,W+2,C) -> padded out
+// (N,OH+2,OW+2,C) + dense argmax tap (N,OH,OW,C) u8.  Out-of-image taps are
+// -inf (torch max_pool2d padding semantics — the halo zeros must NOT win).
+// One thread per (n, oy, ox, c8): 8 channels vectorized.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void maxpool3s2_fwd_kernel(
+    const __hip_bfloat16* __restrict__ in, __hip_bfloat16* __restrict__ out,
+    unsigned char* __restrict__ arg, int N, int H, int W, int OH, int OW,
+    int C) {
+    const int PW = W + 2, PH = H + 2;
+    const int C8 = C / 8;
+    long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long total = (long)N * OH * OW * C8;
+    if (idx >= total) return;
+    int c8 = (int)(idx % C8);
+    long t = idx / C8;
+    int ox = (int)(t % OW);
+    t /= OW;
+    int oy = (int)(t % OH);
+    long n = t / OH;
+
+    float best[8];
+    int bidx[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) { best[e] = -1e30f; bidx[e] = 0; }
+#pragma unroll
+    for (int ky = 0; ky < 3; ++ky) {
+        int y = 2 * oy - 1 + ky;
+        if (y < 0 || y >= H) continue;
+#pragma unroll
+        for (int kx = 0; kx < 3; ++kx) {
+            int x = 2 * ox - 1 + kx;
+            if (x < 0 || x >= W) continue;
+            ibf8u v;
+            v.v = iload8(in + ((n * PH + y + 1) * PW + x + 1) * C + c8 * 8);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                float f = bf2f(v.e[e]);
+                if (f > best[e]) { best[e] = f; bidx[e] = ky * 3 + kx; }
+            }
+        }
+    }
+    const int POW = OW + 2, POH = OH + 2;
+    ibf8u o;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) o.e[e] = f2bf(best[e]);
+    *reinterpret_cast<bf16x8*>(
+        out + ((n * POH + oy + 1) * POW + ox + 1) * C + c8 * 8) = o.v;
+    long abase = (((long)n * OH + oy) * OW + ox) * C + c8 * 8;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) arg[abase + e] = (unsigned char)bidx[e];
+}
+
+// backward: one thread per INPUT real pixel x c8; gathers from the <= 4
+// windows that can contain it (atomic-free).
+__global__ __launch_bounds__(256) void maxpool3s2_bwd_kernel(
+    const __hip_bfloat16* __restrict__ dOut,  // padded (N,OH+2,OW+2,C)
+    const unsigned char* __restrict__ arg,    // dense (N,OH,OW,C)
+    __hip_bfloat16* __restrict__ dIn,         // padded (N,H+2,W+2,C)
+    int N, int H, int W, int OH, int OW, int C) {
+    const int PW = W + 2, PH = H + 2;
+    const int POW = OW + 2, POH = OH + 2;
+    const int C8 = C / 8;
+    long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long total = (long)N * H * W * C8;
+    if (idx >= total) return;
+    int c8 = (int)(idx % C8);
+    long t = idx / C8;
+    int x = (int)(t % W);
+    t /= W;
+    int y = (int)(t % H);
+    long n = t / H;
+
+    float acc[8] = {};
+    int oy_lo = (y - 1 + 1) / 2;      // ceil((y-1)/2) for y>=0
+    if (y == 0) oy_lo = 0;
+    int oy_hi = (y + 1) / 2;
+    int ox_lo = (x == 0) ? 0 : (x + 1 - 1) / 2;
+    int ox_hi = (x + 1) / 2;
+    for (int oy = oy_lo; oy <= oy_hi && oy < OH; ++oy) {
+        int ky = y - (2 * oy - 1);
+        if (ky < 0 || ky > 2) continue;
+        for (int ox = ox_lo; ox <= ox_hi && ox < OW; ++ox) {
+            int kx = x - (2 * ox - 1);
+            if (kx < 0 || kx > 2) continue;
+            int tap = ky * 3 + kx;
+            long abase = (((long)n * OH + oy) * OW + ox) * C + c8 * 8;
+            ibf8u g;
+            g.v = iload8(dOut + ((n * POH + oy + 1) * POW + ox + 1) * C + c8 * 8);
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+                if (arg[abase + e] == tap) acc[e] += bf2f(g.e[e]);
+        }
+    }
+    ibf8u o;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) o.e[e] = f2bf(acc[e]);
+    *reinterpret_cast<bf16x8*>(
+        dIn + ((n * PH + y + 1) * PW + x + 1) * C + c8 * 8) = o.v;
+}
+
+// ---------------------------------------------------------------------------
+// pack_frames: dense u8 HWC frames (M, 84, 84, CIN) -> halo-padded
+// 8-channel u8 (M, 86, 86, 8) (channels CIN.. zero).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void pack_frames_kernel(
+    const unsigned char* __restrict__ in, unsigned char* __restrict__ out,
+    long total, int H, int W, int CIN) {
+    long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= total) return;  // total = M*H*W
+    int x = (int)(idx % W);
+    long t = idx / W;
+    int y = (int)(t % H);
+    long m = t / H;
+    const unsigned char* src = in + (idx * CIN);
+    unsigned char* dst = out + ((m * (H + 2) + y + 1) * (long)(W + 2) + x + 1) * 8;
+    uint2 v{0, 0};
+    for (int c = 0; c < CIN; ++c)
+        reinterpret_cast<unsigned char*>(&v)[c] = src[c];
+    *reinterpret_cast<uint2*>(dst) = v;
+}
+
+// ---------------------------------------------------------------------------
+// pad2dense: padded (N, H+2, W+2, C) bf16 -> dense (N, H*W*C) bf16 with
+// optional relu (the encoder's final relu before flatten+fc).
+// dense2pad: dense bf16 grad -> padded bf16, masked by (act_padded > 0)
+// (backward of that relu); halo rows stay whatever they were — callers
+// pass a zero-initialized padded buffer.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void pad2dense_kernel(
+    const __hip_bfloat16* __restrict__ in, __hip_bfloat16* __restrict__ out,
+    long total, int H, int W, int C, int relu) {
+    long idx = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+    if (idx >= total) return;  // total = N*H*W*C
+    long c = idx % C;
+    long t = idx / C;
+    int x = (int)(t % W);
+    t /= W;
+    int y = (int)(t % H);
+    long n = t / H;
+    bf16x8 v = iload8(in + ((n * (H + 2) + y + 1) * (long)(W + 2) + x + 1) * C + c);
+    if (relu) v = irelu8(v);
+    *reinterpret_cast<bf16x8*>(out + idx) = v;
+}
+
+__global__ __launch_bounds__(256) void dense2pad_mask_kernel(
+    const __hip_bfloat16* __restrict__ dflat,
+    const __hip_bfloat16* __restrict__ act_pad,
+    __hip_bfloat16* __restrict__ out, long total, int H, int W, int C) {
+    long idx = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+    if (idx >= total) return;
+    long c = idx % C;
+    long t = idx / C;
+    int x = (int)(t % W);
+    t /= W;
+    int y = (int)(t % H);
+    long n = t / H;
+    long pidx = ((n * (H + 2) + y + 1) * (long)(W + 2) + x + 1) * C + c;
+    ibf8u g, a, o;
+    g.v = iload8(dflat + idx);
+    a.v = iload8(act_pad + pidx);
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+        o.e[e] = (bf2f(a.e[e]) > 0.f) ? g.e[e] : (__bf16)0.f;
+    *reinterpret_cast<bf16x8*>(out + pidx) = o.v;
+}
+
+// ---------------------------------------------------------------------------
+// Host wrappers
+// ---------------------------------------------------------------------------
+
+static inline int icdiv(long a, long b) { return (int)((a + b - 1) / b); }
+
+// in: padded (N, H+2, W+2, CIN_pad) bf16 or u8 (CIN_pad 8/16/32); out is
+// written into `out` (padded, zero-halo, (N, H+2, W+2, COUT)).
+// epi: 0 plain, 1 +res, 2 *mask, 3 res + mask*acc.
+void conv3p(torch::Tensor in, torch::Tensor Wt, torch::Tensor bias,
+            torch::Tensor res, torch::Tensor mask, torch::Tensor out,
+            int64_t N, int64_t H, int64_t W, bool relu_in, bool has_bias,
+            int64_t epi) {
+    long M = N * H * W;
+    long CIN = in.size(3);
+    long COUT = Wt.size(0) ;
+    bool u8 = in.dtype() == torch::kUInt8;
+    dim3 grid(icdiv(M, 128));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    const void* x = in.data_ptr();
+    auto* w = reinterpret_cast<const __hip_bfloat16*>(Wt.data_ptr());
+    const float* b = has_bias ? bias.data_ptr<float>() : nullptr;
+    auto* rp = (epi == 1 || epi == 3)
+                   ? reinterpret_cast<const __hip_bfloat16*>(res.data_ptr())
+                   : nullptr;
+    auto* mp = (epi == 2 || epi == 3)
+                   ? reinterpret_cast<const __hip_bfloat16*>(mask.data_ptr())
+                   : nullptr;
+    auto* o = reinterpret_cast<__hip_bfloat16*>(out.data_ptr());
+    TORCH_CHECK(Wt.size(1) == 9 * CIN, "Wt K mismatch");
+    TORCH_CHECK(out.size(3) == COUT, "out C mismatch");
+
+#define C3P(U8, CIN_, NCOL_, RELU_, BIAS_, EPI_)                              \
+    hipLaunchKernelGGL((conv3p_kernel<U8, CIN_, NCOL_, RELU_, BIAS_, EPI_>),  \
+                       grid, dim3(256), 0, stream.stream(), x, w, b, rp, mp,  \
+                       o, (int)M, (int)H, (int)W, (int)COUT)
+#define C3P_EPI(U8, CIN_, NCOL_, RELU_, BIAS_)                                \
+    do {                                                                      \
+        if (epi == 0) C3P(U8, CIN_, NCOL_, RELU_, BIAS_, 0);                  \
+        else if (epi == 1) C3P(U8, CIN_, NCOL_, RELU_, BIAS_, 1);             \
+        else if (epi == 2) C3P(U8, CIN_, NCOL_, RELU_, BIAS_, 2);             \
+        else C3P(U8, CIN_, NCOL_, RELU_, BIAS_, 3);                           \
+    } while (0)
+#define C3P_RB(U8, CIN_, NCOL_)                                               \
+    do {                                                                      \
+        if (relu_in && has_bias) C3P_EPI(U8, CIN_, NCOL_, true, true);        \
+        else if (relu_in) C3P_EPI(U8, CIN_, NCOL_, true, false);              \
+        else if (has_bias) C3P_EPI(U8, CIN_, NCOL_, false, true);             \
+        else C3P_EPI(U8, CIN_, NCOL_, false, false);                          \
+    } while (0)
+
+    if (u8) {
+        TORCH_CHECK(CIN == 8, "u8 conv expects 8 padded channels");
+        if (COUT <= 16) C3P_RB(true, 8, 16); else C3P_RB(true, 8, 32);
+    } else if (CIN == 8) {
+        if (COUT <= 16) C3P_RB(false, 8, 16); else C3P_RB(false, 8, 32);
+    } else if (CIN == 16) {
+        if (COUT <= 16) C3P_RB(false, 16, 16); else C3P_RB(false, 16, 32);
+    } else if (CIN == 32) {
+        if (COUT <= 16) C3P_RB(false, 32, 16); else C3P_RB(false, 32, 32);
+    } else {
+        TORCH_CHECK(false, "unsupported CIN ", CIN);
+    }
+#undef C3P_RB
+#undef C3P_EPI
+#undef C3P
+}
+
+std::vector<torch::Tensor> conv3p_wgrad(torch::Tensor dY, torch::Tensor in,
+                                        int64_t N, int64_t H, int64_t W,
+                                        bool relu_in) {
+    long M = N * H * W;
+    long CIN = in.size(3);
+    long COUT = dY.size(3);
+    long K = 9 * CIN;
+    bool u8 = in.dtype() == torch::kUInt8;
+    auto dWt = torch::zeros({COUT, K}, dY.options().dtype(torch::kFloat32));
+    auto db = torch::zeros({COUT}, dY.options().dtype(torch::kFloat32));
+    long target_chunks = 1024;
+    long rows_per_chunk = std::max(32L, (M + target_chunks - 1) / target_chunks);
+    rows_per_chunk = ((rows_per_chunk + 31) / 32) * 32;
+    dim3 grid(icdiv(M, rows_per_chunk));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    auto* dy = reinterpret_cast<const __hip_bfloat16*>(dY.data_ptr());
+    const void* x = in.data_ptr();
+#define WG(U8, CIN_, RELU_)                                                   \
+    hipLaunchKernelGGL((conv3p_wgrad_kernel<U8, CIN_, RELU_>), grid,          \
+                       dim3(256), 0, stream.stream(), dy, x,                  \
+                       dWt.data_ptr<float>(), db.data_ptr<float>(), (int)M,   \
+                       (int)H, (int)W, (int)COUT, (int)rows_per_chunk)
+    if (u8) { TORCH_CHECK(CIN == 8); WG(true, 8, false); }
+    else if (CIN == 8) { if (relu_in) WG(false, 8, true); else WG(false, 8, false); }
+    else if (CIN == 16) { if (relu_in) WG(false, 16, true); else WG(false, 16, false); }
+    else if (CIN == 32) { if (relu_in) WG(false, 32, true); else WG(false, 32, false); }
+    else TORCH_CHECK(false, "unsupported CIN ", CIN);
+#undef WG
+    return {dWt, db};
+}
+
+std::vector<torch::Tensor> maxpool3s2_fwd(torch::Tensor in, int64_t N,
+                                          int64_t H, int64_t W) {
+    long C = in.size(3);
+    long OH = H / 2, OW = W / 2;
+    if (H % 2 == 1) { OH = (H + 1) / 2; OW = (W + 1) / 2; }
+    auto out = torch::zeros({N, OH + 2, OW + 2, C},
+                            in.options().dtype(torch::kBFloat16));
+    auto arg = torch::empty({N, OH, OW, C}, in.options().dtype(torch::kUInt8));
+    long total = N * OH * OW * (C / 8);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(maxpool3s2_fwd_kernel, dim3(icdiv(total, 256)),
+                       dim3(256), 0, stream.stream(),
+                       reinterpret_cast<const __hip_bfloat16*>(in.data_ptr()),
+                       reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                       arg.data_ptr<unsigned char>(), (int)N, (int)H, (int)W,
+                       (int)OH, (int)OW, (int)C);
+    return {out, arg};
+}
+
+void maxpool3s2_bwd(torch::Tensor dOut, torch::Tensor arg, torch::Tensor dIn,
+                    int64_t N, int64_t H, int64_t W, int64_t OH, int64_t OW) {
+    long C = dIn.size(3);
+    long total = N * H * W * (C / 8);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(maxpool3s2_bwd_kernel, dim3(icdiv(total, 256)),
+                       dim3(256), 0, stream.stream(),
+                       reinterpret_cast<const __hip_bfloat16*>(dOut.data_ptr()),
+                       arg.data_ptr<unsigned char>(),
+                       reinterpret_cast<__hip_bfloat16*>(dIn.data_ptr()),
+                       (int)N, (int)H, (int)W, (int)OH, (int)OW, (int)C);
+}
+
+torch::Tensor pack_frames(torch::Tensor frames, int64_t H, int64_t W) {
+    long M = frames.size(0);
+    long CIN = frames.size(3);
+    TORCH_CHECK(frames.dtype() == torch::kUInt8 && CIN <= 8);
+    auto out = torch::zeros({M, H + 2, W + 2, 8},
+                            frames.options().dtype(torch::kUInt8));
+    long total = M * H * W;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(pack_frames_kernel, dim3(icdiv(total, 256)), dim3(256),
+                       0, stream.stream(),
+                       frames.data_ptr<unsigned char>(),
+                       out.data_ptr<unsigned char>(), total, (int)H, (int)W,
+                       (int)CIN);
+    return out;
+}
+
+torch::Tensor pad2dense(torch::Tensor in, int64_t N, int64_t H, int64_t W,
+                        bool relu) {
+    long C = in.size(3);
+    auto out = torch::empty({N, H * W * C},
+                            in.options().dtype(torch::kBFloat16));
+    long total = N * H * W * C;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(pad2dense_kernel, dim3(icdiv(total / 8, 256)),
+                       dim3(256), 0, stream.stream(),
+                       reinterpret_cast<const __hip_bfloat16*>(in.data_ptr()),
+                       reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                       total, (int)H, (int)W, (int)C, relu ? 1 : 0);
+    return out;
+}
+
+void dense2pad_mask(torch::Tensor dflat, torch::Tensor act_pad,
+                    torch::Tensor out, int64_t N, int64_t H, int64_t W) {
+    long C = act_pad.size(3);
+    long total = N * H * W * C;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(dense2pad_mask_kernel, dim3(icdiv(total / 8, 256)),
+                       dim3(256), 0, stream.stream(),
+                       reinterpret_cast<const __hip_bfloat16*>(dflat.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(act_pad.data_ptr()),
+                       reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
+                       total, (int)H, (int)W, (int)C);
+}

@@ -1,0 +1,203 @@
+"""IMPALA-deep ResNet encoder on the gfx950 HIP kernels (configs[4]).
+
+Runs the ImpalaCNN (models/encoders.py — conv 3x3 + maxpool 3x3 s2 + 2
+pre-activation residual blocks per stage, channels (16, 32, 32), then
+relu -> flatten -> fc 3872->512) through ops/hip/impala_kernels.hip:
+halo-padded NHWC activations, MFMA implicit-GEMM 3x3 convs whose dgrad is
+the same kernel with flipped prepacked weights, fused pre-activation
+ReLU / residual epilogues, and an atomic-free maxpool backward.
+
+The engine (ops/engine.py) delegates encoder fwd/bwd here when
+config.encoder == 'impala'; LSTM / heads / loss / optimizer paths are
+shared with the flagship nature config.
+"""
+
+from typing import List, Optional, Tuple
+
+import torch
+
+# stage geometry: (conv-input H, pooled H) per stage; 84 -> 42 -> 21 -> 11
+STAGES = ((84, 42), (42, 21), (21, 11))
+CHANS = (16, 32, 32)
+
+
+def _pack_fwd(w: torch.Tensor, cp: int, dev) -> torch.Tensor:
+    """(COUT, CIN, 3, 3) -> (COUT, 9*cp) bf16, k-order (ky, kx, c),
+    input channels zero-padded to cp."""
+    cout, cin = w.shape[:2]
+    wt = torch.zeros(cout, 3, 3, cp, device=dev)
+    wt[:, :, :, :cin] = w.detach().to(dev).permute(0, 2, 3, 1)
+    return wt.reshape(cout, 9 * cp).bfloat16().contiguous()
+
+
+def _pack_dgrad(w: torch.Tensor, dev) -> torch.Tensor:
+    """(COUT, CIN, 3, 3) -> (CIN, 9*COUT) bf16: spatially flipped,
+    transposed — dgrad of a 3x3 s1 p1 conv as a forward conv over the
+    halo-padded upstream gradient."""
+    wd = w.detach().to(dev).flip(2, 3).permute(1, 2, 3, 0)
+    return wd.reshape(w.shape[1], -1).bfloat16().contiguous()
+
+
+class ImpalaPack:
+    """Prepacked weights for one ImpalaCNN (15 convs + biases)."""
+
+    def __init__(self, enc, device, with_bwd: bool):
+        self.enc = enc
+        self.device = device
+        self.with_bwd = with_bwd
+        self.refresh()
+
+    def _convs(self):
+        """Yield (name, conv, cin_pad) over the 15 convs in forward order:
+        per stage: stage conv, res1.conv1, res1.conv2, res2.conv1, res2.conv2."""
+        for si, stage in enumerate(self.enc.stages):
+            cp = max(8, stage.conv.weight.shape[1])
+            yield f"s{si}c", stage.conv, cp
+            for ri, res in enumerate((stage.res1, stage.res2)):
+                c = res.conv1.weight.shape[1]
+                yield f"s{si}r{ri}a", res.conv1, c
+                yield f"s{si}r{ri}b", res.conv2, c
+
+    def refresh(self):
+        dev = self.device
+        self.wt = {}
+        self.bias = {}
+        self.wd = {}
+        for name, conv, cp in self._convs():
+            self.wt[name] = _pack_fwd(conv.weight, cp, dev)
+            self.bias[name] = conv.bias.detach().to(dev).float().contiguous()
+            if self.with_bwd:
+                self.wd[name] = _pack_dgrad(conv.weight, dev)
+        # fc: torch flattens NCHW (32, 11, 11); our pad2dense flattens HWC
+        fc = self.enc.fc
+        hd = fc.weight.shape[0]
+        wf = fc.weight.detach().to(dev).reshape(hd, 32, 11, 11)
+        self.wft = (wf.permute(0, 2, 3, 1).reshape(hd, 32 * 121)
+                    .bfloat16().contiguous())
+        self.bf = fc.bias.detach().to(dev).float().contiguous()
+        if self.with_bwd:
+            self.wf_kn = self.wft.t().contiguous()
+
+
+def _zeros_pad(M, H, C, dev):
+    return torch.zeros(M, H + 2, H + 2, C, device=dev, dtype=torch.bfloat16)
+
+
+def encoder_fwd(m, pack: ImpalaPack, obs_hwc_u8: torch.Tensor,
+                want_stash: bool):
+    """obs: (M, 84, 84, C<=8) u8 dense -> latent (M, hidden) bf16 (+stash).
+
+    Stash layout (one dict): padded activations and pool argmaxes needed by
+    encoder_bwd."""
+    dev = pack.device
+    M = obs_hwc_u8.shape[0]
+    xp = m.pack_frames(obs_hwc_u8, 84, 84)          # (M, 86, 86, 8) u8
+    st = {"xp": xp} if want_stash else None
+    x = xp
+    empty = torch.Tensor()
+    for si, (hin, hout) in enumerate(STAGES):
+        c = CHANS[si]
+        conv_out = _zeros_pad(M, hin, c, dev)
+        m.conv3p(x, pack.wt[f"s{si}c"], pack.bias[f"s{si}c"], empty, empty,
+                 conv_out, M, hin, hin, False, True, 0)
+        pooled, arg = m.maxpool3s2_fwd(conv_out, M, hin, hin)
+        if want_stash:
+            st[f"s{si}c_in"] = x
+            st[f"s{si}c_out"] = conv_out
+            st[f"s{si}arg"] = arg
+        x = pooled
+        for ri in range(2):
+            y1 = _zeros_pad(M, hout, c, dev)
+            m.conv3p(x, pack.wt[f"s{si}r{ri}a"], pack.bias[f"s{si}r{ri}a"],
+                     empty, empty, y1, M, hout, hout, True, True, 0)
+            out = _zeros_pad(M, hout, c, dev)
+            m.conv3p(y1, pack.wt[f"s{si}r{ri}b"], pack.bias[f"s{si}r{ri}b"],
+                     x, empty, out, M, hout, hout, True, True, 1)
+            if want_stash:
+                st[f"s{si}r{ri}x"] = x
+                st[f"s{si}r{ri}y1"] = y1
+            x = out
+    if want_stash:
+        st["s_out"] = x                                # (M, 13, 13, 32)
+    flat = m.pad2dense(x, M, 11, 11, True)             # (M, 3872) bf16
+    latent = m.gemm_bias_act(flat, pack.wft, pack.bf, 1, False)
+    if want_stash:
+        st["flat"] = flat
+    return latent, st
+
+
+def encoder_bwd(m, pack: ImpalaPack, st: dict, dlat: torch.Tensor,
+                latent: torch.Tensor):
+    """dlat: (M, hidden) bf16 -> conv/fc gradients.
+
+    Returns {name: (dW f32 torch-layout, db f32)} for the 15 convs plus
+    ('fc', (dWf, dbf))."""
+    dev = pack.device
+    M = dlat.shape[0]
+    empty = torch.Tensor()
+    grads = {}
+
+    # fc (+ its relu) then the flatten relu back onto the padded grid
+    dflat = m.gemm_dgrad(dlat, latent, pack.wf_kn, True)
+    dWf, dbf = m.gemm_wgrad(dlat, latent, st["flat"], True, True)
+    hd = pack.wft.shape[0]
+    grads["fc"] = (dWf.view(hd, 11, 11, 32).permute(0, 3, 1, 2)
+                   .reshape(hd, 3872).contiguous(), dbf)
+
+    dx = _zeros_pad(M, 11, 32, dev)
+    m.dense2pad_mask(dflat, st["s_out"], dx, M, 11, 11)
+
+    for si in (2, 1, 0):
+        hin, hout = STAGES[si]
+        c = CHANS[si]
+        for ri in (1, 0):
+            x, y1 = st[f"s{si}r{ri}x"], st[f"s{si}r{ri}y1"]
+            nb, nm = f"s{si}r{ri}b", f"s{si}r{ri}a"
+            dW2, db2 = m.conv3p_wgrad(dx, y1, M, hout, hout, True)
+            grads[nb] = (self_conv_grad(dW2, c, c), db2)
+            dy1 = _zeros_pad(M, hout, c, dev)
+            m.conv3p(dx, pack.wd[nb], empty, empty, y1, dy1,
+                     M, hout, hout, False, False, 2)
+            dW1, db1 = m.conv3p_wgrad(dy1, x, M, hout, hout, True)
+            grads[nm] = (self_conv_grad(dW1, c, c), db1)
+            dx_new = _zeros_pad(M, hout, c, dev)
+            m.conv3p(dy1, pack.wd[nm], empty, dx, x, dx_new,
+                     M, hout, hout, False, False, 3)
+            dx = dx_new
+        # maxpool backward: dx (pooled grid) -> conv-out grid
+        dconv = _zeros_pad(M, hin, c, dev)
+        m.maxpool3s2_bwd(dx, st[f"s{si}arg"], dconv, M, hin, hin, hout, hout)
+        # stage conv wgrad (+ dgrad, except stage 0 whose input is data)
+        sin = st[f"s{si}c_in"]
+        cin = 4 if si == 0 else CHANS[si - 1]
+        dWc, dbc = m.conv3p_wgrad(dconv, sin, M, hin, hin, False)
+        grads[f"s{si}c"] = (self_conv_grad(dWc, c, cin), dbc)
+        if si > 0:
+            dx = _zeros_pad(M, hin, CHANS[si - 1], dev)
+            m.conv3p(dconv, pack.wd[f"s{si}c"], empty, empty, empty, dx,
+                     M, hin, hin, False, False, 0)
+    return grads
+
+
+def self_conv_grad(dWt: torch.Tensor, cout: int, cin: int) -> torch.Tensor:
+    """(COUT, 9*cp) f32 -> (COUT, CIN, 3, 3) torch conv-weight layout."""
+    cp = dWt.shape[1] // 9
+    return (dWt.view(cout, 3, 3, cp)[:, :, :, :cin]
+            .permute(0, 3, 1, 2).contiguous())
+
+
+def write_grads(enc, grads):
+    """Copy encoder_bwd outputs into the nn.Module .grad tensors."""
+    names = {}
+    for si, stage in enumerate(enc.stages):
+        names[f"s{si}c"] = stage.conv
+        for ri, res in enumerate((stage.res1, stage.res2)):
+            names[f"s{si}r{ri}a"] = res.conv1
+            names[f"s{si}r{ri}b"] = res.conv2
+    for name, conv in names.items():
+        dw, db = grads[name]
+        conv.weight.grad.copy_(dw.view(conv.weight.shape))
+        conv.bias.grad.copy_(db)
+    dwf, dbf = grads["fc"]
+    enc.fc.weight.grad.copy_(dwf.view(enc.fc.weight.shape))
+    enc.fc.bias.grad.copy_(dbf)

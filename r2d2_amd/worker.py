@@ -545,7 +545,7 @@ class Learner:
         hip_ops.ext(required=True)
         self.hip_engine = True
         c = self.cfg
-        if (c.encoder == "nature" and c.hidden_dim == 512
+        if (c.encoder in ("nature", "impala") and c.hidden_dim == 512
                 and len(c.obs_shape) == 3 and c.obs_shape[0] == 4
                 and self.device.type == "cuda"):
             from .ops.engine import HipNetworkEngine
