@@ -189,18 +189,25 @@ def test_impala_encoder_full_fwd_bwd_vs_autograd():
 
     x_eager = (obs.float() / 255.0).permute(0, 3, 1, 2)
     ref_lat = enc(x_eager)
-    close(latent, ref_lat, rtol=5e-2,
-          atol=6e-2 * float(ref_lat.abs().max()), name="impala latent")
+    close(latent, ref_lat.detach(), rtol=5e-2,
+          atol=6e-2 * float(ref_lat.detach().abs().max()), name="impala latent")
 
     dlat = torch.randn(Mn, 512, device="cuda")
     ref_lat.backward(dlat)
     grads = imp.encoder_bwd(M_, pack, st, dlat.bfloat16(), latent)
 
+    # grads compare by relative Frobenius error: an fp32 autograd reference
+    # makes different ReLU-mask decisions than the bf16 path for
+    # pre-activations that round across zero, so single elements can
+    # legitimately differ; the aggregate must still agree tightly.
+    def rel_fro(a, b, tol, name):
+        a, b = a.float().flatten(), b.float().flatten()
+        err = (a - b).norm() / (b.norm() + 1e-8)
+        assert err < tol, f"{name}: rel fro {err:.4f}"
+
     dwf, dbf = grads["fc"]
-    close(dwf.view_as(enc.fc.weight), enc.fc.weight.grad, rtol=6e-2,
-          atol=6e-2 * float(enc.fc.weight.grad.abs().max()), name="fc dW")
-    close(dbf, enc.fc.bias.grad, rtol=6e-2,
-          atol=6e-2 * float(enc.fc.bias.grad.abs().max()), name="fc db")
+    rel_fro(dwf, enc.fc.weight.grad.reshape(-1), 0.08, "fc dW")
+    rel_fro(dbf, enc.fc.bias.grad, 0.08, "fc db")
     name_map = {}
     for si, stage in enumerate(enc.stages):
         name_map[f"s{si}c"] = stage.conv
@@ -209,12 +216,9 @@ def test_impala_encoder_full_fwd_bwd_vs_autograd():
             name_map[f"s{si}r{ri}b"] = res.conv2
     for name, conv in name_map.items():
         dw, db = grads[name]
-        ref = conv.weight.grad
-        close(dw.view_as(ref), ref, rtol=8e-2,
-              atol=8e-2 * max(1e-3, float(ref.abs().max())), name=f"{name} dW")
-        close(db, conv.bias.grad, rtol=8e-2,
-              atol=8e-2 * max(1e-3, float(conv.bias.grad.abs().max())),
-              name=f"{name} db")
+        rel_fro(dw.view_as(conv.weight.grad), conv.weight.grad, 0.10,
+                f"{name} dW")
+        rel_fro(db, conv.bias.grad, 0.10, f"{name} db")
 
 
 def test_impala_engine_train_step():
