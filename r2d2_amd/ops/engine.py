@@ -221,7 +221,8 @@ class HipNetworkEngine:
         """obs: (M, 84, 84, C) uint8 -> latent (M, 512) bf16 + stashes."""
         m = self.m
         if self.impala:
-            return impala_ops.encoder_fwd(m, pack.imp, obs_hwc_u8, want_stash)
+            return impala_ops.encoder_fwd(m, pack.imp, obs_hwc_u8, want_stash,
+                                          xp=getattr(self, "_xp", None))
         M = obs_hwc_u8.shape[0]
         a1 = m.conv_fwd(obs_hwc_u8, pack.w1t, pack.b1, 1, M, 84, 84, 20, 20, True)
         a2 = m.conv_fwd(a1, pack.w2t, pack.b2, 2, M, 20, 20, 9, 9, True)
@@ -292,6 +293,9 @@ class HipNetworkEngine:
 
         self._mark("start")
         # ---- forward ---------------------------------------------------
+        if self.impala:
+            # pack frames once; online and target share the padded u8 tensor
+            self._xp = impala_ops.pack_obs(m, self.online.imp, obs_hwc)
         lat_o, enc_stash = self._encoder_fwd(self.online, obs_hwc)
         self._mark("enc_online")
         lat_t, _ = self._encoder_fwd(self.target, obs_hwc, want_stash=False)

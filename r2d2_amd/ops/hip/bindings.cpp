@@ -78,11 +78,12 @@ void conv3p(torch::Tensor in, torch::Tensor Wt, torch::Tensor bias,
 std::vector<torch::Tensor> conv3p_wgrad(torch::Tensor dY, torch::Tensor in,
                                         int64_t N, int64_t H, int64_t W,
                                         bool relu_in);
-std::vector<torch::Tensor> maxpool3s2_fwd(torch::Tensor in, int64_t N,
-                                          int64_t H, int64_t W);
+void maxpool3s2_fwd(torch::Tensor in, torch::Tensor out, torch::Tensor arg,
+                    int64_t N, int64_t H, int64_t W);
 void maxpool3s2_bwd(torch::Tensor dOut, torch::Tensor arg, torch::Tensor dIn,
                     int64_t N, int64_t H, int64_t W, int64_t OH, int64_t OW);
-torch::Tensor pack_frames(torch::Tensor frames, int64_t H, int64_t W);
+void pack_frames(torch::Tensor frames, torch::Tensor out, int64_t H,
+                 int64_t W);
 torch::Tensor pad2dense(torch::Tensor in, int64_t N, int64_t H, int64_t W,
                         bool relu);
 void dense2pad_mask(torch::Tensor dflat, torch::Tensor act_pad,

@@ -187,7 +187,10 @@ __global__ __launch_bounds__(256) void conv3p_kernel(
 // full-K staging scheme as conv_kernels.hip conv_wgrad (dY and patches are
 // each read exactly once per 32-row tile).
 // ---------------------------------------------------------------------------
-template <bool IN_U8, int CIN, bool RELU_IN>
+// NCO2: true when COUT > 16 (two cout fragments per wave, 4-way K split);
+// false for COUT <= 16 (single cout fragment, 2-way K split x 2-way M split
+// — halves the padded-K waste that dominates the small-channel stages).
+template <bool IN_U8, int CIN, bool RELU_IN, bool NCO2>
 __global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
     const __hip_bfloat16* __restrict__ dY,   // (N, H+2, W+2, COUT) padded
     const void* __restrict__ in,             // (N, H+2, W+2, CIN) padded
@@ -196,27 +199,35 @@ __global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
     int M, int H, int W, int COUT, int rows_per_chunk) {
     constexpr int K = 9 * CIN;
     constexpr int KROW = 3 * CIN;
-    constexpr int KHALF = ((K / 4 + 31) / 32) * 32;
+    constexpr int KSPLIT = NCO2 ? 4 : 2;
+    constexpr int KHALF = ((K / KSPLIT + 15) / 16) * 16;
     constexpr int KFRAG = KHALF / 16;
+    constexpr int NCO = NCO2 ? 2 : 1;
     const int PW = W + 2;
     const int PH = H + 2;
-    __shared__ __hip_bfloat16 s_dy[32][32 + 8];
-    __shared__ __hip_bfloat16 s_a[32][K + 8];
+    constexpr int TROWS = NCO2 ? 32 : 64;
+    __shared__ __hip_bfloat16 s_dy[TROWS][32 + 8];
+    __shared__ __hip_bfloat16 s_a[TROWS][K + 8];
     int wave = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
     long mstart = (long)blockIdx.x * rows_per_chunk;
     long mend = min((long)M, mstart + rows_per_chunk);
     int frow = lane & 15;
     int mseg = (lane >> 4) * 8;
+    // wave roles: NCO2 -> all 4 waves on one 32-row tile, K split 4 ways;
+    // else -> waves (wk, wm): K split 2 ways x two 32-row tiles of a 64-row
+    // stage (single cout fragment for COUT <= 16).
+    const int wk = NCO2 ? wave : (wave & 1);
+    const int wm = NCO2 ? 0 : (wave >> 1);
 
-    f32x4 acc[2][KFRAG] = {};
+    f32x4 acc[NCO][KFRAG] = {};
     float bias_acc = 0.f;
 
-    for (long m0 = mstart; m0 < mend; m0 += 32) {
+    for (long m0 = mstart; m0 < mend; m0 += TROWS) {
         __syncthreads();
         {
             int t = threadIdx.x;
-            if (t < 32 * 4) {  // 32 rows x 32 cols of dY, 8 cols per thread
+            if (t < TROWS * 4) {  // TROWS rows x 32 cols of dY, 8/thread
                 int mrow = t / 4;
                 int col = (t % 4) * 8;
                 long gm = m0 + mrow;
@@ -236,7 +247,7 @@ __global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
                 }
                 *reinterpret_cast<bf16x8*>(&s_dy[mrow][col]) = v;
             }
-            for (int e8 = t; e8 < 32 * (K / 8); e8 += 256) {
+            for (int e8 = t; e8 < TROWS * (K / 8); e8 += 256) {
                 int mrow = e8 / (K / 8);
                 int k = (e8 % (K / 8)) * 8;
                 long gm = m0 + mrow;
@@ -262,44 +273,44 @@ __global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
         }
         __syncthreads();
 
-        bf16x8 fa;  // 16 cout cols only (COUT <= 32 -> 2 frag rows)
-        bf16x8 fa2;
+        const int mbase = wm * 32 + mseg;
+        bf16x8 fa[NCO];
 #pragma unroll
-        for (int e = 0; e < 8; ++e) {
-            fa[e] = *(const __bf16*)&s_dy[mseg + e][frow];
-            fa2[e] = *(const __bf16*)&s_dy[mseg + e][16 + frow];
-        }
+        for (int i = 0; i < NCO; ++i)
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+                fa[i][e] = *(const __bf16*)&s_dy[mbase + e][i * 16 + frow];
 #pragma unroll
         for (int kf = 0; kf < KFRAG; ++kf) {
-            int kcol = wave * KHALF + kf * 16 + frow;
+            int kcol = wk * KHALF + kf * 16 + frow;
             bf16x8 fb = izero();
             if (kcol < K) {
 #pragma unroll
                 for (int e = 0; e < 8; ++e)
-                    fb[e] = *(const __bf16*)&s_a[mseg + e][kcol];
+                    fb[e] = *(const __bf16*)&s_a[mbase + e][kcol];
             }
-            acc[0][kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                fa, fb, acc[0][kf], 0, 0, 0);
-            acc[1][kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                fa2, fb, acc[1][kf], 0, 0, 0);
+#pragma unroll
+            for (int i = 0; i < NCO; ++i)
+                acc[i][kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    fa[i], fb, acc[i][kf], 0, 0, 0);
         }
 
         if (threadIdx.x < 32) {
             int c = threadIdx.x;
-            for (int mr = 0; mr < 32; ++mr) bias_acc += bf2f(s_dy[mr][c]);
+            for (int mr = 0; mr < TROWS; ++mr) bias_acc += bf2f(s_dy[mr][c]);
         }
     }
 
     int ccol = lane & 15;
     int crow = (lane >> 4) * 4;
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
+    for (int i = 0; i < NCO; ++i)
 #pragma unroll
         for (int kf = 0; kf < KFRAG; ++kf)
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 long co = i * 16 + crow + r;
-                long kk = wave * KHALF + kf * 16 + ccol;
+                long kk = wk * KHALF + kf * 16 + ccol;
                 if (co < COUT && kk < K)
                     atomicAdd(&dWt[co * K + kk], acc[i][kf][r]);
             }
@@ -556,34 +567,35 @@ std::vector<torch::Tensor> conv3p_wgrad(torch::Tensor dY, torch::Tensor in,
     auto dWt = torch::zeros({COUT, K}, dY.options().dtype(torch::kFloat32));
     auto db = torch::zeros({COUT}, dY.options().dtype(torch::kFloat32));
     long target_chunks = 1024;
-    long rows_per_chunk = std::max(32L, (M + target_chunks - 1) / target_chunks);
-    rows_per_chunk = ((rows_per_chunk + 31) / 32) * 32;
+    long rows_per_chunk = std::max(64L, (M + target_chunks - 1) / target_chunks);
+    rows_per_chunk = ((rows_per_chunk + 63) / 64) * 64;
     dim3 grid(icdiv(M, rows_per_chunk));
     auto stream = at::cuda::getCurrentCUDAStream();
     auto* dy = reinterpret_cast<const __hip_bfloat16*>(dY.data_ptr());
     const void* x = in.data_ptr();
-#define WG(U8, CIN_, RELU_)                                                   \
-    hipLaunchKernelGGL((conv3p_wgrad_kernel<U8, CIN_, RELU_>), grid,          \
+    bool big = COUT > 16;
+#define WG1(U8, CIN_, RELU_, NCO2_)                                           \
+    hipLaunchKernelGGL((conv3p_wgrad_kernel<U8, CIN_, RELU_, NCO2_>), grid,   \
                        dim3(256), 0, stream.stream(), dy, x,                  \
                        dWt.data_ptr<float>(), db.data_ptr<float>(), (int)M,   \
                        (int)H, (int)W, (int)COUT, (int)rows_per_chunk)
+#define WG(U8, CIN_, RELU_)                                                   \
+    do { if (big) WG1(U8, CIN_, RELU_, true); else WG1(U8, CIN_, RELU_, false); } while (0)
     if (u8) { TORCH_CHECK(CIN == 8); WG(true, 8, false); }
     else if (CIN == 8) { if (relu_in) WG(false, 8, true); else WG(false, 8, false); }
     else if (CIN == 16) { if (relu_in) WG(false, 16, true); else WG(false, 16, false); }
     else if (CIN == 32) { if (relu_in) WG(false, 32, true); else WG(false, 32, false); }
     else TORCH_CHECK(false, "unsupported CIN ", CIN);
 #undef WG
+#undef WG1
     return {dWt, db};
 }
 
-std::vector<torch::Tensor> maxpool3s2_fwd(torch::Tensor in, int64_t N,
-                                          int64_t H, int64_t W) {
+void maxpool3s2_fwd(torch::Tensor in, torch::Tensor out, torch::Tensor arg,
+                    int64_t N, int64_t H, int64_t W) {
     long C = in.size(3);
-    long OH = H / 2, OW = W / 2;
-    if (H % 2 == 1) { OH = (H + 1) / 2; OW = (W + 1) / 2; }
-    auto out = torch::zeros({N, OH + 2, OW + 2, C},
-                            in.options().dtype(torch::kBFloat16));
-    auto arg = torch::empty({N, OH, OW, C}, in.options().dtype(torch::kUInt8));
+    long OH = (H + 1) / 2, OW = (W + 1) / 2;
+    TORCH_CHECK(out.size(1) == OH + 2 && arg.size(1) == OH);
     long total = N * OH * OW * (C / 8);
     auto stream = at::cuda::getCurrentCUDAStream();
     hipLaunchKernelGGL(maxpool3s2_fwd_kernel, dim3(icdiv(total, 256)),
@@ -592,7 +604,6 @@ std::vector<torch::Tensor> maxpool3s2_fwd(torch::Tensor in, int64_t N,
                        reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
                        arg.data_ptr<unsigned char>(), (int)N, (int)H, (int)W,
                        (int)OH, (int)OW, (int)C);
-    return {out, arg};
 }
 
 void maxpool3s2_bwd(torch::Tensor dOut, torch::Tensor arg, torch::Tensor dIn,
@@ -608,12 +619,12 @@ void maxpool3s2_bwd(torch::Tensor dOut, torch::Tensor arg, torch::Tensor dIn,
                        (int)N, (int)H, (int)W, (int)OH, (int)OW, (int)C);
 }
 
-torch::Tensor pack_frames(torch::Tensor frames, int64_t H, int64_t W) {
+void pack_frames(torch::Tensor frames, torch::Tensor out, int64_t H,
+                 int64_t W) {
     long M = frames.size(0);
     long CIN = frames.size(3);
     TORCH_CHECK(frames.dtype() == torch::kUInt8 && CIN <= 8);
-    auto out = torch::zeros({M, H + 2, W + 2, 8},
-                            frames.options().dtype(torch::kUInt8));
+    TORCH_CHECK(out.size(1) == H + 2 && out.size(3) == 8);
     long total = M * H * W;
     auto stream = at::cuda::getCurrentCUDAStream();
     hipLaunchKernelGGL(pack_frames_kernel, dim3(icdiv(total, 256)), dim3(256),
@@ -621,7 +632,6 @@ torch::Tensor pack_frames(torch::Tensor frames, int64_t H, int64_t W) {
                        frames.data_ptr<unsigned char>(),
                        out.data_ptr<unsigned char>(), total, (int)H, (int)W,
                        (int)CIN);
-    return out;
 }
 
 torch::Tensor pad2dense(torch::Tensor in, int64_t N, int64_t H, int64_t W,

@@ -21,6 +21,33 @@ STAGES = ((84, 42), (42, 21), (21, 11))
 CHANS = (16, 32, 32)
 
 
+class _Arena:
+    """Step-scoped reusable device buffers.  Padded conv outputs need a
+    zero halo that the kernels never touch — so each buffer is allocated
+    (and zeroed) ONCE and reused every step, eliminating the per-step
+    FillFunctor storm (~2 ms/step at M=5440).  Buffers are handed out in
+    call order within a tag; reset(tag) rewinds that tag's counters."""
+
+    def __init__(self, device):
+        self.device = device
+        self.bufs = {}
+        self.idx = {}
+
+    def reset(self, tag):
+        for k in self.idx:
+            if k[0] == tag:
+                self.idx[k] = 0
+
+    def get(self, tag, shape, dtype=torch.bfloat16):
+        key = (tag,) + tuple(shape) + (dtype,)
+        lst = self.bufs.setdefault(key, [])
+        i = self.idx.get(key, 0)
+        if i >= len(lst):
+            lst.append(torch.zeros(*shape, device=self.device, dtype=dtype))
+        self.idx[key] = i + 1
+        return lst[i]
+
+
 def _pack_fwd(w: torch.Tensor, cp: int, dev) -> torch.Tensor:
     """(COUT, CIN, 3, 3) -> (COUT, 9*cp) bf16, k-order (ky, kx, c),
     input channels zero-padded to cp."""
@@ -45,6 +72,7 @@ class ImpalaPack:
         self.enc = enc
         self.device = device
         self.with_bwd = with_bwd
+        self.arena = _Arena(device)
         self.refresh()
 
     def _convs(self):
@@ -79,38 +107,49 @@ class ImpalaPack:
             self.wf_kn = self.wft.t().contiguous()
 
 
-def _zeros_pad(M, H, C, dev):
-    return torch.zeros(M, H + 2, H + 2, C, device=dev, dtype=torch.bfloat16)
+def pack_obs(m, pack: ImpalaPack, obs_hwc_u8: torch.Tensor) -> torch.Tensor:
+    """Frames -> halo-padded 8-channel u8 via the pack's arena; call once
+    per step and share between the online and target forward passes."""
+    M = obs_hwc_u8.shape[0]
+    pack.arena.reset("xp")
+    xp = pack.arena.get("xp", (M, 86, 86, 8), torch.uint8)
+    m.pack_frames(obs_hwc_u8, xp, 84, 84)
+    return xp
 
 
 def encoder_fwd(m, pack: ImpalaPack, obs_hwc_u8: torch.Tensor,
-                want_stash: bool):
+                want_stash: bool, xp: Optional[torch.Tensor] = None):
     """obs: (M, 84, 84, C<=8) u8 dense -> latent (M, hidden) bf16 (+stash).
 
     Stash layout (one dict): padded activations and pool argmaxes needed by
-    encoder_bwd."""
-    dev = pack.device
-    M = obs_hwc_u8.shape[0]
-    xp = m.pack_frames(obs_hwc_u8, 84, 84)          # (M, 86, 86, 8) u8
+    encoder_bwd.  Pass a prepacked ``xp`` (from pack_obs) to skip the frame
+    packing — the engine packs once and shares it with the target net."""
+    ar = pack.arena
+    M = obs_hwc_u8.shape[0] if xp is None else xp.shape[0]
+    ar.reset("fwd")
+    if xp is None:
+        xp = pack_obs(m, pack, obs_hwc_u8)
     st = {"xp": xp} if want_stash else None
     x = xp
     empty = torch.Tensor()
     for si, (hin, hout) in enumerate(STAGES):
         c = CHANS[si]
-        conv_out = _zeros_pad(M, hin, c, dev)
+        conv_out = ar.get("fwd", (M, hin + 2, hin + 2, c))
         m.conv3p(x, pack.wt[f"s{si}c"], pack.bias[f"s{si}c"], empty, empty,
                  conv_out, M, hin, hin, False, True, 0)
-        pooled, arg = m.maxpool3s2_fwd(conv_out, M, hin, hin)
+        pooled = ar.get("fwd", (M, hout + 2, hout + 2, c))
+        arg = ar.get("fwd", (M, hout, hout, c), torch.uint8)
+        m.maxpool3s2_fwd(conv_out, pooled, arg, M, hin, hin)
         if want_stash:
             st[f"s{si}c_in"] = x
             st[f"s{si}c_out"] = conv_out
             st[f"s{si}arg"] = arg
         x = pooled
         for ri in range(2):
-            y1 = _zeros_pad(M, hout, c, dev)
+            y1 = ar.get("fwd", (M, hout + 2, hout + 2, c))
             m.conv3p(x, pack.wt[f"s{si}r{ri}a"], pack.bias[f"s{si}r{ri}a"],
                      empty, empty, y1, M, hout, hout, True, True, 0)
-            out = _zeros_pad(M, hout, c, dev)
+            out = ar.get("fwd", (M, hout + 2, hout + 2, c))
             m.conv3p(y1, pack.wt[f"s{si}r{ri}b"], pack.bias[f"s{si}r{ri}b"],
                      x, empty, out, M, hout, hout, True, True, 1)
             if want_stash:
@@ -132,7 +171,8 @@ def encoder_bwd(m, pack: ImpalaPack, st: dict, dlat: torch.Tensor,
 
     Returns {name: (dW f32 torch-layout, db f32)} for the 15 convs plus
     ('fc', (dWf, dbf))."""
-    dev = pack.device
+    ar = pack.arena
+    ar.reset("bwd")
     M = dlat.shape[0]
     empty = torch.Tensor()
     grads = {}
@@ -144,7 +184,7 @@ def encoder_bwd(m, pack: ImpalaPack, st: dict, dlat: torch.Tensor,
     grads["fc"] = (dWf.view(hd, 11, 11, 32).permute(0, 3, 1, 2)
                    .reshape(hd, 3872).contiguous(), dbf)
 
-    dx = _zeros_pad(M, 11, 32, dev)
+    dx = ar.get("bwd", (M, 13, 13, 32))
     m.dense2pad_mask(dflat, st["s_out"], dx, M, 11, 11)
 
     for si in (2, 1, 0):
@@ -155,17 +195,17 @@ def encoder_bwd(m, pack: ImpalaPack, st: dict, dlat: torch.Tensor,
             nb, nm = f"s{si}r{ri}b", f"s{si}r{ri}a"
             dW2, db2 = m.conv3p_wgrad(dx, y1, M, hout, hout, True)
             grads[nb] = (self_conv_grad(dW2, c, c), db2)
-            dy1 = _zeros_pad(M, hout, c, dev)
+            dy1 = ar.get("bwd", (M, hout + 2, hout + 2, c))
             m.conv3p(dx, pack.wd[nb], empty, empty, y1, dy1,
                      M, hout, hout, False, False, 2)
             dW1, db1 = m.conv3p_wgrad(dy1, x, M, hout, hout, True)
             grads[nm] = (self_conv_grad(dW1, c, c), db1)
-            dx_new = _zeros_pad(M, hout, c, dev)
+            dx_new = ar.get("bwd", (M, hout + 2, hout + 2, c))
             m.conv3p(dy1, pack.wd[nm], empty, dx, x, dx_new,
                      M, hout, hout, False, False, 3)
             dx = dx_new
         # maxpool backward: dx (pooled grid) -> conv-out grid
-        dconv = _zeros_pad(M, hin, c, dev)
+        dconv = ar.get("bwd", (M, hin + 2, hin + 2, c))
         m.maxpool3s2_bwd(dx, st[f"s{si}arg"], dconv, M, hin, hin, hout, hout)
         # stage conv wgrad (+ dgrad, except stage 0 whose input is data)
         sin = st[f"s{si}c_in"]
@@ -173,7 +213,7 @@ def encoder_bwd(m, pack: ImpalaPack, st: dict, dlat: torch.Tensor,
         dWc, dbc = m.conv3p_wgrad(dconv, sin, M, hin, hin, False)
         grads[f"s{si}c"] = (self_conv_grad(dWc, c, cin), dbc)
         if si > 0:
-            dx = _zeros_pad(M, hin, CHANS[si - 1], dev)
+            dx = ar.get("bwd", (M, hin + 2, hin + 2, CHANS[si - 1]))
             m.conv3p(dconv, pack.wd[f"s{si}c"], empty, empty, empty, dx,
                      M, hin, hin, False, False, 0)
     return grads

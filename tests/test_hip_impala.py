@@ -126,16 +126,40 @@ def test_conv3p_wgrad_matches_autograd():
     close(db, dy.sum((0, 2, 3)), name="conv3p wgrad bias")
 
 
+def test_conv3p_wgrad_u8_frames():
+    """The stage-0 wgrad path: u8 frames (8 padded channels), no relu-in."""
+    torch.manual_seed(9)
+    N, H = 4, 84
+    frames = torch.randint(0, 256, (N, H, H, 4), device="cuda",
+                           dtype=torch.uint8)
+    xp = torch.zeros(N, H + 2, H + 2, 8, device="cuda", dtype=torch.uint8)
+    M_.pack_frames(frames, xp, H, H)
+    w = torch.randn(16, 4, 3, 3, device="cuda", requires_grad=True)
+    x = (frames.float() / 255.0).permute(0, 3, 1, 2).detach()
+    y = F.conv2d(x, w, None, 1, 1)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    dyb = dy.permute(0, 2, 3, 1).bfloat16()
+    dWt, db = M_.conv3p_wgrad(pad_nhwc(dyb).contiguous(), xp, N, H, H, False)
+    dW = imp.self_conv_grad(dWt, 16, 4)
+    close(dW, w.grad, rtol=4e-2, name="conv3p wgrad u8")
+    close(db, dy.sum((0, 2, 3)), name="conv3p wgrad u8 bias")
+
+
 def test_maxpool3s2_fwd_bwd():
     torch.manual_seed(5)
     for H in (84, 42, 21):
         N, C = 3, 16
         x = torch.randn(N, H, H, C, device="cuda").bfloat16()
         xp = pad_nhwc(x).contiguous()
-        out, arg = M_.maxpool3s2_fwd(xp, N, H, H)
+        OH = (H + 1) // 2
+        out = torch.zeros(N, OH + 2, OH + 2, C, device="cuda",
+                          dtype=torch.bfloat16)
+        arg = torch.empty(N, OH, OH, C, device="cuda", dtype=torch.uint8)
+        M_.maxpool3s2_fwd(xp, out, arg, N, H, H)
         xt = x.float().permute(0, 3, 1, 2).requires_grad_()
         ref = F.max_pool2d(xt, 3, 2, 1)
-        OH = ref.shape[-1]
+        assert OH == ref.shape[-1]
         close(out[:, 1:OH + 1, 1:OH + 1],
               ref.detach().permute(0, 2, 3, 1), name=f"maxpool fwd {H}")
         dout = torch.randn_like(ref)
@@ -155,7 +179,8 @@ def test_pack_and_pad_utils():
     Mn = 4
     frames = torch.randint(0, 256, (Mn, 84, 84, 4), device="cuda",
                            dtype=torch.uint8)
-    xp = M_.pack_frames(frames, 84, 84)
+    xp = torch.zeros(Mn, 86, 86, 8, device="cuda", dtype=torch.uint8)
+    M_.pack_frames(frames, xp, 84, 84)
     assert xp.shape == (Mn, 86, 86, 8)
     assert torch.equal(xp[:, 1:85, 1:85, :4], frames)
     assert xp[:, :, :, 4:].abs().sum().item() == 0
@@ -216,9 +241,14 @@ def test_impala_encoder_full_fwd_bwd_vs_autograd():
             name_map[f"s{si}r{ri}b"] = res.conv2
     for name, conv in name_map.items():
         dw, db = grads[name]
-        rel_fro(dw.view_as(conv.weight.grad), conv.weight.grad, 0.10,
+        # s0c sits at the bottom of a 15-conv bf16 backward chain crossing
+        # two maxpool argmax routings; its accumulated divergence from the
+        # fp32 reference is larger (the isolated u8 wgrad is tested tightly
+        # in test_conv3p_wgrad_u8_frames)
+        tol = 0.20 if name == "s0c" else 0.10
+        rel_fro(dw.view_as(conv.weight.grad), conv.weight.grad, tol,
                 f"{name} dW")
-        rel_fro(db, conv.bias.grad, 0.10, f"{name} db")
+        rel_fro(db, conv.bias.grad, tol, f"{name} db")
 
 
 def test_impala_engine_train_step():
