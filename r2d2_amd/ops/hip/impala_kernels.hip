@@ -102,18 +102,22 @@ __global__ __launch_bounds__(256) void conv3p_kernel(
     int frow = lane & 15;
     int kseg = (lane >> 4) * 8;
 
+    // 32-bit index math: M <= 5440*84*84 < 2^31, so unsigned division
+    // lowers to multiply-shift instead of a 64-bit libcall (8 divisions per
+    // lane in the epilogue otherwise dominate these small-K kernels).
+    const unsigned HW = (unsigned)(H * W);
     long abase[2];
     bool avalid[2];
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-        long r = row0 + i * 16 + frow;
-        avalid[i] = r < M;
+        unsigned r = (unsigned)row0 + i * 16 + frow;
+        avalid[i] = r < (unsigned)M;
         if (avalid[i]) {
-            long n = r / (H * W);
-            int p = (int)(r % (H * W));
-            int oy = p / W, ox = p % W;
+            unsigned n = r / HW;
+            unsigned p = r % HW;
+            int oy = p / (unsigned)W, ox = p % (unsigned)W;
             // window top-left in padded coords = (oy, ox)
-            abase[i] = ((n * PH + oy) * PW + ox) * CIN;
+            abase[i] = (((long)n * PH + oy) * PW + ox) * CIN;
         } else {
             abase[i] = 0;
         }
@@ -164,13 +168,13 @@ __global__ __launch_bounds__(256) void conv3p_kernel(
         for (int j = 0; j < NB; ++j)
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                long rr = row0 + i * 16 + crow + r;
+                unsigned rr = (unsigned)row0 + i * 16 + crow + r;
                 int cc = j * 16 + ccol;
-                if (rr < M && cc < COUT) {
-                    long n = rr / (H * W);
-                    int p = (int)(rr % (H * W));
-                    int oy = p / W, ox = p % W;
-                    long oidx = ((n * PH + oy + 1) * PW + ox + 1) * COUT + cc;
+                if (rr < (unsigned)M && cc < COUT) {
+                    unsigned n = rr / HW;
+                    unsigned p = rr % HW;
+                    int oy = p / (unsigned)W, ox = p % (unsigned)W;
+                    long oidx = (((long)n * PH + oy + 1) * PW + ox + 1) * COUT + cc;
                     float v = acc[i][j][r];
                     if (HAS_BIAS) v += bias[cc];
                     if (EPI == 2 || EPI == 3)
@@ -233,10 +237,10 @@ __global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
                 long gm = m0 + mrow;
                 bf16x8 v = izero();
                 if (gm < mend) {
-                    long n = gm / (H * W);
-                    int p = (int)(gm % (H * W));
-                    int oy = p / W, ox = p % W;
-                    long base = ((n * PH + oy + 1) * PW + ox + 1) * COUT;
+                    unsigned n = (unsigned)gm / (unsigned)(H * W);
+                    unsigned p = (unsigned)gm % (unsigned)(H * W);
+                    int oy = p / (unsigned)W, ox = p % (unsigned)W;
+                    long base = (((long)n * PH + oy + 1) * PW + ox + 1) * COUT;
                     ibf8u u;
 #pragma unroll
                     for (int e = 0; e < 8; ++e)
@@ -253,10 +257,10 @@ __global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
                 long gm = m0 + mrow;
                 bf16x8 w = izero();
                 if (gm < mend) {
-                    long n = gm / (H * W);
-                    int p = (int)(gm % (H * W));
-                    int oy = p / W, ox = p % W;
-                    long base = ((n * PH + oy) * PW + ox) * CIN;
+                    unsigned n = (unsigned)gm / (unsigned)(H * W);
+                    unsigned p = (unsigned)gm % (unsigned)(H * W);
+                    int oy = p / (unsigned)W, ox = p % (unsigned)W;
+                    long base = (((long)n * PH + oy) * PW + ox) * CIN;
                     int dy_ = k / KROW, rem = k % KROW;
                     long off = base + (long)dy_ * PW * CIN + rem;
                     if (IN_U8)
@@ -332,15 +336,15 @@ __global__ __launch_bounds__(256) void maxpool3s2_fwd_kernel(
     int C) {
     const int PW = W + 2, PH = H + 2;
     const int C8 = C / 8;
-    long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    long total = (long)N * OH * OW * C8;
+    unsigned idx = blockIdx.x * blockDim.x + threadIdx.x;
+    unsigned total = (unsigned)((long)N * OH * OW * C8);
     if (idx >= total) return;
-    int c8 = (int)(idx % C8);
-    long t = idx / C8;
-    int ox = (int)(t % OW);
-    t /= OW;
-    int oy = (int)(t % OH);
-    long n = t / OH;
+    int c8 = (int)(idx % (unsigned)C8);
+    unsigned t = idx / (unsigned)C8;
+    int ox = (int)(t % (unsigned)OW);
+    t /= (unsigned)OW;
+    int oy = (int)(t % (unsigned)OH);
+    unsigned n = t / (unsigned)OH;
 
     float best[8];
     int bidx[8];
@@ -355,7 +359,7 @@ __global__ __launch_bounds__(256) void maxpool3s2_fwd_kernel(
             int x = 2 * ox - 1 + kx;
             if (x < 0 || x >= W) continue;
             ibf8u v;
-            v.v = iload8(in + ((n * PH + y + 1) * PW + x + 1) * C + c8 * 8);
+            v.v = iload8(in + (((long)n * PH + y + 1) * PW + x + 1) * C + c8 * 8);
 #pragma unroll
             for (int e = 0; e < 8; ++e) {
                 float f = bf2f(v.e[e]);
@@ -368,7 +372,7 @@ __global__ __launch_bounds__(256) void maxpool3s2_fwd_kernel(
 #pragma unroll
     for (int e = 0; e < 8; ++e) o.e[e] = f2bf(best[e]);
     *reinterpret_cast<bf16x8*>(
-        out + ((n * POH + oy + 1) * POW + ox + 1) * C + c8 * 8) = o.v;
+        out + (((long)n * POH + oy + 1) * POW + ox + 1) * C + c8 * 8) = o.v;
     long abase = (((long)n * OH + oy) * OW + ox) * C + c8 * 8;
 #pragma unroll
     for (int e = 0; e < 8; ++e) arg[abase + e] = (unsigned char)bidx[e];
@@ -384,15 +388,15 @@ __global__ __launch_bounds__(256) void maxpool3s2_bwd_kernel(
     const int PW = W + 2, PH = H + 2;
     const int POW = OW + 2, POH = OH + 2;
     const int C8 = C / 8;
-    long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    long total = (long)N * H * W * C8;
+    unsigned idx = blockIdx.x * blockDim.x + threadIdx.x;
+    unsigned total = (unsigned)((long)N * H * W * C8);
     if (idx >= total) return;
-    int c8 = (int)(idx % C8);
-    long t = idx / C8;
-    int x = (int)(t % W);
-    t /= W;
-    int y = (int)(t % H);
-    long n = t / H;
+    int c8 = (int)(idx % (unsigned)C8);
+    unsigned t = idx / (unsigned)C8;
+    int x = (int)(t % (unsigned)W);
+    t /= (unsigned)W;
+    int y = (int)(t % (unsigned)H);
+    unsigned n = t / (unsigned)H;
 
     float acc[8] = {};
     int oy_lo = (y - 1 + 1) / 2;      // ceil((y-1)/2) for y>=0
@@ -409,7 +413,7 @@ __global__ __launch_bounds__(256) void maxpool3s2_bwd_kernel(
             int tap = ky * 3 + kx;
             long abase = (((long)n * OH + oy) * OW + ox) * C + c8 * 8;
             ibf8u g;
-            g.v = iload8(dOut + ((n * POH + oy + 1) * POW + ox + 1) * C + c8 * 8);
+            g.v = iload8(dOut + (((long)n * POH + oy + 1) * POW + ox + 1) * C + c8 * 8);
 #pragma unroll
             for (int e = 0; e < 8; ++e)
                 if (arg[abase + e] == tap) acc[e] += bf2f(g.e[e]);
@@ -419,7 +423,7 @@ __global__ __launch_bounds__(256) void maxpool3s2_bwd_kernel(
 #pragma unroll
     for (int e = 0; e < 8; ++e) o.e[e] = f2bf(acc[e]);
     *reinterpret_cast<bf16x8*>(
-        dIn + ((n * PH + y + 1) * PW + x + 1) * C + c8 * 8) = o.v;
+        dIn + (((long)n * PH + y + 1) * PW + x + 1) * C + c8 * 8) = o.v;
 }
 
 // ---------------------------------------------------------------------------
@@ -429,14 +433,14 @@ __global__ __launch_bounds__(256) void maxpool3s2_bwd_kernel(
 __global__ __launch_bounds__(256) void pack_frames_kernel(
     const unsigned char* __restrict__ in, unsigned char* __restrict__ out,
     long total, int H, int W, int CIN) {
-    long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    if (idx >= total) return;  // total = M*H*W
-    int x = (int)(idx % W);
-    long t = idx / W;
-    int y = (int)(t % H);
-    long m = t / H;
+    unsigned idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= (unsigned)total) return;  // total = M*H*W
+    int x = (int)(idx % (unsigned)W);
+    unsigned t = idx / (unsigned)W;
+    int y = (int)(t % (unsigned)H);
+    unsigned m = t / (unsigned)H;
     const unsigned char* src = in + (idx * CIN);
-    unsigned char* dst = out + ((m * (H + 2) + y + 1) * (long)(W + 2) + x + 1) * 8;
+    unsigned char* dst = out + (((long)m * (H + 2) + y + 1) * (long)(W + 2) + x + 1) * 8;
     uint2 v{0, 0};
     for (int c = 0; c < CIN; ++c)
         reinterpret_cast<unsigned char*>(&v)[c] = src[c];
@@ -453,15 +457,15 @@ __global__ __launch_bounds__(256) void pack_frames_kernel(
 __global__ __launch_bounds__(256) void pad2dense_kernel(
     const __hip_bfloat16* __restrict__ in, __hip_bfloat16* __restrict__ out,
     long total, int H, int W, int C, int relu) {
-    long idx = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
-    if (idx >= total) return;  // total = N*H*W*C
-    long c = idx % C;
-    long t = idx / C;
-    int x = (int)(t % W);
-    t /= W;
-    int y = (int)(t % H);
-    long n = t / H;
-    bf16x8 v = iload8(in + ((n * (H + 2) + y + 1) * (long)(W + 2) + x + 1) * C + c);
+    unsigned idx = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
+    if (idx >= (unsigned)total) return;  // total = N*H*W*C
+    unsigned c = idx % (unsigned)C;
+    unsigned t = idx / (unsigned)C;
+    int x = (int)(t % (unsigned)W);
+    t /= (unsigned)W;
+    int y = (int)(t % (unsigned)H);
+    unsigned n = t / (unsigned)H;
+    bf16x8 v = iload8(in + (((long)n * (H + 2) + y + 1) * (long)(W + 2) + x + 1) * C + c);
     if (relu) v = irelu8(v);
     *reinterpret_cast<bf16x8*>(out + idx) = v;
 }
@@ -470,15 +474,15 @@ __global__ __launch_bounds__(256) void dense2pad_mask_kernel(
     const __hip_bfloat16* __restrict__ dflat,
     const __hip_bfloat16* __restrict__ act_pad,
     __hip_bfloat16* __restrict__ out, long total, int H, int W, int C) {
-    long idx = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
-    if (idx >= total) return;
-    long c = idx % C;
-    long t = idx / C;
-    int x = (int)(t % W);
-    t /= W;
-    int y = (int)(t % H);
-    long n = t / H;
-    long pidx = ((n * (H + 2) + y + 1) * (long)(W + 2) + x + 1) * C + c;
+    unsigned idx = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
+    if (idx >= (unsigned)total) return;
+    unsigned c = idx % (unsigned)C;
+    unsigned t = idx / (unsigned)C;
+    int x = (int)(t % (unsigned)W);
+    t /= (unsigned)W;
+    int y = (int)(t % (unsigned)H);
+    unsigned n = t / (unsigned)H;
+    long pidx = (((long)n * (H + 2) + y + 1) * (long)(W + 2) + x + 1) * C + c;
     ibf8u g, a, o;
     g.v = iload8(dflat + idx);
     a.v = iload8(act_pad + pidx);

@@ -277,3 +277,53 @@ def test_impala_engine_train_step():
         losses.append(loss)
     assert losses[-1] < losses[0]
     cfg.apply("mspacman")
+
+
+def test_res_block_bwd_isolated():
+    """One residual block (C=16, the s0r1a config) fwd+bwd via the exact
+    engine call pattern vs autograd — localizes accumulation vs kernel bugs."""
+    torch.manual_seed(12)
+    N, H, C = 6, 42, 16
+    x = torch.randn(N, H, H, C, device="cuda").bfloat16()
+    w1 = torch.randn(C, C, 3, 3, device="cuda", requires_grad=True)
+    b1 = torch.randn(C, device="cuda", requires_grad=True)
+    w2 = torch.randn(C, C, 3, 3, device="cuda", requires_grad=True)
+    b2 = torch.randn(C, device="cuda", requires_grad=True)
+    for t in (w1, w2):
+        t.data *= 0.2
+
+    xp = pad_nhwc(x).contiguous()
+    wt1, wt2 = imp._pack_fwd(w1, C, "cuda"), imp._pack_fwd(w2, C, "cuda")
+    wd1, wd2 = imp._pack_dgrad(w1, "cuda"), imp._pack_dgrad(w2, "cuda")
+    y1 = torch.zeros_like(xp)
+    M_.conv3p(xp, wt1, b1.detach(), EMPTY, EMPTY, y1, N, H, H, True, True, 0)
+    out = torch.zeros_like(xp)
+    M_.conv3p(y1, wt2, b2.detach(), xp, EMPTY, out, N, H, H, True, True, 1)
+
+    xe = x.float().permute(0, 3, 1, 2).detach().requires_grad_()
+    y1e = F.conv2d(F.relu(xe), w1, b1, 1, 1)
+    oute = xe + F.conv2d(F.relu(y1e), w2, b2, 1, 1)
+    close(out[:, 1:H + 1, 1:H + 1], oute.detach().permute(0, 2, 3, 1),
+          name="resblock fwd")
+
+    dout = torch.randn_like(oute)
+    oute.backward(dout)
+    doutp = pad_nhwc(dout.permute(0, 2, 3, 1).bfloat16()).contiguous()
+    # engine call pattern (impala.py encoder_bwd res loop)
+    dW2, db2g = M_.conv3p_wgrad(doutp, y1, N, H, H, True)
+    dy1 = torch.zeros_like(xp)
+    M_.conv3p(doutp, wd2, EMPTY, EMPTY, y1, dy1, N, H, H, False, False, 2)
+    dW1, db1g = M_.conv3p_wgrad(dy1, xp, N, H, H, True)
+    dx = torch.zeros_like(xp)
+    M_.conv3p(dy1, wd1, EMPTY, doutp, xp, dx, N, H, H, False, False, 3)
+
+    def rel(a, b):
+        return float((a.float().flatten() - b.float().flatten()).norm()
+                     / (b.float().flatten().norm() + 1e-8))
+
+    assert rel(imp.self_conv_grad(dW2, C, C), w2.grad) < 0.04, "dW2"
+    assert rel(db2g, b2.grad) < 0.04, "db2"
+    assert rel(imp.self_conv_grad(dW1, C, C), w1.grad) < 0.05, "dW1"
+    assert rel(db1g, b1.grad) < 0.05, "db1"
+    assert rel(dx[:, 1:H + 1, 1:H + 1],
+               xe.grad.permute(0, 2, 3, 1)) < 0.05, "dx"
