@@ -253,10 +253,17 @@ class HipNetworkEngine:
         return q, (adv1, val1)
 
     # ------------------------------------------------------------------
+    _pos_cache = None
+
     def _positions(self, burn, learn, fwd, T):
         """Host-side gather indices.  Returns (learn_pos, tgt_pos) flat int64
         arrays indexing (b * (T+1) + t + 1) rows of Hout, plus per-row b,t
-        (for the dHext scatter)."""
+        (for the dHext scatter).  Cached by batch layout — fixed-shape
+        batches (the common case) skip the python loop and re-upload."""
+        key = (T, burn.numpy().tobytes(), learn.numpy().tobytes(),
+               fwd.numpy().tobytes())
+        if self._pos_cache is not None and self._pos_cache[0] == key:
+            return self._pos_cache[1]
         n = self.cfg.forward_steps
         lp, tp, lbt = [], [], []
         for b, (bu, le, fw) in enumerate(zip(burn.tolist(), learn.tolist(),
@@ -266,7 +273,20 @@ class HipNetworkEngine:
             lp.append(b * (T + 1) + t_learn + 1)
             tp.append(b * (T + 1) + t_tgt + 1)
             lbt.append(b * T + t_learn)
-        return (np.concatenate(lp), np.concatenate(tp), np.concatenate(lbt))
+        dev = self.device
+        lp = np.concatenate(lp)
+        tp = np.concatenate(tp)
+        lbt = np.concatenate(lbt)
+        lp_t = torch.from_numpy(lp).to(dev)
+        tp_t = torch.from_numpy(tp).to(dev)
+        lbt_t = torch.from_numpy(lbt).to(dev)
+        tgt_bt = tp_t - (tp_t // (T + 1)) - 1   # b*(T+1)+t+1 -> b*T + t
+        lens_dev = (burn + learn + fwd).to(torch.int32).to(dev)
+        seg = torch.zeros(len(burn) + 1, dtype=torch.int32)
+        seg[1:] = torch.cumsum(learn.to(torch.int32), 0)
+        out = (lp_t, tp_t, lbt_t, tgt_bt, lens_dev, seg.to(dev))
+        self._pos_cache = (key, out)
+        return out
 
     # ------------------------------------------------------------------
     def train_step(self, batch):
@@ -287,8 +307,10 @@ class HipNetworkEngine:
             obs_hwc = obs_hwc.contiguous()
         la = batch.last_action.to(dev)
         lr = batch.last_reward.to(dev)
-        lens = (batch.burn_in_steps + batch.learning_steps
-                + batch.forward_steps).to(torch.int32).to(dev)
+        lp_t, tp_t, lbt_t, tgt_bt, lens, seg = self._positions(
+            batch.burn_in_steps, batch.learning_steps,
+            batch.forward_steps, T)
+        R = lp_t.shape[0]
         init = batch.hidden.float().contiguous()   # (2, B, H)
 
         self._mark("start")
@@ -310,13 +332,6 @@ class HipNetworkEngine:
             init, init, lens, self.bar, True)
         self._mark("lstm_fwd")
 
-        lp, tp, lbt = self._positions(batch.burn_in_steps,
-                                      batch.learning_steps,
-                                      batch.forward_steps, T)
-        lp_t = torch.from_numpy(lp).to(dev)
-        tp_t = torch.from_numpy(tp).to(dev)
-        R = lp.shape[0]
-
         Ho_flat = Ho.view(-1, H)
         h_learn = Ho_flat.index_select(0, lp_t)
         h_tgt_o = Ho_flat.index_select(0, tp_t)
@@ -330,9 +345,6 @@ class HipNetworkEngine:
         self._mark("heads")
 
         # ---- fused loss + priorities ----------------------------------
-        seg = torch.zeros(B + 1, dtype=torch.int32)
-        seg[1:] = torch.cumsum(batch.learning_steps.to(torch.int32), 0)
-        seg = seg.to(dev)
         loss, dq, abs_td, _ = m.fused_double_q_loss(
             q_learn.contiguous(), q_online_tgt.contiguous(), q_tgt.contiguous(),
             batch.action.view(-1).long(), batch.n_step_reward, batch.gamma,
@@ -361,8 +373,7 @@ class HipNetworkEngine:
         dh_rows = (dh_a.float() + dh_v.float())     # (2R, 512)
         # scatter-add into dHext (B, T, H): learn rows at t, tgt rows at t_tgt
         dHext = torch.zeros(B * T, H, device=dev)
-        dHext.index_add_(0, torch.from_numpy(lbt).to(dev), dh_rows[:R])
-        tgt_bt = tp_t - (tp_t // (T + 1)) - 1       # b*(T+1)+t+1 -> b*T + t
+        dHext.index_add_(0, lbt_t, dh_rows[:R])
         dHext.index_add_(0, tgt_bt, dh_rows[R:])
         dHext = dHext.view(B, T, H)
         self._mark("head_bwd")
