@@ -83,7 +83,12 @@ __device__ __forceinline__ bf16x8 idequant8(const unsigned char* p) {
 // NCOL 16: each wave 32 rows x 16 cols (1 B frag); NCOL 32: 32x32 (2 B
 // frags).  4 waves always stack the M dim -> 128 rows per workgroup.
 // ---------------------------------------------------------------------------
-template <bool IN_U8, int CIN, int NCOL, bool RELU_IN, bool HAS_BIAS, int EPI>
+// HT: compile-time H (=W) when nonzero — the hot IMPALA geometries are
+// instantiated with it so the per-lane row->(n,y,x) divisions lower to
+// multiply-shift by constants (10 divisions per lane otherwise dominate
+// these small-K launches); HT=0 keeps runtime dims for odd test shapes.
+template <bool IN_U8, int CIN, int NCOL, bool RELU_IN, bool HAS_BIAS, int EPI,
+          int HT = 0>
 __global__ __launch_bounds__(256) void conv3p_kernel(
     const void* __restrict__ in,
     const __hip_bfloat16* __restrict__ Wt,
@@ -94,8 +99,10 @@ __global__ __launch_bounds__(256) void conv3p_kernel(
     int M, int H, int W, int COUT) {
     constexpr int K = 9 * CIN;
     constexpr int KROW = 3 * CIN;  // contiguous k per ky
-    const int PW = W + 2;
-    const int PH = H + 2;
+    const int Hc = HT ? HT : H;
+    const int Wc = HT ? HT : W;
+    const int PW = Wc + 2;
+    const int PH = Hc + 2;
     int wave = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
     long row0 = (long)blockIdx.x * 128 + wave * 32;
@@ -105,7 +112,7 @@ __global__ __launch_bounds__(256) void conv3p_kernel(
     // 32-bit index math: M <= 5440*84*84 < 2^31, so unsigned division
     // lowers to multiply-shift instead of a 64-bit libcall (8 divisions per
     // lane in the epilogue otherwise dominate these small-K kernels).
-    const unsigned HW = (unsigned)(H * W);
+    const unsigned HW = (unsigned)(Hc * Wc);
     long abase[2];
     bool avalid[2];
 #pragma unroll
@@ -115,7 +122,7 @@ __global__ __launch_bounds__(256) void conv3p_kernel(
         if (avalid[i]) {
             unsigned n = r / HW;
             unsigned p = r % HW;
-            int oy = p / (unsigned)W, ox = p % (unsigned)W;
+            int oy = p / (unsigned)Wc, ox = p % (unsigned)Wc;
             // window top-left in padded coords = (oy, ox)
             abase[i] = (((long)n * PH + oy) * PW + ox) * CIN;
         } else {
@@ -173,7 +180,7 @@ __global__ __launch_bounds__(256) void conv3p_kernel(
                 if (rr < (unsigned)M && cc < COUT) {
                     unsigned n = rr / HW;
                     unsigned p = rr % HW;
-                    int oy = p / (unsigned)W, ox = p % (unsigned)W;
+                    int oy = p / (unsigned)Wc, ox = p % (unsigned)Wc;
                     long oidx = (((long)n * PH + oy + 1) * PW + ox + 1) * COUT + cc;
                     float v = acc[i][j][r];
                     if (HAS_BIAS) v += bias[cc];
@@ -194,7 +201,7 @@ __global__ __launch_bounds__(256) void conv3p_kernel(
 // NCO2: true when COUT > 16 (two cout fragments per wave, 4-way K split);
 // false for COUT <= 16 (single cout fragment, 2-way K split x 2-way M split
 // — halves the padded-K waste that dominates the small-channel stages).
-template <bool IN_U8, int CIN, bool RELU_IN, bool NCO2>
+template <bool IN_U8, int CIN, bool RELU_IN, bool NCO2, int HT = 0>
 __global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
     const __hip_bfloat16* __restrict__ dY,   // (N, H+2, W+2, COUT) padded
     const void* __restrict__ in,             // (N, H+2, W+2, CIN) padded
@@ -207,8 +214,10 @@ __global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
     constexpr int KHALF = ((K / KSPLIT + 15) / 16) * 16;
     constexpr int KFRAG = KHALF / 16;
     constexpr int NCO = NCO2 ? 2 : 1;
-    const int PW = W + 2;
-    const int PH = H + 2;
+    const int Hc = HT ? HT : H;
+    const int Wc = HT ? HT : W;
+    const int PW = Wc + 2;
+    const int PH = Hc + 2;
     constexpr int TROWS = NCO2 ? 32 : 64;
     __shared__ __hip_bfloat16 s_dy[TROWS][32 + 8];
     __shared__ __hip_bfloat16 s_a[TROWS][K + 8];
@@ -237,9 +246,9 @@ __global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
                 long gm = m0 + mrow;
                 bf16x8 v = izero();
                 if (gm < mend) {
-                    unsigned n = (unsigned)gm / (unsigned)(H * W);
-                    unsigned p = (unsigned)gm % (unsigned)(H * W);
-                    int oy = p / (unsigned)W, ox = p % (unsigned)W;
+                    unsigned n = (unsigned)gm / (unsigned)(Hc * Wc);
+                    unsigned p = (unsigned)gm % (unsigned)(Hc * Wc);
+                    int oy = p / (unsigned)Wc, ox = p % (unsigned)Wc;
                     long base = (((long)n * PH + oy + 1) * PW + ox + 1) * COUT;
                     ibf8u u;
 #pragma unroll
@@ -257,9 +266,9 @@ __global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
                 long gm = m0 + mrow;
                 bf16x8 w = izero();
                 if (gm < mend) {
-                    unsigned n = (unsigned)gm / (unsigned)(H * W);
-                    unsigned p = (unsigned)gm % (unsigned)(H * W);
-                    int oy = p / (unsigned)W, ox = p % (unsigned)W;
+                    unsigned n = (unsigned)gm / (unsigned)(Hc * Wc);
+                    unsigned p = (unsigned)gm % (unsigned)(Hc * Wc);
+                    int oy = p / (unsigned)Wc, ox = p % (unsigned)Wc;
                     long base = (((long)n * PH + oy) * PW + ox) * CIN;
                     int dy_ = k / KROW, rem = k % KROW;
                     long off = base + (long)dy_ * PW * CIN + rem;
@@ -543,6 +552,53 @@ void conv3p(torch::Tensor in, torch::Tensor Wt, torch::Tensor bias,
         else C3P_EPI(U8, CIN_, NCOL_, false, false);                          \
     } while (0)
 
+    // compile-time-H fast path for the IMPALA geometries (divisions by
+    // constants); falls through to the runtime-H chain otherwise
+#define C3PH(U8, CIN_, NCOL_, RELU_, BIAS_, EPI_, HT_)                        \
+    hipLaunchKernelGGL((conv3p_kernel<U8, CIN_, NCOL_, RELU_, BIAS_, EPI_,    \
+                                      HT_>),                                  \
+                       grid, dim3(256), 0, stream.stream(), x, w, b, rp, mp,  \
+                       o, (int)M, (int)H, (int)W, (int)COUT)
+    bool done = true;
+    if (H == 84 && u8 && COUT == 16 && !relu_in && has_bias && epi == 0)
+        C3PH(true, 8, 16, false, true, 0, 84);
+    else if (H == 42 && !u8 && CIN == 16 && COUT == 16 && relu_in && has_bias
+             && epi == 0)
+        C3PH(false, 16, 16, true, true, 0, 42);
+    else if (H == 42 && !u8 && CIN == 16 && COUT == 16 && relu_in && has_bias
+             && epi == 1)
+        C3PH(false, 16, 16, true, true, 1, 42);
+    else if (H == 42 && !u8 && CIN == 16 && COUT == 16 && !relu_in
+             && !has_bias && epi == 2)
+        C3PH(false, 16, 16, false, false, 2, 42);
+    else if (H == 42 && !u8 && CIN == 16 && COUT == 16 && !relu_in
+             && !has_bias && epi == 3)
+        C3PH(false, 16, 16, false, false, 3, 42);
+    else if (H == 42 && !u8 && CIN == 16 && COUT == 32 && !relu_in && has_bias
+             && epi == 0)
+        C3PH(false, 16, 32, false, true, 0, 42);
+    else if (H == 42 && !u8 && CIN == 32 && COUT == 16 && !relu_in
+             && !has_bias && epi == 0)
+        C3PH(false, 32, 16, false, false, 0, 42);
+    else if (H == 21 && !u8 && CIN == 32 && COUT == 32) {
+        if (relu_in && has_bias && epi == 0) C3PH(false, 32, 32, true, true, 0, 21);
+        else if (relu_in && has_bias && epi == 1) C3PH(false, 32, 32, true, true, 1, 21);
+        else if (!relu_in && !has_bias && epi == 0) C3PH(false, 32, 32, false, false, 0, 21);
+        else if (!relu_in && !has_bias && epi == 2) C3PH(false, 32, 32, false, false, 2, 21);
+        else if (!relu_in && !has_bias && epi == 3) C3PH(false, 32, 32, false, false, 3, 21);
+        else done = false;
+    } else if (H == 11 && !u8 && CIN == 32 && COUT == 32) {
+        if (relu_in && has_bias && epi == 0) C3PH(false, 32, 32, true, true, 0, 11);
+        else if (relu_in && has_bias && epi == 1) C3PH(false, 32, 32, true, true, 1, 11);
+        else if (!relu_in && !has_bias && epi == 2) C3PH(false, 32, 32, false, false, 2, 11);
+        else if (!relu_in && !has_bias && epi == 3) C3PH(false, 32, 32, false, false, 3, 11);
+        else done = false;
+    } else {
+        done = false;
+    }
+#undef C3PH
+    if (done) return;
+
     if (u8) {
         TORCH_CHECK(CIN == 8, "u8 conv expects 8 padded channels");
         if (COUT <= 16) C3P_RB(true, 8, 16); else C3P_RB(true, 8, 32);
@@ -585,6 +641,27 @@ std::vector<torch::Tensor> conv3p_wgrad(torch::Tensor dY, torch::Tensor in,
                        (int)H, (int)W, (int)COUT, (int)rows_per_chunk)
 #define WG(U8, CIN_, RELU_)                                                   \
     do { if (big) WG1(U8, CIN_, RELU_, true); else WG1(U8, CIN_, RELU_, false); } while (0)
+#define WGH(U8, CIN_, RELU_, NCO2_, HT_)                                      \
+    hipLaunchKernelGGL((conv3p_wgrad_kernel<U8, CIN_, RELU_, NCO2_, HT_>),    \
+                       grid, dim3(256), 0, stream.stream(), dy, x,            \
+                       dWt.data_ptr<float>(), db.data_ptr<float>(), (int)M,   \
+                       (int)H, (int)W, (int)COUT, (int)rows_per_chunk)
+    bool done = true;
+    if (H == 84 && u8 && !big) WGH(true, 8, false, false, 84);
+    else if (H == 42 && !u8 && CIN == 16 && relu_in && !big)
+        WGH(false, 16, true, false, 42);
+    else if (H == 42 && !u8 && CIN == 16 && !relu_in && big)
+        WGH(false, 16, false, true, 42);
+    else if (H == 21 && !u8 && CIN == 32 && relu_in && big)
+        WGH(false, 32, true, true, 21);
+    else if (H == 21 && !u8 && CIN == 32 && !relu_in && big)
+        WGH(false, 32, false, true, 21);
+    else if (H == 11 && !u8 && CIN == 32 && relu_in && big)
+        WGH(false, 32, true, true, 11);
+    else done = false;
+#undef WGH
+    if (done) return {dWt, db};
+
     if (u8) { TORCH_CHECK(CIN == 8); WG(true, 8, false); }
     else if (CIN == 8) { if (relu_in) WG(false, 8, true); else WG(false, 8, false); }
     else if (CIN == 16) { if (relu_in) WG(false, 16, true); else WG(false, 16, false); }
