@@ -50,6 +50,11 @@ class Config:
 
     # exploration ---------------------------------------------------------
     num_actors: int = 8                          # reference: config.py:21
+    # batched-inference actor driver (VectorActor): one process stepping all
+    # num_actors envs in lockstep with one network forward per tick (on
+    # actor_device), instead of one process per actor
+    vector_actors: bool = False
+    actor_device: str = "cpu"                    # VectorActor inference device
     base_eps: float = 0.4                        # reference: config.py:22
     alpha_eps: float = 7.0                       # reference: config.py:23
     eval_eps: float = 0.001
@@ -125,14 +130,14 @@ PRESETS = {
     "mspacman_gpu_replay": dict(
         game_name="MsPacman", env_type="synthetic", obs_shape=(4, 84, 84),
         action_dim=9, encoder="nature", gpu_replay=True, num_actors=256,
-        buffer_capacity=4_000_000,
+        buffer_capacity=4_000_000, vector_actors=True, actor_device="cuda",
     ),
     # configs[3]: 8x data-parallel learners (parallelism degree comes from
     # torchrun's WORLD_SIZE; the preset is otherwise mspacman_gpu_replay)
     "mspacman_dp": dict(
         game_name="MsPacman", env_type="synthetic", obs_shape=(4, 84, 84),
         action_dim=9, encoder="nature", gpu_replay=True, num_actors=256,
-        buffer_capacity=4_000_000,
+        buffer_capacity=4_000_000, vector_actors=True, actor_device="cuda",
     ),
     # configs[4]: IMPALA-deep ResNet encoder
     "seaquest_impala": dict(

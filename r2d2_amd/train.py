@@ -31,6 +31,12 @@ def _run_actor(epsilon, model, sample_queue, seed):
     actor.run()
 
 
+def _run_vector_actor(epsilons, model, sample_queues, device, seed):
+    from .worker import VectorActor
+    va = VectorActor(epsilons, model, sample_queues, device=device, seed=seed)
+    va.run()
+
+
 def _run_buffer(buffer: ReplayBuffer):
     buffer.run()
 
@@ -54,11 +60,19 @@ def train(seed: int = 0):
     learner = Learner(batch_queue, priority_queue, model)
 
     actor_procs = []
-    for i, eps in enumerate(epsilon_ladder()):
-        p = mp.Process(target=_run_actor,
-                       args=(eps, model, sample_queues[i], seed + 1 + i))
+    if c.vector_actors:
+        # one driver process, all envs in lockstep, batched inference
+        p = mp.Process(target=_run_vector_actor,
+                       args=(epsilon_ladder(), model, sample_queues,
+                             c.actor_device, seed + 1))
         p.start()
         actor_procs.append(p)
+    else:
+        for i, eps in enumerate(epsilon_ladder()):
+            p = mp.Process(target=_run_actor,
+                           args=(eps, model, sample_queues[i], seed + 1 + i))
+            p.start()
+            actor_procs.append(p)
 
     buffer_proc = mp.Process(target=_run_buffer, args=(buffer,))
     buffer_proc.start()
