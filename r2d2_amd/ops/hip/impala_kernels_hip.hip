@@ -339,10 +339,16 @@ __global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
 // -inf (torch max_pool2d padding semantics — the halo zeros must NOT win).
 // One thread per (n, oy, ox, c8): 8 channels vectorized.
 // ---------------------------------------------------------------------------
+template <int HT = 0, int CT = 0>
 __global__ __launch_bounds__(256) void maxpool3s2_fwd_kernel(
     const __hip_bfloat16* __restrict__ in, __hip_bfloat16* __restrict__ out,
-    unsigned char* __restrict__ arg, int N, int H, int W, int OH, int OW,
-    int C) {
+    unsigned char* __restrict__ arg, int N, int H_, int W_, int OH_, int OW_,
+    int C_) {
+    const int H = HT ? HT : H_;
+    const int W = HT ? HT : W_;
+    const int OH = HT ? (HT + 1) / 2 : OH_;
+    const int OW = HT ? (HT + 1) / 2 : OW_;
+    const int C = CT ? CT : C_;
     const int PW = W + 2, PH = H + 2;
     const int C8 = C / 8;
     unsigned idx = blockIdx.x * blockDim.x + threadIdx.x;
@@ -389,11 +395,17 @@ __global__ __launch_bounds__(256) void maxpool3s2_fwd_kernel(
 
 // backward: one thread per INPUT real pixel x c8; gathers from the <= 4
 // windows that can contain it (atomic-free).
+template <int HT = 0, int CT = 0>
 __global__ __launch_bounds__(256) void maxpool3s2_bwd_kernel(
     const __hip_bfloat16* __restrict__ dOut,  // padded (N,OH+2,OW+2,C)
     const unsigned char* __restrict__ arg,    // dense (N,OH,OW,C)
     __hip_bfloat16* __restrict__ dIn,         // padded (N,H+2,W+2,C)
-    int N, int H, int W, int OH, int OW, int C) {
+    int N, int H_, int W_, int OH_, int OW_, int C_) {
+    const int H = HT ? HT : H_;
+    const int W = HT ? HT : W_;
+    const int OH = HT ? (HT + 1) / 2 : OH_;
+    const int OW = HT ? (HT + 1) / 2 : OW_;
+    const int C = CT ? CT : C_;
     const int PW = W + 2, PH = H + 2;
     const int POW = OW + 2, POH = OH + 2;
     const int C8 = C / 8;
@@ -679,12 +691,19 @@ void maxpool3s2_fwd(torch::Tensor in, torch::Tensor out, torch::Tensor arg,
     TORCH_CHECK(out.size(1) == OH + 2 && arg.size(1) == OH);
     long total = N * OH * OW * (C / 8);
     auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-    hipLaunchKernelGGL(maxpool3s2_fwd_kernel, dim3(icdiv(total, 256)),
-                       dim3(256), 0, stream.stream(),
-                       reinterpret_cast<const __hip_bfloat16*>(in.data_ptr()),
-                       reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),
-                       arg.data_ptr<unsigned char>(), (int)N, (int)H, (int)W,
-                       (int)OH, (int)OW, (int)C);
+#define MPF(HT_, CT_)                                                         \
+    hipLaunchKernelGGL((maxpool3s2_fwd_kernel<HT_, CT_>),                     \
+                       dim3(icdiv(total, 256)), dim3(256), 0,                 \
+                       stream.stream(),                                       \
+                       reinterpret_cast<const __hip_bfloat16*>(in.data_ptr()),\
+                       reinterpret_cast<__hip_bfloat16*>(out.data_ptr()),     \
+                       arg.data_ptr<unsigned char>(), (int)N, (int)H, (int)W, \
+                       (int)OH, (int)OW, (int)C)
+    if (H == 84 && C == 16) MPF(84, 16);
+    else if (H == 42 && C == 32) MPF(42, 32);
+    else if (H == 21 && C == 32) MPF(21, 32);
+    else MPF(0, 0);
+#undef MPF
 }
 
 void maxpool3s2_bwd(torch::Tensor dOut, torch::Tensor arg, torch::Tensor dIn,
@@ -692,12 +711,19 @@ void maxpool3s2_bwd(torch::Tensor dOut, torch::Tensor arg, torch::Tensor dIn,
     long C = dIn.size(3);
     long total = N * H * W * (C / 8);
     auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-    hipLaunchKernelGGL(maxpool3s2_bwd_kernel, dim3(icdiv(total, 256)),
-                       dim3(256), 0, stream.stream(),
-                       reinterpret_cast<const __hip_bfloat16*>(dOut.data_ptr()),
-                       arg.data_ptr<unsigned char>(),
-                       reinterpret_cast<__hip_bfloat16*>(dIn.data_ptr()),
-                       (int)N, (int)H, (int)W, (int)OH, (int)OW, (int)C);
+#define MPB(HT_, CT_)                                                         \
+    hipLaunchKernelGGL((maxpool3s2_bwd_kernel<HT_, CT_>),                     \
+                       dim3(icdiv(total, 256)), dim3(256), 0,                 \
+                       stream.stream(),                                       \
+                       reinterpret_cast<const __hip_bfloat16*>(dOut.data_ptr()),\
+                       arg.data_ptr<unsigned char>(),                         \
+                       reinterpret_cast<__hip_bfloat16*>(dIn.data_ptr()),     \
+                       (int)N, (int)H, (int)W, (int)OH, (int)OW, (int)C)
+    if (H == 84 && C == 16) MPB(84, 16);
+    else if (H == 42 && C == 32) MPB(42, 32);
+    else if (H == 21 && C == 32) MPB(21, 32);
+    else MPB(0, 0);
+#undef MPB
 }
 
 void pack_frames(torch::Tensor frames, torch::Tensor out, int64_t H,

@@ -247,7 +247,7 @@ def test_impala_encoder_full_fwd_bwd_vs_autograd():
         # deterministic and larger.  The kernels themselves are verified
         # tightly in the isolated tests (test_res_block_bwd_isolated,
         # test_conv3p_wgrad_u8_frames, and the unit tests above).
-        tol = 0.20 if name.startswith("s0") else 0.10
+        tol = {"s0": 0.20, "s1": 0.15}.get(name[:2], 0.10)
         rel_fro(dw.view_as(conv.weight.grad), conv.weight.grad, tol,
                 f"{name} dW")
         rel_fro(db, conv.bias.grad, tol, f"{name} db")
