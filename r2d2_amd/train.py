@@ -6,6 +6,7 @@ The epsilon ladder is the reference's: eps_i = base_eps^(1 + i/(N-1) * alpha)
 """
 
 import random
+import threading
 
 import numpy as np
 import torch
@@ -41,7 +42,7 @@ def _run_buffer(buffer: ReplayBuffer):
     buffer.run()
 
 
-def train(seed: int = 0):
+def train(seed: int = 0, restart_dead_actors: bool = True):
     torch.manual_seed(seed)
     np.random.seed(seed)
     random.seed(seed)
@@ -59,26 +60,46 @@ def train(seed: int = 0):
     buffer = ReplayBuffer(sample_queues, batch_queue, priority_queue)
     learner = Learner(batch_queue, priority_queue, model)
 
-    actor_procs = []
     if c.vector_actors:
         # one driver process, all envs in lockstep, batched inference
-        p = mp.Process(target=_run_vector_actor,
-                       args=(epsilon_ladder(), model, sample_queues,
-                             c.actor_device, seed + 1))
-        p.start()
-        actor_procs.append(p)
+        spawners = [lambda: mp.Process(
+            target=_run_vector_actor,
+            args=(epsilon_ladder(), model, sample_queues, c.actor_device,
+                  seed + 1))]
     else:
-        for i, eps in enumerate(epsilon_ladder()):
-            p = mp.Process(target=_run_actor,
-                           args=(eps, model, sample_queues[i], seed + 1 + i))
-            p.start()
-            actor_procs.append(p)
+        spawners = [
+            (lambda eps=eps, i=i: mp.Process(
+                target=_run_actor,
+                args=(eps, model, sample_queues[i], seed + 1 + i)))
+            for i, eps in enumerate(epsilon_ladder())]
+    actor_procs = [s() for s in spawners]
+    for p in actor_procs:
+        p.start()
 
     buffer_proc = mp.Process(target=_run_buffer, args=(buffer,))
     buffer_proc.start()
 
+    # actor supervision: the reference silently loses dead actor processes
+    # (SURVEY §5 — throughput degrades with no signal); restart them.
+    stop = threading.Event()
+
+    def _watchdog():
+        while not stop.wait(5.0):
+            for i, p in enumerate(actor_procs):
+                if not p.is_alive():
+                    print(f"[train] actor process {i} died "
+                          f"(exitcode {p.exitcode}); restarting")
+                    actor_procs[i] = spawners[i]()
+                    actor_procs[i].start()
+
+    wd = None
+    if restart_dead_actors:
+        wd = threading.Thread(target=_watchdog, daemon=True)
+        wd.start()
+
     learner.run()
 
+    stop.set()
     buffer_proc.join()
     for p in actor_procs:
         p.terminate()
