@@ -199,40 +199,54 @@ def test_pack_and_pad_utils():
 
 
 def test_impala_encoder_full_fwd_bwd_vs_autograd():
-    """Full encoder: HIP latent + all 16 weight grads vs eager autograd."""
+    """Full encoder: HIP latent + all 16 weight grads vs an eager reference
+    that rounds every intermediate to bf16 (matching the kernel datapath, so
+    maxpool argmax tie-routing decisions agree and the comparison is tight;
+    a pure-fp32 reference legitimately diverges 10-17%% on deep layers via
+    tie misrouting alone)."""
     from r2d2_amd.models.encoders import ImpalaCNN
     torch.manual_seed(7)
     Mn = 12
     enc = ImpalaCNN(4, 512).cuda()
-    for p in enc.parameters():
-        p.grad = torch.zeros_like(p)
+    for p_ in enc.parameters():
+        p_.grad = torch.zeros_like(p_)
     obs = torch.randint(0, 256, (Mn, 84, 84, 4), device="cuda",
                         dtype=torch.uint8)
 
     pack = imp.ImpalaPack(enc, "cuda", with_bwd=True)
     latent, st = imp.encoder_fwd(M_, pack, obs, True)
 
-    x_eager = (obs.float() / 255.0).permute(0, 3, 1, 2)
-    ref_lat = enc(x_eager)
-    close(latent, ref_lat.detach(), rtol=5e-2,
-          atol=6e-2 * float(ref_lat.detach().abs().max()), name="impala latent")
+    def bf(t):  # round to bf16, keep fp32 autograd flow
+        return t.bfloat16().float()
+
+    def ref_forward(x):
+        for stage in enc.stages:
+            x = bf(stage.conv(x))
+            x = bf(F.max_pool2d(x, 3, stride=2, padding=1))
+            for res in (stage.res1, stage.res2):
+                y = bf(res.conv1(bf(F.relu(x))))
+                y = bf(res.conv2(bf(F.relu(y))))
+                x = bf(x + y)
+        x = F.relu(x).flatten(1)
+        return F.relu(enc.fc(bf(x)))
+
+    x_eager = bf(obs.float() / 255.0).permute(0, 3, 1, 2)
+    ref_lat = ref_forward(x_eager)
+    close(latent, ref_lat.detach(), rtol=4e-2,
+          atol=4e-2 * float(ref_lat.detach().abs().max()), name="impala latent")
 
     dlat = torch.randn(Mn, 512, device="cuda")
     ref_lat.backward(dlat)
     grads = imp.encoder_bwd(M_, pack, st, dlat.bfloat16(), latent)
 
-    # grads compare by relative Frobenius error: an fp32 autograd reference
-    # makes different ReLU-mask decisions than the bf16 path for
-    # pre-activations that round across zero, so single elements can
-    # legitimately differ; the aggregate must still agree tightly.
     def rel_fro(a, b, tol, name):
         a, b = a.float().flatten(), b.float().flatten()
         err = (a - b).norm() / (b.norm() + 1e-8)
         assert err < tol, f"{name}: rel fro {err:.4f}"
 
     dwf, dbf = grads["fc"]
-    rel_fro(dwf, enc.fc.weight.grad.reshape(-1), 0.08, "fc dW")
-    rel_fro(dbf, enc.fc.bias.grad, 0.08, "fc db")
+    rel_fro(dwf, enc.fc.weight.grad.reshape(-1), 0.06, "fc dW")
+    rel_fro(dbf, enc.fc.bias.grad, 0.06, "fc db")
     name_map = {}
     for si, stage in enumerate(enc.stages):
         name_map[f"s{si}c"] = stage.conv
@@ -241,16 +255,9 @@ def test_impala_encoder_full_fwd_bwd_vs_autograd():
             name_map[f"s{si}r{ri}b"] = res.conv2
     for name, conv in name_map.items():
         dw, db = grads[name]
-        # stage-0 grads sit at the bottom of a 15-conv bf16 backward chain
-        # crossing two maxpool argmax routings (bf16 rounding creates ties
-        # the fp32 reference resolves differently); their divergence is
-        # deterministic and larger.  The kernels themselves are verified
-        # tightly in the isolated tests (test_res_block_bwd_isolated,
-        # test_conv3p_wgrad_u8_frames, and the unit tests above).
-        tol = {"s0": 0.20, "s1": 0.15}.get(name[:2], 0.10)
-        rel_fro(dw.view_as(conv.weight.grad), conv.weight.grad, tol,
+        rel_fro(dw.view_as(conv.weight.grad), conv.weight.grad, 0.08,
                 f"{name} dW")
-        rel_fro(db, conv.bias.grad, tol, f"{name} db")
+        rel_fro(db, conv.bias.grad, 0.08, f"{name} db")
 
 
 def test_impala_engine_train_step():
