@@ -163,7 +163,7 @@ class HipNetworkEngine:
         self.target = _NetPack(target_net, self.device, self.A, with_bwd=False)
         self.online_net = online_net
         self.target_net = target_net
-        self.bar = torch.zeros(256, dtype=torch.int32, device=self.device)
+        self.bar = torch.zeros(512, dtype=torch.int32, device=self.device)
         self._empty = torch.Tensor()
 
         # flat parameter/grad/Adam-moment buffers: module params become views

@@ -72,7 +72,7 @@ __device__ __forceinline__ float sigmoidf_(float x) {
 // await: wave 0 polls `nflags` flag words lane-parallel (relaxed), one
 // acquire fence, __syncthreads.  Returns false on bounded-spin timeout.
 // ---------------------------------------------------------------------------
-#define LSTM_MAX_FLAGS 128
+#define LSTM_MAX_FLAGS 256
 
 __device__ __forceinline__ void publish_slice(unsigned* flag, unsigned value) {
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // every storing wave
@@ -121,8 +121,8 @@ __device__ __forceinline__ bool await_slices(unsigned* flags, int nflags,
     return ok_s != 0;
 }
 
-// layout of the workspace (int32 words): [0..127] flags, [128] poison,
-// [129..137] legacy barrier words (barrier_bench)
+// layout of the workspace (int32 words): [0..255] flags (2 nets x 2 batch
+// halves x 64 slices max), [256] poison, then legacy barrier words
 struct GridBar {
     unsigned flags[LSTM_MAX_FLAGS];
     unsigned poison;
@@ -170,7 +170,11 @@ __device__ __forceinline__ bool grid_barrier(GridBar* bar, unsigned epoch,
 // ---------------------------------------------------------------------------
 // Forward
 // ---------------------------------------------------------------------------
-template <int H>
+// BROWS: batch rows per workgroup.  64 = classic layout (one wg per 8
+// hidden units, whole batch).  32 = batch-split: twice the workgroups
+// (256 for the dual-net launch -> every CU busy), half the LDS staging
+// per step, flag groups per (net, half).
+template <int H, int BROWS = 64>
 __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
     const __hip_bfloat16* __restrict__ X0,    // (B, T, 4H)
     const __hip_bfloat16* __restrict__ X1,    // or null
@@ -184,11 +188,16 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
     float* __restrict__ Cout0,                // (B, T+1, H)
     float* __restrict__ Cout1,
     __hip_bfloat16* __restrict__ stash0,      // (B, T, 4H) post-nonlin gates
-    GridBar* bar, int B, int T, int nblocks) {
-    constexpr int WGS_PER_NET = H / 8;
-    const int net = blockIdx.x / WGS_PER_NET;
-    const int wid = blockIdx.x % WGS_PER_NET;
+    GridBar* bar, int B, int T, int nblocks, int nhalves) {
+    constexpr int WGS_PER_HALF = H / 8;
+    const int wgs_per_net = WGS_PER_HALF * nhalves;
+    const int net = blockIdx.x / wgs_per_net;
+    const int rem = blockIdx.x % wgs_per_net;
+    const int half = rem / WGS_PER_HALF;
+    const int wid = rem % WGS_PER_HALF;
     const int u0 = wid * 8;
+    const int b0 = half * BROWS;
+    const int Bl = min(B - b0, BROWS);
 
     const __hip_bfloat16* X = net ? X1 : X0;
     const __hip_bfloat16* Whh = net ? Whh1 : Whh0;
@@ -196,14 +205,14 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
     __hip_bfloat16* Hout = net ? Hout1 : Hout0;
     float* Cout = net ? Cout1 : Cout0;
     __hip_bfloat16* stash = net ? nullptr : stash0;
-    unsigned* flags = bar->flags + net * WGS_PER_NET;
+    unsigned* flags = bar->flags + (net * nhalves + half) * WGS_PER_HALF;
     unsigned* myflag = &flags[wid];
 
     __shared__ __hip_bfloat16 s_whh[32][H + 8];
-    __shared__ __hip_bfloat16 s_h[64][H + 8];
-    __shared__ float s_gates[64][32 + 4];
-    __shared__ float s_c[64][8];
-    __shared__ __hip_bfloat16 s_hrow[64][8];
+    __shared__ __hip_bfloat16 s_h[BROWS][H + 8];
+    __shared__ float s_gates[BROWS][32 + 4];
+    __shared__ float s_c[BROWS][8];
+    __shared__ __hip_bfloat16 s_hrow[BROWS][8];
 
     for (int e = threadIdx.x * 8; e < 32 * H; e += blockDim.x * 8) {
         int c = e / H;
@@ -212,36 +221,40 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
         lstore8(&s_whh[c][k], lload8(Whh + (long)(g * H + u0 + j) * H + k));
     }
     // h0 -> Hout[:,0] (this wg's slice); c0 -> LDS-resident cell state
-    for (int p = threadIdx.x; p < B * 8; p += blockDim.x) {
-        int b = p / 8, j = p % 8;
+    for (int p = threadIdx.x; p < Bl * 8; p += blockDim.x) {
+        int bl = p / 8, j = p % 8;
+        int b = b0 + bl;
         int u = u0 + j;
         Hout[((long)b * (T + 1)) * H + u] = f2bf(init[(long)b * H + u]);
         Cout[((long)b * (T + 1)) * H + u] = init[((long)B + b) * H + u];
-        s_c[b][j] = init[((long)B + b) * H + u];
+        s_c[bl][j] = init[((long)B + b) * H + u];
     }
     publish_slice(myflag, 1u);
 
     const int wave = threadIdx.x / WAVE;
     const int lane = threadIdx.x & (WAVE - 1);
-    const int wrow0 = (wave >> 1) * 32;        // rows (batch)
+    // BROWS=64: 2 A-frags per wave (rows (wave>>1)*32 ..); BROWS=32: 1
+    constexpr int NFRAG = BROWS / 32;
+    const int wrow0 = (wave >> 1) * 16 * NFRAG;
     const int wcol0 = (wave & 1) * 16;         // cols (gate local)
     const int frow = lane & 15;
     const int kseg = (lane >> 4) * 8;
 
-    constexpr int CHUNKS = (64 * H) / 8;
+    constexpr int CHUNKS = (BROWS * H) / 8;
 
     for (int t = 0; t < T; ++t) {
-        if (!await_slices(flags, WGS_PER_NET, (unsigned)(t + 1), &bar->poison))
+        if (!await_slices(flags, WGS_PER_HALF, (unsigned)(t + 1),
+                          &bar->poison))
             return;
-        // bulk-stage h_prev (B x H) into LDS
+        // bulk-stage this half's h_prev (Bl x H) into LDS
         {
             const long base = (long)t * H;
 #pragma unroll
             for (int e = threadIdx.x; e < CHUNKS; e += 256) {
                 int row = e / (H / 8);
                 int k8 = (e % (H / 8)) * 8;
-                bf16x8 v = (row < B)
-                    ? lload8(Hout + ((long)row * (T + 1)) * H + base + k8)
+                bf16x8 v = (row < Bl)
+                    ? lload8(Hout + ((long)(b0 + row) * (T + 1)) * H + base + k8)
                     : lzero8();
                 lstore8(&s_h[row][k8], v);
             }
@@ -249,12 +262,12 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
         __syncthreads();
 
         // gates = h_prev @ Whh_slice^T  (all-LDS MFMA)
-        f32x4 acc[2] = {};
+        f32x4 acc[NFRAG] = {};
 #pragma unroll
         for (int k0 = 0; k0 < H; k0 += 32) {
             bf16x8 bfr = lload8(&s_whh[wcol0 + frow][k0 + kseg]);
 #pragma unroll
-            for (int i = 0; i < 2; ++i) {
+            for (int i = 0; i < NFRAG; ++i) {
                 bf16x8 afr = lload8(&s_h[wrow0 + i * 16 + frow][k0 + kseg]);
                 acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                     afr, bfr, acc[i], 0, 0, 0);
@@ -264,7 +277,7 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
             int ccol = lane & 15;
             int crow = (lane >> 4) * 4;
 #pragma unroll
-            for (int i = 0; i < 2; ++i)
+            for (int i = 0; i < NFRAG; ++i)
 #pragma unroll
                 for (int r = 0; r < 4; ++r)
                     s_gates[wrow0 + i * 16 + crow + r][wcol0 + ccol] = acc[i][r];
@@ -274,17 +287,18 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
         {
             int row = threadIdx.x / 4;
             int g = threadIdx.x % 4;
-            if (row < B) {
-                bf16x8 x8 = lload8(X + ((long)row * T + t) * 4 * H + g * H + u0);
+            if (row < Bl) {
+                bf16x8 x8 = lload8(
+                    X + ((long)(b0 + row) * T + t) * 4 * H + g * H + u0);
 #pragma unroll
                 for (int j = 0; j < 8; ++j)
                     s_gates[row][g * 8 + j] += (float)x8[j];
             }
         }
         __syncthreads();
-        for (int p = threadIdx.x; p < B * 8; p += blockDim.x) {
+        for (int p = threadIdx.x; p < Bl * 8; p += blockDim.x) {
             int b = p / 8, j = p % 8;
-            bool active = t < lens[b];
+            bool active = t < lens[b0 + b];
             float i_ = 0.f, f_ = 0.f, g_ = 0.f, o_ = 0.f;
             float c = s_c[b][j], h;
             if (active) {
@@ -308,27 +322,29 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
         // vectorized writers: h slice (16 B/row), c (32 B/row), gate stash
         {
             int tid = threadIdx.x;
-            if (tid < 64 && tid < B) {
-                long off = ((long)tid * (T + 1) + t + 1) * H + u0;
+            if (tid < 64 && tid < Bl) {
+                long off = ((long)(b0 + tid) * (T + 1) + t + 1) * H + u0;
                 lstore8(Hout + off, *reinterpret_cast<bf16x8*>(&s_hrow[tid][0]));
             } else if (tid >= 64 && tid < 192) {
-                int b = (tid - 64) / 2, half = (tid - 64) & 1;
-                if (b < B) {
-                    long off = ((long)b * (T + 1) + t + 1) * H + u0 + half * 4;
+                int b = (tid - 64) / 2, ch = (tid - 64) & 1;
+                if (b < Bl) {
+                    long off = ((long)(b0 + b) * (T + 1) + t + 1) * H + u0
+                               + ch * 4;
                     *reinterpret_cast<float4*>(Cout + off) =
-                        *reinterpret_cast<float4*>(&s_c[b][half * 4]);
+                        *reinterpret_cast<float4*>(&s_c[b][ch * 4]);
                 }
             }
         }
         if (stash) {
             int row = threadIdx.x / 4;
             int g = threadIdx.x % 4;
-            if (row < B) {
+            if (row < Bl) {
                 bf16x8 v;
 #pragma unroll
                 for (int j = 0; j < 8; ++j)
                     v[j] = (__bf16)s_gates[row][g * 8 + j];
-                lstore8(stash + ((long)row * T + t) * 4 * H + g * H + u0, v);
+                lstore8(stash + ((long)(b0 + row) * T + t) * 4 * H + g * H + u0,
+                        v);
             }
         }
         publish_slice(myflag, (unsigned)(t + 2));
@@ -340,7 +356,10 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
 // LDS; dgates_{t+1} streamed through LDS in 256-col pieces; vectorized row
 // staging of stash/C/dHext; flag hand-off on the dgates stream.
 // ---------------------------------------------------------------------------
-template <int H>
+// BROWS as in lstm_fwd_kernel: 32 doubles the workgroup count (batch
+// halves with their own flag groups) and splits the recurrent GEMM's K
+// reduction across wave pairs.
+template <int H, int BROWS = 64>
 __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
     const __hip_bfloat16* __restrict__ stash,  // (B, T, 4H) i,f,g,o
     const float* __restrict__ Cout,            // (B, T+1, H)
@@ -349,18 +368,22 @@ __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
     const __hip_bfloat16* __restrict__ Whh_bwd,// (H, 4H): W_hh^T row-major
     const int* __restrict__ lens,
     __hip_bfloat16* __restrict__ dgates,       // (B, T, 4H) out
-    GridBar* bar, int B, int T, int nblocks) {
+    GridBar* bar, int B, int T, int nblocks, int nhalves) {
     constexpr int WGS = H / 16;
-    const int wid = blockIdx.x;
+    constexpr int KSPLIT = (BROWS == 32) ? 2 : 1;
+    const int half = blockIdx.x / WGS;
+    const int wid = blockIdx.x % WGS;
     const int u0 = wid * 16;
+    const int b0 = half * BROWS;
+    const int Bl = min(B - b0, BROWS);
 
     __shared__ __hip_bfloat16 s_wb[16][4 * H + 8];
-    __shared__ float s_dh[64][16 + 1];
-    __shared__ float s_dc[64][16 + 1];
-    __shared__ float s_rec[64][16 + 1];
-    __shared__ __hip_bfloat16 s_dgout[64][64 + 8];  // this wg's dgates cols
+    __shared__ float s_dh[BROWS][16 + 1];
+    __shared__ float s_dc[BROWS][16 + 1];
+    __shared__ float s_rec[KSPLIT][BROWS][16 + 1];
+    __shared__ __hip_bfloat16 s_dgout[BROWS][64 + 8];  // this wg's dgates cols
 
-    unsigned* flags = bar->flags;
+    unsigned* flags = bar->flags + half * WGS;
     unsigned* myflag = &flags[wid];
 
     for (int e = threadIdx.x * 8; e < 16 * 4 * H; e += blockDim.x * 8) {
@@ -368,7 +391,7 @@ __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
         int k = e % (4 * H);
         lstore8(&s_wb[c][k], lload8(Whh_bwd + (long)(u0 + c) * 4 * H + k));
     }
-    for (int p = threadIdx.x; p < B * 16; p += blockDim.x) {
+    for (int p = threadIdx.x; p < BROWS * 16; p += blockDim.x) {
         s_dh[p / 16][p % 16] = 0.f;
         s_dc[p / 16][p % 16] = 0.f;
     }
@@ -379,9 +402,15 @@ __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
 
     const int wave = threadIdx.x / WAVE;
     const int lane = threadIdx.x & (WAVE - 1);
-    const int wrow0 = wave * 16;
+    // KSPLIT=1: 4 waves x 16-row tiles, full-K.  KSPLIT=2: wave pairs
+    // split the 4H reduction; partials land in s_rec[wk].
+    const int wrow0 = (KSPLIT == 1 ? wave : (wave & 1)) * 16;
+    const int wk = (KSPLIT == 1) ? 0 : (wave >> 1);
     const int frow = lane & 15;
     const int kseg = (lane >> 4) * 8;
+    constexpr int NCHUNK = 4 * H / 32;
+    const int c0 = wk * (NCHUNK / KSPLIT);
+    const int cN = c0 + NCHUNK / KSPLIT;
 
     for (int t = T - 1; t >= 0; --t) {
         unsigned need = (unsigned)(T - t);   // pieces published for t+1
@@ -390,32 +419,32 @@ __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
         // prefetch ring on the dgates stream (loads stay in flight across
         // MFMAs; no per-piece barriers)
         if (t < T - 1) {
-            const int arow = wrow0 + frow;
+            const int arow_l = wrow0 + frow;
             const __hip_bfloat16* dgrow =
-                dgates + ((long)arow * T + t + 1) * 4 * H;
-            const bool rvalid = arow < B;
+                dgates + ((long)(b0 + arow_l) * T + t + 1) * 4 * H;
+            const bool rvalid = arow_l < Bl;
             f32x4 acc = {};
-            bf16x8 a0 = rvalid ? lload8(dgrow + 0 * 32 + kseg) : lzero8();
-            bf16x8 a1 = rvalid ? lload8(dgrow + 1 * 32 + kseg) : lzero8();
-            bf16x8 a2 = rvalid ? lload8(dgrow + 2 * 32 + kseg) : lzero8();
-            bf16x8 a3 = rvalid ? lload8(dgrow + 3 * 32 + kseg) : lzero8();
+            bf16x8 a0 = rvalid ? lload8(dgrow + (c0 + 0) * 32 + kseg) : lzero8();
+            bf16x8 a1 = rvalid ? lload8(dgrow + (c0 + 1) * 32 + kseg) : lzero8();
+            bf16x8 a2 = rvalid ? lload8(dgrow + (c0 + 2) * 32 + kseg) : lzero8();
+            bf16x8 a3 = rvalid ? lload8(dgrow + (c0 + 3) * 32 + kseg) : lzero8();
 #pragma unroll 4
-            for (int kc = 0; kc < 4 * H / 32; kc += 4) {
+            for (int kc = c0; kc < cN; kc += 4) {
                 bf16x8 b0 = lload8(&s_wb[frow][(kc + 0) * 32 + kseg]);
                 bf16x8 b1 = lload8(&s_wb[frow][(kc + 1) * 32 + kseg]);
                 bf16x8 b2 = lload8(&s_wb[frow][(kc + 2) * 32 + kseg]);
                 bf16x8 b3 = lload8(&s_wb[frow][(kc + 3) * 32 + kseg]);
                 acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc, 0, 0, 0);
-                if (kc + 4 < 4 * H / 32)
+                if (kc + 4 < cN)
                     a0 = rvalid ? lload8(dgrow + (kc + 4) * 32 + kseg) : lzero8();
                 acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc, 0, 0, 0);
-                if (kc + 5 < 4 * H / 32)
+                if (kc + 5 < cN)
                     a1 = rvalid ? lload8(dgrow + (kc + 5) * 32 + kseg) : lzero8();
                 acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b2, acc, 0, 0, 0);
-                if (kc + 6 < 4 * H / 32)
+                if (kc + 6 < cN)
                     a2 = rvalid ? lload8(dgrow + (kc + 6) * 32 + kseg) : lzero8();
                 acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b3, acc, 0, 0, 0);
-                if (kc + 7 < 4 * H / 32)
+                if (kc + 7 < cN)
                     a3 = rvalid ? lload8(dgrow + (kc + 7) * 32 + kseg) : lzero8();
             }
             int ccol = lane & 15;
@@ -423,39 +452,42 @@ __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 int row = wrow0 + crow + r;
-                if (row < B) s_rec[row][ccol] = acc[r];
+                if (row < BROWS) s_rec[wk][row][ccol] = acc[r];
             }
         }
         __syncthreads();
-        for (int p = threadIdx.x; p < B * 16; p += blockDim.x) {
+        for (int p = threadIdx.x; p < Bl * 16; p += blockDim.x) {
             int b = p / 16, jl = p % 16;
+            int bg = b0 + b;
             int u = u0 + jl;
             float dh, dc_in;
             bool last = (t == T - 1);
-            bool active_next = !last && ((t + 1) < lens[b]);
-            float ext = dHext[((long)b * T + t) * H + u];
+            bool active_next = !last && ((t + 1) < lens[bg]);
+            float ext = dHext[((long)bg * T + t) * H + u];
+            float rec = s_rec[0][b][jl];
+            if (KSPLIT == 2) rec += s_rec[1][b][jl];
             if (last) {
                 dh = ext;
                 dc_in = 0.f;
             } else if (active_next) {
-                long so1 = ((long)b * T + t + 1) * 4 * H + u;
+                long so1 = ((long)bg * T + t + 1) * 4 * H + u;
                 float f_next = bf2f(stash[so1 + H]);
-                dh = ext + s_rec[b][jl];
+                dh = ext + rec;
                 dc_in = s_dc[b][jl] * f_next;
             } else {
                 dh = ext + s_dh[b][jl];
                 dc_in = s_dc[b][jl];
             }
-            bool active = t < lens[b];
-            long so = ((long)b * T + t) * 4 * H + u;
+            bool active = t < lens[bg];
+            long so = ((long)bg * T + t) * 4 * H + u;
             float di = 0.f, df = 0.f, dg = 0.f, do_ = 0.f;
             if (active) {
                 float i_ = bf2f(stash[so]);
                 float f_ = bf2f(stash[so + H]);
                 float g_ = bf2f(stash[so + 2 * H]);
                 float o_ = bf2f(stash[so + 3 * H]);
-                float tc = tanhf(Cout[((long)b * (T + 1) + t + 1) * H + u]);
-                float c_prev = Cout[((long)b * (T + 1) + t) * H + u];
+                float tc = tanhf(Cout[((long)bg * (T + 1) + t + 1) * H + u]);
+                float c_prev = Cout[((long)bg * (T + 1) + t) * H + u];
                 float dc = dc_in + dh * o_ * (1.f - tc * tc);
                 di = dc * g_ * i_ * (1.f - i_);
                 df = dc * c_prev * f_ * (1.f - f_);
@@ -478,14 +510,15 @@ __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
             int tid = threadIdx.x;
             int b = tid / 4;
             int gh = tid % 4;
-            if (b < B) {
+            if (b < Bl) {
 #pragma unroll
                 for (int rep = 0; rep < 2; ++rep) {
                     int g = gh;
-                    int half = rep;
-                    long off = ((long)b * T + t) * 4 * H + g * H + u0 + half * 8;
+                    long off = ((long)(b0 + b) * T + t) * 4 * H + g * H + u0
+                               + rep * 8;
                     lstore8(dgates + off,
-                            *reinterpret_cast<bf16x8*>(&s_dgout[b][g * 16 + half * 8]));
+                            *reinterpret_cast<bf16x8*>(
+                                &s_dgout[b][g * 16 + rep * 8]));
                 }
             }
         }
@@ -562,7 +595,8 @@ std::vector<torch::Tensor> lstm_fwd(
     auto stash = want_stash ? torch::empty({B, T, H4}, bf) : torch::Tensor();
 
     int wgs = (int)H / 8;
-    int nblocks = wgs * (two ? 2 : 1);
+    int nhalves = B > 32 ? 2 : 1;
+    int nblocks = wgs * (two ? 2 : 1) * nhalves;
     auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
     zero_ws(barrier_ws, stream.stream());
 
@@ -570,19 +604,22 @@ std::vector<torch::Tensor> lstm_fwd(
         return t.defined()
             ? reinterpret_cast<__hip_bfloat16*>(t.data_ptr()) : nullptr;
     };
-    hipLaunchKernelGGL((lstm_fwd_kernel<512>), dim3(nblocks), dim3(256), 0,
-        stream.stream(),
-        reinterpret_cast<const __hip_bfloat16*>(X0.data_ptr()),
-        two ? reinterpret_cast<const __hip_bfloat16*>(X1.data_ptr()) : nullptr,
-        reinterpret_cast<const __hip_bfloat16*>(Whh0.data_ptr()),
-        two ? reinterpret_cast<const __hip_bfloat16*>(Whh1.data_ptr()) : nullptr,
-        init0.data_ptr<float>(),
-        two ? init1.data_ptr<float>() : nullptr,
-        lens.data_ptr<int>(), bp(H0), bp(H1),
-        C0.data_ptr<float>(), two ? C1.data_ptr<float>() : nullptr,
-        want_stash ? bp(stash) : nullptr,
-        reinterpret_cast<GridBar*>(barrier_ws.data_ptr()),
-        (int)B, (int)T, nblocks);
+#define LSTMF(BROWS_)                                                         \
+    hipLaunchKernelGGL((lstm_fwd_kernel<512, BROWS_>), dim3(nblocks),         \
+        dim3(256), 0, stream.stream(),                                        \
+        reinterpret_cast<const __hip_bfloat16*>(X0.data_ptr()),               \
+        two ? reinterpret_cast<const __hip_bfloat16*>(X1.data_ptr()) : nullptr,\
+        reinterpret_cast<const __hip_bfloat16*>(Whh0.data_ptr()),             \
+        two ? reinterpret_cast<const __hip_bfloat16*>(Whh1.data_ptr()) : nullptr,\
+        init0.data_ptr<float>(),                                              \
+        two ? init1.data_ptr<float>() : nullptr,                              \
+        lens.data_ptr<int>(), bp(H0), bp(H1),                                 \
+        C0.data_ptr<float>(), two ? C1.data_ptr<float>() : nullptr,           \
+        want_stash ? bp(stash) : nullptr,                                     \
+        reinterpret_cast<GridBar*>(barrier_ws.data_ptr()),                    \
+        (int)B, (int)T, nblocks, nhalves)
+    if (nhalves == 2) LSTMF(32); else LSTMF(64);
+#undef LSTMF
 
     std::vector<torch::Tensor> out = {H0, C0};
     out.push_back(two ? H1 : torch::Tensor());
@@ -599,19 +636,23 @@ torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
     long H = H4 / 4;
     TORCH_CHECK(H == 512, "lstm_bwd is instantiated for H=512");
     auto dgates = torch::empty({B, T, H4}, stash.options());
-    int nblocks = (int)H / 16;
+    int nhalves = B > 32 ? 2 : 1;
+    int nblocks = (int)H / 16 * nhalves;
     auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
     zero_ws(barrier_ws, stream.stream());
-    hipLaunchKernelGGL((lstm_bwd_kernel<512>), dim3(nblocks), dim3(256), 0,
-        stream.stream(),
-        reinterpret_cast<const __hip_bfloat16*>(stash.data_ptr()),
-        Cout.data_ptr<float>(),
-        reinterpret_cast<const __hip_bfloat16*>(Hout.data_ptr()),
-        dHext.data_ptr<float>(),
-        reinterpret_cast<const __hip_bfloat16*>(Whh_bwd.data_ptr()),
-        lens.data_ptr<int>(),
-        reinterpret_cast<__hip_bfloat16*>(dgates.data_ptr()),
-        reinterpret_cast<GridBar*>(barrier_ws.data_ptr()),
-        (int)B, (int)T, nblocks);
+#define LSTMB(BROWS_)                                                         \
+    hipLaunchKernelGGL((lstm_bwd_kernel<512, BROWS_>), dim3(nblocks),         \
+        dim3(256), 0, stream.stream(),                                        \
+        reinterpret_cast<const __hip_bfloat16*>(stash.data_ptr()),            \
+        Cout.data_ptr<float>(),                                               \
+        reinterpret_cast<const __hip_bfloat16*>(Hout.data_ptr()),             \
+        dHext.data_ptr<float>(),                                              \
+        reinterpret_cast<const __hip_bfloat16*>(Whh_bwd.data_ptr()),          \
+        lens.data_ptr<int>(),                                                 \
+        reinterpret_cast<__hip_bfloat16*>(dgates.data_ptr()),                 \
+        reinterpret_cast<GridBar*>(barrier_ws.data_ptr()),                    \
+        (int)B, (int)T, nblocks, nhalves)
+    if (nhalves == 2) LSTMB(32); else LSTMB(64);
+#undef LSTMB
     return dgates;
 }

@@ -51,7 +51,7 @@ def make_inputs(B, T, seed=0, scale=0.5):
 
 
 def run_kernel_fwd(X, Whh, h0, c0, lens, want_stash=True):
-    bar = torch.zeros(256, dtype=torch.int32, device="cuda")
+    bar = torch.zeros(512, dtype=torch.int32, device="cuda")
     init = torch.stack([h0, c0]).contiguous()
     empty = torch.Tensor()
     outs = M_.lstm_fwd(X, empty, Whh, empty, init, empty, lens, bar, want_stash)
@@ -80,7 +80,7 @@ def test_lstm_dual_network():
     B, T = 16, 10
     X0, Whh0, h00, c00, lens = make_inputs(B, T, seed=1)
     X1, Whh1, h01, c01, _ = make_inputs(B, T, seed=2)
-    bar = torch.zeros(256, dtype=torch.int32, device="cuda")
+    bar = torch.zeros(512, dtype=torch.int32, device="cuda")
     init0 = torch.stack([h00, c00]).contiguous()
     init1 = torch.stack([h01, c01]).contiguous()
     H0, C0, H1, C1, stash = M_.lstm_fwd(X0, X1, Whh0, Whh1, init0, init1,
@@ -99,7 +99,7 @@ def test_lstm_bwd_matches_autograd():
     dHext = (torch.randn(B, T, H, device="cuda") * 0.1)
 
     # kernel backward
-    bar = torch.zeros(256, dtype=torch.int32, device="cuda")
+    bar = torch.zeros(512, dtype=torch.int32, device="cuda")
     Whh_bwd = Whh.t().contiguous()  # (H, 4H)
     dg = M_.lstm_bwd(stash, C0, H0, dHext.contiguous(), Whh_bwd, lens, bar)
     torch.cuda.synchronize()
