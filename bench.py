@@ -186,6 +186,17 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
     elapsed = float(t.item())
 
+    if (os.environ.get("R2D2_ENGINE_TIMING")
+            and getattr(learner, "engine", None) and rank == 0):
+        import sys
+        agg = {}
+        for name, ms in learner.engine.timing_report():
+            agg[name] = agg.get(name, 0.0) + ms
+        for name, ms in sorted(agg.items(), key=lambda kv: -kv[1]):
+            print(f"  [stage] {name:12s} "
+                  f"{ms / (args.warmup + args.steps):8.3f} ms/step",
+                  file=sys.stderr)
+
     ms_per_step = elapsed / args.steps * 1000.0
     n_gpus = world_size if world_size > 1 else args.gpus
     value = c.batch_size * args.steps * n_gpus / elapsed
