@@ -219,16 +219,19 @@ def test_impala_encoder_full_fwd_bwd_vs_autograd():
     def bf(t):  # round to bf16, keep fp32 autograd flow
         return t.bfloat16().float()
 
+    def conv(c, x):  # conv with bf16-rounded weights (kernel datapath)
+        return F.conv2d(x, bf(c.weight), c.bias, 1, 1)
+
     def ref_forward(x):
         for stage in enc.stages:
-            x = bf(stage.conv(x))
+            x = bf(conv(stage.conv, x))
             x = bf(F.max_pool2d(x, 3, stride=2, padding=1))
             for res in (stage.res1, stage.res2):
-                y = bf(res.conv1(bf(F.relu(x))))
-                y = bf(res.conv2(bf(F.relu(y))))
+                y = bf(conv(res.conv1, bf(F.relu(x))))
+                y = bf(conv(res.conv2, bf(F.relu(y))))
                 x = bf(x + y)
         x = F.relu(x).flatten(1)
-        return F.relu(enc.fc(bf(x)))
+        return F.relu(F.linear(bf(x), bf(enc.fc.weight), enc.fc.bias))
 
     x_eager = bf(obs.float() / 255.0).permute(0, 3, 1, 2)
     ref_lat = ref_forward(x_eager)
