@@ -354,7 +354,11 @@ class HipNetworkEngine:
         self._mark("loss")
 
         # ---- backward --------------------------------------------------
+        # weight grads accumulate STRAIGHT into the flat .grad views
+        # (pre-zeroed once) — no per-parameter copy/permute kernels.
         ON = self.online
+        net = self.online_net
+        self.flat_grad.zero_()
         dq_pad = torch.zeros(2 * R, A, device=dev)  # learn rows only get grad
         dq_pad[:R] = dq
         dadv2, dval2 = m.dueling_combine_bwd(dq_pad.contiguous(),
@@ -362,13 +366,19 @@ class HipNetworkEngine:
         # adv path
         dadv1 = m.gemm_dgrad(dadv2, self._empty, ON.wa2_kn, False)
         dh_a = m.gemm_dgrad(dadv1, adv1_o, ON.wa1_kn, True)
-        dWa2, dba2 = m.gemm_wgrad(dadv2, self._empty, adv1_o, False, True)
-        dWa1, dba1 = m.gemm_wgrad(dadv1, adv1_o, h_cat, True, True)
+        m.gemm_wgrad_into(dadv2, self._empty, adv1_o, False,
+                          net.advantage[2].weight.grad,
+                          net.advantage[2].bias.grad)
+        m.gemm_wgrad_into(dadv1, adv1_o, h_cat, True,
+                          net.advantage[0].weight.grad,
+                          net.advantage[0].bias.grad)
         # value path
         dval1 = m.gemm_dgrad(dval2, self._empty, ON.wv2_kn, False)
         dh_v = m.gemm_dgrad(dval1, val1_o, ON.wv1_kn, True)
-        dWv2, dbv2 = m.gemm_wgrad(dval2, self._empty, val1_o, False, True)
-        dWv1, dbv1 = m.gemm_wgrad(dval1, val1_o, h_cat, True, True)
+        m.gemm_wgrad_into(dval2, self._empty, val1_o, False,
+                          net.value[2].weight.grad, net.value[2].bias.grad)
+        m.gemm_wgrad_into(dval1, val1_o, h_cat, True,
+                          net.value[0].weight.grad, net.value[0].bias.grad)
 
         dh_rows = (dh_a.float() + dh_v.float())     # (2R, 512)
         # scatter-add into dHext (B, T, H): learn rows at t, tgt rows at t_tgt
@@ -384,21 +394,22 @@ class HipNetworkEngine:
         dgates_flat = dgates.view(B * T, 4 * H)
 
         h_prev = Ho[:, :T].reshape(B * T, H).contiguous()
-        dWhh, _ = m.gemm_wgrad(dgates_flat, self._empty, h_prev, False, False)
-        dWih_pad, db_lstm = m.gemm_wgrad(dgates_flat, self._empty, rin_o,
-                                         False, True)
+        m.gemm_wgrad_into(dgates_flat, self._empty, h_prev, False,
+                          net.recurrent.weight_hh_l0.grad, self._empty)
+        m.gemm_wgrad_into(dgates_flat, self._empty, rin_o, False,
+                          net.recurrent.weight_ih_l0.grad,
+                          net.recurrent.bias_ih_l0.grad)
+        net.recurrent.bias_hh_l0.grad.copy_(net.recurrent.bias_ih_l0.grad)
         drin = m.gemm_dgrad(dgates_flat, self._empty, ON.wih_kn, False)
         dlat = drin[:, :512].contiguous()
         self._mark("lstm_wgrads")
 
         M = B * T
         if self.impala:
-            imp_grads = impala_ops.encoder_bwd(m, ON.imp, enc_stash,
-                                               dlat, lat_o)
+            impala_ops.encoder_bwd(m, ON.imp, enc_stash, dlat, lat_o)
             self._mark("conv_bwd")
-            return self._write_grads_common(
-                batch, imp_grads, dWih_pad, dWhh, db_lstm,
-                dWa1, dba1, dWa2, dba2, dWv1, dbv1, dWv2, dbv2, loss, prio)
+            self._mark("grad_write")
+            return loss.squeeze(0), prio
 
         a1, a2, a3, flat = enc_stash
         lat_bf = lat_o  # forward output (relu mask source)
@@ -407,8 +418,10 @@ class HipNetworkEngine:
 
         # conv3 backward
         self._mark("fc_bwd")
-        dW3, db3 = m.conv_wgrad(dflat.view(M * 49, 64), a3, a2, 3,
-                                M, 9, 9, 7, 7, 64, 9 * 64)
+        m.conv_wgrad_into(dflat.view(M * 49, 64), a3, a2, 3,
+                          M, 9, 9, 7, 7, 64,
+                          net.encoder.conv3.weight.grad,
+                          net.encoder.conv3.bias.grad)
         d3m = (dflat.view(M * 49, 64)
                * (a3.view(M * 49, 64) > 0).bfloat16()).view(M, 7, 7, 64)
         dyp3 = torch.zeros(M, 11, 11, 64, device=dev, dtype=torch.bfloat16)
@@ -417,8 +430,10 @@ class HipNetworkEngine:
         m.conv_dgrad(dyp3.contiguous(), ON.w3d, ON.taps3, M, 11, 11, 64,
                      9, 9, 64, 0, 0, 1, 2, d_a2)
         # conv2 backward
-        dW2, db2 = m.conv_wgrad(d_a2.view(M * 81, 64), a2, a1, 2,
-                                M, 20, 20, 9, 9, 64, 4 * 4 * 32)
+        m.conv_wgrad_into(d_a2.view(M * 81, 64), a2, a1, 2,
+                          M, 20, 20, 9, 9, 64,
+                          net.encoder.conv2.weight.grad,
+                          net.encoder.conv2.bias.grad)
         d2m = (d_a2.view(M * 81, 64)
                * (a2.view(M * 81, 64) > 0).bfloat16()).view(M, 9, 9, 64)
         dyp2 = torch.zeros(M, 11, 11, 64, device=dev, dtype=torch.bfloat16)
@@ -430,70 +445,17 @@ class HipNetworkEngine:
                 m.conv_dgrad(dyp2c, ON.w2d[(py, px)], ON.taps2[(py, px)],
                              M, 11, 11, 64, 20, 20, 32, py, px, 2, 1, d_a1)
         # conv1 wgrad (no dgrad: input is data)
-        dW1, db1 = m.conv_wgrad(d_a1.view(M * 400, 32), a1, obs_hwc, 1,
-                                M, 84, 84, 20, 20, 32, 8 * 8 * self.C)
+        m.conv_wgrad_into(d_a1.view(M * 400, 32), a1, obs_hwc, 1,
+                          M, 84, 84, 20, 20, 32,
+                          net.encoder.conv1.weight.grad,
+                          net.encoder.conv1.bias.grad)
         self._mark("conv_bwd")
 
-        # ---- write grads into the nn.Module (f32) ----------------------
-        net = self.online_net
+        # fc grad needs the HWC->CHW reorder, the one remaining copy
         enc = net.encoder
+        enc.fc.weight.grad.copy_(
+            dWf.view(512, 7, 7, 64).permute(0, 3, 1, 2).reshape(512, 3136))
+        enc.fc.bias.grad.copy_(dbf)
 
-        def setg(p, g):
-            p.grad.copy_(g.reshape(p.shape))
-
-        def conv_grad(dwt, cout, kh, kw, cin):
-            return dwt.view(cout, kh, kw, cin).permute(0, 3, 1, 2).contiguous()
-
-        setg(enc.conv1.weight, conv_grad(dW1, 32, 8, 8, self.C))
-        setg(enc.conv1.bias, db1)
-        setg(enc.conv2.weight, conv_grad(dW2, 64, 4, 4, 32))
-        setg(enc.conv2.bias, db2)
-        setg(enc.conv3.weight, conv_grad(dW3, 64, 3, 3, 64))
-        setg(enc.conv3.bias, db3)
-        setg(enc.fc.weight, dWf.view(512, 7, 7, 64).permute(0, 3, 1, 2)
-             .reshape(512, 3136).contiguous())
-        setg(enc.fc.bias, dbf)
-        setg(net.recurrent.weight_ih_l0, dWih_pad[:, :ON.kin].contiguous())
-        setg(net.recurrent.weight_hh_l0, dWhh)
-        setg(net.recurrent.bias_ih_l0, db_lstm)
-        setg(net.recurrent.bias_hh_l0, db_lstm.clone())
-        setg(net.advantage[0].weight, dWa1)
-        setg(net.advantage[0].bias, dba1)
-        setg(net.advantage[2].weight, dWa2[:A].contiguous())
-        setg(net.advantage[2].bias, dba2[:A].contiguous())
-        setg(net.value[0].weight, dWv1)
-        setg(net.value[0].bias, dbv1)
-        setg(net.value[2].weight, dWv2[:1].contiguous())
-        setg(net.value[2].bias, dbv2[:1].contiguous())
-
-        self._mark("grad_write")
-        return loss.squeeze(0), prio
-
-    def _write_grads_common(self, batch, imp_grads, dWih_pad, dWhh, db_lstm,
-                            dWa1, dba1, dWa2, dba2, dWv1, dbv1, dWv2, dbv2,
-                            loss, prio):
-        """IMPALA-path grad write: encoder grads via impala_ops, LSTM/head
-        grads identical to the nature path."""
-        net = self.online_net
-        ON = self.online
-        A = self.A
-
-        impala_ops.write_grads(net.encoder, imp_grads)
-
-        def setg(p, g):
-            p.grad.copy_(g.reshape(p.shape))
-
-        setg(net.recurrent.weight_ih_l0, dWih_pad[:, :ON.kin].contiguous())
-        setg(net.recurrent.weight_hh_l0, dWhh)
-        setg(net.recurrent.bias_ih_l0, db_lstm)
-        setg(net.recurrent.bias_hh_l0, db_lstm.clone())
-        setg(net.advantage[0].weight, dWa1)
-        setg(net.advantage[0].bias, dba1)
-        setg(net.advantage[2].weight, dWa2[:A].contiguous())
-        setg(net.advantage[2].bias, dba2[:A].contiguous())
-        setg(net.value[0].weight, dWv1)
-        setg(net.value[0].bias, dbv1)
-        setg(net.value[2].weight, dWv2[:1].contiguous())
-        setg(net.value[2].bias, dbv2[:1].contiguous())
         self._mark("grad_write")
         return loss.squeeze(0), prio

@@ -181,7 +181,7 @@ __global__ __launch_bounds__(256) void gemm_wgrad_kernel(
     const __hip_bfloat16* __restrict__ A,    // (M, K)
     float* __restrict__ dWt,                 // (N, K) f32 accumulate
     float* __restrict__ db,                  // (N,) f32 accumulate
-    int M, int N, int K, int rows_per_chunk) {
+    int M, int N, int K, int rows_per_chunk, int Nout, int Kout) {
     // grid.x: row chunks; grid.y: N tiles of 64; grid.z: K tiles of 64
     __shared__ __hip_bfloat16 s_dy[32][64 + 8];  // [m][n]
     __shared__ __hip_bfloat16 s_a[32][64 + 8];   // [m][k]
@@ -293,12 +293,12 @@ __global__ __launch_bounds__(256) void gemm_wgrad_kernel(
             for (int r = 0; r < 4; ++r) {
                 long nn = ncol0 + wr * 32 + i * 16 + crow + r;
                 long kk = kcol0 + wc * 32 + j * 16 + ccol;
-                if (nn < N && kk < K)
-                    atomicAdd(&dWt[nn * K + kk], acc[i][j][r]);
+                if (nn < Nout && kk < Kout)
+                    atomicAdd(&dWt[nn * Kout + kk], acc[i][j][r]);
             }
     if (HAS_BIAS && threadIdx.x < 64 && blockIdx.z == 0) {
         long c = ncol0 + threadIdx.x;
-        if (c < N) atomicAdd(&db[c], bias_acc);
+        if (c < Nout) atomicAdd(&db[c], bias_acc);
     }
 }
 
@@ -385,9 +385,42 @@ std::vector<torch::Tensor> gemm_wgrad(torch::Tensor dY, torch::Tensor act_out,
     hipLaunchKernelGGL((gemm_wgrad_kernel<RM, HB>), grid, dim3(256), 0,        \
                        stream.stream(), dy, ac, a, dWt.data_ptr<float>(),      \
                        db.data_ptr<float>(), (int)M, (int)N, (int)K,           \
-                       (int)rows_per_chunk)
+                       (int)rows_per_chunk, (int)N, (int)K)
     if (relu_mask) { if (want_bias) LAUNCHW(true, true); else LAUNCHW(true, false); }
     else           { if (want_bias) LAUNCHW(false, true); else LAUNCHW(false, false); }
 #undef LAUNCHW
     return {dWt, db};
+}
+
+// accumulate straight into pre-zeroed .grad tensors (possibly narrower than
+// the padded compute width: Nout <= N, Kout <= K) — removes the per-step
+// grad-copy kernels from the engine's backward.
+void gemm_wgrad_into(torch::Tensor dY, torch::Tensor act_out, torch::Tensor A,
+                     bool relu_mask, torch::Tensor dW_out,
+                     torch::Tensor db_out) {
+    TORCH_CHECK(dY.is_cuda() && dY.dtype() == torch::kBFloat16 && dY.is_contiguous());
+    TORCH_CHECK(A.dtype() == torch::kBFloat16 && A.is_contiguous());
+    TORCH_CHECK(dW_out.dtype() == torch::kFloat32 && dW_out.is_contiguous());
+    long M = dY.size(0), N = dY.size(1), K = A.size(1);
+    long Nout = dW_out.size(0), Kout = dW_out.numel() / Nout;
+    bool want_bias = db_out.defined() && db_out.numel() > 0;
+    long tiles = (long)cdiv(N, 64) * cdiv(K, 64);
+    long target_chunks = std::max(1L, 1024L / std::max(1L, tiles));
+    long rows_per_chunk = std::max(32L, (M + target_chunks - 1) / target_chunks);
+    rows_per_chunk = ((rows_per_chunk + 31) / 32) * 32;
+    dim3 grid(cdiv(M, rows_per_chunk), cdiv(N, 64), cdiv(K, 64));
+    auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+    auto* dy = reinterpret_cast<const __hip_bfloat16*>(dY.data_ptr());
+    auto* ac = relu_mask
+        ? reinterpret_cast<const __hip_bfloat16*>(act_out.data_ptr()) : nullptr;
+    auto* a = reinterpret_cast<const __hip_bfloat16*>(A.data_ptr());
+    float* dbp = want_bias ? db_out.data_ptr<float>() : nullptr;
+#define LAUNCHW(RM, HB)                                                        \
+    hipLaunchKernelGGL((gemm_wgrad_kernel<RM, HB>), grid, dim3(256), 0,        \
+                       stream.stream(), dy, ac, a, dW_out.data_ptr<float>(),   \
+                       dbp, (int)M, (int)N, (int)K,                            \
+                       (int)rows_per_chunk, (int)Nout, (int)Kout)
+    if (relu_mask) { if (want_bias) LAUNCHW(true, true); else LAUNCHW(true, false); }
+    else           { if (want_bias) LAUNCHW(false, true); else LAUNCHW(false, false); }
+#undef LAUNCHW
 }

@@ -167,25 +167,36 @@ def encoder_fwd(m, pack: ImpalaPack, obs_hwc_u8: torch.Tensor,
 
 def encoder_bwd(m, pack: ImpalaPack, st: dict, dlat: torch.Tensor,
                 latent: torch.Tensor):
-    """dlat: (M, hidden) bf16 -> conv/fc gradients.
-
-    Returns {name: (dW f32 torch-layout, db f32)} for the 15 convs plus
-    ('fc', (dWf, dbf))."""
+    """dlat: (M, hidden) bf16 -> conv/fc gradients, accumulated STRAIGHT
+    into the encoder modules' pre-zeroed .grad views (torch layout in the
+    wgrad epilogues — no permute/copy kernels)."""
     ar = pack.arena
     ar.reset("bwd")
     M = dlat.shape[0]
     empty = torch.Tensor()
-    grads = {}
+    enc = pack.enc
+    convs = {}
+    for si, stage in enumerate(enc.stages):
+        convs[f"s{si}c"] = stage.conv
+        for ri, res in enumerate((stage.res1, stage.res2)):
+            convs[f"s{si}r{ri}a"] = res.conv1
+            convs[f"s{si}r{ri}b"] = res.conv2
 
     # fc (+ its relu) then the flatten relu back onto the padded grid
     dflat = m.gemm_dgrad(dlat, latent, pack.wf_kn, True)
     dWf, dbf = m.gemm_wgrad(dlat, latent, st["flat"], True, True)
     hd = pack.wft.shape[0]
-    grads["fc"] = (dWf.view(hd, 11, 11, 32).permute(0, 3, 1, 2)
-                   .reshape(hd, 3872).contiguous(), dbf)
+    enc.fc.weight.grad.copy_(
+        dWf.view(hd, 11, 11, 32).permute(0, 3, 1, 2).reshape(hd, 3872))
+    enc.fc.bias.grad.copy_(dbf)
 
     dx = ar.get("bwd", (M, 13, 13, 32))
     m.dense2pad_mask(dflat, st["s_out"], dx, M, 11, 11)
+
+    def wgrad(name, dY, inp, H, relu_in, cin):
+        conv = convs[name]
+        m.conv3p_wgrad_into(dY, inp, M, H, H, relu_in, cin,
+                            conv.weight.grad, conv.bias.grad)
 
     for si in (2, 1, 0):
         hin, hout = STAGES[si]
@@ -193,13 +204,11 @@ def encoder_bwd(m, pack: ImpalaPack, st: dict, dlat: torch.Tensor,
         for ri in (1, 0):
             x, y1 = st[f"s{si}r{ri}x"], st[f"s{si}r{ri}y1"]
             nb, nm = f"s{si}r{ri}b", f"s{si}r{ri}a"
-            dW2, db2 = m.conv3p_wgrad(dx, y1, M, hout, hout, True)
-            grads[nb] = (self_conv_grad(dW2, c, c), db2)
+            wgrad(nb, dx, y1, hout, True, c)
             dy1 = ar.get("bwd", (M, hout + 2, hout + 2, c))
             m.conv3p(dx, pack.wd[nb], empty, empty, y1, dy1,
                      M, hout, hout, False, False, 2)
-            dW1, db1 = m.conv3p_wgrad(dy1, x, M, hout, hout, True)
-            grads[nm] = (self_conv_grad(dW1, c, c), db1)
+            wgrad(nm, dy1, x, hout, True, c)
             dx_new = ar.get("bwd", (M, hout + 2, hout + 2, c))
             m.conv3p(dy1, pack.wd[nm], empty, dx, x, dx_new,
                      M, hout, hout, False, False, 3)
@@ -210,13 +219,11 @@ def encoder_bwd(m, pack: ImpalaPack, st: dict, dlat: torch.Tensor,
         # stage conv wgrad (+ dgrad, except stage 0 whose input is data)
         sin = st[f"s{si}c_in"]
         cin = 4 if si == 0 else CHANS[si - 1]
-        dWc, dbc = m.conv3p_wgrad(dconv, sin, M, hin, hin, False)
-        grads[f"s{si}c"] = (self_conv_grad(dWc, c, cin), dbc)
+        wgrad(f"s{si}c", dconv, sin, hin, False, cin)
         if si > 0:
             dx = ar.get("bwd", (M, hin + 2, hin + 2, CHANS[si - 1]))
             m.conv3p(dconv, pack.wd[f"s{si}c"], empty, empty, empty, dx,
                      M, hin, hin, False, False, 0)
-    return grads
 
 
 def self_conv_grad(dWt: torch.Tensor, cout: int, cin: int) -> torch.Tensor:
@@ -224,20 +231,3 @@ def self_conv_grad(dWt: torch.Tensor, cout: int, cin: int) -> torch.Tensor:
     cp = dWt.shape[1] // 9
     return (dWt.view(cout, 3, 3, cp)[:, :, :, :cin]
             .permute(0, 3, 1, 2).contiguous())
-
-
-def write_grads(enc, grads):
-    """Copy encoder_bwd outputs into the nn.Module .grad tensors."""
-    names = {}
-    for si, stage in enumerate(enc.stages):
-        names[f"s{si}c"] = stage.conv
-        for ri, res in enumerate((stage.res1, stage.res2)):
-            names[f"s{si}r{ri}a"] = res.conv1
-            names[f"s{si}r{ri}b"] = res.conv2
-    for name, conv in names.items():
-        dw, db = grads[name]
-        conv.weight.grad.copy_(dw.view(conv.weight.shape))
-        conv.bias.grad.copy_(db)
-    dwf, dbf = grads["fc"]
-    enc.fc.weight.grad.copy_(dwf.view(enc.fc.weight.shape))
-    enc.fc.bias.grad.copy_(dbf)
