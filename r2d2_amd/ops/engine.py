@@ -459,3 +459,66 @@ class HipNetworkEngine:
 
         self._mark("grad_write")
         return loss.squeeze(0), prio
+
+
+class HipInference:
+    """K15: the single-step actor-inference fast path through the gfx950
+    kernels (SURVEY §2.3 K15 — the reference runs this on CPU per actor,
+    model.py:65-79).  Used by VectorActor on a cuda device: one batched
+    encoder pass + one LSTM cell step + the dueling heads per tick, with
+    prepacked bf16 weights refreshed on each weight pull."""
+
+    def __init__(self, net, device, config=None):
+        c = config or cfg.get()
+        assert c.encoder in ("nature", "impala") and c.hidden_dim == 512
+        assert tuple(c.obs_shape) == (4, 84, 84)
+        self.cfg = c
+        self.A = net.action_dim
+        self.device = torch.device(device)
+        self.m = hip_ops.ext(required=True)
+        self.net = net
+        self.pack = _NetPack(net, self.device, self.A, with_bwd=False)
+        self.impala = self.pack.impala
+
+    def refresh(self):
+        self.pack.refresh()
+
+    @torch.no_grad()
+    def forward(self, obs_u8_nchw, last_action, last_reward, hidden):
+        """obs: (E, 4, 84, 84) u8 on device; hidden: (h, c) each (1, E, 512)
+        f32.  Returns (q (E, A) f32, (h', c'))."""
+        m = self.m
+        E = obs_u8_nchw.shape[0]
+        obs_hwc = obs_u8_nchw.permute(0, 2, 3, 1).contiguous()
+        if self.impala:
+            lat, _ = impala_ops.encoder_fwd(m, self.pack.imp, obs_hwc, False)
+        else:
+            p = self.pack
+            a1 = m.conv_fwd(obs_hwc, p.w1t, p.b1, 1, E, 84, 84, 20, 20, True)
+            a2 = m.conv_fwd(a1, p.w2t, p.b2, 2, E, 20, 20, 9, 9, True)
+            a3 = m.conv_fwd(a2, p.w3t, p.b3, 3, E, 9, 9, 7, 7, True)
+            lat = m.gemm_bias_act(a3.view(E, 3136), p.wft, p.bf, 1, False)
+        p = self.pack
+        rin = torch.zeros(E, p.kin_pad, device=self.device,
+                          dtype=torch.bfloat16)
+        rin[:, :512] = lat
+        rin[:, 512:512 + self.A] = last_action.bfloat16()
+        rin[:, 512 + self.A] = last_reward.reshape(E).bfloat16()
+        x = m.gemm_bias_act(rin, p.wih_t, p.lstm_bias, 0, True)     # f32
+        h0, c0 = hidden
+        hg = m.gemm_bias_act(h0.reshape(E, 512).bfloat16().contiguous(),
+                             p.whh_t, torch.Tensor(), 0, True)      # f32
+        gates = (x + hg).view(E, 4, 512)
+        i_ = torch.sigmoid(gates[:, 0])
+        f_ = torch.sigmoid(gates[:, 1])
+        g_ = torch.tanh(gates[:, 2])
+        o_ = torch.sigmoid(gates[:, 3])
+        c1 = f_ * c0.reshape(E, 512) + i_ * g_
+        h1 = o_ * torch.tanh(c1)
+        hb = h1.bfloat16().contiguous()
+        adv1 = m.gemm_bias_act(hb, p.wa1t, p.ba1, 1, False)
+        adv2 = m.gemm_bias_act(adv1, p.wa2t, p.ba2, 0, False)
+        val1 = m.gemm_bias_act(hb, p.wv1t, p.bv1, 1, False)
+        val2 = m.gemm_bias_act(val1, p.wv2t, p.bv2, 0, False)
+        q = m.dueling_combine(adv2, val2, self.A)
+        return q, (h1.view(1, E, 512), c1.view(1, E, 512))

@@ -880,6 +880,14 @@ class VectorActor:
         self._obs_host = np.zeros((E,) + tuple(c.obs_shape), dtype=np.uint8)
         self.episode_steps = np.zeros(E, dtype=np.int64)
         self._pending_finish: List[int] = []
+        # K15 fast path: batched single-step inference through the gfx950
+        # kernels when running on a GPU with the extension built
+        self.hip_inf = None
+        if (self.device.type == "cuda" and c.use_hip_kernels
+                and c.encoder in ("nature", "impala") and c.hidden_dim == 512
+                and tuple(c.obs_shape) == (4, 84, 84)):
+            from .ops.engine import HipInference
+            self.hip_inf = HipInference(self.model, self.device)
 
     def _queue(self, i):
         return self.queues[i % len(self.queues)]
@@ -910,7 +918,11 @@ class VectorActor:
             state.last_reward = self.lr_t
             state.hidden_state = (self.h_t, self.c_t)
             with torch.no_grad():
-                if use_amp:
+                if self.hip_inf is not None:
+                    q, (h, c) = self.hip_inf.forward(
+                        self.obs_t, self.la_t, self.lr_t,
+                        (self.h_t, self.c_t))
+                elif use_amp:
                     with torch.autocast("cuda", dtype=torch.bfloat16):
                         q, (h, c) = self.model(state)
                     q = q.float()
@@ -962,4 +974,6 @@ class VectorActor:
 
             if tick % self.update_interval == 0:
                 self.model.load_state_dict(self.shared_model.state_dict())
+                if self.hip_inf is not None:
+                    self.hip_inf.refresh()
         return total_steps
