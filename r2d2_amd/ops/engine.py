@@ -416,12 +416,13 @@ class HipNetworkEngine:
         dflat = m.gemm_dgrad(dlat, lat_bf, ON.wf_kn, True)
         dWf, dbf = m.gemm_wgrad(dlat, lat_bf, flat, True, True)
 
-        # conv3 backward
+        # conv3 backward.  Conv weight grads accumulate in the packed
+        # (COUT, ky, kx, c) layout — torch-layout atomics scatter each
+        # 16-lane wave across 16 cachelines (measured 4-5x slower) — and
+        # are permuted into .grad afterwards (tiny tensors).
         self._mark("fc_bwd")
-        m.conv_wgrad_into(dflat.view(M * 49, 64), a3, a2, 3,
-                          M, 9, 9, 7, 7, 64,
-                          net.encoder.conv3.weight.grad,
-                          net.encoder.conv3.bias.grad)
+        dW3, db3 = m.conv_wgrad(dflat.view(M * 49, 64), a3, a2, 3,
+                                M, 9, 9, 7, 7, 64, 9 * 64)
         d3m = (dflat.view(M * 49, 64)
                * (a3.view(M * 49, 64) > 0).bfloat16()).view(M, 7, 7, 64)
         dyp3 = torch.zeros(M, 11, 11, 64, device=dev, dtype=torch.bfloat16)
@@ -430,10 +431,8 @@ class HipNetworkEngine:
         m.conv_dgrad(dyp3.contiguous(), ON.w3d, ON.taps3, M, 11, 11, 64,
                      9, 9, 64, 0, 0, 1, 2, d_a2)
         # conv2 backward
-        m.conv_wgrad_into(d_a2.view(M * 81, 64), a2, a1, 2,
-                          M, 20, 20, 9, 9, 64,
-                          net.encoder.conv2.weight.grad,
-                          net.encoder.conv2.bias.grad)
+        dW2, db2 = m.conv_wgrad(d_a2.view(M * 81, 64), a2, a1, 2,
+                                M, 20, 20, 9, 9, 64, 4 * 4 * 32)
         d2m = (d_a2.view(M * 81, 64)
                * (a2.view(M * 81, 64) > 0).bfloat16()).view(M, 9, 9, 64)
         dyp2 = torch.zeros(M, 11, 11, 64, device=dev, dtype=torch.bfloat16)
@@ -445,14 +444,21 @@ class HipNetworkEngine:
                 m.conv_dgrad(dyp2c, ON.w2d[(py, px)], ON.taps2[(py, px)],
                              M, 11, 11, 64, 20, 20, 32, py, px, 2, 1, d_a1)
         # conv1 wgrad (no dgrad: input is data)
-        m.conv_wgrad_into(d_a1.view(M * 400, 32), a1, obs_hwc, 1,
-                          M, 84, 84, 20, 20, 32,
-                          net.encoder.conv1.weight.grad,
-                          net.encoder.conv1.bias.grad)
+        dW1, db1 = m.conv_wgrad(d_a1.view(M * 400, 32), a1, obs_hwc, 1,
+                                M, 84, 84, 20, 20, 32, 8 * 8 * self.C)
         self._mark("conv_bwd")
 
-        # fc grad needs the HWC->CHW reorder, the one remaining copy
         enc = net.encoder
+
+        def conv_grad(dwt, cout, kh, kw, cin):
+            return dwt.view(cout, kh, kw, cin).permute(0, 3, 1, 2)
+
+        enc.conv1.weight.grad.copy_(conv_grad(dW1, 32, 8, 8, self.C))
+        enc.conv1.bias.grad.copy_(db1)
+        enc.conv2.weight.grad.copy_(conv_grad(dW2, 64, 4, 4, 32))
+        enc.conv2.bias.grad.copy_(db2)
+        enc.conv3.weight.grad.copy_(conv_grad(dW3, 64, 3, 3, 64))
+        enc.conv3.bias.grad.copy_(db3)
         enc.fc.weight.grad.copy_(
             dWf.view(512, 7, 7, 64).permute(0, 3, 1, 2).reshape(512, 3136))
         enc.fc.bias.grad.copy_(dbf)

@@ -194,9 +194,13 @@ def encoder_bwd(m, pack: ImpalaPack, st: dict, dlat: torch.Tensor,
     m.dense2pad_mask(dflat, st["s_out"], dx, M, 11, 11)
 
     def wgrad(name, dY, inp, H, relu_in, cin):
+        # packed-layout accumulation + permute copy (torch-layout atomics
+        # scatter wave lanes across cachelines — measured 4-5x slower)
         conv = convs[name]
-        m.conv3p_wgrad_into(dY, inp, M, H, H, relu_in, cin,
-                            conv.weight.grad, conv.bias.grad)
+        cout = conv.weight.shape[0]
+        dWt, db = m.conv3p_wgrad(dY, inp, M, H, H, relu_in)
+        conv.weight.grad.copy_(self_conv_grad(dWt, cout, cin))
+        conv.bias.grad.copy_(db)
 
     for si in (2, 1, 0):
         hin, hout = STAGES[si]
