@@ -523,8 +523,8 @@ class HipInference:
         h1 = o_ * torch.tanh(c1)
         hb = h1.bfloat16().contiguous()
         adv1 = m.gemm_bias_act(hb, p.wa1t, p.ba1, 1, False)
-        adv2 = m.gemm_bias_act(adv1, p.wa2t, p.ba2, 0, False)
+        adv2 = m.gemm_bias_act(adv1, p.wa2t, p.ba2, 0, True)
         val1 = m.gemm_bias_act(hb, p.wv1t, p.bv1, 1, False)
-        val2 = m.gemm_bias_act(val1, p.wv2t, p.bv2, 0, False)
+        val2 = m.gemm_bias_act(val1, p.wv2t, p.bv2, 0, True)
         q = m.dueling_combine(adv2, val2, self.A)
         return q, (h1.view(1, E, 512), c1.view(1, E, 512))
