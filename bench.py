@@ -22,9 +22,11 @@ import numpy as np
 import torch
 
 
-def build_batch(c, device, seed, full_len=True):
+def build_batch(c, device, seed, full_len=True, hwc=False):
     """One synthetic device-resident training batch of the benchmark shape:
-    B sequences of burn_in+learning+forward steps of (C,84,84) uint8 frames."""
+    B sequences of burn_in+learning+forward steps of (C,84,84) uint8 frames.
+    ``hwc`` stores frames channels-innermost — the layout the GPU-resident
+    replay keeps and the NHWC conv kernels consume directly."""
     from r2d2_amd.worker import TrainingBatch
 
     g = torch.Generator(device="cpu").manual_seed(seed)
@@ -36,7 +38,9 @@ def build_batch(c, device, seed, full_len=True):
     A = c.action_dim
     sum_learn = int(learn.sum())
 
-    obs = torch.randint(0, 256, (B, T) + tuple(c.obs_shape),
+    shape = (tuple(c.obs_shape[1:]) + (c.obs_shape[0],)) if hwc \
+        else tuple(c.obs_shape)
+    obs = torch.randint(0, 256, (B, T) + shape,
                         dtype=torch.uint8, generator=g)
     la = torch.zeros(B, T, A)
     la[torch.arange(B)[:, None], torch.arange(T)[None, :],
@@ -148,7 +152,11 @@ def main():
         torch.cuda.synchronize()
         batches = None
     else:
-        batches = [build_batch(c, device, seed=1000 + rank * 100 + i)
+        # HWC frames only when the HIP engine is the consumer (the eager
+        # Network expects NCHW; the replay stores HWC natively)
+        hwc = bool(getattr(learner, "engine", None) is not None
+                   and len(c.obs_shape) == 3)
+        batches = [build_batch(c, device, seed=1000 + rank * 100 + i, hwc=hwc)
                    for i in range(args.batches)]
 
     def one_step(i):

@@ -73,6 +73,14 @@ class GpuReplayBuffer:
         self.size = 0
         self.env_steps = 0
         self._ext = hip_ops.ext(required=True)
+        # pinned staging + dedicated copy stream for the frame payload
+        # (actor->learner trajectory push via pinned host
+        # buffers and hipMemcpyAsync on a copy stream, overlapped with
+        # compute on the default stream)
+        self._copy_stream = torch.cuda.Stream(device=dev)
+        self._pin_obs = torch.empty(self.obs_rows, self.obs_store.shape[-1],
+                                    dtype=torch.uint8, pin_memory=True)
+        self._copy_done = torch.cuda.Event()
 
     # ------------------------------------------------------------------
     def __len__(self):
@@ -99,7 +107,16 @@ class GpuReplayBuffer:
             obs_np = np.ascontiguousarray(obs_np.transpose(0, 2, 3, 1))
         obs_flat = torch.from_numpy(obs_np.reshape(rows, -1))
         assert obs_flat.dtype == torch.uint8
-        self.obs_store[slot, :rows].copy_(obs_flat, non_blocking=True)
+        # frames: host memcpy into the pinned buffer, then async H2D on the
+        # copy stream; the default stream waits on the copy event so gather
+        # kernels never read a half-written block
+        self._copy_stream.synchronize()       # pin buffer free to reuse
+        self._pin_obs[:rows].copy_(obs_flat)
+        with torch.cuda.stream(self._copy_stream):
+            self.obs_store[slot, :rows].copy_(self._pin_obs[:rows],
+                                              non_blocking=True)
+        self._copy_done.record(self._copy_stream)
+        torch.cuda.current_stream().wait_event(self._copy_done)
         la_idx = torch.from_numpy(
             np.ascontiguousarray(block.last_action.argmax(1).astype(np.uint8)))
         self.la_store[slot, :rows].copy_(la_idx, non_blocking=True)
