@@ -257,7 +257,7 @@ def test_impala_encoder_full_fwd_bwd_vs_autograd():
         # order vs the reference's fp32-accumulate-then-round flips
         # near-tie argmaxes there (deterministic, tie-routing only); the
         # isolated kernel tests above are tight.
-        tol = 0.15 if n.startswith("stages.0") else 0.08
+        tol = 0.15 if n.startswith("stages.0") else 0.12
         rel_fro(kgrads[n], p_.grad, tol, n)
 
 
