@@ -57,10 +57,6 @@ std::vector<torch::Tensor> conv_wgrad(torch::Tensor dY, torch::Tensor act,
                                       int64_t N, int64_t INH, int64_t INW,
                                       int64_t OH, int64_t OW, int64_t COUT,
                                       int64_t K);
-void conv_wgrad_into(torch::Tensor dY, torch::Tensor act, torch::Tensor in,
-                     int64_t conv_id, int64_t N, int64_t INH, int64_t INW,
-                     int64_t OH, int64_t OW, int64_t COUT,
-                     torch::Tensor dW_out, torch::Tensor db_out);
 
 // lstm_kernels.hip
 void barrier_bench(torch::Tensor barrier_ws, int64_t steps, int64_t nblocks);
@@ -85,9 +81,6 @@ void conv3p(torch::Tensor in, torch::Tensor Wt, torch::Tensor bias,
 std::vector<torch::Tensor> conv3p_wgrad(torch::Tensor dY, torch::Tensor in,
                                         int64_t N, int64_t H, int64_t W,
                                         bool relu_in);
-void conv3p_wgrad_into(torch::Tensor dY, torch::Tensor in, int64_t N,
-                       int64_t H, int64_t W, bool relu_in, int64_t cin_real,
-                       torch::Tensor dW_out, torch::Tensor db_out);
 void maxpool3s2_fwd(torch::Tensor in, torch::Tensor out, torch::Tensor arg,
                     int64_t N, int64_t H, int64_t W);
 void maxpool3s2_bwd(torch::Tensor dOut, torch::Tensor arg, torch::Tensor dIn,
@@ -129,8 +122,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv_fwd", &conv_fwd, "Implicit-GEMM MFMA conv forward (NHWC)");
     m.def("conv_dgrad", &conv_dgrad, "MFMA conv backward-data (tap classes)");
     m.def("conv_wgrad", &conv_wgrad, "MFMA conv backward-weight");
-    m.def("conv_wgrad_into", &conv_wgrad_into,
-          "conv backward-weight, torch-layout, straight into .grad");
     m.def("lstm_fwd", &lstm_fwd,
           "Persistent fused LSTM forward (dual-network, length-masked)");
     m.def("lstm_bwd", &lstm_bwd, "Persistent fused LSTM BPTT backward");
@@ -143,8 +134,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused relu-in / residual / mask epilogues)");
     m.def("conv3p_wgrad", &conv3p_wgrad,
           "IMPALA 3x3 conv backward-weight (padded dY, relu-in patches)");
-    m.def("conv3p_wgrad_into", &conv3p_wgrad_into,
-          "IMPALA conv backward-weight, torch-layout, straight into .grad");
     m.def("maxpool3s2_fwd", &maxpool3s2_fwd,
           "maxpool 3x3 s2 p1 forward + tap argmax");
     m.def("maxpool3s2_bwd", &maxpool3s2_bwd,

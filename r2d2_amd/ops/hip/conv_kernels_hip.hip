@@ -241,11 +241,7 @@ __global__ __launch_bounds__(256) void conv_dgrad_kernel(
 // accumulators cover all k-tiles and are flushed with f32 atomics once per
 // chunk.  Fused ReLU mask + bias grad.
 // ---------------------------------------------------------------------------
-// TORCH_OUT: write dW in torch conv layout (COUT, CIN, KH, KW) instead of
-// the packed (COUT, K=(ky,kx,c)) — lets the engine accumulate straight into
-// pre-zeroed .grad views with no permute/copy kernels.
-template <bool IN_U8, int KH, int KW, int CIN, int S, int NCOT, bool RELU,
-          bool TORCH_OUT = false>
+template <bool IN_U8, int KH, int KW, int CIN, int S, int NCOT, bool RELU>
 __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     const __hip_bfloat16* __restrict__ dY,   // (M, COUT)
     const __hip_bfloat16* __restrict__ act,  // (M, COUT) forward output
@@ -361,18 +357,8 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
             for (int r = 0; r < 4; ++r) {
                 long co = wr * 32 + i * 16 + crow + r;
                 long kk = wc * KHALF + kf * 16 + ccol;
-                if (co < COUT && kk < K) {
-                    long idx;
-                    if (TORCH_OUT) {
-                        int c = (int)(kk % CIN);
-                        int s_ = (int)(kk / CIN);
-                        int kx = s_ % KW, ky = s_ / KW;
-                        idx = ((co * CIN + c) * KH + ky) * KW + kx;
-                    } else {
-                        idx = co * K + kk;
-                    }
-                    atomicAdd(&dWt[idx], acc[i][kf][r]);
-                }
+                if (co < COUT && kk < K)
+                    atomicAdd(&dWt[co * K + kk], acc[i][kf][r]);
             }
     if (threadIdx.x < 64) {
         long c = threadIdx.x;
@@ -485,33 +471,4 @@ std::vector<torch::Tensor> conv_wgrad(torch::Tensor dY, torch::Tensor act,
     else TORCH_CHECK(false, "unknown conv_id");
 #undef WLAUNCH
     return {dWt, db};
-}
-
-// torch-layout accumulation into pre-zeroed .grad tensors
-void conv_wgrad_into(torch::Tensor dY, torch::Tensor act, torch::Tensor in,
-                     int64_t conv_id, int64_t N, int64_t INH, int64_t INW,
-                     int64_t OH, int64_t OW, int64_t COUT,
-                     torch::Tensor dW_out, torch::Tensor db_out) {
-    long M = N * OH * OW;
-    TORCH_CHECK(dW_out.dtype() == torch::kFloat32 && dW_out.is_contiguous());
-    long target_chunks = 1024;
-    long rows_per_chunk = std::max(32L, (M + target_chunks - 1) / target_chunks);
-    rows_per_chunk = ((rows_per_chunk + 31) / 32) * 32;
-    dim3 grid(ccdiv(M, rows_per_chunk));
-    auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-    auto* dy = reinterpret_cast<const __hip_bfloat16*>(dY.data_ptr());
-    auto* ac = reinterpret_cast<const __hip_bfloat16*>(act.data_ptr());
-    const void* x = in.data_ptr();
-#define WLAUNCH(U8, KH_, KW_, CIN_, S_, NCOT_)                                 \
-    hipLaunchKernelGGL((conv_wgrad_kernel<U8, KH_, KW_, CIN_, S_, NCOT_,       \
-                                          true, true>),                        \
-                       grid, dim3(256), 0, stream.stream(), dy, ac, x,         \
-                       dW_out.data_ptr<float>(), db_out.data_ptr<float>(),     \
-                       (int)M, (int)INH, (int)INW, (int)OH, (int)OW,           \
-                       (int)COUT, (int)rows_per_chunk)
-    if (conv_id == 1) WLAUNCH(true, 8, 8, 4, 4, 1);
-    else if (conv_id == 2) WLAUNCH(false, 4, 4, 32, 2, 2);
-    else if (conv_id == 3) WLAUNCH(false, 3, 3, 64, 1, 2);
-    else TORCH_CHECK(false, "unknown conv_id");
-#undef WLAUNCH
 }

@@ -201,16 +201,13 @@ __global__ __launch_bounds__(256) void conv3p_kernel(
 // NCO2: true when COUT > 16 (two cout fragments per wave, 4-way K split);
 // false for COUT <= 16 (single cout fragment, 2-way K split x 2-way M split
 // — halves the padded-K waste that dominates the small-channel stages).
-// TORCH_OUT: write (COUT, CINR, 3, 3) torch layout (CINR = real input
-// channels, <= padded CIN) straight into a pre-zeroed .grad view.
-template <bool IN_U8, int CIN, bool RELU_IN, bool NCO2, int HT = 0,
-          bool TORCH_OUT = false>
+template <bool IN_U8, int CIN, bool RELU_IN, bool NCO2, int HT = 0>
 __global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
     const __hip_bfloat16* __restrict__ dY,   // (N, H+2, W+2, COUT) padded
     const void* __restrict__ in,             // (N, H+2, W+2, CIN) padded
     float* __restrict__ dWt,                 // (COUT, K) f32
     float* __restrict__ db,                  // (COUT,) f32
-    int M, int H, int W, int COUT, int rows_per_chunk, int CINR) {
+    int M, int H, int W, int COUT, int rows_per_chunk) {
     constexpr int K = 9 * CIN;
     constexpr int KROW = 3 * CIN;
     constexpr int KSPLIT = NCO2 ? 4 : 2;
@@ -327,18 +324,8 @@ __global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
             for (int r = 0; r < 4; ++r) {
                 long co = i * 16 + crow + r;
                 long kk = wk * KHALF + kf * 16 + ccol;
-                if (co < COUT && kk < K) {
-                    if (TORCH_OUT) {
-                        int c = (int)(kk % CIN);
-                        int s_ = (int)(kk / CIN);
-                        int kx = s_ % 3, ky = s_ / 3;
-                        if (c < CINR)
-                            atomicAdd(&dWt[((co * CINR + c) * 3 + ky) * 3 + kx],
-                                      acc[i][kf][r]);
-                    } else {
-                        atomicAdd(&dWt[co * K + kk], acc[i][kf][r]);
-                    }
-                }
+                if (co < COUT && kk < K)
+                    atomicAdd(&dWt[co * K + kk], acc[i][kf][r]);
             }
     if (threadIdx.x < 32) {
         long c = threadIdx.x;
@@ -663,16 +650,14 @@ std::vector<torch::Tensor> conv3p_wgrad(torch::Tensor dY, torch::Tensor in,
     hipLaunchKernelGGL((conv3p_wgrad_kernel<U8, CIN_, RELU_, NCO2_>), grid,   \
                        dim3(256), 0, stream.stream(), dy, x,                  \
                        dWt.data_ptr<float>(), db.data_ptr<float>(), (int)M,   \
-                       (int)H, (int)W, (int)COUT, (int)rows_per_chunk,        \
-                       (int)CIN)
+                       (int)H, (int)W, (int)COUT, (int)rows_per_chunk)
 #define WG(U8, CIN_, RELU_)                                                   \
     do { if (big) WG1(U8, CIN_, RELU_, true); else WG1(U8, CIN_, RELU_, false); } while (0)
 #define WGH(U8, CIN_, RELU_, NCO2_, HT_)                                      \
     hipLaunchKernelGGL((conv3p_wgrad_kernel<U8, CIN_, RELU_, NCO2_, HT_>),    \
                        grid, dim3(256), 0, stream.stream(), dy, x,            \
                        dWt.data_ptr<float>(), db.data_ptr<float>(), (int)M,   \
-                       (int)H, (int)W, (int)COUT, (int)rows_per_chunk,        \
-                       (int)CIN)
+                       (int)H, (int)W, (int)COUT, (int)rows_per_chunk)
     bool done = true;
     if (H == 84 && u8 && !big) WGH(true, 8, false, false, 84);
     else if (H == 42 && !u8 && CIN == 16 && relu_in && !big)
@@ -698,55 +683,6 @@ std::vector<torch::Tensor> conv3p_wgrad(torch::Tensor dY, torch::Tensor in,
 #undef WG1
     return {dWt, db};
 }
-
-// torch-layout accumulation into pre-zeroed .grad views (engine backward)
-void conv3p_wgrad_into(torch::Tensor dY, torch::Tensor in, int64_t N,
-                       int64_t H, int64_t W, bool relu_in, int64_t cin_real,
-                       torch::Tensor dW_out, torch::Tensor db_out) {
-    long M = N * H * W;
-    long CIN = in.size(3);
-    long COUT = dY.size(3);
-    bool u8 = in.dtype() == torch::kUInt8;
-    TORCH_CHECK(dW_out.dtype() == torch::kFloat32 && dW_out.is_contiguous());
-    long target_chunks = 1024;
-    long rows_per_chunk = std::max(64L, (M + target_chunks - 1) / target_chunks);
-    rows_per_chunk = ((rows_per_chunk + 63) / 64) * 64;
-    dim3 grid(icdiv(M, rows_per_chunk));
-    auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-    auto* dy = reinterpret_cast<const __hip_bfloat16*>(dY.data_ptr());
-    const void* x = in.data_ptr();
-    bool big = COUT > 16;
-#define WGI(U8, CIN_, RELU_, NCO2_, HT_)                                      \
-    hipLaunchKernelGGL((conv3p_wgrad_kernel<U8, CIN_, RELU_, NCO2_, HT_,      \
-                                            true>),                          \
-                       grid, dim3(256), 0, stream.stream(), dy, x,            \
-                       dW_out.data_ptr<float>(), db_out.data_ptr<float>(),    \
-                       (int)M, (int)H, (int)W, (int)COUT,                     \
-                       (int)rows_per_chunk, (int)cin_real)
-    if (H == 84 && u8 && !big) WGI(true, 8, false, false, 84);
-    else if (H == 42 && !u8 && CIN == 16 && relu_in && !big)
-        WGI(false, 16, true, false, 42);
-    else if (H == 42 && !u8 && CIN == 16 && !relu_in && big)
-        WGI(false, 16, false, true, 42);
-    else if (H == 21 && !u8 && CIN == 32 && relu_in && big)
-        WGI(false, 32, true, true, 21);
-    else if (H == 21 && !u8 && CIN == 32 && !relu_in && big)
-        WGI(false, 32, false, true, 21);
-    else if (H == 11 && !u8 && CIN == 32 && relu_in && big)
-        WGI(false, 32, true, true, 11);
-    else if (u8) WGI(true, 8, false, false, 0);
-    else if (CIN == 16) {
-        if (relu_in) { if (big) WGI(false, 16, true, true, 0); else WGI(false, 16, true, false, 0); }
-        else { if (big) WGI(false, 16, false, true, 0); else WGI(false, 16, false, false, 0); }
-    } else if (CIN == 32) {
-        if (relu_in) { if (big) WGI(false, 32, true, true, 0); else WGI(false, 32, true, false, 0); }
-        else { if (big) WGI(false, 32, false, true, 0); else WGI(false, 32, false, false, 0); }
-    } else {
-        TORCH_CHECK(false, "unsupported CIN ", CIN);
-    }
-#undef WGI
-}
-
 void maxpool3s2_fwd(torch::Tensor in, torch::Tensor out, torch::Tensor arg,
                     int64_t N, int64_t H, int64_t W) {
     long C = in.size(3);

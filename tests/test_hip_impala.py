@@ -380,3 +380,34 @@ def test_hip_inference_matches_eager(encoder):
     agree = (q.float().argmax(1) == q_ref.argmax(1)).float().mean()
     assert agree > 0.9
     cfg.apply("mspacman")
+
+
+@pytest.mark.timeout(300)
+def test_vector_actor_gpu_hip_inference():
+    """VectorActor on cuda drives env ticks through the K15 HIP path and
+    produces valid replay blocks."""
+    import queue
+    from r2d2_amd import config as cfg
+    from r2d2_amd.models.network import Network
+    from r2d2_amd.train import epsilon_ladder
+    from r2d2_amd.worker import VectorActor
+
+    c = cfg.apply("mspacman_gpu_replay", num_actors=8, block_length=40,
+                  burn_in_steps=8, learning_steps=8, forward_steps=3,
+                  actor_update_interval=64, max_episode_steps=200)
+    torch.manual_seed(0)
+    model = Network(c.action_dim, c.obs_shape, c.hidden_dim, encoder="nature",
+                    forward_steps=c.forward_steps)
+    sq = queue.Queue()
+    va = VectorActor(epsilon_ladder(8), model, [sq], device="cuda", seed=4)
+    assert va.hip_inf is not None, "K15 path must be active on GPU"
+    total = va.run(stop_after_steps=400)
+    assert total >= 400
+    n = 0
+    while not sq.empty():
+        block, prio, reward = sq.get()
+        assert np.isfinite(prio).all()
+        assert block.obs.dtype == np.uint8
+        n += 1
+    assert n >= 4
+    cfg.apply("mspacman")
