@@ -159,11 +159,18 @@ def main():
         batches = [build_batch(c, device, seed=1000 + rank * 100 + i, hwc=hwc)
                    for i in range(args.batches)]
 
+    pending = [None]
+
     def one_step(i):
         if replay is not None:
-            batch = replay.sample()
+            # prefetch: the NEXT batch samples/gathers on the replay's side
+            # stream while this step trains (one step of priority staleness,
+            # far inside the reference's own <=12-batch staleness)
+            batch = pending[0] if pending[0] is not None else replay.sample()
+            tok = replay.sample_async()
             loss, prio = learner.train_step(batch)
             replay.update_priorities(batch.idxes, prio, batch.old_ptr)
+            pending[0] = replay.sample_wait(tok)
         else:
             loss, _ = learner.train_step(batches[i % len(batches)])
         return loss

@@ -81,6 +81,15 @@ class GpuReplayBuffer:
         self._pin_obs = torch.empty(self.obs_rows, self.obs_store.shape[-1],
                                     dtype=torch.uint8, pin_memory=True)
         self._copy_done = torch.cuda.Event()
+        # side stream for prefetch sampling (sample_async): sampling of the
+        # NEXT batch overlaps the current training step.  Tree reads and
+        # writes are fenced both ways: the sample stream waits on the last
+        # tree write (_ev_tree), and tree writers wait on the last sample's
+        # reads (_ev_sampled) — one step of priority staleness, well inside
+        # the reference's own <=12-batch staleness.
+        self._sample_stream = torch.cuda.Stream(device=dev)
+        self._ev_tree = torch.cuda.Event()
+        self._ev_sampled = torch.cuda.Event()
 
     # ------------------------------------------------------------------
     def __len__(self):
@@ -99,6 +108,7 @@ class GpuReplayBuffer:
         steps = block.action.shape[0]
         nseq = block.num_sequences
         dev = self.device
+        torch.cuda.current_stream().wait_event(self._ev_sampled)
 
         # store frames HWC (channels innermost) — the conv kernels are NHWC,
         # so gathered batches feed conv1 with no device-side permute
@@ -167,27 +177,39 @@ class GpuReplayBuffer:
         self.block_ptr = (self.block_ptr + 1) % self.num_blocks
 
     # ------------------------------------------------------------------
-    def sample(self, batch_size: Optional[int] = None):
-        """Returns a device-resident TrainingBatch; one small D2H copy for the
-        per-sample lengths (overlap by calling from a prefetch thread/stream)."""
-        from ..worker import TrainingBatch
-
+    def sample_async(self, batch_size: Optional[int] = None):
+        """Launch sampling + gather on the side stream and start the small
+        D2H of per-sample lengths; returns a token for sample_wait().  Call
+        right after launching the training step to overlap the two."""
         B = batch_size or self.batch_size
         dev = self.device
-        jitter = torch.rand(B, device=dev)
-        idx, prio, weight = self._ext.sumtree_sample(
-            self.tree, self.leaf_offset, self.num_levels, jitter, B, self.beta)
-        meta, seg = self._ext.replay_gather_meta(
-            idx, self.burn_s, self.learn_s, self.fwd_s, self.obs_start_s,
-            self.learn_off_s, self.spb)
-        outs = self._ext.replay_gather_batch(
-            self.obs_store, self.la_store, self.lr_store, self.act_store,
-            self.nsr_store, self.gam_store, self.hid_store, idx, meta, seg,
-            weight, self.T, self.A, self.learn_len, self.H, self.spb)
-        obs, la, lr, act, nsr, gam, w_rep, hid = outs
+        self._ev_tree.record(torch.cuda.current_stream())
+        with torch.cuda.stream(self._sample_stream):
+            self._sample_stream.wait_event(self._ev_tree)
+            jitter = torch.rand(B, device=dev)
+            idx, prio, weight = self._ext.sumtree_sample(
+                self.tree, self.leaf_offset, self.num_levels, jitter, B,
+                self.beta)
+            meta, seg = self._ext.replay_gather_meta(
+                idx, self.burn_s, self.learn_s, self.fwd_s, self.obs_start_s,
+                self.learn_off_s, self.spb)
+            outs = self._ext.replay_gather_batch(
+                self.obs_store, self.la_store, self.lr_store, self.act_store,
+                self.nsr_store, self.gam_store, self.hid_store, idx, meta,
+                seg, weight, self.T, self.A, self.learn_len, self.H, self.spb)
+            meta_h = meta[:3].to("cpu", non_blocking=True)
+            seg_h = seg.to("cpu", non_blocking=True)
+            self._ev_sampled.record(self._sample_stream)
+        return (idx, outs, meta_h, seg_h, self.block_ptr, self.env_steps)
 
-        meta_h = meta[:3].cpu()  # sync point (small)
-        seg_h = seg.cpu()
+    def sample_wait(self, token):
+        """Complete a sample_async(): host-syncs the side stream only."""
+        from ..worker import TrainingBatch
+
+        idx, outs, meta_h, seg_h, old_ptr, env_steps = token
+        self._ev_sampled.synchronize()
+        obs, la, lr, act, nsr, gam, w_rep, hid = outs
+        B = idx.shape[0]
         R = int(seg_h[-1])
         c_, h_, w_ = self.obs_shape
         batch = TrainingBatch(
@@ -197,13 +219,21 @@ class GpuReplayBuffer:
             burn_in_steps=meta_h[0].long(), learning_steps=meta_h[1].long(),
             forward_steps=meta_h[2].long(),
             idxes=idx, is_weights=w_rep[:R],
-            old_ptr=self.block_ptr, env_steps=self.env_steps)
+            old_ptr=old_ptr, env_steps=env_steps)
+        # training-stream consumers must see the gathered tensors
+        torch.cuda.current_stream().wait_event(self._ev_sampled)
         return batch
+
+    def sample(self, batch_size: Optional[int] = None):
+        """Synchronous sampling (sample_async + sample_wait)."""
+        return self.sample_wait(self.sample_async(batch_size))
 
     # ------------------------------------------------------------------
     def update_priorities(self, idxes: torch.Tensor, priorities: torch.Tensor,
                           old_ptr: int):
-        """idxes/priorities stay on device; ring-stale mask applied in-kernel."""
+        """idxes/priorities stay on device; ring-stale mask applied in-kernel.
+        Waits on any in-flight sample's tree reads (side stream)."""
+        torch.cuda.current_stream().wait_event(self._ev_sampled)
         self._ext.sumtree_update(self.tree, self.leaf_offset,
                                  idxes.to(torch.int64),
                                  priorities.float(), self.alpha,
