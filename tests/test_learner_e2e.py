@@ -96,3 +96,26 @@ def test_learning_reduces_loss_on_fixed_batch():
             first = float(loss)
         last = float(loss)
     assert last < first
+
+
+def test_seeded_determinism():
+    """Fixed seed -> identical loss trace across two independent runs
+    (mirrors the reference's train.py:10-13 seeding)."""
+    def run():
+        c, model, rb, actor, learner, sq = build_stack(seed=7)
+        actor.rng = np.random.default_rng(7)
+        actor.stop_after_steps = 200
+        actor.run()
+        while not sq.empty():
+            rb.add(*sq.get())
+        losses = []
+        for _ in range(4):
+            batch = rb.sample_batch()
+            loss, prio = learner.train_step(batch)
+            rb.update_priorities(batch.idxes, prio, batch.old_ptr, float(loss))
+            losses.append(float(loss))
+        return losses
+
+    a = run()
+    b = run()
+    assert a == b, (a, b)
