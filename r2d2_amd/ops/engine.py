@@ -265,18 +265,21 @@ class HipNetworkEngine:
         if self._pos_cache is not None and self._pos_cache[0] == key:
             return self._pos_cache[1]
         n = self.cfg.forward_steps
-        lp, tp, lbt = [], [], []
-        for b, (bu, le, fw) in enumerate(zip(burn.tolist(), learn.tolist(),
-                                             fwd.tolist())):
-            t_learn = np.arange(le) + bu
-            t_tgt = np.minimum(t_learn + n, bu + le + fw - 1)
-            lp.append(b * (T + 1) + t_learn + 1)
-            tp.append(b * (T + 1) + t_tgt + 1)
-            lbt.append(b * T + t_learn)
+        bu = burn.numpy().astype(np.int64)
+        le = learn.numpy().astype(np.int64)
+        fw = fwd.numpy().astype(np.int64)
+        R = int(le.sum())
+        # vectorized ragged expansion (no per-sample python loop)
+        b_of = np.repeat(np.arange(len(le), dtype=np.int64), le)
+        starts = np.zeros(len(le), dtype=np.int64)
+        starts[1:] = np.cumsum(le[:-1])
+        within = np.arange(R, dtype=np.int64) - np.repeat(starts, le)
+        t_learn = bu[b_of] + within
+        t_tgt = np.minimum(t_learn + n, (bu + le + fw - 1)[b_of])
         dev = self.device
-        lp = np.concatenate(lp)
-        tp = np.concatenate(tp)
-        lbt = np.concatenate(lbt)
+        lp = b_of * (T + 1) + t_learn + 1
+        tp = b_of * (T + 1) + t_tgt + 1
+        lbt = b_of * T + t_learn
         lp_t = torch.from_numpy(lp).to(dev)
         tp_t = torch.from_numpy(tp).to(dev)
         lbt_t = torch.from_numpy(lbt).to(dev)

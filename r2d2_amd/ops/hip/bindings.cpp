@@ -61,6 +61,8 @@ std::vector<torch::Tensor> conv_wgrad(torch::Tensor dY, torch::Tensor act,
 // lstm_kernels.hip
 void barrier_bench(torch::Tensor barrier_ws, int64_t steps, int64_t nblocks);
 void handoff_bench(torch::Tensor barrier_ws, int64_t steps, int64_t nblocks);
+void handoff_counter_bench(torch::Tensor barrier_ws, int64_t steps,
+                           int64_t nblocks);
 std::vector<torch::Tensor> lstm_fwd(
     torch::Tensor X0, torch::Tensor X1, torch::Tensor Whh0, torch::Tensor Whh1,
     torch::Tensor init0, torch::Tensor init1, torch::Tensor lens,
@@ -127,6 +129,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("lstm_bwd", &lstm_bwd, "Persistent fused LSTM BPTT backward");
     m.def("barrier_bench", &barrier_bench, "grid barrier microbench");
     m.def("handoff_bench", &handoff_bench, "producer-flag handoff microbench");
+    m.def("handoff_counter_bench", &handoff_counter_bench,
+          "counter-aggregated handoff microbench");
     m.def("dueling_combine", &dueling_combine, "q = V + A - mean(A)");
     m.def("dueling_combine_bwd", &dueling_combine_bwd, "dueling combine backward");
     m.def("conv3p", &conv3p,

@@ -526,6 +526,33 @@ __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
     }
 }
 
+// counter-aggregated handoff microbench: ONE atomic counter per group —
+// publish = single atomicAdd, await = poll one word until nblocks*(t+1).
+__global__ void handoff_counter_bench_kernel(GridBar* bar, int steps,
+                                             int nblocks) {
+    unsigned* ctr = &bar->flags[0];
+    for (int t = 0; t < steps; ++t) {
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+            __hip_atomic_fetch_add(ctr, 1u, __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+        }
+        if (threadIdx.x == 0) {
+            long spins = 0;
+            while (__hip_atomic_load(ctr, __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT)
+                   < (unsigned)(nblocks * (t + 1))) {
+                __builtin_amdgcn_s_sleep(1);
+                if (++spins > (long)2e8) return;
+            }
+            __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+        }
+        __syncthreads();
+    }
+}
+
 // barrier-only microbench
 __global__ void barrier_bench_kernel(GridBar* bar, int steps, int nblocks) {
     for (int t = 0; t < steps; ++t)
@@ -561,6 +588,16 @@ void barrier_bench(torch::Tensor barrier_ws, int64_t steps, int64_t nblocks) {
     zero_ws(barrier_ws, stream.stream());
     hipLaunchKernelGGL(barrier_bench_kernel, dim3((int)nblocks), dim3(256), 0,
                        stream.stream(),
+                       reinterpret_cast<GridBar*>(barrier_ws.data_ptr()),
+                       (int)steps, (int)nblocks);
+}
+
+void handoff_counter_bench(torch::Tensor barrier_ws, int64_t steps,
+                           int64_t nblocks) {
+    auto stream = at::cuda::getCurrentCUDAStream();
+    zero_ws(barrier_ws, stream.stream());
+    hipLaunchKernelGGL(handoff_counter_bench_kernel, dim3((int)nblocks),
+                       dim3(256), 0, stream.stream(),
                        reinterpret_cast<GridBar*>(barrier_ws.data_ptr()),
                        (int)steps, (int)nblocks);
 }
