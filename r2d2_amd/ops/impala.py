@@ -12,7 +12,7 @@ config.encoder == 'impala'; LSTM / heads / loss / optimizer paths are
 shared with the flagship nature config.
 """
 
-from typing import List, Optional, Tuple
+from typing import Optional
 
 import torch
 
