@@ -57,8 +57,12 @@ def train(seed: int = 0, restart_dead_actors: bool = True):
     batch_queue = mp.Queue(c.batch_queue_size)
     priority_queue = mp.Queue(c.batch_queue_size)
 
-    buffer = ReplayBuffer(sample_queues, batch_queue, priority_queue)
+    use_gpu_replay = c.gpu_replay and torch.cuda.is_available()
+    buffer = None if use_gpu_replay else ReplayBuffer(
+        sample_queues, batch_queue, priority_queue)
     learner = Learner(batch_queue, priority_queue, model)
+    if use_gpu_replay and c.use_hip_kernels:
+        learner.enable_hip_engine()
 
     if c.vector_actors:
         # one driver process, all envs in lockstep, batched inference
@@ -76,8 +80,10 @@ def train(seed: int = 0, restart_dead_actors: bool = True):
     for p in actor_procs:
         p.start()
 
-    buffer_proc = mp.Process(target=_run_buffer, args=(buffer,))
-    buffer_proc.start()
+    buffer_proc = None
+    if buffer is not None:
+        buffer_proc = mp.Process(target=_run_buffer, args=(buffer,))
+        buffer_proc.start()
 
     # actor supervision: the reference silently loses dead actor processes
     # (SURVEY §5 — throughput degrades with no signal); restart them.
@@ -97,10 +103,16 @@ def train(seed: int = 0, restart_dead_actors: bool = True):
         wd = threading.Thread(target=_watchdog, daemon=True)
         wd.start()
 
-    learner.run()
+    if use_gpu_replay:
+        # configs[2]/[3]: the learner owns the GPU-resident replay; no
+        # host buffer process
+        learner.run_with_gpu_replay(sample_queues)
+    else:
+        learner.run()
 
     stop.set()
-    buffer_proc.join()
+    if buffer_proc is not None:
+        buffer_proc.join()
     for p in actor_procs:
         p.terminate()
 

@@ -672,6 +672,106 @@ class Learner:
             if self.num_updates % self.save_interval == 0:
                 self.save(start_time)
 
+    def run_with_gpu_replay(self, sample_queue_list):
+        """The configs[2]/[3] topology: the learner process OWNS the
+        GPU-resident prioritized replay (replay/gpu_replay.py).  Actor
+        blocks arrive on mp.Queues and are ingested into the HBM block
+        store by a background thread (pinned staging, copy stream);
+        sampling, batch assembly, and priority updates never leave the
+        device.  The separate host ReplayBuffer process (reference
+        worker.py:77-138) disappears — this method is also the logger and
+        the termination condition."""
+        from .replay.gpu_replay import GpuReplayBuffer
+
+        c = self.cfg
+        replay = GpuReplayBuffer(device=self.device)
+        stats = {"episode_reward": 0.0, "num_episodes": 0, "sum_loss": 0.0,
+                 "last_updates": 0, "last_size": 0}
+        lock = threading.Lock()
+        stop = threading.Event()
+
+        def _ingest():
+            while not stop.is_set():
+                got = False
+                for q in sample_queue_list:
+                    try:
+                        block, prio, reward = q.get_nowait()
+                    except queue_mod.Empty:
+                        continue
+                    with lock:
+                        replay.ingest(block, prio)
+                        if reward is not None:
+                            stats["episode_reward"] += reward
+                            stats["num_episodes"] += 1
+                    got = True
+                if not got:
+                    time.sleep(0.005)
+
+        threading.Thread(target=_ingest, daemon=True).start()
+        start_time = time.time() - self._resumed_minutes * 60.0
+        os.makedirs(self.model_dir, exist_ok=True)
+        last_log = time.time()
+        while len(replay) < c.learning_starts:
+            time.sleep(0.2)
+            if time.time() - last_log > c.log_interval:
+                print(f"filling GPU replay: {len(replay)}/{c.learning_starts}")
+                last_log = time.time()
+
+        while self.num_updates < c.training_steps:
+            with lock:
+                batch = replay.sample()
+            loss, priorities = self.train_step(batch)
+            if not torch.is_tensor(priorities):   # eager path returns numpy
+                priorities = torch.as_tensor(priorities, device=self.device)
+            with lock:
+                replay.update_priorities(batch.idxes, priorities,
+                                         batch.old_ptr)
+            stats["sum_loss"] += float(loss)
+            self.env_steps = replay.env_steps
+            if self.num_updates % 4 == 0:
+                self.store_weights()
+            if self.num_updates % self.target_net_update_interval == 0:
+                self.target_net.load_state_dict(self.online_net.state_dict())
+                if self.engine is not None:
+                    self.engine.refresh_target()
+            if self.num_updates % self.save_interval == 0:
+                self.save(start_time)
+            now = time.time()
+            if now - last_log > c.log_interval:
+                interval = now - last_log
+                delta = self.num_updates - stats["last_updates"]
+                rec = {
+                    "wall_s": round(now - start_time, 1),
+                    "buffer_size": len(replay),
+                    "buffer_fill_rate":
+                        (len(replay) - stats["last_size"]) / interval,
+                    "env_steps": self.env_steps,
+                    "training_steps": self.num_updates,
+                    "training_rate": delta / interval,
+                    "seq_samples_per_s": delta / interval * replay.batch_size,
+                    "loss": stats["sum_loss"] / max(1, delta),
+                }
+                if stats["num_episodes"]:
+                    rec["avg_episode_return"] = (stats["episode_reward"]
+                                                 / stats["num_episodes"])
+                    stats["episode_reward"] = 0.0
+                    stats["num_episodes"] = 0
+                print(f"buffer size: {rec['buffer_size']}  "
+                      f"env steps: {rec['env_steps']}  "
+                      f"updates: {rec['training_steps']} "
+                      f"({rec['training_rate']:.1f}/s, "
+                      f"{rec['seq_samples_per_s']:.0f} seq/s)  "
+                      f"loss: {rec['loss']:.4f}")
+                if self.cfg.metrics_path:
+                    import json
+                    with open(self.cfg.metrics_path, "a") as f:
+                        f.write(json.dumps(rec) + "\n")
+                stats["last_updates"] = self.num_updates
+                stats["last_size"] = len(replay)
+                stats["sum_loss"] = 0.0
+                last_log = now
+        stop.set()
+
     def save(self, start_time):
         """Write the reference 4-tuple checkpoint (worker.py:380-381 /
         test.py:27) plus a ``.train.pth`` sidecar (optimizer + target net)

@@ -414,3 +414,44 @@ def test_vector_actor_gpu_hip_inference():
         n += 1
     assert n >= 4
     cfg.apply("mspacman")
+
+
+@pytest.mark.timeout(420)
+def test_train_gpu_replay_topology():
+    """configs[2] end-to-end on one GPU: VectorActor (K15 inference) feeds
+    the learner process's GPU-resident replay; the learner trains through
+    the HIP engine.  Run in-process with a queue and a driver thread."""
+    import queue
+    import threading
+    from r2d2_amd import config as cfg
+    from r2d2_amd.models.network import Network
+    from r2d2_amd.train import epsilon_ladder
+    from r2d2_amd.worker import Learner, VectorActor
+
+    c = cfg.apply("mspacman_gpu_replay", num_actors=8, block_length=40,
+                  burn_in_steps=8, learning_steps=8, forward_steps=3,
+                  batch_size=16, buffer_capacity=8_000, learning_starts=400,
+                  training_steps=25, max_episode_steps=300, log_interval=5,
+                  actor_update_interval=64, save_interval=10_000)
+    torch.manual_seed(0)
+    model = Network(c.action_dim, c.obs_shape, c.hidden_dim, encoder="nature",
+                    forward_steps=c.forward_steps)
+    model.share_memory()
+    sq = queue.Queue()
+    learner = Learner(None, None, model, model_dir="/tmp/r2d2_test_models")
+    learner.enable_hip_engine()
+    assert learner.engine is not None
+
+    va = VectorActor(epsilon_ladder(8), model, [sq], device="cuda", seed=1)
+    stop = threading.Event()
+
+    def drive():
+        while not stop.is_set():
+            va.run(stop_after_steps=200)
+
+    t = threading.Thread(target=drive, daemon=True)
+    t.start()
+    learner.run_with_gpu_replay([sq])
+    stop.set()
+    assert learner.num_updates == c.training_steps
+    cfg.apply("mspacman")
