@@ -84,17 +84,21 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
     int kseg = (lane >> 4) * 8;
 
     // per-lane patch base addresses for its two A rows
+    // 32-bit index math: M <= B*T*84*84 < 2^31, so unsigned division
+    // lowers to multiply-shift instead of 64-bit libcalls
+    const unsigned OHW = (unsigned)(OH * OW);
     long abase[2];
     bool avalid[2];
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-        long r = row0 + i * 16 + frow;
-        avalid[i] = r < M;
+        unsigned r = (unsigned)row0 + i * 16 + frow;
+        avalid[i] = r < (unsigned)M;
         if (avalid[i]) {
-            long n = r / (OH * OW);
-            int p = (int)(r % (OH * OW));
-            int oy = p / OW, ox = p % OW;
-            abase[i] = ((n * INH + (long)oy * S) * INW + (long)ox * S) * CIN;
+            unsigned n = r / OHW;
+            unsigned p = r % OHW;
+            int oy = p / (unsigned)OW, ox = p % (unsigned)OW;
+            abase[i] = (((long)n * INH + (long)oy * S) * INW
+                        + (long)ox * S) * CIN;
         } else {
             abase[i] = 0;
         }
@@ -155,7 +159,10 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
 // y = y0 + yy*S.  dYp is the zero-PADDED upstream gradient (pre-masked by
 // the ReLU of this conv's output).
 // ---------------------------------------------------------------------------
-template <int TAPS, int NCOL>
+// CO_T: compile-time COUT (both nature dgrads have COUT=64) — the k ->
+// (tap, co) split per lane per k-iteration otherwise runs runtime
+// divisions (PMC: 72:1 VALU:MFMA on this kernel).
+template <int TAPS, int NCOL, int CO_T = 0>
 __global__ __launch_bounds__(256) void conv_dgrad_kernel(
     const __hip_bfloat16* __restrict__ dYp,  // (N, PH, PW, COUT)
     const __hip_bfloat16* __restrict__ Wd,   // (CIN, TAPS*COUT)
@@ -163,7 +170,8 @@ __global__ __launch_bounds__(256) void conv_dgrad_kernel(
     const int* __restrict__ taps,            // (TAPS, 2): dy, dx
     int Mc, int YY, int XX, int y0, int x0, int S, int pad,
     int PH, int PW, int COUT, int XH, int XW, int CIN) {
-    const int K = TAPS * COUT;
+    const int COUTc = CO_T ? CO_T : COUT;
+    const int K = TAPS * COUTc;
     int wave = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
     int wr = (NCOL == 32) ? wave : (wave >> 1);
@@ -173,26 +181,27 @@ __global__ __launch_bounds__(256) void conv_dgrad_kernel(
     int frow = lane & 15;
     int kseg = (lane >> 4) * 8;
 
-    // per-lane (n, y, x) for its two rows
+    // per-lane (n, y, x) for its two rows (32-bit division)
+    const unsigned YX = (unsigned)(YY * XX);
     long nbase[2];
     int yv[2], xv[2];
     bool avalid[2];
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-        long r = row0 + i * 16 + frow;
-        avalid[i] = r < Mc;
-        long n = avalid[i] ? r / (YY * XX) : 0;
-        int p = avalid[i] ? (int)(r % (YY * XX)) : 0;
-        yv[i] = y0 + (p / XX) * S;
-        xv[i] = x0 + (p % XX) * S;
-        nbase[i] = n * PH * PW;
+        unsigned r = (unsigned)row0 + i * 16 + frow;
+        avalid[i] = r < (unsigned)Mc;
+        unsigned n = avalid[i] ? r / YX : 0;
+        unsigned p = avalid[i] ? r % YX : 0;
+        yv[i] = y0 + (int)(p / (unsigned)XX) * S;
+        xv[i] = x0 + (int)(p % (unsigned)XX) * S;
+        nbase[i] = (long)n * PH * PW;
     }
 
     f32x4 acc[2][2] = {};
     for (int k0 = 0; k0 < K; k0 += 32) {
         int k = k0 + kseg;
-        int t = k / COUT;
-        int co = k % COUT;
+        int t = (unsigned)k / (unsigned)COUTc;
+        int co = (unsigned)k % (unsigned)COUTc;
         int dy = taps[2 * t], dx = taps[2 * t + 1];
         bf16x8 a[2], b[2];
 #pragma unroll
@@ -226,10 +235,12 @@ __global__ __launch_bounds__(256) void conv_dgrad_kernel(
                 long rr = row0 + i * 16 + crow + r;
                 long cc = col0 + j * 16 + ccol;
                 if (rr < Mc && cc < CIN) {
-                    long n = rr / (YY * XX);
-                    int p = (int)(rr % (YY * XX));
-                    int y = y0 + (p / XX) * S, x = x0 + (p % XX) * S;
-                    dX[((n * XH + y) * XW + x) * CIN + cc] = f2bf(acc[i][j][r]);
+                    unsigned n = (unsigned)rr / YX;
+                    unsigned p = (unsigned)rr % YX;
+                    int y = y0 + (int)(p / (unsigned)XX) * S;
+                    int x = x0 + (int)(p % (unsigned)XX) * S;
+                    dX[(((long)n * XH + y) * XW + x) * CIN + cc] =
+                        f2bf(acc[i][j][r]);
                 }
             }
 }
@@ -300,10 +311,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
                 long gm = m0 + mrow;
                 bf16x8 w = czero();
                 if (gm < mend) {
-                    long n = gm / (OH * OW);
-                    int p = (int)(gm % (OH * OW));
-                    int oy = p / OW, ox = p % OW;
-                    long base = ((n * INH + (long)oy * S) * INW
+                    unsigned n = (unsigned)gm / (unsigned)(OH * OW);
+                    unsigned p = (unsigned)gm % (unsigned)(OH * OW);
+                    int oy = p / (unsigned)OW, ox = p % (unsigned)OW;
+                    long base = (((long)n * INH + (long)oy * S) * INW
                                  + (long)ox * S) * CIN;
                     int dy_ = k / KWC, rem = k % KWC;
                     long off = base + (long)dy_ * INW * CIN + rem;
@@ -424,21 +435,19 @@ torch::Tensor conv_dgrad(torch::Tensor dYp, torch::Tensor Wd, torch::Tensor taps
     auto* w = reinterpret_cast<const __hip_bfloat16*>(Wd.data_ptr());
     auto* dx = reinterpret_cast<__hip_bfloat16*>(dX.data_ptr());
     const int* tp = taps.data_ptr<int>();
-#define DLAUNCH(T)                                                             \
-    if (CIN <= 32)                                                             \
-        hipLaunchKernelGGL((conv_dgrad_kernel<T, 32>), grid, dim3(256), 0,     \
-                       stream.stream(), dy, w, dx, tp, (int)Mc, (int)YY,       \
-                       (int)XX, (int)y0, (int)x0, (int)S, (int)pad, (int)PH,   \
-                       (int)PW, (int)COUT, (int)XH, (int)XW, (int)CIN);        \
-    else                                                                       \
-        hipLaunchKernelGGL((conv_dgrad_kernel<T, 64>), grid, dim3(256), 0,     \
+#define DL1(T, NC, CO)                                                         \
+    hipLaunchKernelGGL((conv_dgrad_kernel<T, NC, CO>), grid, dim3(256), 0,     \
                        stream.stream(), dy, w, dx, tp, (int)Mc, (int)YY,       \
                        (int)XX, (int)y0, (int)x0, (int)S, (int)pad, (int)PH,   \
                        (int)PW, (int)COUT, (int)XH, (int)XW, (int)CIN)
-    if (TAPS == 4) DLAUNCH(4);
-    else if (TAPS == 9) DLAUNCH(9);
+#define DLAUNCH(T)                                                             \
+    if (CIN <= 32) { if (COUT == 64) DL1(T, 32, 64); else DL1(T, 32, 0); }     \
+    else { if (COUT == 64) DL1(T, 64, 64); else DL1(T, 64, 0); }
+    if (TAPS == 4) { DLAUNCH(4); }
+    else if (TAPS == 9) { DLAUNCH(9); }
     else TORCH_CHECK(false, "unsupported tap count");
 #undef DLAUNCH
+#undef DL1
     return dX;
 }
 
