@@ -366,22 +366,25 @@ class HipNetworkEngine:
         dq_pad[:R] = dq
         dadv2, dval2 = m.dueling_combine_bwd(dq_pad.contiguous(),
                                              PAD_HEAD, PAD_HEAD)
+        self._mark("hb_duel")
         # adv path
         dadv1 = m.gemm_dgrad(dadv2, self._empty, ON.wa2_kn, False)
         dh_a = m.gemm_dgrad(dadv1, adv1_o, ON.wa1_kn, True)
+        # value path
+        dval1 = m.gemm_dgrad(dval2, self._empty, ON.wv2_kn, False)
+        dh_v = m.gemm_dgrad(dval1, val1_o, ON.wv1_kn, True)
+        self._mark("hb_dgrads")
         m.gemm_wgrad_into(dadv2, self._empty, adv1_o, False,
                           net.advantage[2].weight.grad,
                           net.advantage[2].bias.grad)
         m.gemm_wgrad_into(dadv1, adv1_o, h_cat, True,
                           net.advantage[0].weight.grad,
                           net.advantage[0].bias.grad)
-        # value path
-        dval1 = m.gemm_dgrad(dval2, self._empty, ON.wv2_kn, False)
-        dh_v = m.gemm_dgrad(dval1, val1_o, ON.wv1_kn, True)
         m.gemm_wgrad_into(dval2, self._empty, val1_o, False,
                           net.value[2].weight.grad, net.value[2].bias.grad)
         m.gemm_wgrad_into(dval1, val1_o, h_cat, True,
                           net.value[0].weight.grad, net.value[0].bias.grad)
+        self._mark("hb_wgrads")
 
         dh_rows = (dh_a.float() + dh_v.float())     # (2R, 512)
         # scatter-add into dHext (B, T, H): learn rows at t, tgt rows at t_tgt
