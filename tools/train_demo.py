@@ -21,7 +21,7 @@ def main(updates=600):
     c = cfg.apply("mspacman_gpu_replay", num_actors=64,
                   buffer_capacity=400_000, learning_starts=20_000,
                   training_steps=updates, log_interval=5,
-                  save_interval=200, actor_update_interval=400,
+                  save_interval=100_000, actor_update_interval=400,
                   metrics_path="gpurun_out/train_demo_metrics.jsonl")
     torch.manual_seed(0)
     model = Network(c.action_dim, c.obs_shape, c.hidden_dim,
