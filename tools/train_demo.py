@@ -17,8 +17,8 @@ from r2d2_amd.worker import Learner, VectorActor  # noqa: E402
 import queue  # noqa: E402
 
 
-def main(updates=600):
-    c = cfg.apply("mspacman_gpu_replay", num_actors=64,
+def main(updates=600, actors=64):
+    c = cfg.apply("mspacman_gpu_replay", num_actors=actors,
                   buffer_capacity=400_000, learning_starts=20_000,
                   training_steps=updates, log_interval=5,
                   save_interval=100_000, actor_update_interval=400,
@@ -49,4 +49,5 @@ def main(updates=600):
 
 
 if __name__ == "__main__":
-    main(int(sys.argv[1]) if len(sys.argv) > 1 else 600)
+    main(int(sys.argv[1]) if len(sys.argv) > 1 else 600,
+         int(sys.argv[2]) if len(sys.argv) > 2 else 64)
