@@ -193,6 +193,131 @@ __global__ __launch_bounds__(256) void conv3p_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// conv3p_band: band-staged variant of conv3p for the hot IMPALA geometries.
+// A workgroup owns a TH-row band of ONE image: the (TH+2) x (W+2) x CIN
+// input slab is staged through LDS once (u8 dequant / pre-activation ReLU
+// applied at stage time instead of once per tap), weights are preloaded to
+// registers, and all row->(y, x) arithmetic is by compile-time constants.
+// Epilogues identical to conv3p_kernel.
+// ---------------------------------------------------------------------------
+template <bool IN_U8, int CIN, int COUT_T, int HT, int TH, bool RELU_IN,
+          bool HAS_BIAS, int EPI>
+__global__ __launch_bounds__(256) void conv3p_band_kernel(
+    const void* __restrict__ in,
+    const __hip_bfloat16* __restrict__ Wt,
+    const float* __restrict__ bias,
+    const __hip_bfloat16* __restrict__ res,
+    const __hip_bfloat16* __restrict__ mask,
+    __hip_bfloat16* __restrict__ out, int N) {
+    constexpr int K = 9 * CIN;
+    constexpr int KROW = 3 * CIN;
+    constexpr int PW = HT + 2;
+    constexpr int PH = HT + 2;
+    constexpr int NBANDS = (HT + TH - 1) / TH;
+    constexpr int KITERS = (K + 31) / 32;
+    // COUT 16: 4 waves stack rows (128/iter); COUT 32: 2x2 wave grid
+    // (64 rows x 32 cols per iter); every wave owns ONE 16-col B fragment.
+    constexpr int RPI = (COUT_T == 32) ? 64 : 128;   // rows per iter
+    constexpr int SLAB = (TH + 2) * PW * CIN;
+
+    __shared__ __hip_bfloat16 s_in[SLAB];
+
+    const int band = blockIdx.x % NBANDS;
+    const long n = blockIdx.x / NBANDS;
+    const int y0 = band * TH;                         // first real out row
+    const int th_eff = (y0 + TH <= HT) ? TH : (HT - y0);
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int frow = lane & 15;
+    const int kseg = (lane >> 4) * 8;
+
+    // ---- stage the slab: padded rows [y0, y0 + th_eff + 2) ----
+    {
+        const int rows = th_eff + 2;
+        const long gbase = (n * PH + y0) * (long)PW * CIN;
+        for (int e = threadIdx.x * 8; e < rows * PW * CIN;
+             e += blockDim.x * 8) {
+            bf16x8 v;
+            if (IN_U8) {
+                v = idequant8(
+                    reinterpret_cast<const unsigned char*>(in) + gbase + e);
+            } else {
+                v = iload8(
+                    reinterpret_cast<const __hip_bfloat16*>(in) + gbase + e);
+                if (RELU_IN) v = irelu8(v);
+            }
+            *reinterpret_cast<bf16x8*>(&s_in[e]) = v;
+        }
+    }
+
+    // ---- preload weights (this wave's 16-col B fragment, all k-iters) ----
+    const int wr = (COUT_T == 32) ? (wave >> 1) : wave;
+    const int wc = (COUT_T == 32) ? (wave & 1) : 0;
+    bf16x8 wfrag[KITERS];
+#pragma unroll
+    for (int ki = 0; ki < KITERS; ++ki) {
+        int c = wc * 16 + frow;
+        int k = ki * 32 + kseg;
+        wfrag[ki] = (c < COUT_T && k < K)
+                        ? iload8(Wt + (long)c * K + k) : izero();
+    }
+    __syncthreads();
+
+    // ---- iterate row-tiles of the band ----
+    const int npix = th_eff * HT;
+    for (int p0 = wr * 32; p0 < npix; p0 += RPI) {
+        // two 16-row fragments per wave
+        bool lval[2];
+        long lbase[2];
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            int lp = p0 + i * 16 + frow;
+            lval[i] = lp < npix;
+            int yl = lp / HT, x = lp % HT;            // consts: mul-shift
+            lbase[i] = ((long)yl * PW + x) * CIN;
+        }
+        f32x4 acc[2] = {};
+#pragma unroll
+        for (int ki = 0; ki < KITERS; ++ki) {
+            int k = ki * 32 + kseg;
+            bool kval = k < K;
+            int dy = kval ? k / KROW : 0;
+            int rem = kval ? k % KROW : 0;
+            int off = dy * PW * CIN + rem;
+            bf16x8 a[2];
+#pragma unroll
+            for (int i = 0; i < 2; ++i)
+                a[i] = (lval[i] && kval) ? iload8(&s_in[lbase[i] + off])
+                                         : izero();
+#pragma unroll
+            for (int i = 0; i < 2; ++i)
+                acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a[i], wfrag[ki], acc[i], 0, 0, 0);
+        }
+        int ccol = lane & 15;
+        int crow = (lane >> 4) * 4;
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int lp = p0 + i * 16 + crow + r;
+                int cc = wc * 16 + ccol;
+                if (lp < npix && cc < COUT_T) {
+                    int yl = lp / HT, x = lp % HT;
+                    long oidx = ((n * PH + y0 + yl + 1) * PW + x + 1)
+                                    * COUT_T + cc;
+                    float v = acc[i][r];
+                    if (HAS_BIAS) v += bias[cc];
+                    if (EPI == 2 || EPI == 3)
+                        v = (bf2f(mask[oidx]) > 0.f) ? v : 0.f;
+                    if (EPI == 1 || EPI == 3) v += bf2f(res[oidx]);
+                    out[oidx] = f2bf(v);
+                }
+            }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // conv3p_wgrad: dWt(COUT, K=9*CIN) += dY^T @ patches, dY and input both in
 // the halo-padded layout; RELU_IN applies relu on patch load.  Same LDS
 // full-K staging scheme as conv_kernels.hip conv_wgrad (dY and patches are
@@ -564,51 +689,52 @@ void conv3p(torch::Tensor in, torch::Tensor Wt, torch::Tensor bias,
         else C3P_EPI(U8, CIN_, NCOL_, false, false);                          \
     } while (0)
 
-    // compile-time-H fast path for the IMPALA geometries (divisions by
-    // constants); falls through to the runtime-H chain otherwise
-#define C3PH(U8, CIN_, NCOL_, RELU_, BIAS_, EPI_, HT_)                        \
-    hipLaunchKernelGGL((conv3p_kernel<U8, CIN_, NCOL_, RELU_, BIAS_, EPI_,    \
-                                      HT_>),                                  \
-                       grid, dim3(256), 0, stream.stream(), x, w, b, rp, mp,  \
-                       o, (int)M, (int)H, (int)W, (int)COUT)
+    // band-staged fast path for the IMPALA geometries: input slab through
+    // LDS once (dequant / pre-activation ReLU at stage time), weights in
+    // registers, compile-time geometry
+#define C3B(U8, CIN_, CO_, HT_, TH_, RELU_, BIAS_, EPI_)                      \
+    hipLaunchKernelGGL((conv3p_band_kernel<U8, CIN_, CO_, HT_, TH_, RELU_,    \
+                                           BIAS_, EPI_>),                     \
+                       dim3((int)(N * ((HT_ + TH_ - 1) / TH_))), dim3(256),   \
+                       0, stream.stream(), x, w, b, rp, mp, o, (int)N)
     bool done = true;
     if (H == 84 && u8 && COUT == 16 && !relu_in && has_bias && epi == 0)
-        C3PH(true, 8, 16, false, true, 0, 84);
+        C3B(true, 8, 16, 84, 6, false, true, 0);
     else if (H == 42 && !u8 && CIN == 16 && COUT == 16 && relu_in && has_bias
              && epi == 0)
-        C3PH(false, 16, 16, true, true, 0, 42);
+        C3B(false, 16, 16, 42, 12, true, true, 0);
     else if (H == 42 && !u8 && CIN == 16 && COUT == 16 && relu_in && has_bias
              && epi == 1)
-        C3PH(false, 16, 16, true, true, 1, 42);
+        C3B(false, 16, 16, 42, 12, true, true, 1);
     else if (H == 42 && !u8 && CIN == 16 && COUT == 16 && !relu_in
              && !has_bias && epi == 2)
-        C3PH(false, 16, 16, false, false, 2, 42);
+        C3B(false, 16, 16, 42, 12, false, false, 2);
     else if (H == 42 && !u8 && CIN == 16 && COUT == 16 && !relu_in
              && !has_bias && epi == 3)
-        C3PH(false, 16, 16, false, false, 3, 42);
+        C3B(false, 16, 16, 42, 12, false, false, 3);
     else if (H == 42 && !u8 && CIN == 16 && COUT == 32 && !relu_in && has_bias
              && epi == 0)
-        C3PH(false, 16, 32, false, true, 0, 42);
+        C3B(false, 16, 32, 42, 12, false, true, 0);
     else if (H == 42 && !u8 && CIN == 32 && COUT == 16 && !relu_in
              && !has_bias && epi == 0)
-        C3PH(false, 32, 16, false, false, 0, 42);
+        C3B(false, 32, 16, 42, 12, false, false, 0);
     else if (H == 21 && !u8 && CIN == 32 && COUT == 32) {
-        if (relu_in && has_bias && epi == 0) C3PH(false, 32, 32, true, true, 0, 21);
-        else if (relu_in && has_bias && epi == 1) C3PH(false, 32, 32, true, true, 1, 21);
-        else if (!relu_in && !has_bias && epi == 0) C3PH(false, 32, 32, false, false, 0, 21);
-        else if (!relu_in && !has_bias && epi == 2) C3PH(false, 32, 32, false, false, 2, 21);
-        else if (!relu_in && !has_bias && epi == 3) C3PH(false, 32, 32, false, false, 3, 21);
+        if (relu_in && has_bias && epi == 0) C3B(false, 32, 32, 21, 21, true, true, 0);
+        else if (relu_in && has_bias && epi == 1) C3B(false, 32, 32, 21, 21, true, true, 1);
+        else if (!relu_in && !has_bias && epi == 0) C3B(false, 32, 32, 21, 21, false, false, 0);
+        else if (!relu_in && !has_bias && epi == 2) C3B(false, 32, 32, 21, 21, false, false, 2);
+        else if (!relu_in && !has_bias && epi == 3) C3B(false, 32, 32, 21, 21, false, false, 3);
         else done = false;
     } else if (H == 11 && !u8 && CIN == 32 && COUT == 32) {
-        if (relu_in && has_bias && epi == 0) C3PH(false, 32, 32, true, true, 0, 11);
-        else if (relu_in && has_bias && epi == 1) C3PH(false, 32, 32, true, true, 1, 11);
-        else if (!relu_in && !has_bias && epi == 2) C3PH(false, 32, 32, false, false, 2, 11);
-        else if (!relu_in && !has_bias && epi == 3) C3PH(false, 32, 32, false, false, 3, 11);
+        if (relu_in && has_bias && epi == 0) C3B(false, 32, 32, 11, 11, true, true, 0);
+        else if (relu_in && has_bias && epi == 1) C3B(false, 32, 32, 11, 11, true, true, 1);
+        else if (!relu_in && !has_bias && epi == 2) C3B(false, 32, 32, 11, 11, false, false, 2);
+        else if (!relu_in && !has_bias && epi == 3) C3B(false, 32, 32, 11, 11, false, false, 3);
         else done = false;
     } else {
         done = false;
     }
-#undef C3PH
+#undef C3B
     if (done) return;
 
     if (u8) {
