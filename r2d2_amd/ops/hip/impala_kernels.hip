@@ -459,143 +459,6 @@ __global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// conv3p_band_wgrad: band-staged backward-weight.  Each workgroup loops
-// over a chunk of (image, band) slabs: input slab + local dY tile staged
-// once per band (dequant / relu-in applied at stage time — the full-K
-// scheme in conv3p_wgrad_kernel re-stages every input element ~9x), MFMA
-// accumulators live in registers across ALL the workgroup's bands and are
-// flushed with one atomic pass at the end.
-//   dWt(COUT, K) += dY_band^T @ patch_band   (reduction over band pixels)
-// ---------------------------------------------------------------------------
-template <bool IN_U8, int CIN, int COUT_T, int HT, int TH, bool RELU_IN>
-__global__ __launch_bounds__(256) void conv3p_band_wgrad_kernel(
-    const __hip_bfloat16* __restrict__ dY,   // (N, HT+2, HT+2, COUT) padded
-    const void* __restrict__ in,             // (N, HT+2, HT+2, CIN) padded
-    float* __restrict__ dWt,                 // (COUT, K) f32
-    float* __restrict__ db,                  // (COUT,) f32
-    int N, int bands_per_wg) {
-    constexpr int K = 9 * CIN;
-    constexpr int KROW = 3 * CIN;
-    constexpr int PW = HT + 2;
-    constexpr int PH = HT + 2;
-    constexpr int NBANDS = (HT + TH - 1) / TH;
-    constexpr int KHALF = ((K / 4 + 15) / 16) * 16;
-    constexpr int KFRAG = KHALF / 16;
-    constexpr int NCO = COUT_T / 16;
-    constexpr int SLAB = (TH + 2) * PW * CIN;
-
-    __shared__ __hip_bfloat16 s_in[SLAB];
-    // +8 column pad: keeps 16 B alignment for the vectorized stage writes
-    // and staggers the column-gather reads across LDS banks
-    __shared__ __hip_bfloat16 s_dy[TH * HT + 8][COUT_T + 8];
-
-    const int wave = threadIdx.x / WAVE;
-    const int lane = threadIdx.x & (WAVE - 1);
-    const int frow = lane & 15;
-    const int mseg = (lane >> 4) * 8;
-    const long total_bands = (long)N * NBANDS;
-    const long b_start = (long)blockIdx.x * bands_per_wg;
-    const long b_end = min(total_bands, b_start + bands_per_wg);
-
-    f32x4 acc[NCO][KFRAG] = {};
-    float bias_acc = 0.f;
-
-    for (long bb = b_start; bb < b_end; ++bb) {
-        const int band = (int)(bb % NBANDS);
-        const long n = bb / NBANDS;
-        const int y0 = band * TH;
-        const int th_eff = (y0 + TH <= HT) ? TH : (HT - y0);
-        const int npix = th_eff * HT;
-        __syncthreads();
-        {   // input slab
-            const long gbase = (n * PH + y0) * (long)PW * CIN;
-            const int rows = th_eff + 2;
-            for (int e = threadIdx.x * 8; e < rows * PW * CIN;
-                 e += blockDim.x * 8) {
-                bf16x8 v;
-                if (IN_U8) {
-                    v = idequant8(
-                        reinterpret_cast<const unsigned char*>(in) + gbase + e);
-                } else {
-                    v = iload8(
-                        reinterpret_cast<const __hip_bfloat16*>(in) + gbase + e);
-                    if (RELU_IN) v = irelu8(v);
-                }
-                *reinterpret_cast<bf16x8*>(&s_in[e]) = v;
-            }
-        }
-        {   // dY tile, dense local (p, co)
-            for (int e = threadIdx.x * 8; e < npix * COUT_T;
-                 e += blockDim.x * 8) {
-                int pp = e / COUT_T;
-                int co = e % COUT_T;
-                int yl = pp / HT, x = pp % HT;
-                bf16x8 v = iload8(
-                    dY + ((n * PH + y0 + yl + 1) * (long)PW + x + 1) * COUT_T
-                        + co);
-                *reinterpret_cast<bf16x8*>(&s_dy[pp][co]) = v;
-            }
-        }
-        __syncthreads();
-
-        for (int pc = 0; pc < npix; pc += 32) {
-            bf16x8 fa[NCO];
-#pragma unroll
-            for (int i = 0; i < NCO; ++i)
-#pragma unroll
-                for (int e = 0; e < 8; ++e) {
-                    int pp = pc + mseg + e;
-                    fa[i][e] = (pp < npix)
-                                   ? *(const __bf16*)&s_dy[pp][i * 16 + frow]
-                                   : (__bf16)0.f;
-                }
-#pragma unroll
-            for (int kf = 0; kf < KFRAG; ++kf) {
-                int kcol = wave * KHALF + kf * 16 + frow;
-                bf16x8 fb = izero();
-                if (kcol < K) {
-                    int dyk = kcol / KROW;
-                    int rem = kcol % KROW;
-#pragma unroll
-                    for (int e = 0; e < 8; ++e) {
-                        int pp = pc + mseg + e;
-                        if (pp < npix) {
-                            int yl = pp / HT, x = pp % HT;
-                            fb[e] = s_in[((yl + dyk) * PW + x) * CIN + rem];
-                        }
-                    }
-                }
-#pragma unroll
-                for (int i = 0; i < NCO; ++i)
-                    acc[i][kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        fa[i], fb, acc[i][kf], 0, 0, 0);
-            }
-        }
-        {
-            const int bco = threadIdx.x % COUT_T;
-            const int pstep = blockDim.x / COUT_T;
-            for (int pp = threadIdx.x / COUT_T; pp < npix; pp += pstep)
-                bias_acc += bf2f(s_dy[pp][bco]);
-        }
-    }
-
-    int ccol = lane & 15;
-    int crow = (lane >> 4) * 4;
-#pragma unroll
-    for (int i = 0; i < NCO; ++i)
-#pragma unroll
-        for (int kf = 0; kf < KFRAG; ++kf)
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                long co = i * 16 + crow + r;
-                long kk = wave * KHALF + kf * 16 + ccol;
-                if (co < COUT_T && kk < K)
-                    atomicAdd(&dWt[co * K + kk], acc[i][kf][r]);
-            }
-    atomicAdd(&db[threadIdx.x % COUT_T], bias_acc);
-}
-
-// ---------------------------------------------------------------------------
 // maxpool 3x3 stride 2 pad 1: padded in (N,H+2,W+2,C) -> padded out
 // (N,OH+2,OW+2,C) + dense argmax tap (N,OH,OW,C) u8.  Out-of-image taps are
 // -inf (torch max_pool2d padding semantics — the halo zeros must NOT win).
@@ -916,32 +779,6 @@ std::vector<torch::Tensor> conv3p_wgrad(torch::Tensor dY, torch::Tensor in,
                        (int)H, (int)W, (int)COUT, (int)rows_per_chunk)
 #define WG(U8, CIN_, RELU_)                                                   \
     do { if (big) WG1(U8, CIN_, RELU_, true); else WG1(U8, CIN_, RELU_, false); } while (0)
-#define WGB(U8, CIN_, CO_, HT_, TH_, RELU_)                                   \
-    do {                                                                      \
-        long tb = (long)N * ((HT_ + TH_ - 1) / TH_);                          \
-        long bpw = std::max(1L, (tb + 1023) / 1024);                          \
-        hipLaunchKernelGGL((conv3p_band_wgrad_kernel<U8, CIN_, CO_, HT_,      \
-                                                     TH_, RELU_>),            \
-                           dim3(icdiv(tb, bpw)), dim3(256), 0,                \
-                           stream.stream(), dy, x, dWt.data_ptr<float>(),     \
-                           db.data_ptr<float>(), (int)N, (int)bpw);           \
-    } while (0)
-    bool bdone = true;
-    if (H == 84 && u8 && COUT == 16 && !relu_in) WGB(true, 8, 16, 84, 6, false);
-    else if (H == 42 && !u8 && CIN == 16 && COUT == 16 && relu_in)
-        WGB(false, 16, 16, 42, 12, true);
-    else if (H == 42 && !u8 && CIN == 16 && COUT == 32 && !relu_in)
-        WGB(false, 16, 32, 42, 12, false);
-    else if (H == 21 && !u8 && CIN == 32 && COUT == 32 && relu_in)
-        WGB(false, 32, 32, 21, 21, true);
-    else if (H == 21 && !u8 && CIN == 32 && COUT == 32 && !relu_in)
-        WGB(false, 32, 32, 21, 21, false);
-    else if (H == 11 && !u8 && CIN == 32 && COUT == 32 && relu_in)
-        WGB(false, 32, 32, 11, 11, true);
-    else bdone = false;
-#undef WGB
-    if (bdone) return {dWt, db};
-
 #define WGH(U8, CIN_, RELU_, NCO2_, HT_)                                      \
     hipLaunchKernelGGL((conv3p_wgrad_kernel<U8, CIN_, RELU_, NCO2_, HT_>),    \
                        grid, dim3(256), 0, stream.stream(), dy, x,            \
