@@ -447,11 +447,15 @@ def test_train_gpu_replay_topology():
 
     def drive():
         while not stop.is_set():
-            va.run(stop_after_steps=200)
+            va.run(stop_after_steps=100)
 
-    t = threading.Thread(target=drive, daemon=True)
+    t = threading.Thread(target=drive)
     t.start()
-    learner.run_with_gpu_replay([sq])
-    stop.set()
+    try:
+        learner.run_with_gpu_replay([sq])
+    finally:
+        stop.set()
+        t.join(timeout=60)   # join before interpreter teardown (a daemon
+        torch.cuda.synchronize()  # thread mid-CUDA-call aborts at exit)
     assert learner.num_updates == c.training_steps
     cfg.apply("mspacman")
