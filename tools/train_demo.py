@@ -39,11 +39,16 @@ def main(updates=600, actors=64):
 
     def drive():
         while not stop.is_set():
-            va.run(stop_after_steps=2000)
+            va.run(stop_after_steps=500)
 
-    threading.Thread(target=drive, daemon=True).start()
-    learner.run_with_gpu_replay([sq])
-    stop.set()
+    t = threading.Thread(target=drive)
+    t.start()
+    try:
+        learner.run_with_gpu_replay([sq])
+    finally:
+        stop.set()
+        t.join(timeout=60)
+        torch.cuda.synchronize()
     print(f"demo done: {learner.num_updates} updates, "
           f"{learner.env_steps} env steps")
 
