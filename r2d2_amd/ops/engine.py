@@ -19,8 +19,6 @@ every optimizer step (`refresh_online`), and on target-net sync
 (`refresh_target`).
 """
 
-from typing import Optional
-
 import os
 
 import numpy as np

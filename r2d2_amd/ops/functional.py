@@ -14,8 +14,6 @@ Reference semantics reproduced here:
   (reference: worker.py:268-276).
 """
 
-from typing import Tuple
-
 import numpy as np
 import torch
 

@@ -12,7 +12,7 @@ Ingest takes host `Block`s (from CPU actors) through pinned staging buffers
 on a dedicated copy stream.
 """
 
-from typing import List, Optional
+from typing import Optional
 
 import numpy as np
 import torch

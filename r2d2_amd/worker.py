@@ -29,7 +29,7 @@ import queue as queue_mod
 import threading
 import time
 from dataclasses import dataclass
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 import numpy as np
 import torch
