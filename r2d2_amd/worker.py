@@ -1003,11 +1003,17 @@ class VectorActor:
         self.h_t[0, i].zero_(); self.c_t[0, i].zero_()
         self.episode_steps[i] = 0
 
+    _started = False
+
     def run(self, stop_after_steps: Optional[int] = None):
         """Drive all envs until stop_after_steps total env steps (None =
-        forever).  Returns total env steps taken."""
-        for i in range(self.E):
-            self._reset_env(i)
+        forever).  Returns total env steps taken.  Re-entrant: repeated
+        calls RESUME the envs and local buffers where the previous call
+        stopped (resetting would discard partially-filled blocks)."""
+        if not self._started:
+            for i in range(self.E):
+                self._reset_env(i)
+            self._started = True
         total_steps = 0
         tick = 0
         use_amp = (self.device.type == "cuda" and self.cfg.dtype == "bf16")
