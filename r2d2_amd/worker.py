@@ -698,11 +698,14 @@ class Learner:
                         block, prio, reward = q.get_nowait()
                     except queue_mod.Empty:
                         continue
-                    with lock:
-                        replay.ingest(block, prio)
-                        if reward is not None:
-                            stats["episode_reward"] += reward
-                            stats["num_episodes"] += 1
+                    try:
+                        with lock:
+                            replay.ingest(block, prio)
+                            if reward is not None:
+                                stats["episode_reward"] += reward
+                                stats["num_episodes"] += 1
+                    except Exception as e:   # keep the feed alive; surface it
+                        print(f"[learner] replay ingest error: {e!r}")
                     got = True
                 if not got:
                     time.sleep(0.005)
