@@ -90,6 +90,15 @@ class GpuReplayBuffer:
         self._sample_stream = torch.cuda.Stream(device=dev)
         self._ev_tree = torch.cuda.Event()
         self._ev_sampled = torch.cuda.Event()
+        # rotating pinned landing buffers for the small per-sample metadata
+        # D2H (a pageable destination would silently make the copy — and
+        # the host — block inside sample_async)
+        B0 = self.batch_size
+        self._pin_meta = [torch.empty(3, B0, dtype=torch.int32,
+                                      pin_memory=True) for _ in range(2)]
+        self._pin_seg = [torch.empty(B0 + 1, dtype=torch.int32,
+                                     pin_memory=True) for _ in range(2)]
+        self._pin_i = 0
 
     # ------------------------------------------------------------------
     def __len__(self):
@@ -197,8 +206,15 @@ class GpuReplayBuffer:
                 self.obs_store, self.la_store, self.lr_store, self.act_store,
                 self.nsr_store, self.gam_store, self.hid_store, idx, meta,
                 seg, weight, self.T, self.A, self.learn_len, self.H, self.spb)
-            meta_h = meta[:3].to("cpu", non_blocking=True)
-            seg_h = seg.to("cpu", non_blocking=True)
+            if B == self.batch_size:
+                self._pin_i ^= 1
+                meta_h = self._pin_meta[self._pin_i]
+                seg_h = self._pin_seg[self._pin_i]
+                meta_h.copy_(meta[:3], non_blocking=True)
+                seg_h.copy_(seg, non_blocking=True)
+            else:
+                meta_h = meta[:3].cpu()
+                seg_h = seg.cpu()
             self._ev_sampled.record(self._sample_stream)
         return (idx, outs, meta_h, seg_h, self.block_ptr, self.env_steps)
 
