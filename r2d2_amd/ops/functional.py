@@ -96,7 +96,10 @@ def mixed_td_priority_np(abs_td: np.ndarray, lengths: np.ndarray,
     (reference: worker.py:268-276, the numpy loop)."""
     out = np.empty(len(lengths), dtype=np.float32)
     ofs = 0
+    # int() per element: uint8 lengths (the Block storage dtype) would
+    # otherwise wrap the running offset at 256 for full 400-step blocks
     for i, L in enumerate(lengths):
+        L = int(L)
         seg = abs_td[ofs: ofs + L]
         out[i] = eta * seg.max() + (1.0 - eta) * seg.mean()
         ofs += L

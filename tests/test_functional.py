@@ -91,3 +91,18 @@ def test_double_q_target():
     expect = Fn.value_rescale(
         r + g * Fn.inverse_value_rescale(qt[torch.arange(B), a_star]))
     assert torch.allclose(out, expect, atol=1e-6)
+
+
+def test_mixed_td_priority_full_block_uint8_lengths():
+    """Regression: uint8 segment lengths from a FULL 400-step block must not
+    wrap the running offset at 256 (found live — the reference stores
+    per-sequence learning steps as uint8, worker.py Block)."""
+    import numpy as np
+    from r2d2_amd.ops import functional as Fn
+
+    lengths = np.full(10, 40, dtype=np.uint8)   # sums to 400 > 255
+    abs_td = np.arange(400, dtype=np.float32)
+    out = Fn.mixed_td_priority_np(abs_td, lengths, eta=0.9)
+    assert out.shape == (10,)
+    # last segment covers [360, 400): max 399, mean 379.5
+    assert abs(out[-1] - (0.9 * 399 + 0.1 * 379.5)) < 1e-3

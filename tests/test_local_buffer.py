@@ -107,3 +107,27 @@ def test_priorities_follow_td_magnitude():
                 np.zeros(3, dtype=np.float32), np.zeros((2, 8), dtype=np.float32))
     _, prios, _ = buf.finish(np.zeros(3, dtype=np.float32))
     np.testing.assert_allclose(prios, 0.0)
+
+
+def test_full_400_step_block_finish():
+    """Regression: a FULL reference-sized block (400 steps, 10 sequences of
+    uint8 learning lengths) finishes without overflow and with valid
+    priorities (found live via the training demo)."""
+    from r2d2_amd import config as cfg
+    cfg.apply("mspacman")
+    c = cfg.get()
+    rng = np.random.default_rng(0)
+    lb = LocalBuffer(c.action_dim)
+    obs = rng.integers(0, 256, size=tuple(c.obs_shape), dtype=np.uint8)
+    lb.reset(obs)
+    for t in range(c.block_length):
+        lb.add(int(rng.integers(c.action_dim)), float(rng.normal()),
+               rng.integers(0, 256, size=tuple(c.obs_shape), dtype=np.uint8),
+               rng.normal(size=c.action_dim).astype(np.float32),
+               np.zeros((2, c.hidden_dim), dtype=np.float32))
+    block, prio, reward = lb.finish(
+        rng.normal(size=c.action_dim).astype(np.float32))
+    assert block.num_sequences == 10
+    assert int(block.learning_steps.sum()) == 400
+    assert np.isfinite(prio).all() and prio.shape == (10,)
+    cfg.apply("mspacman")
