@@ -27,12 +27,17 @@ def epsilon_ladder(num_actors=None, base_eps=None, alpha=None):
     return [base ** (1 + i / (n - 1) * a) for i in range(n)]
 
 
-def _run_actor(epsilon, model, sample_queue, seed):
+def _run_actor(epsilon, model, sample_queue, seed, config_dict=None):
+    if config_dict:   # spawn-context children start with default config
+        cfg.apply(**config_dict)
     actor = Actor(epsilon, model, sample_queue, seed=seed)
     actor.run()
 
 
-def _run_vector_actor(epsilons, model, sample_queues, device, seed):
+def _run_vector_actor(epsilons, model, sample_queues, device, seed,
+                      config_dict=None):
+    if config_dict:
+        cfg.apply(**config_dict)
     from .worker import VectorActor
     va = VectorActor(epsilons, model, sample_queues, device=device, seed=seed)
     va.run()
@@ -49,13 +54,22 @@ def train(seed: int = 0, restart_dead_actors: bool = True):
     torch.set_num_threads(1)
 
     c = cfg.get()
+    # CUDA is initialized in this process (the learner) before actors are
+    # started: forked children cannot re-initialize CUDA, so a GPU
+    # VectorActor child must be SPAWNED (with the live config handed over —
+    # spawned interpreters start from the default preset).
+    use_spawn = c.vector_actors and c.actor_device == "cuda"
+    ctx = mp.get_context("spawn" if use_spawn else "fork")
+    from dataclasses import asdict
+    config_dict = asdict(c) if use_spawn else None
+
     model = Network(c.action_dim, c.obs_shape, c.hidden_dim, encoder=c.encoder,
                     forward_steps=c.forward_steps, mlp_hidden=c.mlp_hidden)
     model.share_memory()
 
-    sample_queues = [mp.Queue() for _ in range(c.num_actors)]
-    batch_queue = mp.Queue(c.batch_queue_size)
-    priority_queue = mp.Queue(c.batch_queue_size)
+    sample_queues = [ctx.Queue() for _ in range(c.num_actors)]
+    batch_queue = ctx.Queue(c.batch_queue_size)
+    priority_queue = ctx.Queue(c.batch_queue_size)
 
     use_gpu_replay = c.gpu_replay and torch.cuda.is_available()
     buffer = None if use_gpu_replay else ReplayBuffer(
@@ -66,15 +80,16 @@ def train(seed: int = 0, restart_dead_actors: bool = True):
 
     if c.vector_actors:
         # one driver process, all envs in lockstep, batched inference
-        spawners = [lambda: mp.Process(
+        spawners = [lambda: ctx.Process(
             target=_run_vector_actor,
             args=(epsilon_ladder(), model, sample_queues, c.actor_device,
-                  seed + 1))]
+                  seed + 1, config_dict))]
     else:
         spawners = [
-            (lambda eps=eps, i=i: mp.Process(
+            (lambda eps=eps, i=i: ctx.Process(
                 target=_run_actor,
-                args=(eps, model, sample_queues[i], seed + 1 + i)))
+                args=(eps, model, sample_queues[i], seed + 1 + i,
+                      config_dict)))
             for i, eps in enumerate(epsilon_ladder())]
     actor_procs = [s() for s in spawners]
     for p in actor_procs:
@@ -82,7 +97,7 @@ def train(seed: int = 0, restart_dead_actors: bool = True):
 
     buffer_proc = None
     if buffer is not None:
-        buffer_proc = mp.Process(target=_run_buffer, args=(buffer,))
+        buffer_proc = ctx.Process(target=_run_buffer, args=(buffer,))
         buffer_proc.start()
 
     # actor supervision: the reference silently loses dead actor processes
