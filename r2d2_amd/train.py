@@ -43,11 +43,18 @@ def _run_vector_actor(epsilons, model, sample_queues, device, seed,
     va.run()
 
 
-def _run_buffer(buffer: ReplayBuffer):
+def _run_buffer(buffer: ReplayBuffer = None, queues=None, config_dict=None):
+    if buffer is None:
+        # spawn context: a ReplayBuffer (threading.Lock inside) cannot be
+        # pickled — build it in the child from the handed-over config
+        cfg.apply(**config_dict)
+        sq, bq, pq = queues
+        buffer = ReplayBuffer(sq, bq, pq)
     buffer.run()
 
 
-def train(seed: int = 0, restart_dead_actors: bool = True):
+def train(seed: int = 0, restart_dead_actors: bool = True,
+          _force_spawn: bool = False):
     torch.manual_seed(seed)
     np.random.seed(seed)
     random.seed(seed)
@@ -58,7 +65,7 @@ def train(seed: int = 0, restart_dead_actors: bool = True):
     # started: forked children cannot re-initialize CUDA, so a GPU
     # VectorActor child must be SPAWNED (with the live config handed over —
     # spawned interpreters start from the default preset).
-    use_spawn = c.vector_actors and c.actor_device == "cuda"
+    use_spawn = (c.vector_actors and c.actor_device == "cuda") or _force_spawn
     ctx = mp.get_context("spawn" if use_spawn else "fork")
     from dataclasses import asdict
     config_dict = asdict(c) if use_spawn else None
@@ -97,7 +104,12 @@ def train(seed: int = 0, restart_dead_actors: bool = True):
 
     buffer_proc = None
     if buffer is not None:
-        buffer_proc = ctx.Process(target=_run_buffer, args=(buffer,))
+        if use_spawn:
+            buffer_proc = ctx.Process(target=_run_buffer, kwargs=dict(
+                queues=(sample_queues, batch_queue, priority_queue),
+                config_dict=config_dict))
+        else:
+            buffer_proc = ctx.Process(target=_run_buffer, args=(buffer,))
         buffer_proc.start()
 
     # actor supervision: the reference silently loses dead actor processes

@@ -31,3 +31,14 @@ def test_train_to_completion(tmp_path, vector, monkeypatch):
     train(seed=0)
     # learner finished all updates; metrics JSONL was emitted
     assert (tmp_path / "metrics.jsonl").exists()
+
+
+@pytest.mark.timeout(300)
+def test_train_spawn_context(tmp_path, monkeypatch):
+    """The spawn-context actor path (used when vector actors run on cuda):
+    spawned children must receive the live config and shared model."""
+    monkeypatch.chdir(tmp_path)
+    tiny_cfg(tmp_path, vector_actors=True)
+    from r2d2_amd.train import train
+    train(seed=0, _force_spawn=True)
+    assert (tmp_path / "metrics.jsonl").exists()
