@@ -230,13 +230,12 @@ class HipNetworkEngine:
         return latent, (a1, a2, a3, flat)
 
     def _lstm_input(self, pack, latent, last_action, last_reward):
-        """Build padded rin and the input-GEMM X (Mrows, 4H)."""
+        """Build padded rin (one fused kernel) and the input-GEMM X."""
         M = latent.shape[0]
-        rin = torch.zeros(M, pack.kin_pad, device=self.device,
-                          dtype=torch.bfloat16)
-        rin[:, :512] = latent
-        rin[:, 512:512 + self.A] = last_action.reshape(M, self.A).bfloat16()
-        rin[:, 512 + self.A] = last_reward.reshape(M).bfloat16()
+        rin = self.m.assemble_rin(
+            latent.contiguous(),
+            last_action.reshape(M, self.A).float().contiguous(),
+            last_reward.reshape(M).float().contiguous(), pack.kin_pad)
         X = self.m.gemm_bias_act(rin, pack.wih_t, pack.lstm_bias, 0, False)
         return rin, X
 
@@ -404,8 +403,8 @@ class HipNetworkEngine:
                           net.recurrent.weight_ih_l0.grad,
                           net.recurrent.bias_ih_l0.grad)
         net.recurrent.bias_hh_l0.grad.copy_(net.recurrent.bias_ih_l0.grad)
-        drin = m.gemm_dgrad(dgates_flat, self._empty, ON.wih_kn, False)
-        dlat = drin[:, :512].contiguous()
+        # only the latent slice of the padded LSTM-input grad is needed
+        dlat = m.gemm_dgrad(dgates_flat, self._empty, ON.wih_kn, False, 512)
         self._mark("lstm_wgrads")
 
         M = B * T
@@ -509,11 +508,10 @@ class HipInference:
             a3 = m.conv_fwd(a2, p.w3t, p.b3, 3, E, 9, 9, 7, 7, True)
             lat = m.gemm_bias_act(a3.view(E, 3136), p.wft, p.bf, 1, False)
         p = self.pack
-        rin = torch.zeros(E, p.kin_pad, device=self.device,
-                          dtype=torch.bfloat16)
-        rin[:, :512] = lat
-        rin[:, 512:512 + self.A] = last_action.bfloat16()
-        rin[:, 512 + self.A] = last_reward.reshape(E).bfloat16()
+        rin = m.assemble_rin(lat.contiguous(),
+                             last_action.float().contiguous(),
+                             last_reward.reshape(E).float().contiguous(),
+                             p.kin_pad)
         x = m.gemm_bias_act(rin, p.wih_t, p.lstm_bias, 0, True)     # f32
         h0, c0 = hidden
         hg = m.gemm_bias_act(h0.reshape(E, 512).bfloat16().contiguous(),

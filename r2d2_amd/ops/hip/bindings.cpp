@@ -35,7 +35,7 @@ std::vector<torch::Tensor> replay_gather_batch(
 torch::Tensor gemm_bias_act(torch::Tensor A, torch::Tensor Wt,
                             torch::Tensor bias, int64_t act, bool out_f32);
 torch::Tensor gemm_dgrad(torch::Tensor dY, torch::Tensor act_out,
-                         torch::Tensor W, bool relu_mask);
+                         torch::Tensor W, bool relu_mask, int64_t k_out);
 std::vector<torch::Tensor> gemm_wgrad(torch::Tensor dY, torch::Tensor act_out,
                                       torch::Tensor A, bool relu_mask,
                                       bool want_bias);
@@ -59,6 +59,8 @@ std::vector<torch::Tensor> conv_wgrad(torch::Tensor dY, torch::Tensor act,
                                       int64_t K);
 
 // lstm_kernels.hip
+torch::Tensor assemble_rin(torch::Tensor latent, torch::Tensor la,
+                           torch::Tensor lr, int64_t kin_pad);
 void barrier_bench(torch::Tensor barrier_ws, int64_t steps, int64_t nblocks);
 void handoff_bench(torch::Tensor barrier_ws, int64_t steps, int64_t nblocks);
 void handoff_counter_bench(torch::Tensor barrier_ws, int64_t steps,
@@ -117,7 +119,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("replay_gather_batch", &replay_gather_batch,
           "On-device padded batch assembly from the GPU block store");
     m.def("gemm_bias_act", &gemm_bias_act, "MFMA GEMM + bias + activation");
-    m.def("gemm_dgrad", &gemm_dgrad, "MFMA GEMM backward-data (+ fused ReLU mask)");
+    m.def("gemm_dgrad", &gemm_dgrad,
+          "MFMA GEMM backward-data (+ fused ReLU mask; optional dense "
+          "k_out-column output)",
+          pybind11::arg("dY"), pybind11::arg("act_out"), pybind11::arg("W"),
+          pybind11::arg("relu_mask"), pybind11::arg("k_out") = 0);
     m.def("gemm_wgrad", &gemm_wgrad, "MFMA GEMM backward-weight (+ bias grad)");
     m.def("gemm_wgrad_into", &gemm_wgrad_into,
           "GEMM backward-weight accumulated straight into .grad views");
@@ -127,6 +133,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("lstm_fwd", &lstm_fwd,
           "Persistent fused LSTM forward (dual-network, length-masked)");
     m.def("lstm_bwd", &lstm_bwd, "Persistent fused LSTM BPTT backward");
+    m.def("assemble_rin", &assemble_rin,
+          "fused LSTM-input row assembly (latent | action | reward | pad)");
     m.def("barrier_bench", &barrier_bench, "grid barrier microbench");
     m.def("handoff_bench", &handoff_bench, "producer-flag handoff microbench");
     m.def("handoff_counter_bench", &handoff_counter_bench,

@@ -115,7 +115,7 @@ __global__ __launch_bounds__(256) void gemm_dgrad_kernel(
     const __hip_bfloat16* __restrict__ act,     // (M, N) or null
     const __hip_bfloat16* __restrict__ W,       // (K, N) row-major
     __hip_bfloat16* __restrict__ dA,            // (M, K)
-    int M, int N, int K) {
+    int M, int N, int K, int Kout) {
     int wave = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
     int wr = wave >> 1, wc = wave & 1;
@@ -163,8 +163,8 @@ __global__ __launch_bounds__(256) void gemm_dgrad_kernel(
             for (int r = 0; r < 4; ++r) {
                 long rr = row0 + i * 16 + crow + r;
                 long cc = col0 + j * 16 + ccol;
-                if (rr < M && cc < K)
-                    dA[rr * K + cc] = f2bf(acc[i][j][r]);
+                if (rr < M && cc < Kout)
+                    dA[rr * Kout + cc] = f2bf(acc[i][j][r]);
             }
 }
 
@@ -339,13 +339,17 @@ torch::Tensor gemm_bias_act(torch::Tensor A, torch::Tensor Wt,
 }
 
 torch::Tensor gemm_dgrad(torch::Tensor dY, torch::Tensor act_out,
-                         torch::Tensor W, bool relu_mask) {
+                         torch::Tensor W, bool relu_mask, int64_t k_out) {
     TORCH_CHECK(dY.is_cuda() && dY.dtype() == torch::kBFloat16 && dY.is_contiguous());
     TORCH_CHECK(W.dtype() == torch::kBFloat16 && W.is_contiguous());
     long M = dY.size(0), N = dY.size(1), K = W.size(0);
     TORCH_CHECK(W.size(1) == N && N % 32 == 0, "N must be a multiple of 32");
-    auto dA = torch::empty({M, K}, dY.options());
-    dim3 grid(cdiv(M, 64), cdiv(K, 64));
+    // k_out: emit only the first k_out input-feature columns as a DENSE
+    // (M, k_out) tensor (e.g. the latent slice of the padded LSTM input)
+    long Kout = (k_out > 0) ? k_out : K;
+    TORCH_CHECK(Kout <= K);
+    auto dA = torch::empty({M, Kout}, dY.options());
+    dim3 grid(cdiv(M, 64), cdiv(Kout, 64));
     auto stream = at::cuda::getCurrentCUDAStream();
     auto* dy = reinterpret_cast<const __hip_bfloat16*>(dY.data_ptr());
     auto* ac = relu_mask
@@ -354,10 +358,12 @@ torch::Tensor gemm_dgrad(torch::Tensor dY, torch::Tensor act_out,
     auto* da = reinterpret_cast<__hip_bfloat16*>(dA.data_ptr());
     if (relu_mask)
         hipLaunchKernelGGL((gemm_dgrad_kernel<true>), grid, dim3(256), 0,
-                           stream.stream(), dy, ac, w, da, (int)M, (int)N, (int)K);
+                           stream.stream(), dy, ac, w, da, (int)M, (int)N,
+                           (int)K, (int)Kout);
     else
         hipLaunchKernelGGL((gemm_dgrad_kernel<false>), grid, dim3(256), 0,
-                           stream.stream(), dy, ac, w, da, (int)M, (int)N, (int)K);
+                           stream.stream(), dy, ac, w, da, (int)M, (int)N,
+                           (int)K, (int)Kout);
     return dA;
 }
 

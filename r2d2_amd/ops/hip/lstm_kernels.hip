@@ -597,6 +597,39 @@ __global__ void handoff_counter_bench_kernel(GridBar* bar, int steps,
     }
 }
 
+// ---------------------------------------------------------------------------
+// assemble_rin: build the padded LSTM input rows (latent | one-hot action |
+// reward | zero pad) in ONE launch (replaces a zeros fill + three slice
+// copies per network per step).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void assemble_rin_kernel(
+    const __hip_bfloat16* __restrict__ latent,  // (M, 512)
+    const float* __restrict__ la,               // (M, A)
+    const float* __restrict__ lr,               // (M,)
+    __hip_bfloat16* __restrict__ rin,           // (M, KP)
+    int M, int A, int KP) {
+    unsigned idx = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
+    if (idx >= (unsigned)M * KP) return;
+    int k = (int)(idx % (unsigned)KP);
+    unsigned m = idx / (unsigned)KP;
+    if (k + 8 <= 512) {
+        *reinterpret_cast<bf16x8*>(rin + idx) =
+            lload8(latent + (long)m * 512 + k);
+        return;
+    }
+    bf16x8 v;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+        int kk = k + e;
+        float f = 0.f;
+        if (kk < 512) f = bf2f(latent[(long)m * 512 + kk]);
+        else if (kk < 512 + A) f = la[(long)m * A + (kk - 512)];
+        else if (kk == 512 + A) f = lr[m];
+        v[e] = (__bf16)f;
+    }
+    *reinterpret_cast<bf16x8*>(rin + idx) = v;
+}
+
 // barrier-only microbench
 __global__ void barrier_bench_kernel(GridBar* bar, int steps, int nblocks) {
     for (int t = 0; t < steps; ++t)
@@ -625,6 +658,26 @@ static void zero_ws(torch::Tensor& ws, hipStream_t stream) {
     TORCH_CHECK(ws.numel() * ws.element_size() >= (long)sizeof(GridBar),
                 "barrier workspace too small (need >= 256 int32)");
     hipMemsetAsync(ws.data_ptr(), 0, sizeof(GridBar), stream);
+}
+
+torch::Tensor assemble_rin(torch::Tensor latent, torch::Tensor la,
+                           torch::Tensor lr, int64_t kin_pad) {
+    long M = latent.size(0);
+    long A = la.size(1);
+    TORCH_CHECK(latent.dtype() == torch::kBFloat16 && latent.is_contiguous());
+    TORCH_CHECK(la.dtype() == torch::kFloat32 && la.is_contiguous());
+    TORCH_CHECK(kin_pad % 32 == 0 && 512 + A + 1 <= kin_pad);
+    auto rin = torch::empty({M, kin_pad}, latent.options());
+    long total = M * kin_pad / 8;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(assemble_rin_kernel,
+                       dim3((int)((total + 255) / 256)), dim3(256), 0,
+                       stream.stream(),
+                       reinterpret_cast<const __hip_bfloat16*>(latent.data_ptr()),
+                       la.data_ptr<float>(), lr.data_ptr<float>(),
+                       reinterpret_cast<__hip_bfloat16*>(rin.data_ptr()),
+                       (int)M, (int)A, (int)kin_pad);
+    return rin;
 }
 
 void barrier_bench(torch::Tensor barrier_ws, int64_t steps, int64_t nblocks) {
