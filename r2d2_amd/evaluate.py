@@ -160,5 +160,21 @@ def _maybe_plot(results, game_name, out_dir):
     plt.close(fig)
 
 
+def main():
+    import argparse
+    ap = argparse.ArgumentParser(description="Evaluate saved checkpoints "
+                                 "(reference test.py harness)")
+    ap.add_argument("--game", type=str, default=None)
+    ap.add_argument("--preset", type=str, default=None)
+    ap.add_argument("--model-dir", type=str, default="models")
+    ap.add_argument("--episodes", type=int, default=5)
+    ap.add_argument("--out-dir", type=str, default=None)
+    args = ap.parse_args()
+    if args.preset:
+        cfg.apply(args.preset)
+    test(game_name=args.game, model_dir=args.model_dir,
+         num_episodes=args.episodes, out_dir=args.out_dir)
+
+
 if __name__ == "__main__":
-    test()
+    main()
