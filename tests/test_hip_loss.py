@@ -95,3 +95,29 @@ def test_extreme_values_value_rescale():
     expect_target = Fn.value_rescale(Fn.inverse_value_rescale(x))
     assert torch.allclose(expect_target, x, atol=1e-2, rtol=1e-4)
     assert torch.isfinite(loss).all() and torch.isfinite(prio).all()
+
+
+@pytest.mark.timeout(240)
+def test_learner_hybrid_path_unsupported_config():
+    """GPU configs outside the full engine's coverage (e.g. hidden_dim 256)
+    run eager forward + the fused loss/priority kernels (Learner.train_step
+    hybrid branch) — loss finite, priorities on-device."""
+    import numpy as np
+    from r2d2_amd import config as cfg
+    from r2d2_amd.models.network import Network
+    from r2d2_amd.worker import Learner
+    from bench import build_batch
+
+    c = cfg.apply("mspacman", hidden_dim=256, batch_size=8, burn_in_steps=8,
+                  learning_steps=8, forward_steps=3)
+    torch.manual_seed(0)
+    model = Network(c.action_dim, c.obs_shape, c.hidden_dim, encoder="nature",
+                    forward_steps=c.forward_steps)
+    learner = Learner(None, None, model)
+    learner.enable_hip_engine()
+    assert learner.hip_engine and learner.engine is None
+    batch = build_batch(c, torch.device("cuda:0"), seed=5)
+    loss, prio = learner.train_step(batch)
+    assert np.isfinite(float(loss))
+    assert torch.is_tensor(prio) and prio.is_cuda and torch.isfinite(prio).all()
+    cfg.apply("mspacman")
