@@ -17,7 +17,8 @@ void sumtree_update(torch::Tensor tree, int64_t leaf_offset, torch::Tensor idxes
                     int64_t cur_ptr, int64_t seq_per_block, int64_t num_blocks);
 std::vector<torch::Tensor> sumtree_sample(torch::Tensor tree, int64_t leaf_offset,
                                           int64_t num_levels, torch::Tensor jitter,
-                                          int64_t n, double beta);
+                                          int64_t n, double beta,
+                                          int64_t max_idx);
 
 // replay_gather.hip
 std::vector<torch::Tensor> replay_gather_meta(

@@ -63,7 +63,7 @@ __global__ void sumtree_sample_kernel(
     long* __restrict__ out_idx,         // (n,)
     float* __restrict__ out_prio,       // (n,)
     float* __restrict__ out_weight,     // (n,)
-    int n, float beta) {
+    int n, float beta, long max_idx) {
     int i = threadIdx.x;
     __shared__ double warp_min[16];
     double prio = 0.0;
@@ -83,7 +83,10 @@ __global__ void sumtree_sample_kernel(
             }
         }
         idx = node - leaf_offset;
-        prio = tree[node];
+        // fp-edge guard: a rounding overshoot in the subtract path could
+        // land in the zero-padded leaves past the last real sequence
+        if (idx > max_idx) idx = max_idx;
+        prio = tree[leaf_offset + idx];
         out_idx[i] = idx;
         out_prio[i] = (float)prio;
     }
@@ -132,7 +135,8 @@ void sumtree_update(torch::Tensor tree, int64_t leaf_offset, torch::Tensor idxes
 
 std::vector<torch::Tensor> sumtree_sample(torch::Tensor tree, int64_t leaf_offset,
                                           int64_t num_levels, torch::Tensor jitter,
-                                          int64_t n, double beta) {
+                                          int64_t n, double beta,
+                                          int64_t max_idx) {
     TORCH_CHECK(tree.is_cuda() && tree.dtype() == torch::kFloat64);
     TORCH_CHECK(n <= 1024, "sample batch must be <= 1024");
     auto opts_l = tree.options().dtype(torch::kInt64);
@@ -147,6 +151,7 @@ std::vector<torch::Tensor> sumtree_sample(torch::Tensor tree, int64_t leaf_offse
                        stream.stream(), tree.data_ptr<double>(), leaf_offset,
                        (int)num_levels, jitter.data_ptr<float>(),
                        idx.data_ptr<long>(), prio.data_ptr<float>(),
-                       weight.data_ptr<float>(), (int)n, (float)beta);
+                       weight.data_ptr<float>(), (int)n, (float)beta,
+                       (long)max_idx);
     return {idx, prio, weight};
 }

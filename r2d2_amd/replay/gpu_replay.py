@@ -198,7 +198,7 @@ class GpuReplayBuffer:
             jitter = torch.rand(B, device=dev)
             idx, prio, weight = self._ext.sumtree_sample(
                 self.tree, self.leaf_offset, self.num_levels, jitter, B,
-                self.beta)
+                self.beta, self.num_sequences - 1)
             meta, seg = self._ext.replay_gather_meta(
                 idx, self.burn_s, self.learn_s, self.fwd_s, self.obs_start_s,
                 self.learn_off_s, self.spb)

@@ -56,7 +56,8 @@ def test_sumtree_update_and_sample_match_cpu():
     # sampling with known jitter reproduces the CPU descent exactly
     jitter = torch.rand(64, device="cuda")
     out_idx, out_prio, out_w = m.sumtree_sample(tree, leaf_offset, levels,
-                                                jitter, 64, 0.6)
+                                                jitter, 64, 0.6,
+                                                (1 << levels) - 1)
     torch.cuda.synchronize()
     total = tree_np[0]
     j = jitter.cpu().numpy().astype(np.float64)
