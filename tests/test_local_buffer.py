@@ -131,3 +131,69 @@ def test_full_400_step_block_finish():
     assert int(block.learning_steps.sum()) == 400
     assert np.isfinite(prio).all() and prio.shape == (10,)
     cfg.apply("mspacman")
+
+
+# ---------------------------------------------------------------------------
+# property-based invariants (hypothesis): random episode/block interleavings
+# ---------------------------------------------------------------------------
+
+from hypothesis import given, settings, strategies as st
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    n=st.integers(1, 4),
+    burn=st.integers(0, 6),
+    learn=st.integers(1, 6),
+    spb=st.integers(1, 4),
+    cuts=st.lists(st.integers(1, 30), min_size=1, max_size=4),
+    done_last=st.booleans(),
+    seed=st.integers(0, 10_000),
+)
+def test_localbuffer_invariants_random(n, burn, learn, spb, cuts, done_last,
+                                       seed):
+    """For arbitrary (n-step, burn-in, learning, block) geometry and random
+    mid-episode block cuts: every finished block satisfies the structural
+    invariants the replay/gather stack relies on."""
+    rng = np.random.default_rng(seed)
+    block_len = learn * spb
+    A, H = 3, 4
+    buf = LocalBuffer(A, forward_steps=n, burn_in_steps=burn,
+                      learning_steps=learn, gamma=0.9, hidden_dim=H,
+                      block_length=block_len)
+    buf.reset(np.zeros((2,), dtype=np.uint8))
+    carried = 0
+    for ci, steps in enumerate(cuts):
+        steps = min(steps, block_len)
+        for t in range(steps):
+            buf.add(int(rng.integers(A)), float(rng.normal()),
+                    rng.integers(0, 255, size=(2,)).astype(np.uint8),
+                    rng.normal(size=A).astype(np.float32),
+                    rng.normal(size=(2, H)).astype(np.float32))
+        last = ci == len(cuts) - 1
+        done = last and done_last
+        block, prios, reward = buf.finish(
+            None if done else rng.normal(size=A).astype(np.float32))
+        S = len(block.action)
+        assert S == steps
+        # obs rows = carried burn-in + steps + 1
+        assert block.obs.shape[0] == block.burn_in_steps[0] + S + 1
+        assert block.burn_in_steps[0] == carried
+        # sequence layout: learning sums to S, forward ends at the block end
+        assert int(block.learning_steps[:block.num_sequences].sum()) == S
+        assert block.forward_steps[block.num_sequences - 1] == 1
+        assert (block.forward_steps[:block.num_sequences] >= 1).all()
+        assert (block.forward_steps[:block.num_sequences] <= n).all()
+        # gamma: zero only at terminal steps
+        g = block.gamma
+        if done:
+            assert g[-1] == 0.0
+        else:
+            assert (g > 0).all()
+        assert np.isfinite(prios).all() and (prios >= 0).all()
+        assert prios.shape == (spb,)
+        # priorities beyond num_sequences are zero (killed tree slots)
+        assert (prios[block.num_sequences:] == 0).all()
+        # burn-in carry for the next block
+        carried = min(burn, carried + S)
+        assert reward is None or done
