@@ -66,3 +66,37 @@ def test_update_overwrite():
     assert abs(tree.total - (15 + 100)) < 1e-9
     idx, _ = tree.sample(64)
     assert (idx == 3).mean() > 0.5
+
+
+def test_property_random_ops_keep_tree_consistent():
+    """hypothesis: any interleaving of updates/samples keeps root == sum of
+    leaves and samples in-range (SURVEY §4 sum-tree property tests)."""
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=30, deadline=None)
+    @given(cap=st.integers(2, 200),
+           ops=st.lists(st.tuples(st.integers(0, 1), st.integers(0, 10_000)),
+                        min_size=1, max_size=20),
+           seed=st.integers(0, 99999))
+    def run(cap, ops, seed):
+        rng = np.random.default_rng(seed)
+        tree = PriorityTree(cap, prio_exponent=0.9, is_exponent=0.6,
+                            rng=np.random.default_rng(seed + 1))
+        touched = False
+        for kind, s in ops:
+            if kind == 0:
+                k = int(rng.integers(1, min(cap, 16) + 1))
+                idx = rng.choice(cap, size=k, replace=False)
+                td = rng.random(k).astype(np.float64) + 1e-3
+                tree.update(idx, td)
+                touched = True
+            elif touched:
+                n = int(rng.integers(1, 17))
+                idxes, w = tree.sample(n)
+                assert ((idxes >= 0) & (idxes < cap)).all()
+                assert np.isfinite(w).all() and (w > 0).all() and (w <= 1 + 1e-9).all()
+            leaves = tree.levels[-1][:tree.num_leaves]
+            np.testing.assert_allclose(tree.levels[0][0], leaves.sum(),
+                                       rtol=1e-9)
+
+    run()
