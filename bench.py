@@ -98,9 +98,15 @@ def main():
                     help="auto: HIP kernels on GPU when built, eager otherwise")
     ap.add_argument("--batches", type=int, default=4,
                     help="distinct synthetic batches rotated through")
-    ap.add_argument("--gpu-replay", action="store_true",
+    ap.add_argument("--gpu-replay", dest="gpu_replay", action="store_true",
+                    default=None,
                     help="sample each step from the GPU-resident prioritized "
-                         "replay (sum-tree + on-device gather in the timed loop)")
+                         "replay (sum-tree + on-device gather in the timed "
+                         "loop).  DEFAULT on GPU with the HIP engine — the "
+                         "headline number times the full learner loop "
+                         "(sample + gather + train + priority update)")
+    ap.add_argument("--no-gpu-replay", dest="gpu_replay", action="store_false",
+                    help="time prebuilt device batches instead")
     ap.add_argument("--replay-transitions", type=int, default=160_000,
                     help="synthetic transitions pre-filled into the GPU replay")
     args = ap.parse_args()
@@ -140,7 +146,11 @@ def main():
         except AttributeError:
             pass  # engine not built yet; eager path
     replay = None
-    if args.gpu_replay and have_cuda:
+    use_replay = args.gpu_replay
+    if use_replay is None:   # default: full loop when the HIP engine runs
+        use_replay = bool(have_cuda and learner.hip_engine
+                          and len(c.obs_shape) == 3)
+    if use_replay and have_cuda:
         assert learner.hip_engine, "--gpu-replay needs the HIP engine"
         from r2d2_amd.replay.gpu_replay import GpuReplayBuffer
         replay = GpuReplayBuffer(device=device,

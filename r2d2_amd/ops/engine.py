@@ -163,6 +163,12 @@ class HipNetworkEngine:
         self.target_net = target_net
         self.bar = torch.zeros(512, dtype=torch.int32, device=self.device)
         self._empty = torch.Tensor()
+        # persistent-LSTM watchdog: the kernels poison GridBar.poison (int32
+        # word 256 of `bar`) on a bounded-spin timeout and return with
+        # partially written outputs.  The host snapshots that word right
+        # after each launch (async, no sync) and raises at the NEXT step —
+        # a poisoned run must never silently corrupt gradients.
+        self._poison_pin = torch.zeros(2, dtype=torch.int32, pin_memory=True)
 
         # flat parameter/grad/Adam-moment buffers: module params become views
         # so the whole network is clipped + stepped by TWO kernels
@@ -295,6 +301,11 @@ class HipNetworkEngine:
         m = self.m
         c = self.cfg
         dev = self.device
+        if bool(self._poison_pin.any()):
+            raise RuntimeError(
+                "persistent LSTM kernel poisoned on a previous step "
+                "(bounded-spin timeout) — outputs/gradients of that step "
+                "are invalid; aborting instead of training on them")
         B, T = batch.obs.shape[:2]
         A, H = self.A, self.H
 
@@ -330,6 +341,7 @@ class HipNetworkEngine:
         Ho, Co, Ht, Ct, stash = m.lstm_fwd(
             Xo, Xt, self.online.whh_t, self.target.whh_t,
             init, init, lens, self.bar, True)
+        self._poison_pin[0:1].copy_(self.bar[256:257], non_blocking=True)
         self._mark("lstm_fwd")
 
         Ho_flat = Ho.view(-1, H)
@@ -393,6 +405,7 @@ class HipNetworkEngine:
 
         dgates = m.lstm_bwd(stash, Co, Ho, dHext.contiguous(), ON.whh_bwd,
                             lens, self.bar)
+        self._poison_pin[1:2].copy_(self.bar[256:257], non_blocking=True)
         self._mark("lstm_bwd")
         dgates_flat = dgates.view(B * T, 4 * H)
 

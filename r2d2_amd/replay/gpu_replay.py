@@ -70,6 +70,7 @@ class GpuReplayBuffer:
                                 dtype=torch.float64, device=dev)
 
         self.block_ptr = 0
+        self.blocks_written = 0   # min(total ingested, num_blocks)
         self.size = 0
         self.env_steps = 0
         self._ext = hip_ops.ext(required=True)
@@ -183,6 +184,7 @@ class GpuReplayBuffer:
         if self.size > self.capacity:
             self.size = min(self.size, self.capacity)
         self.env_steps += int(learn.sum())
+        self.blocks_written = min(self.blocks_written + 1, self.num_blocks)
         self.block_ptr = (self.block_ptr + 1) % self.num_blocks
 
     # ------------------------------------------------------------------
@@ -196,9 +198,15 @@ class GpuReplayBuffer:
         with torch.cuda.stream(self._sample_stream):
             self._sample_stream.wait_event(self._ev_tree)
             jitter = torch.rand(B, device=dev)
+            # fp-edge descent overshoot is clamped to the highest WRITTEN
+            # leaf, not tree capacity — a partially filled ring must never
+            # surface an unwritten zero-priority leaf (its floored priority
+            # would explode the IS weight)
+            max_leaf = min(self.blocks_written * self.spb,
+                           self.num_sequences) - 1
             idx, prio, weight = self._ext.sumtree_sample(
                 self.tree, self.leaf_offset, self.num_levels, jitter, B,
-                self.beta, self.num_sequences - 1)
+                self.beta, max_leaf)
             meta, seg = self._ext.replay_gather_meta(
                 idx, self.burn_s, self.learn_s, self.fwd_s, self.obs_start_s,
                 self.learn_off_s, self.spb)

@@ -968,7 +968,12 @@ class VectorActor:
         self.queues = sample_queues
         self.buffers = [LocalBuffer(self.action_dim) for _ in range(E)]
         self.rng = np.random.default_rng(seed)
-        self.update_interval = max(1, c.actor_update_interval // E)
+        # weight-pull cadence measured in TOTAL env-steps across the vector
+        # (the reference pulls every actor_update_interval per-actor steps,
+        # worker.py:560; per-tick pulling at E=256 was 1.6x too frequent and
+        # paid a full load_state_dict + prepack refresh every tick)
+        self.update_interval = c.actor_update_interval
+        self._steps_since_pull = 0
         self.max_episode_steps = c.max_episode_steps
         self.block_length = c.block_length
         H = c.hidden_dim
@@ -982,7 +987,11 @@ class VectorActor:
         self.c_t = torch.zeros(1, E, H, device=self.device)
         self._obs_host = np.zeros((E,) + tuple(c.obs_shape), dtype=np.uint8)
         self.episode_steps = np.zeros(E, dtype=np.int64)
-        self._pending_finish: List[int] = []
+        # (env_idx, need_reset): blocks finished one tick late with the next
+        # forward's bootstrap q; need_reset marks episode truncation at
+        # max_episode_steps (finish, then reset like the reference's loop
+        # exit at worker.py:526)
+        self._pending_finish: List[tuple] = []
         # K15 fast path: batched single-step inference through the gfx950
         # kernels when running on a GPU with the extension built
         self.hip_inf = None
@@ -1040,25 +1049,38 @@ class VectorActor:
             q_cpu = q.cpu().numpy()                       # (E, A)
             hid_cpu = torch.stack((h[0], c[0]), dim=1).float().cpu().numpy()
 
-            for i in self._pending_finish:                # late block cuts
-                self._queue(i).put(self.buffers[i].finish(q_cpu[i]))
+            skip = set()       # envs reset this tick: idle for one step
+            for i, need_reset in self._pending_finish:    # late block cuts
+                data = self.buffers[i].finish(q_cpu[i])
+                if self.epsilons[i] > 0.01:
+                    data[2] = None   # non-done cut: no return report
+                self._queue(i).put(data)
+                if need_reset:       # truncation at max_episode_steps
+                    self._reset_env(i)
+                    skip.add(i)
             self._pending_finish.clear()
 
             greedy = q_cpu.argmax(axis=1)
             explore = self.rng.random(self.E) < self.epsilons
             rand_a = self.rng.integers(0, self.action_dim, self.E)
             actions = np.where(explore, rand_a, greedy)
+            if skip:
+                # reset envs don't step this tick; action slot 0 matches the
+                # LocalBuffer.reset() initial one-hot
+                actions[list(skip)] = 0
 
             rewards = np.zeros(self.E, dtype=np.float32)
             dones = np.zeros(self.E, dtype=bool)
             for i, env in enumerate(self.envs):
+                if i in skip:
+                    continue
                 a = int(actions[i])
                 next_obs, r, done, _ = env.step(a)
                 self.buffers[i].add(a, float(r), next_obs, q_cpu[i], hid_cpu[i])
                 self._obs_host[i] = next_obs
                 rewards[i] = r
                 dones[i] = done
-            total_steps += self.E
+            total_steps += self.E - len(skip)
             self.episode_steps += 1
             tick += 1
 
@@ -1069,20 +1091,33 @@ class VectorActor:
             self.la_t[np.arange(self.E), actions] = 1.0
             self.lr_t.copy_(torch.from_numpy(rewards).unsqueeze(1))
             self.h_t, self.c_t = h.detach(), c.detach()
+            for i in skip:   # _reset_env ran before the h/c advance: re-zero
+                self.h_t[0, i].zero_()
+                self.c_t[0, i].zero_()
+                self.episode_steps[i] = 0
 
             for i in range(self.E):
+                if i in skip:
+                    continue
                 if dones[i]:
-                    data = self.buffers[i].finish()
-                    if self.epsilons[i] > 0.01:
-                        data[2] = None   # only near-greedy actors report
-                    self._queue(i).put(data)
+                    # episode end: report the return from every actor
+                    # (the reference filters only the non-done block cut,
+                    # worker.py:555-557)
+                    self._queue(i).put(self.buffers[i].finish())
                     self._reset_env(i)
                 elif (len(self.buffers[i]) == self.block_length
                       or self.episode_steps[i] >= self.max_episode_steps):
-                    self._pending_finish.append(i)
+                    self._pending_finish.append(
+                        (i, self.episode_steps[i] >= self.max_episode_steps))
 
-            if tick % self.update_interval == 0:
-                self.model.load_state_dict(self.shared_model.state_dict())
-                if self.hip_inf is not None:
-                    self.hip_inf.refresh()
+            self._steps_since_pull += self.E - len(skip)
+            if self._steps_since_pull >= self.update_interval:
+                self._steps_since_pull = 0
+                self.pull_weights()
         return total_steps
+
+    def pull_weights(self):
+        """Refresh inference weights from the learner's published copy."""
+        self.model.load_state_dict(self.shared_model.state_dict())
+        if self.hip_inf is not None:
+            self.hip_inf.refresh()

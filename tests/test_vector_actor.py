@@ -81,3 +81,33 @@ def test_vector_actor_block_cut_matches_single_actor():
             got_full = True
             assert block.num_sequences == c.block_length // c.learning_steps
     assert got_full
+
+
+@pytest.mark.timeout(180)
+def test_vector_actor_truncates_at_max_episode_steps():
+    """Hitting max_episode_steps must finish the block once and RESET the
+    env (the reference ends the episode at the cap, worker.py:526) — not
+    re-queue a 1-step block every subsequent tick."""
+    c = setup(block_length=40, learning_steps=4, burn_in_steps=4,
+              forward_steps=2, max_episode_steps=12)
+    torch.manual_seed(0)
+    model = Network(c.action_dim, c.obs_shape, c.hidden_dim, encoder="mlp",
+                    forward_steps=c.forward_steps, mlp_hidden=c.mlp_hidden)
+    sq = queue.Queue()
+    va = VectorActor([0.3, 0.3], model, [sq], device="cpu", seed=3)
+    ticks = 200
+    va.run(stop_after_steps=2 * ticks - 20)
+    assert (va.episode_steps <= c.max_episode_steps).all()
+    n_blocks, sizes = 0, []
+    while not sq.empty():
+        block, prio, reward = sq.get()
+        S = len(block.action)
+        assert 1 <= S <= c.max_episode_steps
+        sizes.append(S)
+        n_blocks += 1
+    # every env produces at most ~1 block per episode (<= cap steps), so the
+    # block count is bounded by 2 envs * ticks/episode_len episodes (+ slack);
+    # the old bug produced ~1 one-step block per tick per truncated env
+    assert n_blocks <= 2 * (ticks // 10) + 8
+    # truncation must not dominate with degenerate 1-step blocks
+    assert np.mean(sizes) > 3
