@@ -182,12 +182,28 @@ class HipNetworkEngine:
         self.exp_avg_sq = torch.zeros(n_total, dtype=torch.float32, device=dev)
         self.norm_buf = torch.zeros(1, dtype=torch.float32, device=dev)
         self.adam_t = 0
+        # contiguous flat-grad ranges per backward stage, in COMPLETION
+        # order (heads -> lstm -> encoder).  The DP learner launches each
+        # segment's RCCL all-reduce as soon as the manual backward finishes
+        # filling it, overlapping communication with the remaining backward
+        # (the comm runs on RCCL's stream; later compute keeps the default
+        # stream busy).
+        named = list(online_net.named_parameters())
+        seg_of = lambda name: ("heads" if name.split(".")[0]
+                               in ("advantage", "value")
+                               else "lstm" if name.startswith("recurrent")
+                               else "encoder")
+        self.seg_ranges = {}
         ofs = 0
-        for p in params:
+        for (name, _), p in zip(named, params):
             n = p.numel()
             self.flat_param[ofs:ofs + n].copy_(p.data.to(dev).view(-1))
             p.data = self.flat_param[ofs:ofs + n].view(p.shape)
             p.grad = self.flat_grad[ofs:ofs + n].view(p.shape)
+            s = seg_of(name)
+            lo, hi = self.seg_ranges.get(s, (ofs, ofs))
+            assert hi == ofs, "segment params must be flat-contiguous"
+            self.seg_ranges[s] = (min(lo, ofs), ofs + n)
             ofs += n
         self.timing = bool(os.environ.get("R2D2_ENGINE_TIMING"))
         self._events = []
@@ -285,19 +301,26 @@ class HipNetworkEngine:
         lbt = b_of * T + t_learn
         lp_t = torch.from_numpy(lp).to(dev)
         tp_t = torch.from_numpy(tp).to(dev)
-        lbt_t = torch.from_numpy(lbt).to(dev)
-        tgt_bt = tp_t - (tp_t // (T + 1)) - 1   # b*(T+1)+t+1 -> b*T + t
+        # inverse map for the dHext scatter kernel: (b*T + t) -> learn row
+        # index, -1 where no learning position sits (learn t's are unique
+        # per sample, so the map is collision-free)
+        row_of = np.full(len(le) * T, -1, dtype=np.int32)
+        row_of[lbt] = np.arange(R, dtype=np.int32)
+        row_of_t = torch.from_numpy(row_of).to(dev)
         lens_dev = (burn + learn + fwd).to(torch.int32).to(dev)
         seg = torch.zeros(len(burn) + 1, dtype=torch.int32)
         seg[1:] = torch.cumsum(learn.to(torch.int32), 0)
-        out = (lp_t, tp_t, lbt_t, tgt_bt, lens_dev, seg.to(dev))
+        out = (lp_t, tp_t, row_of_t, lens_dev, seg.to(dev))
         self._pos_cache = (key, out)
         return out
 
     # ------------------------------------------------------------------
-    def train_step(self, batch):
+    def train_step(self, batch, grad_hook=None):
         """Full fused update.  Fills .grad on the online nn.Module params and
-        returns (loss tensor, per-sequence priority tensor, both on device)."""
+        returns (loss tensor, per-sequence priority tensor, both on device).
+        ``grad_hook(segment)`` fires when a flat-grad segment ("heads",
+        "lstm", "encoder") is complete — the DP learner uses it to overlap
+        the RCCL all-reduce with the rest of the manual backward."""
         m = self.m
         c = self.cfg
         dev = self.device
@@ -318,7 +341,7 @@ class HipNetworkEngine:
             obs_hwc = obs_hwc.contiguous()
         la = batch.last_action.to(dev)
         lr = batch.last_reward.to(dev)
-        lp_t, tp_t, lbt_t, tgt_bt, lens, seg = self._positions(
+        lp_t, tp_t, row_of, lens, seg = self._positions(
             batch.burn_in_steps, batch.learning_steps,
             batch.forward_steps, T)
         R = lp_t.shape[0]
@@ -368,39 +391,40 @@ class HipNetworkEngine:
         # ---- backward --------------------------------------------------
         # weight grads accumulate STRAIGHT into the flat .grad views
         # (pre-zeroed once) — no per-parameter copy/permute kernels.
+        # Only the R learning rows carry gradient (the target-position rows
+        # of q_cat enter the loss detached, reference worker.py:346), so the
+        # whole head backward runs on R rows, not 2R.
         ON = self.online
         net = self.online_net
         self.flat_grad.zero_()
-        dq_pad = torch.zeros(2 * R, A, device=dev)  # learn rows only get grad
-        dq_pad[:R] = dq
-        dadv2, dval2 = m.dueling_combine_bwd(dq_pad.contiguous(),
+        adv1_l, val1_l, h_l = adv1_o[:R], val1_o[:R], h_cat[:R]
+        dadv2, dval2 = m.dueling_combine_bwd(dq.contiguous(),
                                              PAD_HEAD, PAD_HEAD)
         self._mark("hb_duel")
         # adv path
         dadv1 = m.gemm_dgrad(dadv2, self._empty, ON.wa2_kn, False)
-        dh_a = m.gemm_dgrad(dadv1, adv1_o, ON.wa1_kn, True)
+        dh_a = m.gemm_dgrad(dadv1, adv1_l, ON.wa1_kn, True)
         # value path
         dval1 = m.gemm_dgrad(dval2, self._empty, ON.wv2_kn, False)
-        dh_v = m.gemm_dgrad(dval1, val1_o, ON.wv1_kn, True)
+        dh_v = m.gemm_dgrad(dval1, val1_l, ON.wv1_kn, True)
         self._mark("hb_dgrads")
-        m.gemm_wgrad_into(dadv2, self._empty, adv1_o, False,
+        m.gemm_wgrad_into(dadv2, self._empty, adv1_l, False,
                           net.advantage[2].weight.grad,
                           net.advantage[2].bias.grad)
-        m.gemm_wgrad_into(dadv1, adv1_o, h_cat, True,
+        m.gemm_wgrad_into(dadv1, adv1_l, h_l, True,
                           net.advantage[0].weight.grad,
                           net.advantage[0].bias.grad)
-        m.gemm_wgrad_into(dval2, self._empty, val1_o, False,
+        m.gemm_wgrad_into(dval2, self._empty, val1_l, False,
                           net.value[2].weight.grad, net.value[2].bias.grad)
-        m.gemm_wgrad_into(dval1, val1_o, h_cat, True,
+        m.gemm_wgrad_into(dval1, val1_l, h_l, True,
                           net.value[0].weight.grad, net.value[0].bias.grad)
+        if grad_hook is not None:
+            grad_hook("heads")
         self._mark("hb_wgrads")
 
-        dh_rows = (dh_a.float() + dh_v.float())     # (2R, 512)
-        # scatter-add into dHext (B, T, H): learn rows at t, tgt rows at t_tgt
-        dHext = torch.zeros(B * T, H, device=dev)
-        dHext.index_add_(0, lbt_t, dh_rows[:R])
-        dHext.index_add_(0, tgt_bt, dh_rows[R:])
-        dHext = dHext.view(B, T, H)
+        # dHext (B, T, H): dh_a+dh_v scattered to the learning positions in
+        # one kernel via the cached inverse map (no zeros fill / index_add_)
+        dHext = m.scatter_dh(dh_a, dh_v, row_of, B * T).view(B, T, H)
         self._mark("head_bwd")
 
         dgates = m.lstm_bwd(stash, Co, Ho, dHext.contiguous(), ON.whh_bwd,
@@ -416,6 +440,8 @@ class HipNetworkEngine:
                           net.recurrent.weight_ih_l0.grad,
                           net.recurrent.bias_ih_l0.grad)
         net.recurrent.bias_hh_l0.grad.copy_(net.recurrent.bias_ih_l0.grad)
+        if grad_hook is not None:
+            grad_hook("lstm")
         # only the latent slice of the padded LSTM-input grad is needed
         dlat = m.gemm_dgrad(dgates_flat, self._empty, ON.wih_kn, False, 512)
         self._mark("lstm_wgrads")
@@ -424,12 +450,18 @@ class HipNetworkEngine:
         if self.impala:
             impala_ops.encoder_bwd(m, ON.imp, enc_stash, dlat, lat_o)
             self._mark("conv_bwd")
+            if grad_hook is not None:
+                grad_hook("encoder")
             self._mark("grad_write")
             return loss.squeeze(0), prio
 
         a1, a2, a3, flat = enc_stash
         lat_bf = lat_o  # forward output (relu mask source)
-        dflat = m.gemm_dgrad(dlat, lat_bf, ON.wf_kn, True)
+        # FC dgrad with BOTH relu backwards fused: the mask of the FC's own
+        # relu (lat) on load, conv3's relu mask (a3) on store — dflat leaves
+        # this kernel pre-masked, so no elementwise mask/pad passes run on
+        # the conv gradients at all.
+        dflat = m.gemm_dgrad(dlat, lat_bf, ON.wf_kn, True, 0, a3)
         dWf, dbf = m.gemm_wgrad(dlat, lat_bf, flat, True, True)
 
         # conv3 backward.  Conv weight grads accumulate in the packed
@@ -437,29 +469,25 @@ class HipNetworkEngine:
         # 16-lane wave across 16 cachelines (measured 4-5x slower) — and
         # are permuted into .grad afterwards (tiny tensors).
         self._mark("fc_bwd")
-        dW3, db3 = m.conv_wgrad(dflat.view(M * 49, 64), a3, a2, 3,
+        dflat3 = dflat.view(M * 49, 64)   # pre-masked by a3
+        dW3, db3 = m.conv_wgrad(dflat3, a3, a2, 3,
                                 M, 9, 9, 7, 7, 64, 9 * 64)
-        d3m = (dflat.view(M * 49, 64)
-               * (a3.view(M * 49, 64) > 0).bfloat16()).view(M, 7, 7, 64)
-        dyp3 = torch.zeros(M, 11, 11, 64, device=dev, dtype=torch.bfloat16)
-        dyp3[:, 2:9, 2:9] = d3m
+        # dense bounds-checked dgrad (no zero-padded staging copies); the
+        # output mask fuses conv2's relu backward into the d_a2 store
         d_a2 = torch.empty(M, 9, 9, 64, device=dev, dtype=torch.bfloat16)
-        m.conv_dgrad(dyp3.contiguous(), ON.w3d, ON.taps3, M, 11, 11, 64,
-                     9, 9, 64, 0, 0, 1, 2, d_a2)
-        # conv2 backward
-        dW2, db2 = m.conv_wgrad(d_a2.view(M * 81, 64), a2, a1, 2,
+        m.conv_dgrad_dense(dflat3, ON.w3d, ON.taps3, a2,
+                           M, 7, 7, 64, 9, 9, 64, 0, 0, 1, d_a2)
+        # conv2 backward (d_a2 pre-masked by a2)
+        d_a2f = d_a2.view(M * 81, 64)
+        dW2, db2 = m.conv_wgrad(d_a2f, a2, a1, 2,
                                 M, 20, 20, 9, 9, 64, 4 * 4 * 32)
-        d2m = (d_a2.view(M * 81, 64)
-               * (a2.view(M * 81, 64) > 0).bfloat16()).view(M, 9, 9, 64)
-        dyp2 = torch.zeros(M, 11, 11, 64, device=dev, dtype=torch.bfloat16)
-        dyp2[:, 1:10, 1:10] = d2m
         d_a1 = torch.empty(M, 20, 20, 32, device=dev, dtype=torch.bfloat16)
-        dyp2c = dyp2.contiguous()
         for py in range(2):
             for px in range(2):
-                m.conv_dgrad(dyp2c, ON.w2d[(py, px)], ON.taps2[(py, px)],
-                             M, 11, 11, 64, 20, 20, 32, py, px, 2, 1, d_a1)
-        # conv1 wgrad (no dgrad: input is data)
+                m.conv_dgrad_dense(d_a2f, ON.w2d[(py, px)],
+                                   ON.taps2[(py, px)], a1,
+                                   M, 9, 9, 64, 20, 20, 32, py, px, 2, d_a1)
+        # conv1 wgrad (no dgrad: input is data; d_a1 pre-masked by a1)
         dW1, db1 = m.conv_wgrad(d_a1.view(M * 400, 32), a1, obs_hwc, 1,
                                 M, 84, 84, 20, 20, 32, 8 * 8 * self.C)
         self._mark("conv_bwd")
@@ -478,6 +506,8 @@ class HipNetworkEngine:
         enc.fc.weight.grad.copy_(
             dWf.view(512, 7, 7, 64).permute(0, 3, 1, 2).reshape(512, 3136))
         enc.fc.bias.grad.copy_(dbf)
+        if grad_hook is not None:
+            grad_hook("encoder")
 
         self._mark("grad_write")
         return loss.squeeze(0), prio

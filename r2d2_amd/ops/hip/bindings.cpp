@@ -36,13 +36,15 @@ std::vector<torch::Tensor> replay_gather_batch(
 torch::Tensor gemm_bias_act(torch::Tensor A, torch::Tensor Wt,
                             torch::Tensor bias, int64_t act, bool out_f32);
 torch::Tensor gemm_dgrad(torch::Tensor dY, torch::Tensor act_out,
-                         torch::Tensor W, bool relu_mask, int64_t k_out);
+                         torch::Tensor W, bool relu_mask, int64_t k_out,
+                         torch::Tensor out_mask);
 std::vector<torch::Tensor> gemm_wgrad(torch::Tensor dY, torch::Tensor act_out,
                                       torch::Tensor A, bool relu_mask,
                                       bool want_bias);
 void gemm_wgrad_into(torch::Tensor dY, torch::Tensor act_out, torch::Tensor A,
                      bool relu_mask, torch::Tensor dW_out,
-                     torch::Tensor db_out);
+                     torch::Tensor db_out,
+                     int64_t kd0, int64_t kd1, int64_t kd2);
 
 // conv_kernels.hip
 torch::Tensor conv_fwd(torch::Tensor in, torch::Tensor Wt, torch::Tensor bias,
@@ -53,6 +55,12 @@ torch::Tensor conv_dgrad(torch::Tensor dYp, torch::Tensor Wd, torch::Tensor taps
                          int64_t XH, int64_t XW, int64_t CIN,
                          int64_t y0, int64_t x0, int64_t S, int64_t pad,
                          torch::Tensor dX);
+torch::Tensor conv_dgrad_dense(torch::Tensor dY, torch::Tensor Wd,
+                               torch::Tensor taps, torch::Tensor actx,
+                               int64_t N, int64_t OH, int64_t OW, int64_t COUT,
+                               int64_t XH, int64_t XW, int64_t CIN,
+                               int64_t y0, int64_t x0, int64_t S,
+                               torch::Tensor dX);
 std::vector<torch::Tensor> conv_wgrad(torch::Tensor dY, torch::Tensor act,
                                       torch::Tensor in, int64_t conv_id,
                                       int64_t N, int64_t INH, int64_t INW,
@@ -71,6 +79,8 @@ std::vector<torch::Tensor> lstm_fwd(
     torch::Tensor init0, torch::Tensor init1, torch::Tensor lens,
     torch::Tensor barrier_ws, bool want_stash);
 torch::Tensor dueling_combine(torch::Tensor adv, torch::Tensor val, int64_t A);
+torch::Tensor scatter_dh(torch::Tensor dh_a, torch::Tensor dh_v,
+                         torch::Tensor row_of, int64_t BT);
 std::vector<torch::Tensor> dueling_combine_bwd(torch::Tensor dq, int64_t PADA,
                                                int64_t PADV);
 torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
@@ -122,14 +132,25 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gemm_bias_act", &gemm_bias_act, "MFMA GEMM + bias + activation");
     m.def("gemm_dgrad", &gemm_dgrad,
           "MFMA GEMM backward-data (+ fused ReLU mask; optional dense "
-          "k_out-column output)",
+          "k_out-column output; optional fused output ReLU mask)",
           pybind11::arg("dY"), pybind11::arg("act_out"), pybind11::arg("W"),
-          pybind11::arg("relu_mask"), pybind11::arg("k_out") = 0);
+          pybind11::arg("relu_mask"), pybind11::arg("k_out") = 0,
+          pybind11::arg("out_mask") = torch::Tensor());
     m.def("gemm_wgrad", &gemm_wgrad, "MFMA GEMM backward-weight (+ bias grad)");
     m.def("gemm_wgrad_into", &gemm_wgrad_into,
-          "GEMM backward-weight accumulated straight into .grad views");
+          "GEMM backward-weight accumulated straight into .grad views "
+          "(optional (d0,d1,d2)->(d2,d0,d1) k-axis store permutation)",
+          pybind11::arg("dY"), pybind11::arg("act_out"), pybind11::arg("A"),
+          pybind11::arg("relu_mask"), pybind11::arg("dW_out"),
+          pybind11::arg("db_out"), pybind11::arg("kd0") = 0,
+          pybind11::arg("kd1") = 0, pybind11::arg("kd2") = 0);
     m.def("conv_fwd", &conv_fwd, "Implicit-GEMM MFMA conv forward (NHWC)");
     m.def("conv_dgrad", &conv_dgrad, "MFMA conv backward-data (tap classes)");
+    m.def("conv_dgrad_dense", &conv_dgrad_dense,
+          "MFMA conv backward-data from dense pre-masked dY (bounds-checked "
+          "taps, optional fused output ReLU mask)");
+    m.def("scatter_dh", &scatter_dh,
+          "dHext (BT, H) from head-backward rows via inverse position map");
     m.def("conv_wgrad", &conv_wgrad, "MFMA conv backward-weight");
     m.def("lstm_fwd", &lstm_fwd,
           "Persistent fused LSTM forward (dual-network, length-masked)");

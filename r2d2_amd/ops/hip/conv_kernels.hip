@@ -246,13 +246,112 @@ __global__ __launch_bounds__(256) void conv_dgrad_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Dgrad from the DENSE (unpadded, pre-masked) upstream gradient
+// dY (N, OH, OW, COUT): taps landing outside [0,OH)x[0,OW) contribute zero
+// via a bounds check instead of a zero-padded staging copy (drops the
+// dyp zeros-fill + interior copy + separate ReLU-mask pass from the engine
+// backward).  OUT_MASK applies the ReLU mask of the conv BELOW
+// ((act_x > 0) at the dX address) on store, so the next layer's dgrad and
+// wgrad consume a pre-masked tensor directly.
+// ---------------------------------------------------------------------------
+template <int TAPS, int NCOL, int CO_T, bool OUT_MASK>
+__global__ __launch_bounds__(256) void conv_dgrad_dense_kernel(
+    const __hip_bfloat16* __restrict__ dY,   // (N, OH, OW, COUT)
+    const __hip_bfloat16* __restrict__ Wd,   // (CIN, TAPS*COUT)
+    const __hip_bfloat16* __restrict__ actx, // (N, XH, XW, CIN) or null
+    __hip_bfloat16* __restrict__ dX,         // (N, XH, XW, CIN)
+    const int* __restrict__ taps,            // (TAPS, 2): dy, dx
+    int Mc, int YY, int XX, int y0, int x0, int S,
+    int OH, int OW, int COUT, int XH, int XW, int CIN) {
+    const int COUTc = CO_T ? CO_T : COUT;
+    const int K = TAPS * COUTc;
+    int wave = threadIdx.x / WAVE;
+    int lane = threadIdx.x & (WAVE - 1);
+    int wr = (NCOL == 32) ? wave : (wave >> 1);
+    int wc = (NCOL == 32) ? 0 : (wave & 1);
+    long row0 = (long)blockIdx.x * (NCOL == 32 ? 128 : 64) + wr * 32;
+    long col0 = (long)blockIdx.y * 64 + wc * 32;
+    int frow = lane & 15;
+    int kseg = (lane >> 4) * 8;
+
+    const unsigned YX = (unsigned)(YY * XX);
+    long nbase[2];
+    int yv[2], xv[2];
+    bool avalid[2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+        unsigned r = (unsigned)row0 + i * 16 + frow;
+        avalid[i] = r < (unsigned)Mc;
+        unsigned n = avalid[i] ? r / YX : 0;
+        unsigned p = avalid[i] ? r % YX : 0;
+        yv[i] = y0 + (int)(p / (unsigned)XX) * S;
+        xv[i] = x0 + (int)(p % (unsigned)XX) * S;
+        nbase[i] = (long)n * OH * OW;
+    }
+
+    f32x4 acc[2][2] = {};
+    for (int k0 = 0; k0 < K; k0 += 32) {
+        int k = k0 + kseg;
+        int t = (unsigned)k / (unsigned)COUTc;
+        int co = (unsigned)k % (unsigned)COUTc;
+        int dy = taps[2 * t], dx = taps[2 * t + 1];
+        bf16x8 a[2], b[2];
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            // parity class guarantees S | (yv-dy); bounds replace padding
+            int ny = yv[i] - dy, nx = xv[i] - dx;
+            int oy = ny / S, ox = nx / S;
+            bool ok = avalid[i] && ny >= 0 && nx >= 0 && oy < OH && ox < OW;
+            a[i] = ok
+                ? cload_bf16x8(dY + (nbase[i] + (long)oy * OW + ox) * COUT + co)
+                : czero();
+            long c = col0 + i * 16 + frow;
+            b[i] = (c < CIN) ? cload_bf16x8(Wd + c * K + k0 + kseg) : czero();
+        }
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+            for (int j = 0; j < 2; ++j)
+                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+
+    int ccol = lane & 15;
+    int crow = (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                long rr = row0 + i * 16 + crow + r;
+                long cc = col0 + j * 16 + ccol;
+                if (rr < Mc && cc < CIN) {
+                    unsigned n = (unsigned)rr / YX;
+                    unsigned p = (unsigned)rr % YX;
+                    int y = y0 + (int)(p / (unsigned)XX) * S;
+                    int x = x0 + (int)(p % (unsigned)XX) * S;
+                    long addr = (((long)n * XH + y) * XW + x) * CIN + cc;
+                    float v = acc[i][j][r];
+                    if (OUT_MASK)
+                        v = (bf2f(actx[addr]) > 0.f) ? v : 0.f;
+                    dX[addr] = f2bf(v);
+                }
+            }
+}
+
+// ---------------------------------------------------------------------------
 // Wgrad: dWt(COUT, K) += sum_rows relu_mask(dY)[row][co] * patch[row][k]
 // The FULL K extent (<= 576) and full COUT (<= 64) are staged per 32-row
 // tile, so dY and the patches are each read exactly once; per-wave
 // accumulators cover all k-tiles and are flushed with f32 atomics once per
 // chunk.  Fused ReLU mask + bias grad.
 // ---------------------------------------------------------------------------
-template <bool IN_U8, int KH, int KW, int CIN, int S, int NCOT, bool RELU>
+// TORCH_LAYOUT: remap each (ky, kx, ci)-packed k store to torch's
+// (ci, ky, kx) weight order, so gradients accumulate straight into the
+// module's .grad view (compile-time dims -> cheap index math).
+template <bool IN_U8, int KH, int KW, int CIN, int S, int NCOT, bool RELU,
+          bool TORCH_LAYOUT = false>
 __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     const __hip_bfloat16* __restrict__ dY,   // (M, COUT)
     const __hip_bfloat16* __restrict__ act,  // (M, COUT) forward output
@@ -368,8 +467,15 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
             for (int r = 0; r < 4; ++r) {
                 long co = wr * 32 + i * 16 + crow + r;
                 long kk = wc * KHALF + kf * 16 + ccol;
-                if (co < COUT && kk < K)
+                if (co < COUT && kk < K) {
+                    if (TORCH_LAYOUT) {
+                        int ci = (int)(kk % CIN);
+                        int t_ = (int)(kk / CIN);
+                        int kx = t_ % KW, ky = t_ / KW;
+                        kk = ((long)ci * KH + ky) * KW + kx;
+                    }
                     atomicAdd(&dWt[co * K + kk], acc[i][kf][r]);
+                }
             }
     if (threadIdx.x < 64) {
         long c = threadIdx.x;
@@ -448,6 +554,48 @@ torch::Tensor conv_dgrad(torch::Tensor dYp, torch::Tensor Wd, torch::Tensor taps
     else TORCH_CHECK(false, "unsupported tap count");
 #undef DLAUNCH
 #undef DL1
+    return dX;
+}
+
+// dense-input dgrad: dY is the UNPADDED, PRE-MASKED upstream gradient
+// (N, OH, OW, COUT); actx (optional) fuses the ReLU mask of the conv below
+// into the dX store.
+torch::Tensor conv_dgrad_dense(torch::Tensor dY, torch::Tensor Wd,
+                               torch::Tensor taps, torch::Tensor actx,
+                               int64_t N, int64_t OH, int64_t OW, int64_t COUT,
+                               int64_t XH, int64_t XW, int64_t CIN,
+                               int64_t y0, int64_t x0, int64_t S,
+                               torch::Tensor dX) {
+    long YY = (XH - 1 - y0) / S + 1;
+    long XX = (XW - 1 - x0) / S + 1;
+    long Mc = N * YY * XX;
+    long TAPS = taps.size(0);
+    bool has_m = actx.defined() && actx.numel() > 0;
+    if (has_m) TORCH_CHECK(actx.numel() == dX.numel());
+    dim3 grid(ccdiv(Mc, CIN <= 32 ? 128 : 64), ccdiv(CIN, 64));
+    auto stream = at::cuda::getCurrentCUDAStream();
+    auto* dy = reinterpret_cast<const __hip_bfloat16*>(dY.data_ptr());
+    auto* w = reinterpret_cast<const __hip_bfloat16*>(Wd.data_ptr());
+    auto* am = has_m
+        ? reinterpret_cast<const __hip_bfloat16*>(actx.data_ptr()) : nullptr;
+    auto* dx = reinterpret_cast<__hip_bfloat16*>(dX.data_ptr());
+    const int* tp = taps.data_ptr<int>();
+#define DD1(T, NC, CO, OM)                                                     \
+    hipLaunchKernelGGL((conv_dgrad_dense_kernel<T, NC, CO, OM>), grid,         \
+                       dim3(256), 0, stream.stream(), dy, w, am, dx, tp,       \
+                       (int)Mc, (int)YY, (int)XX, (int)y0, (int)x0, (int)S,    \
+                       (int)OH, (int)OW, (int)COUT, (int)XH, (int)XW, (int)CIN)
+#define DDM(T, NC, CO)                                                         \
+    do { if (has_m) DD1(T, NC, CO, true); else DD1(T, NC, CO, false); } while (0)
+#define DDLAUNCH(T)                                                            \
+    if (CIN <= 32) { if (COUT == 64) DDM(T, 32, 64); else DDM(T, 32, 0); }     \
+    else { if (COUT == 64) DDM(T, 64, 64); else DDM(T, 64, 0); }
+    if (TAPS == 4) { DDLAUNCH(4); }
+    else if (TAPS == 9) { DDLAUNCH(9); }
+    else TORCH_CHECK(false, "unsupported tap count");
+#undef DDLAUNCH
+#undef DDM
+#undef DD1
     return dX;
 }
 

@@ -108,13 +108,17 @@ __global__ __launch_bounds__(256) void gemm_bias_act_kernel(
 //   dA[m][k] = sum_n dY[m][n] * W[k][n], W prepacked row-major (K, N).
 //   RELU_MASK: multiply dY by (act_out > 0) on load (fused ReLU backward,
 //   act_out is the forward output of this layer, same shape as dY).
+//   OUT_MASK: multiply dA by (outm > 0) on store — fuses the ReLU backward
+//   of the layer BELOW (e.g. conv3's mask applied to the FC dgrad output),
+//   so no separate elementwise mask pass runs on the big dX tensors.
 // ---------------------------------------------------------------------------
-template <bool RELU_MASK>
+template <bool RELU_MASK, bool OUT_MASK = false>
 __global__ __launch_bounds__(256) void gemm_dgrad_kernel(
     const __hip_bfloat16* __restrict__ dY,      // (M, N)
     const __hip_bfloat16* __restrict__ act,     // (M, N) or null
     const __hip_bfloat16* __restrict__ W,       // (K, N) row-major
     __hip_bfloat16* __restrict__ dA,            // (M, K)
+    const __hip_bfloat16* __restrict__ outm,    // (M, Kout) or null
     int M, int N, int K, int Kout) {
     int wave = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
@@ -163,8 +167,12 @@ __global__ __launch_bounds__(256) void gemm_dgrad_kernel(
             for (int r = 0; r < 4; ++r) {
                 long rr = row0 + i * 16 + crow + r;
                 long cc = col0 + j * 16 + ccol;
-                if (rr < M && cc < Kout)
-                    dA[rr * Kout + cc] = f2bf(acc[i][j][r]);
+                if (rr < M && cc < Kout) {
+                    float v = acc[i][j][r];
+                    if (OUT_MASK)
+                        v = (bf2f(outm[rr * Kout + cc]) > 0.f) ? v : 0.f;
+                    dA[rr * Kout + cc] = f2bf(v);
+                }
             }
 }
 
@@ -174,14 +182,19 @@ __global__ __launch_bounds__(256) void gemm_dgrad_kernel(
 // Also accumulates db(N) = sum_m dY[m][n] when HAS_BIAS.
 //   RELU_MASK as in dgrad (applied to dY).
 // ---------------------------------------------------------------------------
-template <bool RELU_MASK, bool HAS_BIAS>
+// KPERM: the k axis is a packed (d0, d1, d2) index (e.g. conv taps (ky, kx,
+// cin) or the FC's flattened (h, w, c)); remap each store to torch's
+// (d2, d0, d1) order so gradients accumulate STRAIGHT into the module
+// .grad views with no per-step permute-copy kernels.
+template <bool RELU_MASK, bool HAS_BIAS, bool KPERM = false>
 __global__ __launch_bounds__(256) void gemm_wgrad_kernel(
     const __hip_bfloat16* __restrict__ dY,   // (M, N)
     const __hip_bfloat16* __restrict__ act,  // (M, N) or null
     const __hip_bfloat16* __restrict__ A,    // (M, K)
     float* __restrict__ dWt,                 // (N, K) f32 accumulate
     float* __restrict__ db,                  // (N,) f32 accumulate
-    int M, int N, int K, int rows_per_chunk, int Nout, int Kout) {
+    int M, int N, int K, int rows_per_chunk, int Nout, int Kout,
+    int kd0 = 0, int kd1 = 0, int kd2 = 0) {
     // grid.x: row chunks; grid.y: N tiles of 64; grid.z: K tiles of 64
     __shared__ __hip_bfloat16 s_dy[32][64 + 8];  // [m][n]
     __shared__ __hip_bfloat16 s_a[32][64 + 8];   // [m][k]
@@ -293,8 +306,16 @@ __global__ __launch_bounds__(256) void gemm_wgrad_kernel(
             for (int r = 0; r < 4; ++r) {
                 long nn = ncol0 + wr * 32 + i * 16 + crow + r;
                 long kk = kcol0 + wc * 32 + j * 16 + ccol;
-                if (nn < Nout && kk < Kout)
+                if (nn < Nout && kk < Kout) {
+                    if (KPERM) {
+                        int d2 = (int)(kk % kd2);
+                        int t_ = (int)(kk / kd2);
+                        int d1 = t_ % kd1;
+                        int d0 = t_ / kd1;
+                        kk = ((long)d2 * kd0 + d0) * kd1 + d1;
+                    }
                     atomicAdd(&dWt[nn * Kout + kk], acc[i][j][r]);
+                }
             }
     if (HAS_BIAS && threadIdx.x < 64 && blockIdx.z == 0) {
         long c = ncol0 + threadIdx.x;
@@ -339,7 +360,8 @@ torch::Tensor gemm_bias_act(torch::Tensor A, torch::Tensor Wt,
 }
 
 torch::Tensor gemm_dgrad(torch::Tensor dY, torch::Tensor act_out,
-                         torch::Tensor W, bool relu_mask, int64_t k_out) {
+                         torch::Tensor W, bool relu_mask, int64_t k_out,
+                         torch::Tensor out_mask) {
     TORCH_CHECK(dY.is_cuda() && dY.dtype() == torch::kBFloat16 && dY.is_contiguous());
     TORCH_CHECK(W.dtype() == torch::kBFloat16 && W.is_contiguous());
     long M = dY.size(0), N = dY.size(1), K = W.size(0);
@@ -348,6 +370,11 @@ torch::Tensor gemm_dgrad(torch::Tensor dY, torch::Tensor act_out,
     // (M, k_out) tensor (e.g. the latent slice of the padded LSTM input)
     long Kout = (k_out > 0) ? k_out : K;
     TORCH_CHECK(Kout <= K);
+    bool has_om = out_mask.defined() && out_mask.numel() > 0;
+    if (has_om)
+        TORCH_CHECK(out_mask.dtype() == torch::kBFloat16
+                    && out_mask.is_contiguous()
+                    && out_mask.numel() == M * Kout);
     auto dA = torch::empty({M, Kout}, dY.options());
     dim3 grid(cdiv(M, 64), cdiv(Kout, 64));
     auto stream = at::cuda::getCurrentCUDAStream();
@@ -356,14 +383,15 @@ torch::Tensor gemm_dgrad(torch::Tensor dY, torch::Tensor act_out,
         ? reinterpret_cast<const __hip_bfloat16*>(act_out.data_ptr()) : nullptr;
     auto* w = reinterpret_cast<const __hip_bfloat16*>(W.data_ptr());
     auto* da = reinterpret_cast<__hip_bfloat16*>(dA.data_ptr());
-    if (relu_mask)
-        hipLaunchKernelGGL((gemm_dgrad_kernel<true>), grid, dim3(256), 0,
-                           stream.stream(), dy, ac, w, da, (int)M, (int)N,
-                           (int)K, (int)Kout);
-    else
-        hipLaunchKernelGGL((gemm_dgrad_kernel<false>), grid, dim3(256), 0,
-                           stream.stream(), dy, ac, w, da, (int)M, (int)N,
-                           (int)K, (int)Kout);
+    auto* om = has_om
+        ? reinterpret_cast<const __hip_bfloat16*>(out_mask.data_ptr()) : nullptr;
+#define LAUNCHD(RM, OM)                                                        \
+    hipLaunchKernelGGL((gemm_dgrad_kernel<RM, OM>), grid, dim3(256), 0,        \
+                       stream.stream(), dy, ac, w, da, om, (int)M, (int)N,     \
+                       (int)K, (int)Kout)
+    if (relu_mask) { if (has_om) LAUNCHD(true, true); else LAUNCHD(true, false); }
+    else           { if (has_om) LAUNCHD(false, true); else LAUNCHD(false, false); }
+#undef LAUNCHD
     return dA;
 }
 
@@ -391,7 +419,7 @@ std::vector<torch::Tensor> gemm_wgrad(torch::Tensor dY, torch::Tensor act_out,
     hipLaunchKernelGGL((gemm_wgrad_kernel<RM, HB>), grid, dim3(256), 0,        \
                        stream.stream(), dy, ac, a, dWt.data_ptr<float>(),      \
                        db.data_ptr<float>(), (int)M, (int)N, (int)K,           \
-                       (int)rows_per_chunk, (int)N, (int)K)
+                       (int)rows_per_chunk, (int)N, (int)K, 0, 0, 0)
     if (relu_mask) { if (want_bias) LAUNCHW(true, true); else LAUNCHW(true, false); }
     else           { if (want_bias) LAUNCHW(false, true); else LAUNCHW(false, false); }
 #undef LAUNCHW
@@ -403,13 +431,16 @@ std::vector<torch::Tensor> gemm_wgrad(torch::Tensor dY, torch::Tensor act_out,
 // grad-copy kernels from the engine's backward.
 void gemm_wgrad_into(torch::Tensor dY, torch::Tensor act_out, torch::Tensor A,
                      bool relu_mask, torch::Tensor dW_out,
-                     torch::Tensor db_out) {
+                     torch::Tensor db_out,
+                     int64_t kd0, int64_t kd1, int64_t kd2) {
     TORCH_CHECK(dY.is_cuda() && dY.dtype() == torch::kBFloat16 && dY.is_contiguous());
     TORCH_CHECK(A.dtype() == torch::kBFloat16 && A.is_contiguous());
     TORCH_CHECK(dW_out.dtype() == torch::kFloat32 && dW_out.is_contiguous());
     long M = dY.size(0), N = dY.size(1), K = A.size(1);
     long Nout = dW_out.size(0), Kout = dW_out.numel() / Nout;
     bool want_bias = db_out.defined() && db_out.numel() > 0;
+    bool kperm = kd2 > 0;
+    if (kperm) TORCH_CHECK(kd0 * kd1 * kd2 == Kout, "kperm dims mismatch");
     long tiles = (long)cdiv(N, 64) * cdiv(K, 64);
     long target_chunks = std::max(1L, 1024L / std::max(1L, tiles));
     long rows_per_chunk = std::max(32L, (M + target_chunks - 1) / target_chunks);
@@ -421,12 +452,16 @@ void gemm_wgrad_into(torch::Tensor dY, torch::Tensor act_out, torch::Tensor A,
         ? reinterpret_cast<const __hip_bfloat16*>(act_out.data_ptr()) : nullptr;
     auto* a = reinterpret_cast<const __hip_bfloat16*>(A.data_ptr());
     float* dbp = want_bias ? db_out.data_ptr<float>() : nullptr;
-#define LAUNCHW(RM, HB)                                                        \
-    hipLaunchKernelGGL((gemm_wgrad_kernel<RM, HB>), grid, dim3(256), 0,        \
+#define LAUNCHW(RM, HB, KP)                                                    \
+    hipLaunchKernelGGL((gemm_wgrad_kernel<RM, HB, KP>), grid, dim3(256), 0,    \
                        stream.stream(), dy, ac, a, dW_out.data_ptr<float>(),   \
                        dbp, (int)M, (int)N, (int)K,                            \
-                       (int)rows_per_chunk, (int)Nout, (int)Kout)
-    if (relu_mask) { if (want_bias) LAUNCHW(true, true); else LAUNCHW(true, false); }
-    else           { if (want_bias) LAUNCHW(false, true); else LAUNCHW(false, false); }
+                       (int)rows_per_chunk, (int)Nout, (int)Kout,              \
+                       (int)kd0, (int)kd1, (int)kd2)
+#define LAUNCHW2(RM, HB)                                                       \
+    do { if (kperm) LAUNCHW(RM, HB, true); else LAUNCHW(RM, HB, false); } while (0)
+    if (relu_mask) { if (want_bias) LAUNCHW2(true, true); else LAUNCHW2(true, false); }
+    else           { if (want_bias) LAUNCHW2(false, true); else LAUNCHW2(false, false); }
+#undef LAUNCHW2
 #undef LAUNCHW
 }

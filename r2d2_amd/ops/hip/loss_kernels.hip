@@ -205,6 +205,57 @@ __global__ void dueling_combine_bwd_kernel(
         dval[(long)r * PADV + lane] = f2bf((lane == 0) ? s : 0.f);
 }
 
+// ---------------------------------------------------------------------------
+// scatter_dh: build the LSTM's upstream gradient dHext (BT, H) from the
+// head-backward rows in ONE pass:
+//   dHext[bt] = dh_a[row_of[bt]] + dh_v[row_of[bt]]   (row_of[bt] >= 0)
+//             = 0                                      (no learning row at bt)
+// Learn positions are unique per (b, t), so no atomics; replaces a zeros
+// fill + two index_add_ launches + two bf16->f32 casts.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void scatter_dh_kernel(
+    const __hip_bfloat16* __restrict__ dh_a,  // (R, H)
+    const __hip_bfloat16* __restrict__ dh_v,  // (R, H)
+    const int* __restrict__ row_of,           // (BT,) learn row or -1
+    float* __restrict__ out,                  // (BT, H)
+    long total, int H) {
+    long idx = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+    if (idx >= total) return;
+    int bt = (int)(idx / H);
+    int k = (int)(idx % H);
+    int r = row_of[bt];
+    float4 v{0.f, 0.f, 0.f, 0.f};
+    if (r >= 0) {
+        long off = (long)r * H + k;
+        v.x = bf2f(dh_a[off]) + bf2f(dh_v[off]);
+        v.y = bf2f(dh_a[off + 1]) + bf2f(dh_v[off + 1]);
+        v.z = bf2f(dh_a[off + 2]) + bf2f(dh_v[off + 2]);
+        v.w = bf2f(dh_a[off + 3]) + bf2f(dh_v[off + 3]);
+    }
+    *reinterpret_cast<float4*>(out + idx) = v;
+}
+
+torch::Tensor scatter_dh(torch::Tensor dh_a, torch::Tensor dh_v,
+                         torch::Tensor row_of, int64_t BT) {
+    TORCH_CHECK(dh_a.is_cuda() && dh_a.dtype() == torch::kBFloat16
+                && dh_a.is_contiguous() && dh_v.is_contiguous());
+    TORCH_CHECK(row_of.dtype() == torch::kInt32 && row_of.numel() == BT);
+    long H = dh_a.size(1);
+    TORCH_CHECK(H % 4 == 0);
+    auto out = torch::empty({BT, H},
+                            dh_a.options().dtype(torch::kFloat32));
+    long total = BT * H;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(scatter_dh_kernel,
+                       dim3((int)((total / 4 + 255) / 256)), dim3(256), 0,
+                       stream.stream(),
+                       reinterpret_cast<const __hip_bfloat16*>(dh_a.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(dh_v.data_ptr()),
+                       row_of.data_ptr<int>(), out.data_ptr<float>(),
+                       total, (int)H);
+    return out;
+}
+
 torch::Tensor dueling_combine(torch::Tensor adv, torch::Tensor val, int64_t A) {
     int R = adv.size(0), PADA = adv.size(1), PADV = val.size(1);
     auto q = torch::empty({R, A}, adv.options());

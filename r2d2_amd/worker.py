@@ -585,12 +585,28 @@ class Learner:
         batch.to(self.device)
 
         if self.engine is not None:
-            # full HIP path: manual backward fills the flat grad buffer
-            loss, prio = self.engine.train_step(batch)
+            # full HIP path: manual backward fills the flat grad buffer.
+            # DP: each flat-grad segment all-reduces over RCCL as soon as
+            # the manual backward completes it (heads -> lstm -> encoder),
+            # overlapping the collective with the remaining backward; the
+            # works are drained before the fused clip+Adam step.
             if self.reducer is not None:
                 import torch.distributed as dist
-                self.engine.flat_grad.div_(self.reducer.world_size)
-                dist.all_reduce(self.engine.flat_grad)
+                works = []
+                ws = self.reducer.world_size
+
+                def _seg_reduce(seg):
+                    lo, hi = self.engine.seg_ranges[seg]
+                    g = self.engine.flat_grad[lo:hi]
+                    g.div_(ws)
+                    works.append(dist.all_reduce(g, async_op=True))
+
+                loss, prio = self.engine.train_step(batch,
+                                                    grad_hook=_seg_reduce)
+                for w in works:
+                    w.wait()
+            else:
+                loss, prio = self.engine.train_step(batch)
             self.engine.optimizer_step(
                 lr=self.optimizer.param_groups[0]["lr"],
                 eps=self.optimizer.param_groups[0]["eps"],

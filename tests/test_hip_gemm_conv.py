@@ -221,3 +221,94 @@ def test_conv_wgrad(conv_id, cin, cout, k, s, inhw):
     ref_dw = prepack_w(w32.grad)  # (COUT, K) in (ky,kx,c) order, f32
     close(dWt, ref_dw, rtol=5e-2, atol=0.5, name=f"conv{conv_id} wgrad")
     close(db, b32.grad, rtol=5e-2, atol=0.5, name=f"conv{conv_id} bgrad")
+
+
+# -------------------------------------------------------------------------
+# Dense (unpadded) dgrad with fused output mask; gemm output mask; scatter
+# -------------------------------------------------------------------------
+
+def test_conv3_dgrad_dense_with_out_mask():
+    """Bounds-checked dense dgrad == padded-staging dgrad; the output mask
+    fuses the next conv's ReLU backward into the dX store."""
+    torch.manual_seed(3)
+    N, CIN, COUT, K, XH = 5, 64, 64, 3, 9
+    OH = XH - K + 1  # 7
+    w = (torch.randn(COUT, CIN, K, K, device="cuda") * 0.2).bfloat16()
+    x32 = torch.randn(N, CIN, XH, XH, device="cuda", requires_grad=True)
+    out = F.conv2d(x32, w.float(), stride=1)
+    dy32 = torch.randn_like(out)
+    out.backward(dy32)
+
+    dy_dense = dy32.bfloat16().permute(0, 2, 3, 1).contiguous()  # (N,OH,OW,CO)
+    Wd = w.permute(1, 2, 3, 0).reshape(CIN, K * K * COUT).contiguous()
+    taps = torch.tensor([[dy, dx] for dy in range(K) for dx in range(K)],
+                        dtype=torch.int32, device="cuda")
+    dX = torch.zeros(N, XH, XH, CIN, device="cuda").bfloat16()
+    M_.conv_dgrad_dense(dy_dense.view(-1, COUT), Wd, taps,
+                        torch.empty(0, device="cuda"),
+                        N, OH, OH, COUT, XH, XH, CIN, 0, 0, 1, dX)
+    ref = x32.grad.permute(0, 2, 3, 1)
+    close(dX, ref, rtol=5e-2, name="conv3 dgrad dense")
+
+    # with output mask: dX masked by (actx > 0) at the store
+    actx = torch.randn(N, XH, XH, CIN, device="cuda").bfloat16()
+    dXm = torch.zeros_like(dX)
+    M_.conv_dgrad_dense(dy_dense.view(-1, COUT), Wd, taps, actx,
+                        N, OH, OH, COUT, XH, XH, CIN, 0, 0, 1, dXm)
+    refm = ref * (actx.float() > 0)
+    close(dXm, refm, rtol=5e-2, name="conv3 dgrad dense masked")
+
+
+def test_conv2_dgrad_dense_parity_classes():
+    torch.manual_seed(2)
+    N, CIN, COUT, K, S, XH = 4, 32, 64, 4, 2, 20
+    OH = (XH - K) // S + 1  # 9
+    w = (torch.randn(COUT, CIN, K, K, device="cuda") * 0.2).bfloat16()
+    x32 = torch.randn(N, CIN, XH, XH, device="cuda", requires_grad=True)
+    out = F.conv2d(x32, w.float(), stride=S)
+    dy32 = torch.randn_like(out)
+    out.backward(dy32)
+
+    dy_dense = (dy32.bfloat16().permute(0, 2, 3, 1).contiguous()
+                .view(-1, COUT))
+    dX = torch.zeros(N, XH, XH, CIN, device="cuda").bfloat16()
+    w_nhwc = w.permute(0, 2, 3, 1)
+    for py in range(S):
+        for px in range(S):
+            tap_list = [(dy_, dx_) for dy_ in range(py, K, S)
+                        for dx_ in range(px, K, S)]
+            taps = torch.tensor(tap_list, dtype=torch.int32, device="cuda")
+            Wd = torch.stack([w_nhwc[:, d, x_, :] for d, x_ in tap_list], dim=0)
+            Wd = (Wd.permute(2, 0, 1)
+                  .reshape(CIN, len(tap_list) * COUT).contiguous())
+            M_.conv_dgrad_dense(dy_dense, Wd, taps,
+                                torch.empty(0, device="cuda"),
+                                N, OH, OH, COUT, XH, XH, CIN, py, px, S, dX)
+    ref = x32.grad.permute(0, 2, 3, 1)
+    close(dX, ref, rtol=5e-2, name="conv2 dgrad dense")
+
+
+def test_gemm_dgrad_out_mask():
+    torch.manual_seed(4)
+    M, K, N = 128, 96, 64
+    dY = torch.randn(M, N, device="cuda").bfloat16()
+    W_kn = torch.randn(K, N, device="cuda").bfloat16()
+    om = torch.randn(M, K, device="cuda").bfloat16()
+    dA = M_.gemm_dgrad(dY, torch.empty(0, device="cuda"), W_kn, False, 0, om)
+    ref = (dY.float() @ W_kn.float().t()) * (om.float() > 0)
+    close(dA, ref, rtol=5e-2, name="gemm dgrad out-mask")
+
+
+def test_scatter_dh_matches_index_add():
+    torch.manual_seed(5)
+    B, T, H, R = 7, 11, 64, 40
+    dh_a = torch.randn(R, H, device="cuda").bfloat16()
+    dh_v = torch.randn(R, H, device="cuda").bfloat16()
+    # unique positions
+    pos = torch.randperm(B * T, device="cuda")[:R].to(torch.int64)
+    row_of = torch.full((B * T,), -1, dtype=torch.int32, device="cuda")
+    row_of[pos] = torch.arange(R, dtype=torch.int32, device="cuda")
+    out = M_.scatter_dh(dh_a, dh_v, row_of, B * T)
+    ref = torch.zeros(B * T, H, device="cuda")
+    ref.index_add_(0, pos, dh_a.float() + dh_v.float())
+    close(out, ref, rtol=1e-3, atol=1e-3, name="scatter_dh")
