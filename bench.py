@@ -118,7 +118,11 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
 
     have_cuda = torch.cuda.is_available()
-    device = torch.device(f"cuda:{local_rank}" if have_cuda else "cpu")
+    # ranks beyond the visible GPU count share GPU 0 — lets the exact
+    # multi-rank RCCL code path (init, reduce, max-over-ranks) be validated
+    # on a 1-GPU box before an 8-GPU node is available
+    dev_idx = local_rank % max(1, torch.cuda.device_count()) if have_cuda else 0
+    device = torch.device(f"cuda:{dev_idx}" if have_cuda else "cpu")
     if have_cuda:
         torch.cuda.set_device(device)
 

@@ -58,6 +58,33 @@ def _worker(rank, world_size, port, fail_q):
         raise
 
 
+@pytest.mark.timeout(300)
+def test_torchrun_multirank_bench_path():
+    """The EXACT multi-rank entry the driver uses — torch.distributed.run
+    with --nproc-per-node 2 on 127.0.0.1 — must run bench.py end-to-end
+    (rendezvous, per-rank Learner, max-over-ranks timing, rank-0 JSON).
+    Uses the cartpole preset so the CPU step is fast; the launch/rendezvous/
+    reduce code path is identical to the GPU run."""
+    import json
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29517", "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1", "--batches", "1",
+         "--preset", "cartpole"],
+        cwd=repo, capture_output=True, text=True, timeout=240)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert line, out.stdout
+    rec = json.loads(line[-1])
+    assert rec["n_gpus"] == 2 and rec["config"]["parallelism"] == "dp2"
+    assert rec["value"] > 0
+
+
 @pytest.mark.timeout(120)
 def test_bucketed_allreduce_matches_mean_grad():
     ctx = mp.get_context("spawn")
