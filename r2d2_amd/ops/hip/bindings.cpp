@@ -37,7 +37,7 @@ torch::Tensor gemm_bias_act(torch::Tensor A, torch::Tensor Wt,
                             torch::Tensor bias, int64_t act, bool out_f32);
 torch::Tensor gemm_dgrad(torch::Tensor dY, torch::Tensor act_out,
                          torch::Tensor W, bool relu_mask, int64_t k_out,
-                         torch::Tensor out_mask);
+                         c10::optional<torch::Tensor> out_mask);
 std::vector<torch::Tensor> gemm_wgrad(torch::Tensor dY, torch::Tensor act_out,
                                       torch::Tensor A, bool relu_mask,
                                       bool want_bias);
@@ -135,7 +135,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "k_out-column output; optional fused output ReLU mask)",
           pybind11::arg("dY"), pybind11::arg("act_out"), pybind11::arg("W"),
           pybind11::arg("relu_mask"), pybind11::arg("k_out") = 0,
-          pybind11::arg("out_mask") = torch::Tensor());
+          pybind11::arg("out_mask") = pybind11::none());
     m.def("gemm_wgrad", &gemm_wgrad, "MFMA GEMM backward-weight (+ bias grad)");
     m.def("gemm_wgrad_into", &gemm_wgrad_into,
           "GEMM backward-weight accumulated straight into .grad views "

@@ -361,7 +361,9 @@ torch::Tensor gemm_bias_act(torch::Tensor A, torch::Tensor Wt,
 
 torch::Tensor gemm_dgrad(torch::Tensor dY, torch::Tensor act_out,
                          torch::Tensor W, bool relu_mask, int64_t k_out,
-                         torch::Tensor out_mask) {
+                         c10::optional<torch::Tensor> out_mask_opt) {
+    torch::Tensor out_mask =
+        out_mask_opt.has_value() ? *out_mask_opt : torch::Tensor();
     TORCH_CHECK(dY.is_cuda() && dY.dtype() == torch::kBFloat16 && dY.is_contiguous());
     TORCH_CHECK(W.dtype() == torch::kBFloat16 && W.is_contiguous());
     long M = dY.size(0), N = dY.size(1), K = W.size(0);
