@@ -128,8 +128,12 @@ def main():
 
     if world_size > 1:
         import torch.distributed as dist
-        dist.init_process_group("nccl" if have_cuda else "gloo",
-                                rank=rank, world_size=world_size)
+        # R2D2_DIST_BACKEND=gloo lets the multi-rank GPU code path run with
+        # several ranks SHARING one GPU (RCCL refuses duplicate devices);
+        # the real 8-GPU run uses RCCL ("nccl" on ROCm) unchanged.
+        backend = os.environ.get("R2D2_DIST_BACKEND") or (
+            "nccl" if have_cuda else "gloo")
+        dist.init_process_group(backend, rank=rank, world_size=world_size)
 
     c = cfg.apply(args.preset, device="cuda" if have_cuda else "cpu")
     use_hip = args.engine == "hip" or (args.engine == "auto" and have_cuda)

@@ -79,9 +79,18 @@ class GpuReplayBuffer:
         # buffers and hipMemcpyAsync on a copy stream, overlapped with
         # compute on the default stream)
         self._copy_stream = torch.cuda.Stream(device=dev)
-        self._pin_obs = torch.empty(self.obs_rows, self.obs_store.shape[-1],
-                                    dtype=torch.uint8, pin_memory=True)
-        self._copy_done = torch.cuda.Event()
+        # rotating pinned staging buffers: the host only waits for the
+        # buffer N-ago to drain (no per-block stream sync), and ALL ingest
+        # device work runs on the copy stream — the training stream never
+        # waits on ingest (the sample stream fences on _ev_ingest instead)
+        self._n_pin = 4
+        self._pin_obs = [torch.empty(self.obs_rows, self.obs_store.shape[-1],
+                                     dtype=torch.uint8, pin_memory=True)
+                         for _ in range(self._n_pin)]
+        self._pin_ev = [torch.cuda.Event() for _ in range(self._n_pin)]
+        self._pin_used = [False] * self._n_pin
+        self._pin_cursor = 0
+        self._ev_ingest = torch.cuda.Event()
         # side stream for prefetch sampling (sample_async): sampling of the
         # NEXT batch overlaps the current training step.  Tree reads and
         # writes are fenced both ways: the sample stream waits on the last
@@ -118,7 +127,6 @@ class GpuReplayBuffer:
         steps = block.action.shape[0]
         nseq = block.num_sequences
         dev = self.device
-        torch.cuda.current_stream().wait_event(self._ev_sampled)
 
         # store frames HWC (channels innermost) — the conv kernels are NHWC,
         # so gathered batches feed conv1 with no device-side permute
@@ -127,32 +135,20 @@ class GpuReplayBuffer:
             obs_np = np.ascontiguousarray(obs_np.transpose(0, 2, 3, 1))
         obs_flat = torch.from_numpy(obs_np.reshape(rows, -1))
         assert obs_flat.dtype == torch.uint8
-        # frames: host memcpy into the pinned buffer, then async H2D on the
-        # copy stream; the default stream waits on the copy event so gather
-        # kernels never read a half-written block
-        self._copy_stream.synchronize()       # pin buffer free to reuse
-        self._pin_obs[:rows].copy_(obs_flat)
-        with torch.cuda.stream(self._copy_stream):
-            self.obs_store[slot, :rows].copy_(self._pin_obs[:rows],
-                                              non_blocking=True)
-        self._copy_done.record(self._copy_stream)
-        torch.cuda.current_stream().wait_event(self._copy_done)
-        la_idx = torch.from_numpy(
-            np.ascontiguousarray(block.last_action.argmax(1).astype(np.uint8)))
-        self.la_store[slot, :rows].copy_(la_idx, non_blocking=True)
-        self.lr_store[slot, :rows].copy_(
-            torch.from_numpy(block.last_reward), non_blocking=True)
-        self.act_store[slot, :steps].copy_(
-            torch.from_numpy(block.action), non_blocking=True)
-        self.nsr_store[slot, :steps].copy_(
-            torch.from_numpy(block.n_step_reward), non_blocking=True)
-        self.gam_store[slot, :steps].copy_(
-            torch.from_numpy(block.gamma), non_blocking=True)
-        base = slot * self.spb
-        self.hid_store[base: base + nseq].copy_(
-            torch.from_numpy(block.hidden.reshape(nseq, -1)), non_blocking=True)
+        # frames: host memcpy into the next rotating pinned buffer (waiting
+        # only for THAT buffer's previous H2D, n_pin blocks ago), then every
+        # device-side ingest op — H2D copies, metadata, tree update — runs
+        # on the copy stream.  The training stream never waits on ingest;
+        # sample_async fences its gathers on _ev_ingest instead.
+        pi = self._pin_cursor
+        self._pin_cursor = (pi + 1) % self._n_pin
+        if self._pin_used[pi]:
+            self._pin_ev[pi].synchronize()
+        pin = self._pin_obs[pi]
+        pin[:rows].copy_(obs_flat)
+        self._pin_used[pi] = True
 
-        # per-sequence metadata
+        # per-sequence metadata (host side)
         burn = block.burn_in_steps.astype(np.int32)
         learn = block.learning_steps.astype(np.int32)
         fwd = block.forward_steps.astype(np.int32)
@@ -165,20 +161,44 @@ class GpuReplayBuffer:
         meta[2, :nseq] = fwd
         meta[3, :nseq] = obs_start
         meta[4, :nseq] = learn_off
-        mt = torch.from_numpy(meta).to(dev, non_blocking=True)
-        sl = slice(base, base + self.spb)
-        self.burn_s[sl] = mt[0]
-        self.learn_s[sl] = mt[1]
-        self.fwd_s[sl] = mt[2]
-        self.obs_start_s[sl] = mt[3]
-        self.learn_off_s[sl] = mt[4]
+        base = slot * self.spb
 
-        # priorities (zero for unused seq slots kills them in the tree)
-        idxes = torch.arange(base, base + self.spb, dtype=torch.int64, device=dev)
-        prio = torch.from_numpy(
-            np.ascontiguousarray(priorities.astype(np.float32))).to(dev, non_blocking=True)
-        self._ext.sumtree_update(self.tree, self.leaf_offset, idxes, prio,
-                                 self.alpha, 0, 0, self.spb, self.num_blocks)
+        with torch.cuda.stream(self._copy_stream):
+            # don't overwrite a slot an in-flight sample may be gathering
+            self._copy_stream.wait_event(self._ev_sampled)
+            self.obs_store[slot, :rows].copy_(pin[:rows], non_blocking=True)
+            self._pin_ev[pi].record(self._copy_stream)
+            la_idx = torch.from_numpy(np.ascontiguousarray(
+                block.last_action.argmax(1).astype(np.uint8)))
+            self.la_store[slot, :rows].copy_(la_idx, non_blocking=True)
+            self.lr_store[slot, :rows].copy_(
+                torch.from_numpy(block.last_reward), non_blocking=True)
+            self.act_store[slot, :steps].copy_(
+                torch.from_numpy(block.action), non_blocking=True)
+            self.nsr_store[slot, :steps].copy_(
+                torch.from_numpy(block.n_step_reward), non_blocking=True)
+            self.gam_store[slot, :steps].copy_(
+                torch.from_numpy(block.gamma), non_blocking=True)
+            self.hid_store[base: base + nseq].copy_(
+                torch.from_numpy(block.hidden.reshape(nseq, -1)),
+                non_blocking=True)
+            mt = torch.from_numpy(meta).to(dev, non_blocking=True)
+            sl = slice(base, base + self.spb)
+            self.burn_s[sl] = mt[0]
+            self.learn_s[sl] = mt[1]
+            self.fwd_s[sl] = mt[2]
+            self.obs_start_s[sl] = mt[3]
+            self.learn_off_s[sl] = mt[4]
+
+            # priorities (zero for unused seq slots kills them in the tree)
+            idxes = torch.arange(base, base + self.spb, dtype=torch.int64,
+                                 device=dev)
+            prio = torch.from_numpy(np.ascontiguousarray(
+                priorities.astype(np.float32))).to(dev, non_blocking=True)
+            self._ext.sumtree_update(self.tree, self.leaf_offset, idxes, prio,
+                                     self.alpha, 0, 0, self.spb,
+                                     self.num_blocks)
+            self._ev_ingest.record(self._copy_stream)
 
         self.size += int(learn.sum())
         if self.size > self.capacity:
@@ -197,6 +217,9 @@ class GpuReplayBuffer:
         self._ev_tree.record(torch.cuda.current_stream())
         with torch.cuda.stream(self._sample_stream):
             self._sample_stream.wait_event(self._ev_tree)
+            # gathers must see fully-ingested blocks (frames + metadata +
+            # tree), which the copy stream publishes via _ev_ingest
+            self._sample_stream.wait_event(self._ev_ingest)
             jitter = torch.rand(B, device=dev)
             # fp-edge descent overshoot is clamped to the highest WRITTEN
             # leaf, not tree capacity — a partially filled ring must never

@@ -35,11 +35,12 @@ def _run_actor(epsilon, model, sample_queue, seed, config_dict=None):
 
 
 def _run_vector_actor(epsilons, model, sample_queues, device, seed,
-                      config_dict=None):
+                      config_dict=None, weight_bus=None):
     if config_dict:
         cfg.apply(**config_dict)
     from .worker import VectorActor
-    va = VectorActor(epsilons, model, sample_queues, device=device, seed=seed)
+    va = VectorActor(epsilons, model, sample_queues, device=device, seed=seed,
+                     weight_bus=weight_bus)
     va.run()
 
 
@@ -86,11 +87,14 @@ def train(seed: int = 0, restart_dead_actors: bool = True,
         learner.enable_hip_engine()
 
     if c.vector_actors:
-        # one driver process, all envs in lockstep, batched inference
+        # one driver process, all envs in lockstep, batched inference.
+        # The learner's WeightBus (device-resident prepacked weights +
+        # version word) travels to the spawned child via CUDA IPC — pulls
+        # become same-GPU slice copies instead of CPU state_dict loads.
         spawners = [lambda: ctx.Process(
             target=_run_vector_actor,
             args=(epsilon_ladder(), model, sample_queues, c.actor_device,
-                  seed + 1, config_dict))]
+                  seed + 1, config_dict, learner.weight_bus))]
     else:
         spawners = [
             (lambda eps=eps, i=i: ctx.Process(

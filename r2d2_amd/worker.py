@@ -551,15 +551,26 @@ class Learner:
             from .ops.engine import HipNetworkEngine
             self.engine = HipNetworkEngine(self.online_net, self.target_net,
                                            self.device, c)
+            # device-resident weight distribution to same-GPU VectorActors
+            # (replaces the reference's GPU->CPU shared-memory
+            # publish + per-pull repack, worker.py:306-307,560-566).  The
+            # spec comes from the target pack — same tensor names/shapes as
+            # the actor-side inference pack (with_bwd=False).
+            from .parallel.weight_bus import WeightBus, pack_tensors
+            self.weight_bus = WeightBus(pack_tensors(self.engine.target),
+                                        self.device)
 
     hip_engine = False
     engine = None
+    weight_bus = None
 
     # -- weight publication -------------------------------------------------
 
     def store_weights(self):
         state = {k: v.cpu() for k, v in self.online_net.state_dict().items()}
         self.shared_model.load_state_dict(state)
+        if self.weight_bus is not None and self.engine is not None:
+            self.weight_bus.publish(self.engine.online)
 
     # -- data staging -------------------------------------------------------
 
@@ -736,15 +747,22 @@ class Learner:
                 print(f"filling GPU replay: {len(replay)}/{c.learning_starts}")
                 last_log = time.time()
 
+        # side-stream prefetch: the NEXT batch samples/gathers on the
+        # replay's sample stream while the current step trains (one step of
+        # priority staleness — far inside the reference's own <=12-batch
+        # staleness)
+        pending = None
         while self.num_updates < c.training_steps:
             with lock:
-                batch = replay.sample()
+                batch = pending if pending is not None else replay.sample()
+                tok = replay.sample_async()
             loss, priorities = self.train_step(batch)
             if not torch.is_tensor(priorities):   # eager path returns numpy
                 priorities = torch.as_tensor(priorities, device=self.device)
             with lock:
                 replay.update_priorities(batch.idxes, priorities,
                                          batch.old_ptr)
+                pending = replay.sample_wait(tok)
             stats["sum_loss"] += float(loss)
             self.env_steps = replay.env_steps
             if self.num_updates % 4 == 0:
@@ -962,7 +980,7 @@ class VectorActor:
 
     def __init__(self, epsilons: List[float], model: Network, sample_queues,
                  num_envs: Optional[int] = None, device: str = "cpu",
-                 seed: Optional[int] = None):
+                 seed: Optional[int] = None, weight_bus=None):
         c = cfg.get()
         self.cfg = c
         E = num_envs or len(epsilons)
@@ -1011,11 +1029,16 @@ class VectorActor:
         # K15 fast path: batched single-step inference through the gfx950
         # kernels when running on a GPU with the extension built
         self.hip_inf = None
+        self.weight_bus = weight_bus
+        self._bus_ver = 0
         if (self.device.type == "cuda" and c.use_hip_kernels
                 and c.encoder in ("nature", "impala") and c.hidden_dim == 512
                 and tuple(c.obs_shape) == (4, 84, 84)):
             from .ops.engine import HipInference
             self.hip_inf = HipInference(self.model, self.device)
+            if self.weight_bus is not None:   # catch up if already published
+                self._bus_ver = self.weight_bus.pull_into(
+                    self.hip_inf.pack, 0)
 
     def _queue(self, i):
         return self.queues[i % len(self.queues)]
@@ -1133,7 +1156,15 @@ class VectorActor:
         return total_steps
 
     def pull_weights(self):
-        """Refresh inference weights from the learner's published copy."""
+        """Refresh inference weights from the learner's published copy.
+        With a WeightBus on a cuda device this is a device-to-device slice
+        copy straight into the prepacked inference tensors (no state_dict,
+        no repack, no host round trip); otherwise the reference's
+        shared-CPU-model path (worker.py:560-566)."""
+        if self.weight_bus is not None and self.hip_inf is not None:
+            self._bus_ver = self.weight_bus.pull_into(self.hip_inf.pack,
+                                                      self._bus_ver)
+            return
         self.model.load_state_dict(self.shared_model.state_dict())
         if self.hip_inf is not None:
             self.hip_inf.refresh()
