@@ -1155,16 +1155,26 @@ class VectorActor:
                 self.pull_weights()
         return total_steps
 
+    _pulls = 0
+    _pull_seconds = 0.0
+
     def pull_weights(self):
         """Refresh inference weights from the learner's published copy.
         With a WeightBus on a cuda device this is a device-to-device slice
         copy straight into the prepacked inference tensors (no state_dict,
         no repack, no host round trip); otherwise the reference's
         shared-CPU-model path (worker.py:560-566)."""
+        t0 = time.perf_counter()
         if self.weight_bus is not None and self.hip_inf is not None:
             self._bus_ver = self.weight_bus.pull_into(self.hip_inf.pack,
                                                       self._bus_ver)
-            return
-        self.model.load_state_dict(self.shared_model.state_dict())
-        if self.hip_inf is not None:
-            self.hip_inf.refresh()
+        else:
+            self.model.load_state_dict(self.shared_model.state_dict())
+            if self.hip_inf is not None:
+                self.hip_inf.refresh()
+        self._pulls += 1
+        self._pull_seconds += time.perf_counter() - t0
+        if self._pulls % 200 == 0:
+            print(f"[vector-actor] weight pulls: {self._pulls}, "
+                  f"avg {self._pull_seconds / self._pulls * 1e3:.2f} ms "
+                  f"({'bus' if self.weight_bus is not None else 'cpu'})")

@@ -80,3 +80,39 @@ def test_engine_training_reduces_loss():
         last = float(loss)
     assert np.isfinite(last)
     assert last < first, (first, last)
+
+
+def test_weight_bus_roundtrip():
+    """WeightBus publish/pull: the learner's packed online weights land
+    bit-exactly in a separate inference pack (same-process; the live demo
+    exercises the cross-process CUDA-IPC path)."""
+    from r2d2_amd.parallel.weight_bus import WeightBus, pack_tensors
+    from r2d2_amd.ops.engine import HipInference
+
+    c, eager, hip = make_learners(seed=3)
+    engine = hip.engine
+    bus = WeightBus(pack_tensors(engine.target), "cuda")
+
+    # a second network with different weights = the actor's inference pack
+    torch.manual_seed(99)
+    from r2d2_amd.models.network import Network
+    other = Network(c.action_dim, c.obs_shape, c.hidden_dim,
+                    encoder=c.encoder, forward_steps=c.forward_steps,
+                    mlp_hidden=c.mlp_hidden).cuda()
+    inf = HipInference(other, torch.device("cuda"), c)
+
+    before = {n: t.clone() for n, t in pack_tensors(inf.pack).items()}
+    bus.publish(engine.online)
+    ver = bus.pull_into(inf.pack, 0)
+    assert ver == 1
+
+    src = pack_tensors(engine.online)
+    dst = pack_tensors(inf.pack)
+    changed = 0
+    for name, t in dst.items():
+        assert torch.equal(t, src[name].view(t.shape)), name
+        if not torch.equal(t, before[name]):
+            changed += 1
+    assert changed > 0
+    # unchanged version -> no-op
+    assert bus.pull_into(inf.pack, ver) == ver
