@@ -429,21 +429,26 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
         }
         __syncthreads();
 
+        // fragment gathers via hardware transpose-reads (common.h)
         bf16x8 fa[2];
 #pragma unroll
-        for (int i = 0; i < 2; ++i) {
-#pragma unroll
-            for (int e = 0; e < 8; ++e)
-                fa[i][e] = *(const __bf16*)&s_dy[mseg + e][wr * 32 + i * 16 + frow];
-        }
+        for (int i = 0; i < 2; ++i)
+            fa[i] = lds_col_frag8<64 + 8>(&s_dy[0][0], mseg,
+                                          wr * 32 + i * 16, lane);
 #pragma unroll
         for (int kf = 0; kf < KFRAG; ++kf) {
-            int kcol = wc * KHALF + kf * 16 + frow;
-            bf16x8 fb = czero();
-            if (kcol < K) {
+            int kcol0 = wc * KHALF + kf * 16;
+            bf16x8 fb;
+            if (kcol0 + 16 <= K) {
+                fb = lds_col_frag8<K + 8>(&s_a[0][0], mseg, kcol0, lane);
+            } else {                       // partial tail column span
+                fb = czero();
+                int kcol = kcol0 + frow;
+                if (kcol < K) {
 #pragma unroll
-                for (int e = 0; e < 8; ++e)
-                    fb[e] = *(const __bf16*)&s_a[mseg + e][kcol];
+                    for (int e = 0; e < 8; ++e)
+                        fb[e] = *(const __bf16*)&s_a[mseg + e][kcol];
+                }
             }
 #pragma unroll
             for (int i = 0; i < 2; ++i)

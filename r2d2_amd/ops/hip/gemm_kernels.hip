@@ -267,17 +267,16 @@ __global__ __launch_bounds__(256) void gemm_wgrad_kernel(
 
         // MFMA: out tile (n, k); gemm-M = n-dim, gemm-N = k-dim, gemm-K = m
         // A-frag: s_dy column (n = row0 + frow), 8 m values
-        // B-frag: s_a column (k), 8 m values — both are LDS column reads
-#pragma unroll
-        for (int ks = 0; ks < 32; ks += 32) {
+        // B-frag: s_a column (k), 8 m values — both gathered by hardware
+        // transpose-reads (ds_read_b64_tr_b16, common.h lds_col_frag8)
+        {
             bf16x8 fa[2], fb[2];
 #pragma unroll
             for (int i = 0; i < 2; ++i) {
-#pragma unroll
-                for (int e = 0; e < 8; ++e) {
-                    fa[i][e] = *(const __bf16*)&s_dy[mseg + e][wr * 32 + i * 16 + frow];
-                    fb[i][e] = *(const __bf16*)&s_a[mseg + e][wc * 32 + i * 16 + frow];
-                }
+                fa[i] = lds_col_frag8<64 + 8>(&s_dy[0][0], mseg,
+                                              wr * 32 + i * 16, lane);
+                fb[i] = lds_col_frag8<64 + 8>(&s_a[0][0], mseg,
+                                              wc * 32 + i * 16, lane);
             }
 #pragma unroll
             for (int i = 0; i < 2; ++i)

@@ -412,20 +412,25 @@ __global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
         __syncthreads();
 
         const int mbase = wm * 32 + mseg;
+        // fragment gathers via hardware transpose-reads (common.h)
         bf16x8 fa[NCO];
 #pragma unroll
         for (int i = 0; i < NCO; ++i)
-#pragma unroll
-            for (int e = 0; e < 8; ++e)
-                fa[i][e] = *(const __bf16*)&s_dy[mbase + e][i * 16 + frow];
+            fa[i] = lds_col_frag8<32 + 8>(&s_dy[0][0], mbase, i * 16, lane);
 #pragma unroll
         for (int kf = 0; kf < KFRAG; ++kf) {
-            int kcol = wk * KHALF + kf * 16 + frow;
-            bf16x8 fb = izero();
-            if (kcol < K) {
+            int kcol0 = wk * KHALF + kf * 16;
+            bf16x8 fb;
+            if (kcol0 + 16 <= K) {
+                fb = lds_col_frag8<K + 8>(&s_a[0][0], mbase, kcol0, lane);
+            } else {
+                fb = izero();
+                int kcol = kcol0 + frow;
+                if (kcol < K) {
 #pragma unroll
-                for (int e = 0; e < 8; ++e)
-                    fb[e] = *(const __bf16*)&s_a[mbase + e][kcol];
+                    for (int e = 0; e < 8; ++e)
+                        fb[e] = *(const __bf16*)&s_a[mbase + e][kcol];
+                }
             }
 #pragma unroll
             for (int i = 0; i < NCO; ++i)
@@ -569,20 +574,26 @@ __global__ __launch_bounds__(256) void conv3p_wgrad_band_kernel(
             __syncthreads();
 
             const int mbase = wm * 32 + mseg;
+            // fragment gathers via hardware transpose-reads (common.h)
             bf16x8 fa[NCO];
 #pragma unroll
             for (int i = 0; i < NCO; ++i)
-#pragma unroll
-                for (int e = 0; e < 8; ++e)
-                    fa[i][e] = *(const __bf16*)&s_dy[mbase + e][i * 16 + frow];
+                fa[i] = lds_col_frag8<32 + 8>(&s_dy[0][0], mbase, i * 16,
+                                              lane);
 #pragma unroll
             for (int kf = 0; kf < KFRAG; ++kf) {
-                int kcol = wk * KHALF + kf * 16 + frow;
-                bf16x8 fb = izero();
-                if (kcol < K) {
+                int kcol0 = wk * KHALF + kf * 16;
+                bf16x8 fb;
+                if (kcol0 + 16 <= K) {
+                    fb = lds_col_frag8<K + 8>(&s_a[0][0], mbase, kcol0, lane);
+                } else {
+                    fb = izero();
+                    int kcol = kcol0 + frow;
+                    if (kcol < K) {
 #pragma unroll
-                    for (int e = 0; e < 8; ++e)
-                        fb[e] = *(const __bf16*)&s_a[mbase + e][kcol];
+                        for (int e = 0; e < 8; ++e)
+                            fb[e] = *(const __bf16*)&s_a[mbase + e][kcol];
+                    }
                 }
 #pragma unroll
                 for (int i = 0; i < NCO; ++i)
