@@ -37,17 +37,132 @@ def _round_up(x, m):
     return (x + m - 1) // m * m
 
 
+def _pack_items(pack, prefix="", out=None):
+    """(name, holder, attr, dict_key) for every floating cuda tensor in a
+    pack — same traversal order as parallel.weight_bus.pack_tensors."""
+    out = [] if out is None else out
+    for name in sorted(vars(pack)):
+        if name.startswith("_"):
+            continue
+        v = vars(pack)[name]
+        p = f"{prefix}{name}"
+        if torch.is_tensor(v):
+            if v.is_cuda and v.is_floating_point():
+                out.append((p, pack, name, None))
+        elif isinstance(v, dict):
+            for k in sorted(v, key=repr):
+                t = v[k]
+                if torch.is_tensor(t) and t.is_cuda and t.is_floating_point():
+                    out.append((f"{p}[{k!r}]", pack, name, k))
+        elif hasattr(v, "__dict__") and hasattr(v, "refresh"):
+            _pack_items(v, p + ".", out)
+    return out
+
+
+def _item_get(holder, attr, key):
+    v = getattr(holder, attr)
+    return v if key is None else v[key]
+
+
+def _item_set(holder, attr, key, t):
+    if key is None:
+        setattr(holder, attr, t)
+    else:
+        getattr(holder, attr)[key] = t
+
+
 class _NetPack:
-    """Prepacked bf16 weights for one network (online or target)."""
+    """Prepacked bf16 weights for one network (online or target).
+
+    ``refresh()`` re-derives every packed tensor from the module weights.
+    When the owning net's parameters are views of the engine's flat f32
+    buffer, ``enable_fast_refresh`` replaces the ~40-launch python repack
+    with ONE gather kernel per dtype: the (purely linear) packing is probed
+    bit-by-bit once at init to recover, for every packed element, its
+    source index in the flat buffer (plus the second source for the
+    b_ih + b_hh LSTM bias sum), and the pack tensors become views of two
+    flat output buffers filled by ops/hip/optim_kernels.hip gather_pack.
+    """
 
     def __init__(self, net, device, A, with_bwd: bool):
         self.net = net
         self.device = device
         self.A = A
         self.with_bwd = with_bwd
+        self._fast = None
         self.refresh()
 
     def refresh(self):
+        if self._fast is not None:
+            m, flat, m1b, m2b, bufb, m1f, m2f, buff = self._fast
+            m.gather_pack(flat, m1b, m2b, bufb)
+            m.gather_pack(flat, m1f, m2f, buff)
+            return
+        self._refresh_py()
+
+    def enable_fast_refresh(self, flat_param, m):
+        """Bit-probe the packing to build index maps, then freeze the pack
+        tensors as views of two flat buffers refreshed by gather_pack."""
+        dev = self.device
+        n = flat_param.numel()
+        save = flat_param.detach().clone()
+        # the only dual-source packed tensor is lstm_bias = b_ih + b_hh;
+        # zero b_hh during probing so probed indices point at b_ih, and add
+        # the (constant) hh offset analytically afterwards
+        bih, bhh = self.net.recurrent.bias_ih_l0, self.net.recurrent.bias_hh_l0
+        ih_ofs = bih.data.storage_offset()
+        hh_ofs = bhh.data.storage_offset()
+        hh_n = bhh.numel()
+
+        idx_plus1 = torch.arange(1, n + 1, dtype=torch.int64, device=dev)
+        acc = {}
+        for b in range(24):
+            bits = ((idx_plus1 >> b) & 1).float()
+            bits[hh_ofs:hh_ofs + hh_n] = 0.0
+            with torch.no_grad():
+                flat_param.copy_(bits)
+            self._refresh_py()
+            for name, holder, attr, key in _pack_items(self):
+                t = _item_get(holder, attr, key)
+                bit = (t.detach().float() != 0).to(torch.int64).reshape(-1)
+                acc[name] = acc.get(name, 0) + (bit << b)
+        with torch.no_grad():
+            flat_param.copy_(save)
+        self._refresh_py()
+
+        items = _pack_items(self)
+        sizes = {torch.bfloat16: 0, torch.float32: 0}
+        layout = []
+        for name, holder, attr, key in items:
+            t = _item_get(holder, attr, key)
+            layout.append((name, holder, attr, key, t.dtype, t.shape,
+                           sizes[t.dtype]))
+            sizes[t.dtype] += t.numel()
+        bufb = torch.zeros(max(1, sizes[torch.bfloat16]),
+                           dtype=torch.bfloat16, device=dev)
+        buff = torch.zeros(max(1, sizes[torch.float32]),
+                           dtype=torch.float32, device=dev)
+        m1b = torch.full((max(1, sizes[torch.bfloat16]),), -1,
+                         dtype=torch.int32, device=dev)
+        m2b = torch.full_like(m1b, -1)
+        m1f = torch.full((max(1, sizes[torch.float32]),), -1,
+                         dtype=torch.int32, device=dev)
+        m2f = torch.full_like(m1f, -1)
+        for name, holder, attr, key, dtype, shape, ofs in layout:
+            t = _item_get(holder, attr, key)
+            nel = t.numel()
+            buf = bufb if dtype == torch.bfloat16 else buff
+            m1 = m1b if dtype == torch.bfloat16 else m1f
+            m2 = m2b if dtype == torch.bfloat16 else m2f
+            buf[ofs:ofs + nel].copy_(t.detach().reshape(-1).to(dtype))
+            map1 = (acc[name] - 1).to(torch.int32)
+            m1[ofs:ofs + nel] = map1
+            if name == "lstm_bias":
+                m2[ofs:ofs + nel] = map1 + (hh_ofs - ih_ofs)
+            _item_set(holder, attr, key, buf[ofs:ofs + nel].view(shape))
+        self._fast = (m, flat_param, m1b, m2b, bufb, m1f, m2f, buff)
+
+    def _refresh_py(self):
         net = self.net
         dev = self.device
         A = self.A
@@ -207,6 +322,9 @@ class HipNetworkEngine:
             ofs += n
         self.timing = bool(os.environ.get("R2D2_ENGINE_TIMING"))
         self._events = []
+        # per-step repack -> ONE gather kernel per dtype (maps bit-probed
+        # from the now-flat param buffer; see _NetPack.enable_fast_refresh)
+        self.online.enable_fast_refresh(self.flat_param, self.m)
 
     def _mark(self, name):
         if self.timing:

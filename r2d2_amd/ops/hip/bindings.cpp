@@ -108,6 +108,8 @@ void dense2pad_mask(torch::Tensor dflat, torch::Tensor act_pad,
                     torch::Tensor out, int64_t N, int64_t H, int64_t W);
 
 // optim_kernels.hip
+void gather_pack(torch::Tensor flat, torch::Tensor m1, torch::Tensor m2,
+                 torch::Tensor out);
 torch::Tensor grad_sumsq(torch::Tensor grad, torch::Tensor norm_buf);
 void adam_step(torch::Tensor param, torch::Tensor grad, torch::Tensor m,
                torch::Tensor v, torch::Tensor norm_buf, double max_norm,
@@ -177,6 +179,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("pad2dense", &pad2dense, "padded NHWC -> dense rows (+relu)");
     m.def("dense2pad_mask", &dense2pad_mask,
           "dense grad -> padded, masked by act>0 (relu backward)");
+    m.def("gather_pack", &gather_pack,
+          "regenerate prepacked weights from the flat param buffer "
+          "(index-map gather, one launch per dtype)");
     m.def("grad_sumsq", &grad_sumsq, "flat gradient squared-norm reduction");
     m.def("adam_step", &adam_step,
           "fused multi-tensor clip + Adam on the flat parameter buffer");

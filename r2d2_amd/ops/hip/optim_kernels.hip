@@ -72,6 +72,51 @@ __global__ void adam_step_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// gather_pack: regenerate ALL prepacked weight tensors from the flat f32
+// parameter buffer in ONE launch per output dtype.  Each packed element is
+// a permutation (map1) of flat_param, optionally plus a second source
+// (map2 — the LSTM bias sum b_ih + b_hh); -1 in map1 means structural zero
+// padding.  Replaces the ~40 permute/cast/zero launches of the python
+// repack that otherwise run EVERY optimizer step (engine refresh_online).
+// ---------------------------------------------------------------------------
+template <bool OUT_BF16>
+__global__ void gather_pack_kernel(
+    const float* __restrict__ flat, const int* __restrict__ m1,
+    const int* __restrict__ m2, void* __restrict__ out, long n) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long stride = (long)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        int a = m1[i];
+        float v = (a >= 0) ? flat[a] : 0.f;
+        int b = m2[i];
+        if (b >= 0) v += flat[b];
+        if (OUT_BF16)
+            reinterpret_cast<__hip_bfloat16*>(out)[i] = f2bf(v);
+        else
+            reinterpret_cast<float*>(out)[i] = v;
+    }
+}
+
+void gather_pack(torch::Tensor flat, torch::Tensor m1, torch::Tensor m2,
+                 torch::Tensor out) {
+    TORCH_CHECK(flat.is_cuda() && flat.dtype() == torch::kFloat32);
+    TORCH_CHECK(m1.dtype() == torch::kInt32 && m2.dtype() == torch::kInt32);
+    long n = out.numel();
+    if (n == 0) return;
+    TORCH_CHECK(m1.numel() == n && m2.numel() == n);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    int blocks = (int)std::min(2048L, (n + 255) / 256);
+    if (out.dtype() == torch::kBFloat16)
+        hipLaunchKernelGGL(gather_pack_kernel<true>, dim3(blocks), dim3(256),
+                           0, stream.stream(), flat.data_ptr<float>(),
+                           m1.data_ptr<int>(), m2.data_ptr<int>(),
+                           out.data_ptr(), n);
+    else
+        hipLaunchKernelGGL(gather_pack_kernel<false>, dim3(blocks), dim3(256),
+                           0, stream.stream(), flat.data_ptr<float>(),
+                           m1.data_ptr<int>(), m2.data_ptr<int>(),
+                           out.data_ptr(), n);
+}
 
 torch::Tensor grad_sumsq(torch::Tensor grad, torch::Tensor norm_buf) {
     TORCH_CHECK(grad.is_cuda() && grad.dtype() == torch::kFloat32
