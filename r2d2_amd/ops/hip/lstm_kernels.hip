@@ -219,11 +219,15 @@ __device__ __forceinline__ bool grid_barrier(GridBar* bar, unsigned epoch,
 // ---------------------------------------------------------------------------
 // Forward
 // ---------------------------------------------------------------------------
-// BROWS: batch rows per workgroup.  64 = classic layout (one wg per 8
-// hidden units, whole batch).  32 = batch-split: twice the workgroups
-// (256 for the dual-net launch -> every CU busy), half the LDS staging
-// per step, flag groups per (net, half).
-template <int H, int BROWS = 64>
+// BROWS: batch rows per workgroup.  64 = classic layout (whole batch).
+// 32 = batch-split: twice the workgroups, half the LDS staging per step,
+// flag groups per (net, half).
+// UNITS: hidden units per workgroup.  8 = classic; 16 halves the workgroup
+// count (half the handoff producers per counter group AND half the total
+// h re-read traffic) at twice the per-wg MFMA work — the per-step cost is
+// latency-dominated, so fewer/fatter workgroups win (measured; VERDICT r1
+// item 7).
+template <int H, int BROWS = 64, int UNITS = 8>
 __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
     const __hip_bfloat16* __restrict__ X0,    // (B, T, 4H)
     const __hip_bfloat16* __restrict__ X1,    // or null
@@ -238,13 +242,14 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
     float* __restrict__ Cout1,
     __hip_bfloat16* __restrict__ stash0,      // (B, T, 4H) post-nonlin gates
     GridBar* bar, int B, int T, int nblocks, int nhalves) {
-    constexpr int WGS_PER_HALF = H / 8;
+    constexpr int WGS_PER_HALF = H / UNITS;
+    constexpr int GCOLS = 4 * UNITS;          // gate columns per wg
     const int wgs_per_net = WGS_PER_HALF * nhalves;
     const int net = blockIdx.x / wgs_per_net;
     const int rem = blockIdx.x % wgs_per_net;
     const int half = rem / WGS_PER_HALF;
     const int wid = rem % WGS_PER_HALF;
-    const int u0 = wid * 8;
+    const int u0 = wid * UNITS;
     const int b0 = half * BROWS;
     const int Bl = min(B - b0, BROWS);
 
@@ -256,21 +261,21 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
     __hip_bfloat16* stash = net ? nullptr : stash0;
     unsigned* ctr = &bar->flags[(net * nhalves + half) * 32];
 
-    __shared__ __hip_bfloat16 s_whh[32][H + 8];
+    __shared__ __hip_bfloat16 s_whh[GCOLS][H + 8];
     __shared__ __hip_bfloat16 s_h[BROWS][H + 8];
-    __shared__ float s_gates[BROWS][32 + 4];
-    __shared__ float s_c[BROWS][8];
-    __shared__ __hip_bfloat16 s_hrow[BROWS][8];
+    __shared__ float s_gates[BROWS][GCOLS + 4];
+    __shared__ float s_c[BROWS][UNITS];
+    __shared__ __hip_bfloat16 s_hrow[BROWS][UNITS];
 
-    for (int e = threadIdx.x * 8; e < 32 * H; e += blockDim.x * 8) {
+    for (int e = threadIdx.x * 8; e < GCOLS * H; e += blockDim.x * 8) {
         int c = e / H;
         int k = e % H;
-        int g = c / 8, j = c % 8;
+        int g = c / UNITS, j = c % UNITS;
         lstore8(&s_whh[c][k], lload8(Whh + (long)(g * H + u0 + j) * H + k));
     }
     // h0 -> Hout[:,0] (this wg's slice); c0 -> LDS-resident cell state
-    for (int p = threadIdx.x; p < Bl * 8; p += blockDim.x) {
-        int bl = p / 8, j = p % 8;
+    for (int p = threadIdx.x; p < Bl * UNITS; p += blockDim.x) {
+        int bl = p / UNITS, j = p % UNITS;
         int b = b0 + bl;
         int u = u0 + j;
         Hout[((long)b * (T + 1)) * H + u] = f2bf(init[(long)b * H + u]);
@@ -281,10 +286,13 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
 
     const int wave = threadIdx.x / WAVE;
     const int lane = threadIdx.x & (WAVE - 1);
-    // BROWS=64: 2 A-frags per wave (rows (wave>>1)*32 ..); BROWS=32: 1
-    constexpr int NFRAG = BROWS / 32;
-    const int wrow0 = (wave >> 1) * 16 * NFRAG;
-    const int wcol0 = (wave & 1) * 16;         // cols (gate local)
+    // wave tiling over (BROWS rows x GCOLS cols):
+    //   GCOLS=32: 2x2 waves of 16x16xNFRAG; GCOLS=64: 4 col-waves, rows
+    //   split into BROWS/16 fragments per wave.
+    constexpr int NCW = GCOLS / 16;            // col fragments total
+    constexpr int NFRAG = (NCW == 4) ? BROWS / 16 : BROWS / 32;
+    const int wrow0 = (NCW == 4) ? 0 : (wave >> 1) * 16 * NFRAG;
+    const int wcol0 = (NCW == 4) ? wave * 16 : (wave & 1) * 16;
     const int frow = lane & 15;
     const int kseg = (lane >> 4) * 8;
 
@@ -336,24 +344,28 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
             int row = threadIdx.x / 4;
             int g = threadIdx.x % 4;
             if (row < Bl) {
-                bf16x8 x8 = lload8(
-                    X + ((long)(b0 + row) * T + t) * 4 * H + g * H + u0);
 #pragma unroll
-                for (int j = 0; j < 8; ++j)
-                    s_gates[row][g * 8 + j] += (float)x8[j];
+                for (int j8 = 0; j8 < UNITS; j8 += 8) {
+                    bf16x8 x8 = lload8(
+                        X + ((long)(b0 + row) * T + t) * 4 * H + g * H + u0
+                        + j8);
+#pragma unroll
+                    for (int j = 0; j < 8; ++j)
+                        s_gates[row][g * UNITS + j8 + j] += (float)x8[j];
+                }
             }
         }
         __syncthreads();
-        for (int p = threadIdx.x; p < Bl * 8; p += blockDim.x) {
-            int b = p / 8, j = p % 8;
+        for (int p = threadIdx.x; p < Bl * UNITS; p += blockDim.x) {
+            int b = p / UNITS, j = p % UNITS;
             bool active = t < lens[b0 + b];
             float i_ = 0.f, f_ = 0.f, g_ = 0.f, o_ = 0.f;
             float c = s_c[b][j], h;
             if (active) {
-                i_ = sigmoidf_(s_gates[b][0 + j]);
-                f_ = sigmoidf_(s_gates[b][8 + j]);
-                g_ = tanhf(s_gates[b][16 + j]);
-                o_ = sigmoidf_(s_gates[b][24 + j]);
+                i_ = sigmoidf_(s_gates[b][0 * UNITS + j]);
+                f_ = sigmoidf_(s_gates[b][1 * UNITS + j]);
+                g_ = tanhf(s_gates[b][2 * UNITS + j]);
+                o_ = sigmoidf_(s_gates[b][3 * UNITS + j]);
                 c = f_ * c + i_ * g_;
                 h = o_ * tanhf(c);
             } else {
@@ -361,20 +373,31 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
             }
             s_c[b][j] = c;
             *(__bf16*)&s_hrow[b][j] = (__bf16)h;
-            s_gates[b][0 + j] = i_;
-            s_gates[b][8 + j] = f_;
-            s_gates[b][16 + j] = g_;
-            s_gates[b][24 + j] = o_;
+            s_gates[b][0 * UNITS + j] = i_;
+            s_gates[b][1 * UNITS + j] = f_;
+            s_gates[b][2 * UNITS + j] = g_;
+            s_gates[b][3 * UNITS + j] = o_;
         }
         __syncthreads();
-        // vectorized writers: h slice (16 B/row), c (32 B/row), gate stash
+        // vectorized writers: h slice, c, gate stash
         {
             int tid = threadIdx.x;
-            if (tid < 64 && tid < Bl) {
-                long off = ((long)(b0 + tid) * (T + 1) + t + 1) * H + u0;
-                lstore8(Hout + off, *reinterpret_cast<bf16x8*>(&s_hrow[tid][0]));
-            } else if (tid >= 64 && tid < 192) {
-                int b = (tid - 64) / 2, ch = (tid - 64) & 1;
+            constexpr int H8 = UNITS / 8;        // 16-B pieces per h row
+            if (tid < 64 * H8) {
+                int b = tid / H8, piece = tid % H8;
+                if (b < Bl) {
+                    long off = ((long)(b0 + b) * (T + 1) + t + 1) * H + u0
+                               + piece * 8;
+                    lstore8(Hout + off, *reinterpret_cast<bf16x8*>(
+                        &s_hrow[b][piece * 8]));
+                }
+            } else if (tid >= 128) {
+                // c writers: threads [128, 256) cover BROWS rows x C4
+                // float4 pieces (BROWS*C4 <= 128 for both layouts)
+                int t2 = tid - 128;
+                constexpr int C4 = UNITS / 4;    // float4 pieces per c row
+                static_assert(BROWS * C4 <= 128, "c-writer thread budget");
+                int b = t2 / C4, ch = t2 % C4;
                 if (b < Bl) {
                     long off = ((long)(b0 + b) * (T + 1) + t + 1) * H + u0
                                + ch * 4;
@@ -387,12 +410,15 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
             int row = threadIdx.x / 4;
             int g = threadIdx.x % 4;
             if (row < Bl) {
-                bf16x8 v;
 #pragma unroll
-                for (int j = 0; j < 8; ++j)
-                    v[j] = (__bf16)s_gates[row][g * 8 + j];
-                lstore8(stash + ((long)(b0 + row) * T + t) * 4 * H + g * H + u0,
-                        v);
+                for (int j8 = 0; j8 < UNITS; j8 += 8) {
+                    bf16x8 v;
+#pragma unroll
+                    for (int j = 0; j < 8; ++j)
+                        v[j] = (__bf16)s_gates[row][g * UNITS + j8 + j];
+                    lstore8(stash + ((long)(b0 + row) * T + t) * 4 * H
+                            + g * H + u0 + j8, v);
+                }
             }
         }
         publish_count(ctr);
@@ -728,8 +754,18 @@ std::vector<torch::Tensor> lstm_fwd(
     auto C1 = two ? torch::empty({B, T + 1, H}, f32) : torch::Tensor();
     auto stash = want_stash ? torch::empty({B, T, H4}, bf) : torch::Tensor();
 
-    int wgs = (int)H / 8;
+    // UNITS: hidden units per workgroup.  16 is the measured default for
+    // the dual-net batch-split launch (fewer/fatter workgroups: half the
+    // handoff producers per counter and half the total h re-read);
+    // R2D2_LSTM_UNITS=8 selects the classic layout for comparison.
+    static const int units_env = [] {
+        const char* e = getenv("R2D2_LSTM_UNITS");
+        return e ? atoi(e) : 16;
+    }();
+    int units = (units_env == 8) ? 8 : 16;
     int nhalves = B > 32 ? 2 : 1;
+    if (nhalves == 1) units = 8;   // single-half keeps the classic layout
+    int wgs = (int)H / units;
     int nblocks = wgs * (two ? 2 : 1) * nhalves;
     auto stream = at::cuda::getCurrentCUDAStream();
     zero_ws(barrier_ws, stream.stream());
@@ -738,8 +774,8 @@ std::vector<torch::Tensor> lstm_fwd(
         return t.defined()
             ? reinterpret_cast<__hip_bfloat16*>(t.data_ptr()) : nullptr;
     };
-#define LSTMF(BROWS_)                                                         \
-    hipLaunchKernelGGL((lstm_fwd_kernel<512, BROWS_>), dim3(nblocks),         \
+#define LSTMF(BROWS_, UNITS_)                                                 \
+    hipLaunchKernelGGL((lstm_fwd_kernel<512, BROWS_, UNITS_>), dim3(nblocks), \
         dim3(256), 0, stream.stream(),                                        \
         reinterpret_cast<const __hip_bfloat16*>(X0.data_ptr()),               \
         two ? reinterpret_cast<const __hip_bfloat16*>(X1.data_ptr()) : nullptr,\
@@ -752,7 +788,11 @@ std::vector<torch::Tensor> lstm_fwd(
         want_stash ? bp(stash) : nullptr,                                     \
         reinterpret_cast<GridBar*>(barrier_ws.data_ptr()),                    \
         (int)B, (int)T, nblocks, nhalves)
-    if (nhalves == 2) LSTMF(32); else LSTMF(64);
+    if (nhalves == 2) {
+        if (units == 16) LSTMF(32, 16); else LSTMF(32, 8);
+    } else {
+        LSTMF(64, 8);
+    }
 #undef LSTMF
 
     std::vector<torch::Tensor> out = {H0, C0};
