@@ -432,8 +432,11 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
 // ---------------------------------------------------------------------------
 // BROWS as in lstm_fwd_kernel: 32 doubles the workgroup count (batch
 // halves with their own flag groups) and splits the recurrent GEMM's K
-// reduction across wave pairs.
-template <int H, int BROWS = 64>
+// reduction across wave pairs.  UNITS=32 (with BROWS=32) halves the
+// workgroup count again — fewer/fatter workgroups, half the handoff
+// producers per counter, waves tile (row half x col half) at full K
+// (s_wb then holds 32 u-rows x 4H ~ 131 KB of the 160 KB LDS).
+template <int H, int BROWS = 64, int UNITS = 16>
 __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
     const __hip_bfloat16* __restrict__ stash,  // (B, T, 4H) i,f,g,o
     const float* __restrict__ Cout,            // (B, T+1, H)
@@ -443,38 +446,42 @@ __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
     const int* __restrict__ lens,
     __hip_bfloat16* __restrict__ dgates,       // (B, T, 4H) out
     GridBar* bar, int B, int T, int nblocks, int nhalves) {
-    constexpr int WGS = H / 16;
-    constexpr int KSPLIT = (BROWS == 32) ? 2 : 1;
+    constexpr int WGS = H / UNITS;
+    constexpr int KSPLIT = (UNITS == 16 && BROWS == 32) ? 2 : 1;
+    constexpr int NCWAVE = UNITS / 16;         // col fragments across waves
     const int half = blockIdx.x / WGS;
     const int wid = blockIdx.x % WGS;
-    const int u0 = wid * 16;
+    const int u0 = wid * UNITS;
     const int b0 = half * BROWS;
     const int Bl = min(B - b0, BROWS);
 
-    __shared__ __hip_bfloat16 s_wb[16][4 * H + 8];
-    __shared__ float s_dh[BROWS][16 + 1];
-    __shared__ float s_dc[BROWS][16 + 1];
-    __shared__ float s_rec[KSPLIT][BROWS][16 + 1];
-    __shared__ __hip_bfloat16 s_dgout[BROWS][64 + 8];  // this wg's dgates cols
+    __shared__ __hip_bfloat16 s_wb[UNITS][4 * H + 8];
+    __shared__ float s_dh[BROWS][UNITS + 1];
+    __shared__ float s_dc[BROWS][UNITS + 1];
+    __shared__ float s_rec[KSPLIT][BROWS][UNITS + 1];
+    __shared__ __hip_bfloat16 s_dgout[BROWS][4 * UNITS + 8];
 
     unsigned* ctr = &bar->flags[half * 32];
 
-    for (int e = threadIdx.x * 8; e < 16 * 4 * H; e += blockDim.x * 8) {
+    for (int e = threadIdx.x * 8; e < UNITS * 4 * H; e += blockDim.x * 8) {
         int c = e / (4 * H);
         int k = e % (4 * H);
         lstore8(&s_wb[c][k], lload8(Whh_bwd + (long)(u0 + c) * 4 * H + k));
     }
-    for (int p = threadIdx.x; p < BROWS * 16; p += blockDim.x) {
-        s_dh[p / 16][p % 16] = 0.f;
-        s_dc[p / 16][p % 16] = 0.f;
+    for (int p = threadIdx.x; p < BROWS * UNITS; p += blockDim.x) {
+        s_dh[p / UNITS][p % UNITS] = 0.f;
+        s_dc[p / UNITS][p % UNITS] = 0.f;
     }
     publish_count(ctr);
 
     const int wave = threadIdx.x / WAVE;
     const int lane = threadIdx.x & (WAVE - 1);
-    // KSPLIT=1: 4 waves x 16-row tiles, full-K.  KSPLIT=2: wave pairs
-    // split the 4H reduction; partials land in s_rec[wk].
-    const int wrow0 = (KSPLIT == 1 ? wave : (wave & 1)) * 16;
+    // UNITS=16, KSPLIT=1: 4 waves x 16-row tiles, full-K.
+    // UNITS=16, KSPLIT=2: wave pairs split the 4H reduction.
+    // UNITS=32: waves tile (row half x col half), full-K.
+    const int wrow0 = (NCWAVE == 2) ? (wave >> 1) * 16
+                      : (KSPLIT == 1 ? wave : (wave & 1)) * 16;
+    const int wcol0 = (NCWAVE == 2) ? (wave & 1) * 16 : 0;
     const int wk = (KSPLIT == 1) ? 0 : (wave >> 1);
     const int frow = lane & 15;
     const int kseg = (lane >> 4) * 8;
@@ -485,7 +492,7 @@ __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
     for (int t = T - 1; t >= 0; --t) {
         unsigned need = (unsigned)(T - t);   // pieces published for t+1
         if (!await_count(ctr, (unsigned)WGS * need, &bar->poison)) return;
-        // rec(B,16) = dgates_{t+1}(B,4H) @ s_wb^T with a 4-deep register
+        // rec(B,UNITS) = dgates_{t+1}(B,4H) @ s_wb^T with a 4-deep register
         // prefetch ring on the dgates stream (loads stay in flight across
         // MFMAs; no per-piece barriers)
         if (t < T - 1) {
@@ -500,10 +507,10 @@ __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
             bf16x8 a3 = rvalid ? lload8(dgrow + (c0 + 3) * 32 + kseg) : lzero8();
 #pragma unroll 4
             for (int kc = c0; kc < cN; kc += 4) {
-                bf16x8 b0 = lload8(&s_wb[frow][(kc + 0) * 32 + kseg]);
-                bf16x8 b1 = lload8(&s_wb[frow][(kc + 1) * 32 + kseg]);
-                bf16x8 b2 = lload8(&s_wb[frow][(kc + 2) * 32 + kseg]);
-                bf16x8 b3 = lload8(&s_wb[frow][(kc + 3) * 32 + kseg]);
+                bf16x8 b0 = lload8(&s_wb[wcol0 + frow][(kc + 0) * 32 + kseg]);
+                bf16x8 b1 = lload8(&s_wb[wcol0 + frow][(kc + 1) * 32 + kseg]);
+                bf16x8 b2 = lload8(&s_wb[wcol0 + frow][(kc + 2) * 32 + kseg]);
+                bf16x8 b3 = lload8(&s_wb[wcol0 + frow][(kc + 3) * 32 + kseg]);
                 acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc, 0, 0, 0);
                 if (kc + 4 < cN)
                     a0 = rvalid ? lload8(dgrow + (kc + 4) * 32 + kseg) : lzero8();
@@ -522,12 +529,12 @@ __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 int row = wrow0 + crow + r;
-                if (row < BROWS) s_rec[wk][row][ccol] = acc[r];
+                if (row < BROWS) s_rec[wk][row][wcol0 + ccol] = acc[r];
             }
         }
         __syncthreads();
-        for (int p = threadIdx.x; p < Bl * 16; p += blockDim.x) {
-            int b = p / 16, jl = p % 16;
+        for (int p = threadIdx.x; p < Bl * UNITS; p += blockDim.x) {
+            int b = p / UNITS, jl = p % UNITS;
             int bg = b0 + b;
             int u = u0 + jl;
             float dh, dc_in;
@@ -569,26 +576,25 @@ __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
                 s_dh[b][jl] = dh;
                 s_dc[b][jl] = dc_in;
             }
-            *(__bf16*)&s_dgout[b][0 + jl] = (__bf16)di;
-            *(__bf16*)&s_dgout[b][16 + jl] = (__bf16)df;
-            *(__bf16*)&s_dgout[b][32 + jl] = (__bf16)dg;
-            *(__bf16*)&s_dgout[b][48 + jl] = (__bf16)do_;
+            *(__bf16*)&s_dgout[b][0 * UNITS + jl] = (__bf16)di;
+            *(__bf16*)&s_dgout[b][1 * UNITS + jl] = (__bf16)df;
+            *(__bf16*)&s_dgout[b][2 * UNITS + jl] = (__bf16)dg;
+            *(__bf16*)&s_dgout[b][3 * UNITS + jl] = (__bf16)do_;
         }
         __syncthreads();
-        // vectorized dgates writes: thread (b, g, half) -> 16 B
+        // vectorized dgates writes: thread (b, g) -> UNITS/8 x 16 B
         {
             int tid = threadIdx.x;
             int b = tid / 4;
-            int gh = tid % 4;
+            int g = tid % 4;
             if (b < Bl) {
 #pragma unroll
-                for (int rep = 0; rep < 2; ++rep) {
-                    int g = gh;
+                for (int rep = 0; rep < UNITS / 8; ++rep) {
                     long off = ((long)(b0 + b) * T + t) * 4 * H + g * H + u0
                                + rep * 8;
                     lstore8(dgates + off,
                             *reinterpret_cast<bf16x8*>(
-                                &s_dgout[b][g * 16 + rep * 8]));
+                                &s_dgout[b][g * UNITS + rep * 8]));
                 }
             }
         }
@@ -810,12 +816,19 @@ torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
     long H = H4 / 4;
     TORCH_CHECK(H == 512, "lstm_bwd is instantiated for H=512");
     auto dgates = torch::empty({B, T, H4}, stash.options());
+    // 32-unit slices by default for the batch-split case (fewer/fatter
+    // workgroups — measured faster; R2D2_LSTM_BWD_UNITS=16 for comparison)
+    static const int bunits_env = [] {
+        const char* e = getenv("R2D2_LSTM_BWD_UNITS");
+        return e ? atoi(e) : 32;
+    }();
     int nhalves = B > 32 ? 2 : 1;
-    int nblocks = (int)H / 16 * nhalves;
+    int units = (nhalves == 2 && bunits_env != 16) ? 32 : 16;
+    int nblocks = (int)H / units * nhalves;
     auto stream = at::cuda::getCurrentCUDAStream();
     zero_ws(barrier_ws, stream.stream());
-#define LSTMB(BROWS_)                                                         \
-    hipLaunchKernelGGL((lstm_bwd_kernel<512, BROWS_>), dim3(nblocks),         \
+#define LSTMB(BROWS_, UNITS_)                                                 \
+    hipLaunchKernelGGL((lstm_bwd_kernel<512, BROWS_, UNITS_>), dim3(nblocks), \
         dim3(256), 0, stream.stream(),                                        \
         reinterpret_cast<const __hip_bfloat16*>(stash.data_ptr()),            \
         Cout.data_ptr<float>(),                                               \
@@ -826,7 +839,11 @@ torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
         reinterpret_cast<__hip_bfloat16*>(dgates.data_ptr()),                 \
         reinterpret_cast<GridBar*>(barrier_ws.data_ptr()),                    \
         (int)B, (int)T, nblocks, nhalves)
-    if (nhalves == 2) LSTMB(32); else LSTMB(64);
+    if (nhalves == 2) {
+        if (units == 32) LSTMB(32, 32); else LSTMB(32, 16);
+    } else {
+        LSTMB(64, 16);
+    }
 #undef LSTMB
     return dgates;
 }
