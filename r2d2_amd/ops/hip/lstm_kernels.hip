@@ -816,14 +816,17 @@ torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
     long H = H4 / 4;
     TORCH_CHECK(H == 512, "lstm_bwd is instantiated for H=512");
     auto dgates = torch::empty({B, T, H4}, stash.options());
-    // 32-unit slices by default for the batch-split case (fewer/fatter
-    // workgroups — measured faster; R2D2_LSTM_BWD_UNITS=16 for comparison)
+    // Measured DEAD END (docs/KERNELS.md): 32-unit bwd slices
+    // (R2D2_LSTM_BWD_UNITS=32) run 1.34 ms vs 0.94 ms at 16 — the 131 KB
+    // s_wb slab and the narrower 16-row dgates A-tiles at full K=2048 cost
+    // more than the halved handoff saves.  16 stays the default (the fwd
+    // kernel, whose weight slab is 4x smaller, DOES win at 16 units).
     static const int bunits_env = [] {
         const char* e = getenv("R2D2_LSTM_BWD_UNITS");
-        return e ? atoi(e) : 32;
+        return e ? atoi(e) : 16;
     }();
     int nhalves = B > 32 ? 2 : 1;
-    int units = (nhalves == 2 && bunits_env != 16) ? 32 : 16;
+    int units = (nhalves == 2 && bunits_env == 32) ? 32 : 16;
     int nblocks = (int)H / units * nhalves;
     auto stream = at::cuda::getCurrentCUDAStream();
     zero_ws(barrier_ws, stream.stream());
