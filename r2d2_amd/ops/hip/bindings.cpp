@@ -89,6 +89,9 @@ torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
                        torch::Tensor barrier_ws);
 
 // impala_kernels.hip
+void conv3p_pool(torch::Tensor in, torch::Tensor Wt, torch::Tensor bias,
+                 torch::Tensor pout, torch::Tensor parg, int64_t N,
+                 int64_t stage);
 void conv3p(torch::Tensor in, torch::Tensor Wt, torch::Tensor bias,
             torch::Tensor res, torch::Tensor mask, torch::Tensor out,
             int64_t N, int64_t H, int64_t W, bool relu_in, bool has_bias,
@@ -168,6 +171,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv3p", &conv3p,
           "IMPALA 3x3 s1 p1 conv on halo-padded NHWC (fwd & dgrad-as-conv, "
           "fused relu-in / residual / mask epilogues)");
+    m.def("conv3p_pool", &conv3p_pool,
+          "IMPALA stage conv FUSED with maxpool 3x3 s2 (conv output stays "
+          "in LDS; emits pooled activations + tap argmax)");
     m.def("conv3p_wgrad", &conv3p_wgrad,
           "IMPALA 3x3 conv backward-weight (padded dY, relu-in patches)");
     m.def("maxpool3s2_fwd", &maxpool3s2_fwd,

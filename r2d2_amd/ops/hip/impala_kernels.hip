@@ -318,6 +318,174 @@ __global__ __launch_bounds__(256) void conv3p_band_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// conv3p_pool_band: stage conv (3x3 s1 p1) FUSED with maxpool 3x3 s2 p1.
+// A workgroup owns PTH pooled rows of ONE image: it computes the
+// TH = 2*PTH+1 conv rows those pooling windows touch into LDS — the
+// full-resolution conv output never reaches HBM (s0: 1.29 GB/step/net of
+// writes + the same of pool reads eliminated; docs/IMPALA_ROOFLINE.md) —
+// then pools from LDS into the padded pooled output + dense tap argmax
+// (same encoding as maxpool3s2_fwd, so maxpool3s2_bwd is unchanged).
+// Adjacent bands recompute one shared conv row (1/TH redundancy).
+// ---------------------------------------------------------------------------
+template <bool IN_U8, int CIN, int COUT_T, int HT, int PTH, bool HAS_BIAS>
+__global__ __launch_bounds__(256) void conv3p_pool_band_kernel(
+    const void* __restrict__ in,              // (N, HT+2, HT+2, CIN)
+    const __hip_bfloat16* __restrict__ Wt,    // (COUT, 9*CIN)
+    const float* __restrict__ bias,
+    __hip_bfloat16* __restrict__ pout,        // (N, POUT+2, POUT+2, COUT)
+    unsigned char* __restrict__ parg,         // (N, POUT, POUT, COUT)
+    int N) {
+    constexpr int K = 9 * CIN;
+    constexpr int KROW = 3 * CIN;
+    constexpr int PW = HT + 2;
+    constexpr int PH = HT + 2;
+    constexpr int POUT = (HT + 1) / 2;
+    constexpr int NBANDS = (POUT + PTH - 1) / PTH;
+    constexpr int KITERS = (K + 31) / 32;
+    constexpr int TH = 2 * PTH + 1;
+    constexpr int RPI = (COUT_T == 32) ? 64 : 128;
+
+    __shared__ __hip_bfloat16 s_in[(TH + 2) * PW * CIN];
+    __shared__ __hip_bfloat16 s_conv[TH][HT][COUT_T];
+
+    const int band = blockIdx.x % NBANDS;
+    const long n = blockIdx.x / NBANDS;
+    const int py0 = band * PTH;
+    const int pth_eff = (py0 + PTH <= POUT) ? PTH : (POUT - py0);
+    // conv rows this band's pooling windows touch, clamped to the image
+    const int c0 = max(0, 2 * py0 - 1);
+    const int c1 = min(HT - 1, 2 * (py0 + pth_eff - 1) + 1);
+    const int crows = c1 - c0 + 1;
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int frow = lane & 15;
+    const int kseg = (lane >> 4) * 8;
+
+    {   // stage the input slab: padded rows [c0, c1 + 2]
+        const long gbase = (n * PH + c0) * (long)PW * CIN;
+        const int rows = crows + 2;
+        for (int e = threadIdx.x * 8; e < rows * PW * CIN;
+             e += blockDim.x * 8) {
+            bf16x8 v;
+            if (IN_U8)
+                v = idequant8(
+                    reinterpret_cast<const unsigned char*>(in) + gbase + e);
+            else
+                v = iload8(
+                    reinterpret_cast<const __hip_bfloat16*>(in) + gbase + e);
+            *reinterpret_cast<bf16x8*>(&s_in[e]) = v;
+        }
+    }
+
+    // preload this wave's 16-col B fragment for all k-iters
+    const int wr = (COUT_T == 32) ? (wave >> 1) : wave;
+    const int wc = (COUT_T == 32) ? (wave & 1) : 0;
+    bf16x8 wfrag[KITERS];
+#pragma unroll
+    for (int ki = 0; ki < KITERS; ++ki) {
+        int c = wc * 16 + frow;
+        int k = ki * 32 + kseg;
+        wfrag[ki] = (c < COUT_T && k < K)
+                        ? iload8(Wt + (long)c * K + k) : izero();
+    }
+    __syncthreads();
+
+    // conv rows c0..c1 -> s_conv (LDS), band-kernel row tiling
+    const int npix = crows * HT;
+    for (int p0 = wr * 32; p0 < npix; p0 += RPI) {
+        bool lval[2];
+        long lbase[2];
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            int lp = p0 + i * 16 + frow;
+            lval[i] = lp < npix;
+            int yl = lp / HT, x = lp % HT;
+            lbase[i] = ((long)yl * PW + x) * CIN;
+        }
+        f32x4 acc[2] = {};
+#pragma unroll
+        for (int ki = 0; ki < KITERS; ++ki) {
+            int k = ki * 32 + kseg;
+            bool kval = k < K;
+            int dy = kval ? k / KROW : 0;
+            int rem = kval ? k % KROW : 0;
+            int off = dy * PW * CIN + rem;
+            bf16x8 a[2];
+#pragma unroll
+            for (int i = 0; i < 2; ++i)
+                a[i] = (lval[i] && kval) ? iload8(&s_in[lbase[i] + off])
+                                         : izero();
+#pragma unroll
+            for (int i = 0; i < 2; ++i)
+                acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a[i], wfrag[ki], acc[i], 0, 0, 0);
+        }
+        int ccol = lane & 15;
+        int crow_ = (lane >> 4) * 4;
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int lp = p0 + i * 16 + crow_ + r;
+                int cc = wc * 16 + ccol;
+                if (lp < npix && cc < COUT_T) {
+                    int yl = lp / HT, x = lp % HT;
+                    float v = acc[i][r];
+                    if (HAS_BIAS) v += bias[cc];
+                    s_conv[yl][x][cc] = f2bf(v);
+                }
+            }
+    }
+    __syncthreads();
+
+    // pool PTH x POUT outputs x COUT/8 channels from LDS
+    constexpr int C8 = COUT_T / 8;
+    constexpr int POOL_THREADS = PTH * POUT * C8;
+    static_assert(POOL_THREADS <= 256, "one pooled output per thread");
+    {
+        int t = threadIdx.x;
+        if (t < pth_eff * POUT * C8) {
+            int c8 = t % C8;
+            int rem = t / C8;
+            int ox = rem % POUT;
+            int oyl = rem / POUT;
+            int oy = py0 + oyl;
+            float best[8];
+            int bidx[8];
+#pragma unroll
+            for (int e = 0; e < 8; ++e) { best[e] = -1e30f; bidx[e] = 0; }
+#pragma unroll
+            for (int ky = 0; ky < 3; ++ky) {
+                int y = 2 * oy - 1 + ky;
+                if (y < 0 || y >= HT) continue;
+#pragma unroll
+                for (int kx = 0; kx < 3; ++kx) {
+                    int x = 2 * ox - 1 + kx;
+                    if (x < 0 || x >= HT) continue;
+                    const __hip_bfloat16* src = &s_conv[y - c0][x][c8 * 8];
+#pragma unroll
+                    for (int e = 0; e < 8; ++e) {
+                        float f = bf2f(src[e]);
+                        if (f > best[e]) { best[e] = f; bidx[e] = ky * 3 + kx; }
+                    }
+                }
+            }
+            constexpr int POW_ = POUT + 2, POH_ = POUT + 2;
+            ibf8u o;
+#pragma unroll
+            for (int e = 0; e < 8; ++e) o.e[e] = f2bf(best[e]);
+            *reinterpret_cast<bf16x8*>(
+                pout + (((long)n * POH_ + oy + 1) * POW_ + ox + 1) * COUT_T
+                + c8 * 8) = o.v;
+            long abase = (((long)n * POUT + oy) * POUT + ox) * COUT_T + c8 * 8;
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+                parg[abase + e] = (unsigned char)bidx[e];
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // conv3p_wgrad: dWt(COUT, K=9*CIN) += dY^T @ patches, dY and input both in
 // the halo-padded layout; RELU_IN applies relu on patch load.  Same LDS
 // full-K staging scheme as conv_kernels.hip conv_wgrad (dY and patches are
@@ -813,6 +981,40 @@ __global__ __launch_bounds__(256) void dense2pad_mask_kernel(
 // ---------------------------------------------------------------------------
 
 static inline int icdiv(long a, long b) { return (int)((a + b - 1) / b); }
+
+// fused stage conv + maxpool: in (padded) -> pooled out (padded) + argmax;
+// the full-resolution conv output stays in LDS.  stage: 0/1/2 selects the
+// IMPALA geometry (84/8->16, 42/16->32, 21/32->32).
+void conv3p_pool(torch::Tensor in, torch::Tensor Wt, torch::Tensor bias,
+                 torch::Tensor pout, torch::Tensor parg, int64_t N,
+                 int64_t stage) {
+    auto stream = at::cuda::getCurrentCUDAStream();
+    const void* x = in.data_ptr();
+    auto* w = reinterpret_cast<const __hip_bfloat16*>(Wt.data_ptr());
+    const float* b = bias.data_ptr<float>();
+    auto* po = reinterpret_cast<__hip_bfloat16*>(pout.data_ptr());
+    auto* pa = parg.data_ptr<unsigned char>();
+#define CPLAUNCH(U8, CIN_, COUT_, HT_)                                        \
+    do {                                                                      \
+        constexpr int POUT_ = (HT_ + 1) / 2;                                  \
+        constexpr int NB_ = (POUT_ + 2) / 3;                                  \
+        hipLaunchKernelGGL((conv3p_pool_band_kernel<U8, CIN_, COUT_, HT_, 3,  \
+                                                    true>),                   \
+                           dim3((int)(N * NB_)), dim3(256), 0,                \
+                           stream.stream(), x, w, b, po, pa, (int)N);         \
+    } while (0)
+    if (stage == 0) {
+        TORCH_CHECK(in.dtype() == torch::kUInt8);
+        CPLAUNCH(true, 8, 16, 84);
+    } else if (stage == 1) {
+        CPLAUNCH(false, 16, 32, 42);
+    } else if (stage == 2) {
+        CPLAUNCH(false, 32, 32, 21);
+    } else {
+        TORCH_CHECK(false, "unknown stage");
+    }
+#undef CPLAUNCH
+}
 
 // in: padded (N, H+2, W+2, CIN_pad) bf16 or u8 (CIN_pad 8/16/32); out is
 // written into `out` (padded, zero-halo, (N, H+2, W+2, COUT)).

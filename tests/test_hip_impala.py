@@ -459,3 +459,42 @@ def test_train_gpu_replay_topology():
         torch.cuda.synchronize()  # thread mid-CUDA-call aborts at exit)
     assert learner.num_updates == c.training_steps
     cfg.apply("mspacman")
+
+
+@pytest.mark.parametrize("stage,cin,cout,ht", [(0, 8, 16, 84),
+                                               (1, 16, 32, 42),
+                                               (2, 32, 32, 21)])
+def test_conv3p_pool_matches_separate_ops(stage, cin, cout, ht):
+    """Fused stage-conv+maxpool == conv3p followed by maxpool3s2_fwd
+    (bit-exact: same bf16 conv values, same tap argmax tie-breaks)."""
+    torch.manual_seed(30 + stage)
+    N = 4
+    pout_h = (ht + 1) // 2
+    w = (torch.randn(cout, cin, 3, 3, device="cuda") * 0.2).bfloat16()
+    b = torch.randn(cout, device="cuda") * 0.1
+    wt = w.permute(0, 2, 3, 1).reshape(cout, -1).contiguous()
+    if stage == 0:
+        xin = torch.zeros(N, ht + 2, ht + 2, cin, dtype=torch.uint8,
+                          device="cuda")
+        xin[:, 1:-1, 1:-1] = torch.randint(
+            0, 256, (N, ht, ht, cin), dtype=torch.uint8, device="cuda")
+    else:
+        xin = torch.zeros(N, ht + 2, ht + 2, cin, device="cuda").bfloat16()
+        xin[:, 1:-1, 1:-1] = torch.randn(N, ht, ht, cin,
+                                         device="cuda").bfloat16()
+
+    # reference: separate kernels
+    conv_out = torch.zeros(N, ht + 2, ht + 2, cout, device="cuda").bfloat16()
+    M_.conv3p(xin, wt, b, EMPTY, EMPTY, conv_out, N, ht, ht, False, True, 0)
+    pooled_ref = torch.zeros(N, pout_h + 2, pout_h + 2, cout,
+                             device="cuda").bfloat16()
+    arg_ref = torch.zeros(N, pout_h, pout_h, cout, dtype=torch.uint8,
+                          device="cuda")
+    M_.maxpool3s2_fwd(conv_out, pooled_ref, arg_ref, N, ht, ht)
+
+    pooled = torch.zeros_like(pooled_ref)
+    arg = torch.zeros_like(arg_ref)
+    M_.conv3p_pool(xin, wt, b, pooled, arg, N, stage)
+    assert torch.equal(pooled, pooled_ref), \
+        (pooled - pooled_ref).abs().max()
+    assert torch.equal(arg, arg_ref)
