@@ -588,8 +588,9 @@ class HipNetworkEngine:
         # are permuted into .grad afterwards (tiny tensors).
         self._mark("fc_bwd")
         dflat3 = dflat.view(M * 49, 64)   # pre-masked by a3
-        dW3, db3 = m.conv_wgrad(dflat3, a3, a2, 3,
-                                M, 9, 9, 7, 7, 64, 9 * 64)
+        # per-image band wgrads: whole input image staged in LDS once
+        # (single dequant per element; conv1 patch traffic 557 -> 154 MB)
+        dW3, db3 = m.conv_wgrad_band(dflat3, a3, a2, 3, M)
         # dense bounds-checked dgrad (no zero-padded staging copies); the
         # output mask fuses conv2's relu backward into the d_a2 store
         d_a2 = torch.empty(M, 9, 9, 64, device=dev, dtype=torch.bfloat16)
@@ -597,8 +598,7 @@ class HipNetworkEngine:
                            M, 7, 7, 64, 9, 9, 64, 0, 0, 1, d_a2)
         # conv2 backward (d_a2 pre-masked by a2)
         d_a2f = d_a2.view(M * 81, 64)
-        dW2, db2 = m.conv_wgrad(d_a2f, a2, a1, 2,
-                                M, 20, 20, 9, 9, 64, 4 * 4 * 32)
+        dW2, db2 = m.conv_wgrad_band(d_a2f, a2, a1, 2, M)
         d_a1 = torch.empty(M, 20, 20, 32, device=dev, dtype=torch.bfloat16)
         for py in range(2):
             for px in range(2):
@@ -606,8 +606,8 @@ class HipNetworkEngine:
                                    ON.taps2[(py, px)], a1,
                                    M, 9, 9, 64, 20, 20, 32, py, px, 2, d_a1)
         # conv1 wgrad (no dgrad: input is data; d_a1 pre-masked by a1)
-        dW1, db1 = m.conv_wgrad(d_a1.view(M * 400, 32), a1, obs_hwc, 1,
-                                M, 84, 84, 20, 20, 32, 8 * 8 * self.C)
+        dW1, db1 = m.conv_wgrad_band(d_a1.view(M * 400, 32), a1, obs_hwc, 1,
+                                     M)
         self._mark("conv_bwd")
 
         enc = net.encoder

@@ -66,6 +66,9 @@ std::vector<torch::Tensor> conv_wgrad(torch::Tensor dY, torch::Tensor act,
                                       int64_t N, int64_t INH, int64_t INW,
                                       int64_t OH, int64_t OW, int64_t COUT,
                                       int64_t K);
+std::vector<torch::Tensor> conv_wgrad_band(torch::Tensor dY, torch::Tensor act,
+                                           torch::Tensor in, int64_t conv_id,
+                                           int64_t N);
 
 // lstm_kernels.hip
 torch::Tensor assemble_rin(torch::Tensor latent, torch::Tensor la,
@@ -157,6 +160,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("scatter_dh", &scatter_dh,
           "dHext (BT, H) from head-backward rows via inverse position map");
     m.def("conv_wgrad", &conv_wgrad, "MFMA conv backward-weight");
+    m.def("conv_wgrad_band", &conv_wgrad_band,
+          "per-image band wgrad (whole input image LDS-staged, one dequant "
+          "per element, register accumulation, one atomic flush per wg)");
     m.def("lstm_fwd", &lstm_fwd,
           "Persistent fused LSTM forward (dual-network, length-masked)");
     m.def("lstm_bwd", &lstm_bwd, "Persistent fused LSTM BPTT backward");

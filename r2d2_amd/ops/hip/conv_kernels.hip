@@ -489,6 +489,156 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// conv_wgrad_band: per-image wgrad.  The WHOLE input image fits in LDS for
+// every Nature-CNN conv (conv1 84x84x4 u8->bf16 56 KB, conv2 20x20x32
+// 26 KB, conv3 9x9x64 10 KB), so one workgroup stages it ONCE (single
+// dequant per input element instead of one per patch overlap), rebuilds
+// each 32-row patch tile LDS->LDS, and accumulates the whole image's
+// contribution in registers with ONE atomic flush at the end.  Fragments
+// gathered by hardware transpose-reads.  Replaces conv_wgrad_kernel's
+// global patch re-reads (conv1: 557 MB -> 154 MB per step).
+// ---------------------------------------------------------------------------
+template <bool IN_U8, int KH, int KW, int CIN, int S, int INH, int INW,
+          int OH, int OW, int COUT_T>
+__global__ __launch_bounds__(256) void conv_wgrad_band_kernel(
+    const __hip_bfloat16* __restrict__ dY,   // (M=N*OH*OW, COUT) PRE-MASKED
+    const __hip_bfloat16* __restrict__ act,  // (M, COUT) forward out (mask)
+    const void* __restrict__ in,             // (N, INH, INW, CIN)
+    float* __restrict__ dWt,                 // (COUT, K) f32
+    float* __restrict__ db,                  // (COUT,) f32
+    int N, int imgs_per_wg) {
+    constexpr int K = KH * KW * CIN;
+    constexpr int KWC = KW * CIN;
+    constexpr int NPIX = OH * OW;
+    constexpr int NCOT = (COUT_T + 31) / 32;          // cout tiles of 32
+    constexpr int NKW = 4 / NCOT;                     // k splits
+    constexpr int KHALF = ((K / NKW + 31) / 32) * 32;
+    constexpr int KFRAG = KHALF / 16;
+
+    __shared__ __hip_bfloat16 s_img[INH * INW * CIN];
+    __shared__ __hip_bfloat16 s_dy[32][64 + 8];
+    __shared__ __hip_bfloat16 s_a[32][K + 8];
+
+    int wave = threadIdx.x / WAVE;
+    int lane = threadIdx.x & (WAVE - 1);
+    int wr = (NCOT == 1) ? 0 : (wave >> 1);
+    int wc = (NCOT == 1) ? wave : (wave & 1);
+    int frow = lane & 15;
+    int mseg = (lane >> 4) * 8;
+
+    f32x4 acc[2][KFRAG] = {};
+    float bias_acc = 0.f;
+
+    const long n0 = (long)blockIdx.x * imgs_per_wg;
+    const long n1 = min((long)N, n0 + imgs_per_wg);
+    for (long n = n0; n < n1; ++n) {
+        __syncthreads();
+        {   // stage the whole input image (dequant once per element)
+            const long gbase = n * INH * INW * CIN;
+            for (int e = threadIdx.x * 8; e < INH * INW * CIN;
+                 e += blockDim.x * 8) {
+                bf16x8 v;
+                if (IN_U8)
+                    v = load_dequant8(
+                        reinterpret_cast<const unsigned char*>(in) + gbase + e);
+                else
+                    v = cload_bf16x8(
+                        reinterpret_cast<const __hip_bfloat16*>(in) + gbase + e);
+                *reinterpret_cast<bf16x8*>(&s_img[e]) = v;
+            }
+        }
+        for (int p0 = 0; p0 < NPIX; p0 += 32) {
+            __syncthreads();
+            {   // dY rows (pre-masked by the producing kernel; `act` kept
+                // for the generic path) + patch rows from the LDS image
+                int t = threadIdx.x;
+                {
+                    int mrow = t / 8;
+                    int col = (t % 8) * 8;
+                    int pp = p0 + mrow;
+                    bf16x8 v = czero();
+                    if (pp < NPIX) {
+                        long gm = n * NPIX + pp;
+#pragma unroll
+                        for (int e = 0; e < 8; ++e) {
+                            float g = (col + e < COUT_T)
+                                ? bf2f(dY[gm * COUT_T + col + e]) : 0.f;
+                            float m_ = (col + e < COUT_T)
+                                ? bf2f(act[gm * COUT_T + col + e]) : 0.f;
+                            v[e] = (__bf16)((m_ > 0.f) ? g : 0.f);
+                        }
+                    }
+                    *reinterpret_cast<bf16x8*>(&s_dy[mrow][col]) = v;
+                }
+                for (int e8 = t; e8 < 32 * (K / 8); e8 += 256) {
+                    int mrow = e8 / (K / 8);
+                    int k = (e8 % (K / 8)) * 8;
+                    int pp = p0 + mrow;
+                    bf16x8 w = czero();
+                    if (pp < NPIX) {
+                        int oy = pp / OW, ox = pp % OW;
+                        int dy_ = k / KWC, rem = k % KWC;
+                        w = *reinterpret_cast<const bf16x8*>(
+                            &s_img[((oy * S + dy_) * INW + ox * S) * CIN + rem]);
+                    }
+                    *reinterpret_cast<bf16x8*>(&s_a[mrow][k]) = w;
+                }
+            }
+            __syncthreads();
+
+            bf16x8 fa[2];
+#pragma unroll
+            for (int i = 0; i < 2; ++i)
+                fa[i] = lds_col_frag8<64 + 8>(&s_dy[0][0], mseg,
+                                              wr * 32 + i * 16, lane);
+#pragma unroll
+            for (int kf = 0; kf < KFRAG; ++kf) {
+                int kcol0 = wc * KHALF + kf * 16;
+                bf16x8 fb;
+                if (kcol0 + 16 <= K) {
+                    fb = lds_col_frag8<K + 8>(&s_a[0][0], mseg, kcol0, lane);
+                } else {
+                    fb = czero();
+                    int kcol = kcol0 + frow;
+                    if (kcol < K) {
+#pragma unroll
+                        for (int e = 0; e < 8; ++e)
+                            fb[e] = *(const __bf16*)&s_a[mseg + e][kcol];
+                    }
+                }
+#pragma unroll
+                for (int i = 0; i < 2; ++i)
+                    acc[i][kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        fa[i], fb, acc[i][kf], 0, 0, 0);
+            }
+            if (threadIdx.x < 64) {
+                int c = threadIdx.x;
+                for (int mr = 0; mr < 32; ++mr)
+                    bias_acc += bf2f(s_dy[mr][c]);
+            }
+        }
+    }
+
+    int ccol = lane & 15;
+    int crow = (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int kf = 0; kf < KFRAG; ++kf)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                long co = wr * 32 + i * 16 + crow + r;
+                long kk = wc * KHALF + kf * 16 + ccol;
+                if (co < COUT_T && kk < K)
+                    atomicAdd(&dWt[co * K + kk], acc[i][kf][r]);
+            }
+    if (threadIdx.x < 64) {
+        long c = threadIdx.x;
+        if (c < COUT_T) atomicAdd(&db[c], bias_acc);
+    }
+}
+
+// ---------------------------------------------------------------------------
 // Host wrappers.  Conv geometry is dispatched over the three Nature-CNN
 // layers (and reusable for any conv matching a template instance).
 // ---------------------------------------------------------------------------
@@ -602,6 +752,36 @@ torch::Tensor conv_dgrad_dense(torch::Tensor dY, torch::Tensor Wd,
 #undef DDM
 #undef DD1
     return dX;
+}
+
+// per-image band wgrad for the three Nature-CNN geometries (compile-time
+// shapes; see conv_wgrad_band_kernel).  N = number of images (B*T).
+std::vector<torch::Tensor> conv_wgrad_band(torch::Tensor dY, torch::Tensor act,
+                                           torch::Tensor in, int64_t conv_id,
+                                           int64_t N) {
+    long COUT = (conv_id == 1) ? 32 : 64;
+    long K = (conv_id == 1) ? 8 * 8 * 4 : (conv_id == 2) ? 4 * 4 * 32
+                                                         : 3 * 3 * 64;
+    auto dWt = torch::zeros({COUT, K}, dY.options().dtype(torch::kFloat32));
+    auto db = torch::zeros({COUT}, dY.options().dtype(torch::kFloat32));
+    int imgs = (int)((N + 1023) / 1024);
+    int grid = (int)((N + imgs - 1) / imgs);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    auto* dy = reinterpret_cast<const __hip_bfloat16*>(dY.data_ptr());
+    auto* ac = reinterpret_cast<const __hip_bfloat16*>(act.data_ptr());
+    const void* x = in.data_ptr();
+#define WBLAUNCH(U8, KH_, KW_, CIN_, S_, INH_, OH_, CO_)                       \
+    hipLaunchKernelGGL((conv_wgrad_band_kernel<U8, KH_, KW_, CIN_, S_, INH_,   \
+                                               INH_, OH_, OH_, CO_>),          \
+                       dim3(grid), dim3(256), 0, stream.stream(), dy, ac, x,   \
+                       dWt.data_ptr<float>(), db.data_ptr<float>(), (int)N,    \
+                       imgs)
+    if (conv_id == 1) WBLAUNCH(true, 8, 8, 4, 4, 84, 20, 32);
+    else if (conv_id == 2) WBLAUNCH(false, 4, 4, 32, 2, 20, 9, 64);
+    else if (conv_id == 3) WBLAUNCH(false, 3, 3, 64, 1, 9, 7, 64);
+    else TORCH_CHECK(false, "unknown conv_id");
+#undef WBLAUNCH
+    return {dWt, db};
 }
 
 std::vector<torch::Tensor> conv_wgrad(torch::Tensor dY, torch::Tensor act,

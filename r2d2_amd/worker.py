@@ -1177,4 +1177,5 @@ class VectorActor:
         if self._pulls % 200 == 0:
             print(f"[vector-actor] weight pulls: {self._pulls}, "
                   f"avg {self._pull_seconds / self._pulls * 1e3:.2f} ms "
-                  f"({'bus' if self.weight_bus is not None else 'cpu'})")
+                  f"({'bus' if self.weight_bus is not None else 'cpu'})",
+                  flush=True)

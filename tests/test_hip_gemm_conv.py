@@ -312,3 +312,28 @@ def test_scatter_dh_matches_index_add():
     ref = torch.zeros(B * T, H, device="cuda")
     ref.index_add_(0, pos, dh_a.float() + dh_v.float())
     close(out, ref, rtol=1e-3, atol=1e-3, name="scatter_dh")
+
+
+@pytest.mark.parametrize("conv_id,cin,cout,k,s,inhw", [
+    (1, 4, 32, 8, 4, 84),
+    (2, 32, 64, 4, 2, 20),
+    (3, 64, 64, 3, 1, 9),
+])
+def test_conv_wgrad_band_matches_chunked(conv_id, cin, cout, k, s, inhw):
+    """Per-image band wgrad == the chunked global-patch wgrad."""
+    torch.manual_seed(20 + conv_id)
+    N = 6
+    ohw = (inhw - k) // s + 1
+    if conv_id == 1:
+        x_in = torch.randint(0, 256, (N, inhw, inhw, cin),
+                             dtype=torch.uint8, device="cuda")
+    else:
+        x_in = torch.randn(N, inhw, inhw, cin, device="cuda").bfloat16()
+    dY = (torch.randn(N * ohw * ohw, cout, device="cuda") * 0.5).bfloat16()
+    act = torch.randn(N * ohw * ohw, cout, device="cuda").bfloat16()
+
+    dWt_ref, db_ref = M_.conv_wgrad(dY, act, x_in, conv_id, N, inhw, inhw,
+                                    ohw, ohw, cout, k * k * cin)
+    dWt, db = M_.conv_wgrad_band(dY, act, x_in, conv_id, N)
+    close(dWt, dWt_ref, rtol=2e-3, atol=1e-2, name=f"band wgrad {conv_id}")
+    close(db, db_ref, rtol=2e-3, atol=1e-2, name=f"band bgrad {conv_id}")
