@@ -355,6 +355,8 @@ class HipNetworkEngine:
         self.target.refresh()
 
     # ------------------------------------------------------------------
+    _fwd_band = os.environ.get("R2D2_CONV_FWD_BAND", "1") != "0"
+
     def _encoder_fwd(self, pack, obs_hwc_u8, want_stash=True):
         """obs: (M, 84, 84, C) uint8 -> latent (M, 512) bf16 + stashes."""
         m = self.m
@@ -362,9 +364,17 @@ class HipNetworkEngine:
             return impala_ops.encoder_fwd(m, pack.imp, obs_hwc_u8, want_stash,
                                           xp=getattr(self, "_xp", None))
         M = obs_hwc_u8.shape[0]
-        a1 = m.conv_fwd(obs_hwc_u8, pack.w1t, pack.b1, 1, M, 84, 84, 20, 20, True)
-        a2 = m.conv_fwd(a1, pack.w2t, pack.b2, 2, M, 20, 20, 9, 9, True)
-        a3 = m.conv_fwd(a2, pack.w3t, pack.b3, 3, M, 9, 9, 7, 7, True)
+        if self._fwd_band:
+            # per-image band forwards: image + weights LDS-resident, one
+            # global read/dequant per input element
+            a1 = m.conv_fwd_band(obs_hwc_u8, pack.w1t, pack.b1, 1, M)
+            a2 = m.conv_fwd_band(a1, pack.w2t, pack.b2, 2, M)
+            a3 = m.conv_fwd_band(a2, pack.w3t, pack.b3, 3, M)
+        else:
+            a1 = m.conv_fwd(obs_hwc_u8, pack.w1t, pack.b1, 1, M, 84, 84,
+                            20, 20, True)
+            a2 = m.conv_fwd(a1, pack.w2t, pack.b2, 2, M, 20, 20, 9, 9, True)
+            a3 = m.conv_fwd(a2, pack.w3t, pack.b3, 3, M, 9, 9, 7, 7, True)
         flat = a3.view(M, 3136)
         latent = m.gemm_bias_act(flat, pack.wft, pack.bf, 1, False)
         return latent, (a1, a2, a3, flat)

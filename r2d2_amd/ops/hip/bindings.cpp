@@ -69,6 +69,8 @@ std::vector<torch::Tensor> conv_wgrad(torch::Tensor dY, torch::Tensor act,
 std::vector<torch::Tensor> conv_wgrad_band(torch::Tensor dY, torch::Tensor act,
                                            torch::Tensor in, int64_t conv_id,
                                            int64_t N);
+torch::Tensor conv_fwd_band(torch::Tensor in, torch::Tensor Wt,
+                            torch::Tensor bias, int64_t conv_id, int64_t N);
 
 // lstm_kernels.hip
 torch::Tensor assemble_rin(torch::Tensor latent, torch::Tensor la,
@@ -163,6 +165,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv_wgrad_band", &conv_wgrad_band,
           "per-image band wgrad (whole input image LDS-staged, one dequant "
           "per element, register accumulation, one atomic flush per wg)");
+    m.def("conv_fwd_band", &conv_fwd_band,
+          "per-image band forward (image + weights LDS-resident)");
     m.def("lstm_fwd", &lstm_fwd,
           "Persistent fused LSTM forward (dual-network, length-masked)");
     m.def("lstm_bwd", &lstm_bwd, "Persistent fused LSTM BPTT backward");

@@ -489,6 +489,116 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// conv_fwd_band: per-image forward.  The whole input image AND the full
+// prepacked weight matrix fit in LDS for every Nature-CNN conv, so the
+// patch field reads LDS (one global read + one dequant per input element
+// instead of one per patch overlap: conv1 557 MB u8 re-dequant -> 154 MB,
+// conv3 9x re-read -> 1x).  Fused bias + ReLU as in conv_fwd_kernel.
+// ---------------------------------------------------------------------------
+template <bool IN_U8, int KH, int KW, int CIN, int S, int INH, int INW,
+          int OH, int OW, int COUT_T>
+__global__ __launch_bounds__(256) void conv_fwd_band_kernel(
+    const void* __restrict__ in,            // (N, INH, INW, CIN)
+    const __hip_bfloat16* __restrict__ Wt,  // (COUT, K)
+    const float* __restrict__ bias,
+    __hip_bfloat16* __restrict__ out,       // (N*OH*OW, COUT)
+    int N, int imgs_per_wg) {
+    constexpr int K = KH * KW * CIN;
+    constexpr int KWC = KW * CIN;
+    constexpr int NPIX = OH * OW;
+    constexpr int NB = (COUT_T == 32) ? 1 : 2;     // B frags per wave
+    constexpr int RPI = (COUT_T == 32) ? 128 : 64; // rows per iter
+
+    __shared__ __hip_bfloat16 s_img[INH * INW * CIN];
+    __shared__ __hip_bfloat16 s_w[COUT_T][K + 8];
+
+    int wave = threadIdx.x / WAVE;
+    int lane = threadIdx.x & (WAVE - 1);
+    int wr = (COUT_T == 32) ? wave : (wave >> 1);
+    int wc = (COUT_T == 32) ? 0 : (wave & 1);
+    int frow = lane & 15;
+    int kseg = (lane >> 4) * 8;
+
+    for (int e = threadIdx.x * 8; e < COUT_T * K; e += blockDim.x * 8) {
+        int c = e / K, k = e % K;
+        *reinterpret_cast<bf16x8*>(&s_w[c][k]) = cload_bf16x8(Wt + c * K + k);
+    }
+
+    const long n0 = (long)blockIdx.x * imgs_per_wg;
+    const long n1 = min((long)N, n0 + imgs_per_wg);
+    for (long n = n0; n < n1; ++n) {
+        __syncthreads();
+        {
+            const long gbase = n * INH * INW * CIN;
+            for (int e = threadIdx.x * 8; e < INH * INW * CIN;
+                 e += blockDim.x * 8) {
+                bf16x8 v;
+                if (IN_U8)
+                    v = load_dequant8(
+                        reinterpret_cast<const unsigned char*>(in) + gbase + e);
+                else
+                    v = cload_bf16x8(
+                        reinterpret_cast<const __hip_bfloat16*>(in) + gbase + e);
+                *reinterpret_cast<bf16x8*>(&s_img[e]) = v;
+            }
+        }
+        __syncthreads();
+
+        for (int p0 = wr * 32; p0 < NPIX; p0 += RPI) {
+            int pbase[2];
+            bool pval[2];
+#pragma unroll
+            for (int i = 0; i < 2; ++i) {
+                int pp = p0 + i * 16 + frow;
+                pval[i] = pp < NPIX;
+                int oy = pp / OW, ox = pp % OW;
+                pbase[i] = ((oy * S) * INW + ox * S) * CIN;
+            }
+            f32x4 acc[2][NB] = {};
+            for (int k0 = 0; k0 < K; k0 += 32) {
+                int k = k0 + kseg;
+                int dy = k / KWC;
+                int rem = k % KWC;
+                int off = dy * INW * CIN + rem;
+                bf16x8 a[2], b[NB];
+#pragma unroll
+                for (int i = 0; i < 2; ++i)
+                    a[i] = pval[i]
+                        ? *reinterpret_cast<const bf16x8*>(
+                              &s_img[pbase[i] + off])
+                        : czero();
+#pragma unroll
+                for (int j = 0; j < NB; ++j)
+                    b[j] = cload_bf16x8(
+                        &s_w[wc * 32 + j * 16 + frow][k0 + kseg]);
+#pragma unroll
+                for (int i = 0; i < 2; ++i)
+#pragma unroll
+                    for (int j = 0; j < NB; ++j)
+                        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a[i], b[j], acc[i][j], 0, 0, 0);
+            }
+            int ccol = lane & 15;
+            int crow = (lane >> 4) * 4;
+#pragma unroll
+            for (int i = 0; i < 2; ++i)
+#pragma unroll
+                for (int j = 0; j < NB; ++j)
+#pragma unroll
+                    for (int r = 0; r < 4; ++r) {
+                        int pp = p0 + i * 16 + crow + r;
+                        int cc = wc * 32 + j * 16 + ccol;
+                        if (pp < NPIX) {
+                            float v = acc[i][j][r] + bias[cc];
+                            out[(n * NPIX + pp) * COUT_T + cc] =
+                                f2bf(fmaxf(v, 0.f));
+                        }
+                    }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // conv_wgrad_band: per-image wgrad.  The WHOLE input image fits in LDS for
 // every Nature-CNN conv (conv1 84x84x4 u8->bf16 56 KB, conv2 20x20x32
 // 26 KB, conv3 9x9x64 10 KB), so one workgroup stages it ONCE (single
@@ -752,6 +862,33 @@ torch::Tensor conv_dgrad_dense(torch::Tensor dY, torch::Tensor Wd,
 #undef DDM
 #undef DD1
     return dX;
+}
+
+// per-image band forward for the three Nature-CNN geometries.
+torch::Tensor conv_fwd_band(torch::Tensor in, torch::Tensor Wt,
+                            torch::Tensor bias, int64_t conv_id, int64_t N) {
+    long COUT = (conv_id == 1) ? 32 : 64;
+    long NPIX = (conv_id == 1) ? 400 : (conv_id == 2) ? 81 : 49;
+    auto out = torch::empty({N * NPIX, COUT},
+                            in.options().dtype(torch::kBFloat16));
+    int imgs = (int)((N + 2047) / 2048);
+    int grid = (int)((N + imgs - 1) / imgs);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    const void* x = in.data_ptr();
+    auto* w = reinterpret_cast<const __hip_bfloat16*>(Wt.data_ptr());
+    const float* b = bias.data_ptr<float>();
+    auto* o = reinterpret_cast<__hip_bfloat16*>(out.data_ptr());
+#define FBLAUNCH(U8, KH_, KW_, CIN_, S_, INH_, OH_, CO_)                       \
+    hipLaunchKernelGGL((conv_fwd_band_kernel<U8, KH_, KW_, CIN_, S_, INH_,     \
+                                             INH_, OH_, OH_, CO_>),            \
+                       dim3(grid), dim3(256), 0, stream.stream(), x, w, b, o,  \
+                       (int)N, imgs)
+    if (conv_id == 1) FBLAUNCH(true, 8, 8, 4, 4, 84, 20, 32);
+    else if (conv_id == 2) FBLAUNCH(false, 4, 4, 32, 2, 20, 9, 64);
+    else if (conv_id == 3) FBLAUNCH(false, 3, 3, 64, 1, 9, 7, 64);
+    else TORCH_CHECK(false, "unknown conv_id");
+#undef FBLAUNCH
+    return out;
 }
 
 // per-image band wgrad for the three Nature-CNN geometries (compile-time

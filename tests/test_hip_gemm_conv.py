@@ -337,3 +337,24 @@ def test_conv_wgrad_band_matches_chunked(conv_id, cin, cout, k, s, inhw):
     dWt, db = M_.conv_wgrad_band(dY, act, x_in, conv_id, N)
     close(dWt, dWt_ref, rtol=2e-3, atol=1e-2, name=f"band wgrad {conv_id}")
     close(db, db_ref, rtol=2e-3, atol=1e-2, name=f"band bgrad {conv_id}")
+
+
+@pytest.mark.parametrize("conv_id", [1, 2, 3])
+def test_conv_fwd_band_matches_classic(conv_id):
+    """Per-image band forward == the classic global-patch forward."""
+    geo = {1: (4, 32, 8, 4, 84), 2: (32, 64, 4, 2, 20), 3: (64, 64, 3, 1, 9)}
+    cin, cout, k, s, inhw = geo[conv_id]
+    torch.manual_seed(40 + conv_id)
+    N = 9
+    ohw = (inhw - k) // s + 1
+    w = (torch.randn(cout, cin, k, k, device="cuda") * 0.2).bfloat16()
+    b = torch.randn(cout, device="cuda") * 0.1
+    if conv_id == 1:
+        x_in = torch.randint(0, 256, (N, inhw, inhw, cin),
+                             dtype=torch.uint8, device="cuda")
+    else:
+        x_in = torch.randn(N, inhw, inhw, cin, device="cuda").bfloat16()
+    ref = M_.conv_fwd(x_in, prepack_w(w), b, conv_id, N, inhw, inhw,
+                      ohw, ohw, True)
+    out = M_.conv_fwd_band(x_in, prepack_w(w), b, conv_id, N)
+    assert torch.equal(out, ref), (out.float() - ref.float()).abs().max()
