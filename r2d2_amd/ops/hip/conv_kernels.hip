@@ -506,7 +506,7 @@ __global__ __launch_bounds__(256) void conv_fwd_band_kernel(
     constexpr int K = KH * KW * CIN;
     constexpr int KWC = KW * CIN;
     constexpr int NPIX = OH * OW;
-    constexpr int NB = (COUT_T == 32) ? 1 : 2;     // B frags per wave
+    constexpr int NB = 2;                          // B frags = 32 cols/wave
     constexpr int RPI = (COUT_T == 32) ? 128 : 64; // rows per iter
 
     __shared__ __hip_bfloat16 s_img[INH * INW * CIN];
