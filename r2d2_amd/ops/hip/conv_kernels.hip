@@ -495,33 +495,37 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
 // instead of one per patch overlap: conv1 557 MB u8 re-dequant -> 154 MB,
 // conv3 9x re-read -> 1x).  Fused bias + ReLU as in conv_fwd_kernel.
 // ---------------------------------------------------------------------------
+// COUT_T: output columns computed per launch (always 32 — a 64-out conv
+// runs as two launches with co0 = 0/32, halving the LDS weight slab so
+// 2-3 workgroups fit per CU); COUT_FULL is the output row stride.
 template <bool IN_U8, int KH, int KW, int CIN, int S, int INH, int INW,
-          int OH, int OW, int COUT_T>
+          int OH, int OW, int COUT_T, int COUT_FULL>
 __global__ __launch_bounds__(256) void conv_fwd_band_kernel(
     const void* __restrict__ in,            // (N, INH, INW, CIN)
-    const __hip_bfloat16* __restrict__ Wt,  // (COUT, K)
+    const __hip_bfloat16* __restrict__ Wt,  // (COUT_FULL, K)
     const float* __restrict__ bias,
-    __hip_bfloat16* __restrict__ out,       // (N*OH*OW, COUT)
-    int N, int imgs_per_wg) {
+    __hip_bfloat16* __restrict__ out,       // (N*OH*OW, COUT_FULL)
+    int N, int imgs_per_wg, int co0) {
     constexpr int K = KH * KW * CIN;
     constexpr int KWC = KW * CIN;
     constexpr int NPIX = OH * OW;
     constexpr int NB = 2;                          // B frags = 32 cols/wave
-    constexpr int RPI = (COUT_T == 32) ? 128 : 64; // rows per iter
+    constexpr int RPI = 128;                       // 4 waves stack rows
 
     __shared__ __hip_bfloat16 s_img[INH * INW * CIN];
     __shared__ __hip_bfloat16 s_w[COUT_T][K + 8];
 
     int wave = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
-    int wr = (COUT_T == 32) ? wave : (wave >> 1);
-    int wc = (COUT_T == 32) ? 0 : (wave & 1);
+    int wr = wave;
+    int wc = 0;
     int frow = lane & 15;
     int kseg = (lane >> 4) * 8;
 
     for (int e = threadIdx.x * 8; e < COUT_T * K; e += blockDim.x * 8) {
         int c = e / K, k = e % K;
-        *reinterpret_cast<bf16x8*>(&s_w[c][k]) = cload_bf16x8(Wt + c * K + k);
+        *reinterpret_cast<bf16x8*>(&s_w[c][k]) =
+            cload_bf16x8(Wt + (long)(co0 + c) * K + k);
     }
 
     const long n0 = (long)blockIdx.x * imgs_per_wg;
@@ -589,8 +593,8 @@ __global__ __launch_bounds__(256) void conv_fwd_band_kernel(
                         int pp = p0 + i * 16 + crow + r;
                         int cc = wc * 32 + j * 16 + ccol;
                         if (pp < NPIX) {
-                            float v = acc[i][j][r] + bias[cc];
-                            out[(n * NPIX + pp) * COUT_T + cc] =
+                            float v = acc[i][j][r] + bias[co0 + cc];
+                            out[(n * NPIX + pp) * COUT_FULL + co0 + cc] =
                                 f2bf(fmaxf(v, 0.f));
                         }
                     }
@@ -878,15 +882,22 @@ torch::Tensor conv_fwd_band(torch::Tensor in, torch::Tensor Wt,
     auto* w = reinterpret_cast<const __hip_bfloat16*>(Wt.data_ptr());
     const float* b = bias.data_ptr<float>();
     auto* o = reinterpret_cast<__hip_bfloat16*>(out.data_ptr());
-#define FBLAUNCH(U8, KH_, KW_, CIN_, S_, INH_, OH_, CO_)                       \
+#define FBLAUNCH(U8, KH_, KW_, CIN_, S_, INH_, OH_, COF_, CO0_)                \
     hipLaunchKernelGGL((conv_fwd_band_kernel<U8, KH_, KW_, CIN_, S_, INH_,     \
-                                             INH_, OH_, OH_, CO_>),            \
+                                             INH_, OH_, OH_, 32, COF_>),       \
                        dim3(grid), dim3(256), 0, stream.stream(), x, w, b, o,  \
-                       (int)N, imgs)
-    if (conv_id == 1) FBLAUNCH(true, 8, 8, 4, 4, 84, 20, 32);
-    else if (conv_id == 2) FBLAUNCH(false, 4, 4, 32, 2, 20, 9, 64);
-    else if (conv_id == 3) FBLAUNCH(false, 3, 3, 64, 1, 9, 7, 64);
-    else TORCH_CHECK(false, "unknown conv_id");
+                       (int)N, imgs, CO0_)
+    if (conv_id == 1) {
+        FBLAUNCH(true, 8, 8, 4, 4, 84, 20, 32, 0);
+    } else if (conv_id == 2) {   // 64-out: two 32-col launches (LDS halved)
+        FBLAUNCH(false, 4, 4, 32, 2, 20, 9, 64, 0);
+        FBLAUNCH(false, 4, 4, 32, 2, 20, 9, 64, 32);
+    } else if (conv_id == 3) {
+        FBLAUNCH(false, 3, 3, 64, 1, 9, 7, 64, 0);
+        FBLAUNCH(false, 3, 3, 64, 1, 9, 7, 64, 32);
+    } else {
+        TORCH_CHECK(false, "unknown conv_id");
+    }
 #undef FBLAUNCH
     return out;
 }
