@@ -298,26 +298,7 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
 
     constexpr int CHUNKS = (BROWS * H) / 8;
 
-    const int ccol_s = lane & 15;
-    const int crow_s = (lane >> 4) * 4;
     for (int t = 0; t < T; ++t) {
-        // prefetch this step's X contribution for the lane's accumulator
-        // cells BEFORE the await — X has no cross-workgroup dependency, so
-        // its global-load latency hides entirely under the handoff spin,
-        // and the separate +X LDS pass (and its barrier) disappears
-        float xv[NFRAG][4];
-#pragma unroll
-        for (int i = 0; i < NFRAG; ++i)
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                int row = wrow0 + i * 16 + crow_s + r;
-                int gc = wcol0 + ccol_s;
-                int g = gc / UNITS, j = gc % UNITS;
-                xv[i][r] = (row < Bl)
-                    ? bf2f(X[((long)(b0 + row) * T + t) * 4 * H
-                             + g * H + u0 + j])
-                    : 0.f;
-            }
         if (!await_count(ctr, (unsigned)WGS_PER_HALF * (t + 1),
                          &bar->poison))
             return;
@@ -349,12 +330,30 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
             }
         }
         {
+            int ccol = lane & 15;
+            int crow = (lane >> 4) * 4;
 #pragma unroll
             for (int i = 0; i < NFRAG; ++i)
 #pragma unroll
                 for (int r = 0; r < 4; ++r)
-                    s_gates[wrow0 + i * 16 + crow_s + r][wcol0 + ccol_s] =
-                        acc[i][r] + xv[i][r];
+                    s_gates[wrow0 + i * 16 + crow + r][wcol0 + ccol] = acc[i][r];
+        }
+        __syncthreads();
+        // + X[t] (vectorized), then nonlinearities + state advance
+        {
+            int row = threadIdx.x / 4;
+            int g = threadIdx.x % 4;
+            if (row < Bl) {
+#pragma unroll
+                for (int j8 = 0; j8 < UNITS; j8 += 8) {
+                    bf16x8 x8 = lload8(
+                        X + ((long)(b0 + row) * T + t) * 4 * H + g * H + u0
+                        + j8);
+#pragma unroll
+                    for (int j = 0; j < 8; ++j)
+                        s_gates[row][g * UNITS + j8 + j] += (float)x8[j];
+                }
+            }
         }
         __syncthreads();
         for (int p = threadIdx.x; p < Bl * UNITS; p += blockDim.x) {
