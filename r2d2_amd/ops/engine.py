@@ -325,6 +325,16 @@ class HipNetworkEngine:
         # per-step repack -> ONE gather kernel per dtype (maps bit-probed
         # from the now-flat param buffer; see _NetPack.enable_fast_refresh)
         self.online.enable_fast_refresh(self.flat_param, self.m)
+        # hipGraph capture of the fwd+bwd launch sequence (~55 launches):
+        # the step is a fixed kernel DAG for the common fixed-layout batch,
+        # so replaying a captured graph removes the per-launch host gaps.
+        # Disabled under timing (events can't record in capture), for DP
+        # (the RCCL segment reduces run between backward stages), and for
+        # ragged layouts (graph keyed on the exact batch layout).
+        self._graph = None
+        self._graph_key = None
+        self._use_graph = (os.environ.get("R2D2_HIP_GRAPH", "1") != "0"
+                           and not self.timing)
 
     def _mark(self, name):
         if self.timing:
@@ -443,20 +453,84 @@ class HipNetworkEngine:
         return out
 
     # ------------------------------------------------------------------
-    def train_step(self, batch, grad_hook=None):
-        """Full fused update.  Fills .grad on the online nn.Module params and
-        returns (loss tensor, per-sequence priority tensor, both on device).
-        ``grad_hook(segment)`` fires when a flat-grad segment ("heads",
-        "lstm", "encoder") is complete — the DP learner uses it to overlap
-        the RCCL all-reduce with the rest of the manual backward."""
-        m = self.m
-        c = self.cfg
+    # -- hipGraph capture ----------------------------------------------
+    def _layout_key(self, batch):
+        return (tuple(batch.obs.shape), batch.obs.dtype,
+                batch.burn_in_steps.numpy().tobytes(),
+                batch.learning_steps.numpy().tobytes(),
+                batch.forward_steps.numpy().tobytes())
+
+    def _copy_into_static(self, batch):
+        sb = self._static_batch
+        for f in ("obs", "last_action", "last_reward", "hidden", "action",
+                  "n_step_reward", "gamma", "is_weights"):
+            getattr(sb, f).copy_(getattr(batch, f), non_blocking=True)
+
+    def _try_capture(self, batch, key):
+        import types
         dev = self.device
+        sb = types.SimpleNamespace(
+            obs=batch.obs.to(dev).contiguous().clone(),
+            last_action=batch.last_action.to(dev).contiguous().clone(),
+            last_reward=batch.last_reward.to(dev).contiguous().clone(),
+            hidden=batch.hidden.to(dev).contiguous().clone(),
+            action=batch.action.to(dev).contiguous().clone(),
+            n_step_reward=batch.n_step_reward.to(dev).contiguous().clone(),
+            gamma=batch.gamma.to(dev).contiguous().clone(),
+            is_weights=batch.is_weights.to(dev).contiguous().clone(),
+            burn_in_steps=batch.burn_in_steps,
+            learning_steps=batch.learning_steps,
+            forward_steps=batch.forward_steps)
+        try:
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):                  # allocator warmup
+                    self._train_step_impl(sb)
+            torch.cuda.current_stream().wait_stream(s)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                loss, prio = self._train_step_impl(sb)
+        except Exception as e:
+            print(f"[engine] hipGraph capture failed ({e!r}); "
+                  f"running uncaptured", flush=True)
+            self._use_graph = False
+            return False
+        self._graph = g
+        self._graph_key = key
+        self._static_batch = sb
+        self._graph_out = (loss, prio)
+        return True
+
+    def train_step(self, batch, grad_hook=None):
+        """Full fused update (hipGraph replay when the batch layout matches
+        the captured graph).  Fills .grad on the online nn.Module params
+        and returns (loss tensor, per-sequence priority tensor, both on
+        device).  ``grad_hook(segment)`` fires when a flat-grad segment
+        ("heads", "lstm", "encoder") is complete — the DP learner uses it
+        to overlap the RCCL all-reduce with the rest of the manual
+        backward (graphs are bypassed in that case so the collectives can
+        interleave)."""
         if bool(self._poison_pin.any()):
             raise RuntimeError(
                 "persistent LSTM kernel poisoned on a previous step "
                 "(bounded-spin timeout) — outputs/gradients of that step "
                 "are invalid; aborting instead of training on them")
+        if self._use_graph and grad_hook is None:
+            key = self._layout_key(batch)
+            if self._graph_key == key:
+                self._copy_into_static(batch)
+                self._graph.replay()
+                return self._graph_out
+            if self._try_capture(batch, key):
+                return self._graph_out
+        return self._train_step_impl(batch, grad_hook)
+
+    def _train_step_impl(self, batch, grad_hook=None):
+        """The uncaptured launch sequence (also the capture body)."""
+        m = self.m
+        c = self.cfg
+        dev = self.device
         B, T = batch.obs.shape[:2]
         A, H = self.A, self.H
 
