@@ -518,11 +518,13 @@ class HipNetworkEngine:
                 "are invalid; aborting instead of training on them")
         if self._use_graph and grad_hook is None:
             key = self._layout_key(batch)
+            if self._graph_key != key:
+                self._try_capture(batch, key)
             if self._graph_key == key:
+                # capture only RECORDS the work — every step (including the
+                # capture step) must replay to actually execute
                 self._copy_into_static(batch)
                 self._graph.replay()
-                return self._graph_out
-            if self._try_capture(batch, key):
                 return self._graph_out
         return self._train_step_impl(batch, grad_hook)
 

@@ -686,10 +686,20 @@ __global__ void handoff_bench_kernel(GridBar* bar, int steps, int nblocks) {
 // Host wrappers
 // ---------------------------------------------------------------------------
 
+// NOT hipMemsetAsync: a captured memset node replays with a garbage fill
+// value from the second hipGraph replay onward (observed on ROCm 7.2,
+// gpurun_out/graph_probe2.log — counters came back as <garbage>+increments
+// and the poison word as the raw garbage).  A plain kernel replays
+// correctly.
+__global__ void zero_gridbar_kernel(unsigned* ws) {
+    if (threadIdx.x < sizeof(GridBar) / 4) ws[threadIdx.x] = 0u;
+}
+
 static void zero_ws(torch::Tensor& ws, hipStream_t stream) {
     TORCH_CHECK(ws.numel() * ws.element_size() >= (long)sizeof(GridBar),
                 "barrier workspace too small (need >= 256 int32)");
-    hipMemsetAsync(ws.data_ptr(), 0, sizeof(GridBar), stream);
+    hipLaunchKernelGGL(zero_gridbar_kernel, dim3(1), dim3(320), 0, stream,
+                       reinterpret_cast<unsigned*>(ws.data_ptr()));
 }
 
 torch::Tensor assemble_rin(torch::Tensor latent, torch::Tensor la,
