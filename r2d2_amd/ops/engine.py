@@ -489,7 +489,10 @@ class HipNetworkEngine:
                     self._train_step_impl(sb)
             torch.cuda.current_stream().wait_stream(s)
             g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
+            # thread_local: the replay-ingest thread keeps streaming actor
+            # blocks on its own streams while we capture — global capture
+            # mode would invalidate them and poison the whole HIP context
+            with torch.cuda.graph(g, capture_error_mode="thread_local"):
                 loss, prio = self._train_step_impl(sb)
         except Exception as e:
             print(f"[engine] hipGraph capture failed ({e!r}); "
