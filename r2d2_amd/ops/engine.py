@@ -325,15 +325,22 @@ class HipNetworkEngine:
         # per-step repack -> ONE gather kernel per dtype (maps bit-probed
         # from the now-flat param buffer; see _NetPack.enable_fast_refresh)
         self.online.enable_fast_refresh(self.flat_param, self.m)
-        # hipGraph capture of the fwd+bwd launch sequence (~55 launches):
-        # the step is a fixed kernel DAG for the common fixed-layout batch,
-        # so replaying a captured graph removes the per-launch host gaps.
-        # Disabled under timing (events can't record in capture), for DP
-        # (the RCCL segment reduces run between backward stages), and for
-        # ragged layouts (graph keyed on the exact batch layout).
+        # hipGraph capture of the fwd+bwd launch sequence — MEASURED
+        # NEGATIVE, default off (R2D2_HIP_GRAPH=1 to enable):
+        # - prebuilt batches: 11.41k vs 11.54k seq/s uncaptured (the ~55
+        #   launches are already async-submitted; replay saves nothing);
+        # - with the GPU-resident replay in the loop it collapses to ~2.8k:
+        #   the graph's private memory pool evicts the caching allocator's
+        #   blocks for the per-sample gather outputs, degenerating into
+        #   hipMalloc churn.
+        # Kept (capture-correct, all tests pass with it on) for future
+        # launch-bound configs; finding a real ROCm defect on the way: a
+        # captured hipMemsetAsync replays a garbage fill value from the
+        # 2nd replay (gpurun_out/graph_probe2.log) — the LSTM workspace
+        # reset is a plain kernel now (zero_gridbar_kernel).
         self._graph = None
         self._graph_key = None
-        self._use_graph = (os.environ.get("R2D2_HIP_GRAPH", "1") != "0"
+        self._use_graph = (os.environ.get("R2D2_HIP_GRAPH", "0") == "1"
                            and not self.timing)
 
     def _mark(self, name):
