@@ -104,6 +104,70 @@ __global__ __launch_bounds__(256) void gemm_bias_act_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// gemm_bias_act2: fat-tile variant for the big GEMMs (FC 3136->512, LSTM
+// gate GEMMs 544->2048, head hidden 512->512).  128x128 block, each wave a
+// 64x64 tile of 4x4 fragments: 16 MFMAs per 8 operand loads per k-chunk
+// (vs 4 per 4 in the 64x64 kernel) — doubles the MFMA issue density while
+// operands stream from L2.  Same accumulation order per output element as
+// the small kernel (bitwise-identical results).
+// ---------------------------------------------------------------------------
+template <int ACT, bool HAS_BIAS, bool OUT_F32>
+__global__ __launch_bounds__(256) void gemm_bias_act2_kernel(
+    const __hip_bfloat16* __restrict__ A,   // (M, K)
+    const __hip_bfloat16* __restrict__ Wt,  // (N, K)
+    const float* __restrict__ bias,         // (N,)
+    void* __restrict__ out,                 // (M, N)
+    int M, int N, int K) {
+    int wave = threadIdx.x / WAVE;
+    int lane = threadIdx.x & (WAVE - 1);
+    int wr = wave >> 1, wc = wave & 1;
+    long row0 = (long)blockIdx.x * 128 + wr * 64;
+    long col0 = (long)blockIdx.y * 128 + wc * 64;
+    int frow = lane & 15;
+    int kseg = (lane >> 4) * 8;
+
+    f32x4 acc[4][4] = {};
+    for (int k0 = 0; k0 < K; k0 += 32) {
+        bf16x8 a[4], b[4];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            long r = row0 + i * 16 + frow;
+            a[i] = (r < M) ? load_bf16x8(A + r * K + k0 + kseg) : zero_bf16x8();
+            long c = col0 + i * 16 + frow;
+            b[i] = (c < N) ? load_bf16x8(Wt + c * K + k0 + kseg) : zero_bf16x8();
+        }
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+
+    int ccol = lane & 15;
+    int crow = (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                long rr = row0 + i * 16 + crow + r;
+                long cc = col0 + j * 16 + ccol;
+                if (rr < M && cc < N) {
+                    float v = acc[i][j][r];
+                    if (HAS_BIAS) v += bias[cc];
+                    if (ACT == 1) v = fmaxf(v, 0.f);
+                    if (OUT_F32)
+                        reinterpret_cast<float*>(out)[rr * N + cc] = v;
+                    else
+                        reinterpret_cast<__hip_bfloat16*>(out)[rr * N + cc] =
+                            f2bf(v);
+                }
+            }
+}
+
+// ---------------------------------------------------------------------------
 // gemm_dgrad: dA(M,K) = dY(M,N) @ W(K,N)^T-with-W-stored-(K,N)... i.e.
 //   dA[m][k] = sum_n dY[m][n] * W[k][n], W prepacked row-major (K, N).
 //   RELU_MASK: multiply dY by (act_out > 0) on load (fused ReLU backward,
@@ -337,16 +401,26 @@ torch::Tensor gemm_bias_act(torch::Tensor A, torch::Tensor Wt,
     bool has_bias = bias.defined() && bias.numel() > 0;
     auto out = torch::empty({M, N}, A.options().dtype(
         out_f32 ? torch::kFloat32 : torch::kBFloat16));
-    dim3 grid(cdiv(M, 64), cdiv(N, 64));
+    // fat-tile kernel for shapes big enough to fill the chip with 128x128
+    // blocks (the small-N head-output GEMMs keep the 64x64 kernel)
+    bool fat = (M >= 1024 && N >= 128);
+    dim3 grid(cdiv(M, fat ? 128 : 64), cdiv(N, fat ? 128 : 64));
     auto stream = at::cuda::getCurrentCUDAStream();
     const float* bptr = has_bias ? bias.data_ptr<float>() : nullptr;
     auto* a = reinterpret_cast<const __hip_bfloat16*>(A.data_ptr());
     auto* w = reinterpret_cast<const __hip_bfloat16*>(Wt.data_ptr());
 
 #define LAUNCH(ACT, HB, OF)                                                    \
-    hipLaunchKernelGGL((gemm_bias_act_kernel<ACT, HB, OF>), grid, dim3(256),   \
-                       0, stream.stream(), a, w, bptr, out.data_ptr(),         \
-                       (int)M, (int)N, (int)K)
+    do {                                                                       \
+        if (fat)                                                               \
+            hipLaunchKernelGGL((gemm_bias_act2_kernel<ACT, HB, OF>), grid,     \
+                               dim3(256), 0, stream.stream(), a, w, bptr,      \
+                               out.data_ptr(), (int)M, (int)N, (int)K);        \
+        else                                                                   \
+            hipLaunchKernelGGL((gemm_bias_act_kernel<ACT, HB, OF>), grid,      \
+                               dim3(256), 0, stream.stream(), a, w, bptr,      \
+                               out.data_ptr(), (int)M, (int)N, (int)K);        \
+    } while (0)
     if (has_bias) {
         if (act == 1) { if (out_f32) LAUNCH(1, true, true); else LAUNCH(1, true, false); }
         else          { if (out_f32) LAUNCH(0, true, true); else LAUNCH(0, true, false); }
