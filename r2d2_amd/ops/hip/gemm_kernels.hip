@@ -240,6 +240,71 @@ __global__ __launch_bounds__(256) void gemm_dgrad_kernel(
             }
 }
 
+// fat-tile dgrad (same 128x128 / 4x4-fragment scheme as gemm_bias_act2)
+template <bool RELU_MASK, bool OUT_MASK = false>
+__global__ __launch_bounds__(256) void gemm_dgrad2_kernel(
+    const __hip_bfloat16* __restrict__ dY,      // (M, N)
+    const __hip_bfloat16* __restrict__ act,     // (M, N) or null
+    const __hip_bfloat16* __restrict__ W,       // (K, N) row-major
+    __hip_bfloat16* __restrict__ dA,            // (M, Kout)
+    const __hip_bfloat16* __restrict__ outm,    // (M, Kout) or null
+    int M, int N, int K, int Kout) {
+    int wave = threadIdx.x / WAVE;
+    int lane = threadIdx.x & (WAVE - 1);
+    int wr = wave >> 1, wc = wave & 1;
+    long row0 = (long)blockIdx.x * 128 + wr * 64;
+    long col0 = (long)blockIdx.y * 128 + wc * 64;
+    int frow = lane & 15;
+    int nseg = (lane >> 4) * 8;
+
+    f32x4 acc[4][4] = {};
+    for (int n0 = 0; n0 < N; n0 += 32) {
+        bf16x8 a[4], b[4];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            long r = row0 + i * 16 + frow;
+            if (r < M) {
+                a[i] = load_bf16x8(dY + r * N + n0 + nseg);
+                if (RELU_MASK) {
+                    bf16x8 mv = load_bf16x8(act + r * N + n0 + nseg);
+#pragma unroll
+                    for (int e = 0; e < 8; ++e) {
+                        float vv = (float)a[i][e];
+                        a[i][e] = (__bf16)(((float)mv[e] > 0.f) ? vv : 0.f);
+                    }
+                }
+            } else {
+                a[i] = zero_bf16x8();
+            }
+            long c = col0 + i * 16 + frow;
+            b[i] = (c < K) ? load_bf16x8(W + c * N + n0 + nseg) : zero_bf16x8();
+        }
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+                acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+    int ccol = lane & 15;
+    int crow = (lane >> 4) * 4;
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                long rr = row0 + i * 16 + crow + r;
+                long cc = col0 + j * 16 + ccol;
+                if (rr < M && cc < Kout) {
+                    float v = acc[i][j][r];
+                    if (OUT_MASK)
+                        v = (bf2f(outm[rr * Kout + cc]) > 0.f) ? v : 0.f;
+                    dA[rr * Kout + cc] = f2bf(v);
+                }
+            }
+}
+
 // ---------------------------------------------------------------------------
 // gemm_wgrad: dWt(N,K) += dY(M,N)^T @ A(M,K), reduction over M with
 // LDS-staged 32-row tiles; row-chunked grid with f32 atomics into dW.
@@ -451,7 +516,8 @@ torch::Tensor gemm_dgrad(torch::Tensor dY, torch::Tensor act_out,
                     && out_mask.is_contiguous()
                     && out_mask.numel() == M * Kout);
     auto dA = torch::empty({M, Kout}, dY.options());
-    dim3 grid(cdiv(M, 64), cdiv(Kout, 64));
+    bool fat = (M >= 1024 && Kout >= 128);
+    dim3 grid(cdiv(M, fat ? 128 : 64), cdiv(Kout, fat ? 128 : 64));
     auto stream = at::cuda::getCurrentCUDAStream();
     auto* dy = reinterpret_cast<const __hip_bfloat16*>(dY.data_ptr());
     auto* ac = relu_mask
@@ -461,9 +527,16 @@ torch::Tensor gemm_dgrad(torch::Tensor dY, torch::Tensor act_out,
     auto* om = has_om
         ? reinterpret_cast<const __hip_bfloat16*>(out_mask.data_ptr()) : nullptr;
 #define LAUNCHD(RM, OM)                                                        \
-    hipLaunchKernelGGL((gemm_dgrad_kernel<RM, OM>), grid, dim3(256), 0,        \
-                       stream.stream(), dy, ac, w, da, om, (int)M, (int)N,     \
-                       (int)K, (int)Kout)
+    do {                                                                       \
+        if (fat)                                                               \
+            hipLaunchKernelGGL((gemm_dgrad2_kernel<RM, OM>), grid, dim3(256),  \
+                               0, stream.stream(), dy, ac, w, da, om, (int)M,  \
+                               (int)N, (int)K, (int)Kout);                     \
+        else                                                                   \
+            hipLaunchKernelGGL((gemm_dgrad_kernel<RM, OM>), grid, dim3(256),   \
+                               0, stream.stream(), dy, ac, w, da, om, (int)M,  \
+                               (int)N, (int)K, (int)Kout);                     \
+    } while (0)
     if (relu_mask) { if (has_om) LAUNCHD(true, true); else LAUNCHD(true, false); }
     else           { if (has_om) LAUNCHD(false, true); else LAUNCHD(false, false); }
 #undef LAUNCHD
