@@ -632,6 +632,169 @@ __global__ __launch_bounds__(256) void conv3p_wgrad_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// conv3p_wgrad_tap: tap-decomposed wgrad for the small-channel 3x3 convs.
+// Instead of rebuilding a (rows, 9*CIN) patch matrix per tile (the band
+// kernel's staging cost dwarfs its ~5 MFMAs per 64-row stage at C=16),
+// treat dW as NINE (COUT x CIN) GEMMs over the PIXEL dimension:
+//   dW[t] += dY^T @ in_shifted_by_tap_t
+// Both operands come straight off LDS via hardware transpose-reads
+// (B-side with per-lane ARBITRARY row addresses = the shifted pixels —
+// the tr16 instruction redistributes per 16-lane group, so rows need no
+// common stride).  Taps are split across the 4 waves (3/2/2/2), so
+// accumulators stay in registers (<= 48 VGPRs at 32x32) and each wave
+// issues ~1 MFMA per 2 transpose-reads.  One atomic flush per wg.
+// ---------------------------------------------------------------------------
+template <bool IN_U8, int CIN, int COUT, int HT, int TH, bool RELU_IN>
+__global__ __launch_bounds__(256) void conv3p_wgrad_tap_kernel(
+    const __hip_bfloat16* __restrict__ dY,   // (N, HT+2, HT+2, COUT) padded
+    const void* __restrict__ in,             // (N, HT+2, HT+2, CIN) padded
+    float* __restrict__ dWt,                 // (COUT, 9*CIN) f32
+    float* __restrict__ db,                  // (COUT,) f32
+    int N, int bands_per_wg) {
+    constexpr int PW = HT + 2;
+    constexpr int PH = HT + 2;
+    constexpr int NBANDS = (HT + TH - 1) / TH;
+    constexpr int NCO = (COUT + 15) / 16;
+    constexpr int NCI = (CIN + 15) / 16;
+    constexpr int LDY = COUT + 8;
+
+    __shared__ __hip_bfloat16 s_in[(TH + 2) * PW * CIN];
+    __shared__ __hip_bfloat16 s_dy[32][LDY];
+
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int i15 = lane & 15;
+    const int mseg = (lane >> 4) * 8;
+
+    int mytaps[3], ntap = 0;
+    for (int tt = wave; tt < 9; tt += 4) mytaps[ntap++] = tt;
+
+    f32x4 acc[3][NCO > 1 ? 2 : 1][NCI > 1 ? 2 : 1] = {};
+    float bias_acc = 0.f;
+
+    const long b_start = (long)blockIdx.x * bands_per_wg;
+    const long b_end = min((long)N * NBANDS, b_start + bands_per_wg);
+    for (long bb = b_start; bb < b_end; ++bb) {
+        const int band = (int)(bb % NBANDS);
+        const long n = bb / NBANDS;
+        const int y0 = band * TH;
+        const int th_eff = (y0 + TH <= HT) ? TH : (HT - y0);
+        const int npix = th_eff * HT;
+        __syncthreads();
+        {   // stage the input slab (dequant / relu-in once per element)
+            const long gbase = (n * PH + y0) * (long)PW * CIN;
+            const int rows = th_eff + 2;
+            for (int e = threadIdx.x * 8; e < rows * PW * CIN;
+                 e += blockDim.x * 8) {
+                bf16x8 v;
+                if (IN_U8)
+                    v = idequant8(
+                        reinterpret_cast<const unsigned char*>(in) + gbase + e);
+                else {
+                    v = iload8(
+                        reinterpret_cast<const __hip_bfloat16*>(in) + gbase + e);
+                    if (RELU_IN) v = irelu8(v);
+                }
+                *reinterpret_cast<bf16x8*>(&s_in[e]) = v;
+            }
+        }
+
+        for (int p0 = 0; p0 < npix; p0 += 32) {
+            __syncthreads();
+            {   // stage 32 pixels of dY
+                int t = threadIdx.x;
+                if (t < 32 * (COUT / 8)) {
+                    int mrow = t / (COUT / 8);
+                    int col = (t % (COUT / 8)) * 8;
+                    int pp = p0 + mrow;
+                    bf16x8 v = izero();
+                    if (pp < npix) {
+                        int yl = pp / HT, x = pp % HT;
+                        v = iload8(dY + ((n * PH + y0 + yl + 1) * (long)PW
+                                         + x + 1) * COUT + col);
+                    }
+                    *reinterpret_cast<bf16x8*>(&s_dy[mrow][col]) = v;
+                }
+            }
+            __syncthreads();
+
+            // A-frags: dY columns (co) over the 8 pixels mseg..mseg+7
+            bf16x8 a[NCO > 1 ? 2 : 1];
+#pragma unroll
+            for (int c2 = 0; c2 < NCO; ++c2)
+                a[c2] = lds_col_frag8<LDY>(&s_dy[0][0], mseg, c2 * 16, lane);
+
+            // per-lane shifted-pixel geometry (two reads: pixels +0..3,
+            // +4..7); OOB pixels clamp to 0 — their dY rows are zero
+            int q1 = p0 + mseg + (i15 >> 2);
+            int q2 = q1 + 4;
+            if (q1 >= npix) q1 = 0;
+            if (q2 >= npix) q2 = 0;
+            const int yl1 = q1 / HT, x1 = q1 % HT;
+            const int yl2 = q2 / HT, x2 = q2 % HT;
+            // piece column inside the ci fragment (clamped for CIN=8)
+            const int pc = (CIN >= 16) ? 4 * (i15 & 3)
+                                       : min(4 * (i15 & 3), CIN - 4);
+
+#pragma unroll
+            for (int ti = 0; ti < 3; ++ti) {
+                if (ti >= ntap) break;
+                const int tt = mytaps[ti];
+                const int dy_ = tt / 3, dx_ = tt % 3;
+#pragma unroll
+                for (int c1 = 0; c1 < NCI; ++c1) {
+                    auto* p1 = (__attribute__((address_space(3))) cmn_bf16x4*)
+                        &s_in[((yl1 + dy_) * PW + x1 + dx_) * CIN
+                              + c1 * 16 + pc];
+                    auto* p2 = (__attribute__((address_space(3))) cmn_bf16x4*)
+                        &s_in[((yl2 + dy_) * PW + x2 + dx_) * CIN
+                              + c1 * 16 + pc];
+                    union {
+                        struct { cmn_bf16x4 lo, hi; } p;
+                        bf16x8 v;
+                    } u;
+                    u.p.lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+                    u.p.hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p2);
+                    bf16x8 b = u.v;
+                    if (CIN < 16 && i15 >= CIN) b = izero();
+#pragma unroll
+                    for (int c2 = 0; c2 < NCO; ++c2)
+                        acc[ti][c2][c1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a[c2], b, acc[ti][c2][c1], 0, 0, 0);
+                }
+            }
+            if (threadIdx.x < COUT) {
+                int c = threadIdx.x;
+                for (int mr = 0; mr < 32; ++mr)
+                    bias_acc += bf2f(s_dy[mr][c]);
+            }
+        }
+    }
+
+    int ccol = lane & 15;
+    int crow = (lane >> 4) * 4;
+#pragma unroll
+    for (int ti = 0; ti < 3; ++ti) {
+        if (ti >= ntap) break;
+        const int tt = mytaps[ti];
+#pragma unroll
+        for (int c2 = 0; c2 < NCO; ++c2)
+#pragma unroll
+            for (int c1 = 0; c1 < NCI; ++c1)
+#pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    int co = c2 * 16 + crow + r;
+                    int ci = c1 * 16 + ccol;
+                    if (co < COUT && ci < CIN)
+                        atomicAdd(&dWt[(long)co * 9 * CIN + tt * CIN + ci],
+                                  acc[ti][c2][c1][r]);
+                }
+    }
+    if (threadIdx.x < COUT)
+        atomicAdd(&db[threadIdx.x], bias_acc);
+}
+
+// ---------------------------------------------------------------------------
 // conv3p_wgrad_band: hybrid band wgrad.  The input slab of one (image,
 // band) is staged through LDS once (dequant / relu-in applied there), and
 // each 32/64-row tile's patch matrix s_a is rebuilt LDS->LDS from the slab
@@ -1161,6 +1324,40 @@ std::vector<torch::Tensor> conv3p_wgrad(torch::Tensor dY, torch::Tensor in,
                            stream.stream(), dy, x, dWt.data_ptr<float>(),     \
                            db.data_ptr<float>(), (int)N, (int)bpw);           \
     } while (0)
+    // tap-decomposed wgrad (default; R2D2_IMPALA_TAP_WGRAD=0 falls back to
+    // the patch-matrix band kernel for comparison)
+    static const bool use_tap = [] {
+        const char* e = getenv("R2D2_IMPALA_TAP_WGRAD");
+        return !(e && e[0] == '0');
+    }();
+#define WGT(U8, CIN_, CO_, HT_, TH_, RELU_)                                   \
+    do {                                                                      \
+        long tb = (long)N * ((HT_ + TH_ - 1) / TH_);                          \
+        long bpw = std::max(1L, (tb + 1023) / 1024);                          \
+        hipLaunchKernelGGL((conv3p_wgrad_tap_kernel<U8, CIN_, CO_, HT_,       \
+                                                    TH_, RELU_>),             \
+                           dim3(icdiv(tb, bpw)), dim3(256), 0,                \
+                           stream.stream(), dy, x, dWt.data_ptr<float>(),     \
+                           db.data_ptr<float>(), (int)N, (int)bpw);           \
+    } while (0)
+    if (use_tap) {
+        bool tdone = true;
+        if (H == 84 && u8 && COUT == 16 && !relu_in)
+            WGT(true, 8, 16, 84, 6, false);
+        else if (H == 42 && !u8 && CIN == 16 && COUT == 16 && relu_in)
+            WGT(false, 16, 16, 42, 12, true);
+        else if (H == 42 && !u8 && CIN == 16 && COUT == 32 && !relu_in)
+            WGT(false, 16, 32, 42, 12, false);
+        else if (H == 21 && !u8 && CIN == 32 && COUT == 32 && relu_in)
+            WGT(false, 32, 32, 21, 21, true);
+        else if (H == 21 && !u8 && CIN == 32 && COUT == 32 && !relu_in)
+            WGT(false, 32, 32, 21, 21, false);
+        else if (H == 11 && !u8 && CIN == 32 && COUT == 32 && relu_in)
+            WGT(false, 32, 32, 11, 11, true);
+        else tdone = false;
+        if (tdone) return {dWt, db};
+    }
+#undef WGT
     bool bdone = true;
     if (H == 84 && u8 && COUT == 16 && !relu_in) WGB(true, 8, 16, 84, 6, false);
     else if (H == 42 && !u8 && CIN == 16 && COUT == 16 && relu_in)
