@@ -447,7 +447,8 @@ __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
     __hip_bfloat16* __restrict__ dgates,       // (B, T, 4H) out
     GridBar* bar, int B, int T, int nblocks, int nhalves) {
     constexpr int WGS = H / UNITS;
-    constexpr int KSPLIT = (UNITS == 16 && BROWS == 32) ? 2 : 1;
+    constexpr int KSPLIT = (UNITS == 16 && BROWS == 32) ? 2
+                           : (UNITS == 16 && BROWS == 16) ? 4 : 1;
     constexpr int NCWAVE = UNITS / 16;         // col fragments across waves
     const int half = blockIdx.x / WGS;
     const int wid = blockIdx.x % WGS;
@@ -480,9 +481,10 @@ __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
     // UNITS=16, KSPLIT=2: wave pairs split the 4H reduction.
     // UNITS=32: waves tile (row half x col half), full-K.
     const int wrow0 = (NCWAVE == 2) ? (wave >> 1) * 16
+                      : (KSPLIT == 4) ? 0
                       : (KSPLIT == 1 ? wave : (wave & 1)) * 16;
     const int wcol0 = (NCWAVE == 2) ? (wave & 1) * 16 : 0;
-    const int wk = (KSPLIT == 1) ? 0 : (wave >> 1);
+    const int wk = (KSPLIT == 1) ? 0 : (KSPLIT == 4) ? wave : (wave >> 1);
     const int frow = lane & 15;
     const int kseg = (lane >> 4) * 8;
     constexpr int NCHUNK = 4 * H / 32;
@@ -542,7 +544,8 @@ __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
             bool active_next = !last && ((t + 1) < lens[bg]);
             float ext = dHext[((long)bg * T + t) * H + u];
             float rec = s_rec[0][b][jl];
-            if (KSPLIT == 2) rec += s_rec[1][b][jl];
+#pragma unroll
+            for (int kk = 1; kk < KSPLIT; ++kk) rec += s_rec[kk][b][jl];
             if (last) {
                 dh = ext;
                 dc_in = 0.f;
@@ -835,8 +838,16 @@ torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
         const char* e = getenv("R2D2_LSTM_BWD_UNITS");
         return e ? atoi(e) : 16;
     }();
+    // R2D2_LSTM_BWD_BROWS=16 quarters the batch (128 wgs, KSPLIT=4)
+    static const int brows_env = [] {
+        const char* e = getenv("R2D2_LSTM_BWD_BROWS");
+        return e ? atoi(e) : 32;
+    }();
     int nhalves = B > 32 ? 2 : 1;
     int units = (nhalves == 2 && bunits_env == 32) ? 32 : 16;
+    int brows = (nhalves == 2 && units == 16 && brows_env == 16 && B > 48)
+                    ? 16 : (nhalves == 2 ? 32 : 64);
+    if (brows == 16) nhalves = 4;
     int nblocks = (int)H / units * nhalves;
     auto stream = at::cuda::getCurrentCUDAStream();
     zero_ws(barrier_ws, stream.stream());
@@ -852,7 +863,8 @@ torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
         reinterpret_cast<__hip_bfloat16*>(dgates.data_ptr()),                 \
         reinterpret_cast<GridBar*>(barrier_ws.data_ptr()),                    \
         (int)B, (int)T, nblocks, nhalves)
-    if (nhalves == 2) {
+    if (brows == 16) LSTMB(16, 16);
+    else if (nhalves == 2) {
         if (units == 32) LSTMB(32, 32); else LSTMB(32, 16);
     } else {
         LSTMB(64, 16);
