@@ -781,9 +781,18 @@ std::vector<torch::Tensor> lstm_fwd(
         const char* e = getenv("R2D2_LSTM_UNITS");
         return e ? atoi(e) : 16;
     }();
+    // R2D2_LSTM_FWD_BROWS=16 quarters the batch (256 wgs for the dual-net
+    // launch = every CU)
+    static const int fbrows_env = [] {
+        const char* e = getenv("R2D2_LSTM_FWD_BROWS");
+        return e ? atoi(e) : 32;
+    }();
     int units = (units_env == 8) ? 8 : 16;
     int nhalves = B > 32 ? 2 : 1;
     if (nhalves == 1) units = 8;   // single-half keeps the classic layout
+    int fbrows = (nhalves == 2 && units == 16 && fbrows_env == 16 && B > 48)
+                     ? 16 : (nhalves == 2 ? 32 : 64);
+    if (fbrows == 16) nhalves = 4;
     int wgs = (int)H / units;
     int nblocks = wgs * (two ? 2 : 1) * nhalves;
     auto stream = at::cuda::getCurrentCUDAStream();
@@ -807,7 +816,8 @@ std::vector<torch::Tensor> lstm_fwd(
         want_stash ? bp(stash) : nullptr,                                     \
         reinterpret_cast<GridBar*>(barrier_ws.data_ptr()),                    \
         (int)B, (int)T, nblocks, nhalves)
-    if (nhalves == 2) {
+    if (fbrows == 16) LSTMF(16, 16);
+    else if (nhalves == 2) {
         if (units == 16) LSTMF(32, 16); else LSTMF(32, 8);
     } else {
         LSTMF(64, 8);
@@ -838,10 +848,12 @@ torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
         const char* e = getenv("R2D2_LSTM_BWD_UNITS");
         return e ? atoi(e) : 16;
     }();
-    // R2D2_LSTM_BWD_BROWS=16 quarters the batch (128 wgs, KSPLIT=4)
+    // batch QUARTERED by default (128 wgs x 2 CUs' worth each, KSPLIT=4):
+    // measured 0.71 vs 0.95 ms at BROWS=32 (the backward is CU-coverage
+    // bound; R2D2_LSTM_BWD_BROWS=32 selects the half-split for comparison)
     static const int brows_env = [] {
         const char* e = getenv("R2D2_LSTM_BWD_BROWS");
-        return e ? atoi(e) : 32;
+        return e ? atoi(e) : 16;
     }();
     int nhalves = B > 32 ? 2 : 1;
     int units = (nhalves == 2 && bunits_env == 32) ? 32 : 16;
