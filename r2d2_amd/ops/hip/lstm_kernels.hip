@@ -241,8 +241,6 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
     float* __restrict__ Cout0,                // (B, T+1, H)
     float* __restrict__ Cout1,
     __hip_bfloat16* __restrict__ stash0,      // (B, T, 4H) post-nonlin gates
-    __hip_bfloat16* __restrict__ hx0,         // (2, B, H) ping-pong exchange
-    __hip_bfloat16* __restrict__ hx1,
     GridBar* bar, int B, int T, int nblocks, int nhalves) {
     constexpr int WGS_PER_HALF = H / UNITS;
     constexpr int GCOLS = 4 * UNITS;          // gate columns per wg
@@ -261,10 +259,6 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
     __hip_bfloat16* Hout = net ? Hout1 : Hout0;
     float* Cout = net ? Cout1 : Cout0;
     __hip_bfloat16* stash = net ? nullptr : stash0;
-    // compact (2, B, H) ping-pong buffer for the cross-workgroup h
-    // exchange: dense 64 KB per phase stays hot in L2, unlike the
-    // (T+1)*H-strided Hout rows the consumers would otherwise re-read
-    __hip_bfloat16* hx = net ? hx1 : hx0;
     unsigned* ctr = &bar->flags[(net * nhalves + half) * 32];
 
     __shared__ __hip_bfloat16 s_whh[GCOLS][H + 8];
@@ -279,14 +273,12 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
         int g = c / UNITS, j = c % UNITS;
         lstore8(&s_whh[c][k], lload8(Whh + (long)(g * H + u0 + j) * H + k));
     }
-    // h0 -> Hout[:,0] + exchange buffer 0; c0 -> LDS-resident cell state
+    // h0 -> Hout[:,0] (this wg's slice); c0 -> LDS-resident cell state
     for (int p = threadIdx.x; p < Bl * UNITS; p += blockDim.x) {
         int bl = p / UNITS, j = p % UNITS;
         int b = b0 + bl;
         int u = u0 + j;
-        float h0v = init[(long)b * H + u];
-        Hout[((long)b * (T + 1)) * H + u] = f2bf(h0v);
-        hx[(long)b * H + u] = f2bf(h0v);
+        Hout[((long)b * (T + 1)) * H + u] = f2bf(init[(long)b * H + u]);
         Cout[((long)b * (T + 1)) * H + u] = init[((long)B + b) * H + u];
         s_c[bl][j] = init[((long)B + b) * H + u];
     }
@@ -310,16 +302,15 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
         if (!await_count(ctr, (unsigned)WGS_PER_HALF * (t + 1),
                          &bar->poison))
             return;
-        // bulk-stage this half's h_prev (Bl x H) into LDS from the dense
-        // exchange buffer (phase t & 1)
+        // bulk-stage this half's h_prev (Bl x H) into LDS
         {
-            const __hip_bfloat16* src = hx + (long)(t & 1) * B * H;
+            const long base = (long)t * H;
 #pragma unroll
             for (int e = threadIdx.x; e < CHUNKS; e += 256) {
                 int row = e / (H / 8);
                 int k8 = (e % (H / 8)) * 8;
                 bf16x8 v = (row < Bl)
-                    ? lload8(src + (long)(b0 + row) * H + k8)
+                    ? lload8(Hout + ((long)(b0 + row) * (T + 1)) * H + base + k8)
                     : lzero8();
                 lstore8(&s_h[row][k8], v);
             }
@@ -388,20 +379,17 @@ __global__ __launch_bounds__(256, 1) void lstm_fwd_kernel(
             s_gates[b][3 * UNITS + j] = o_;
         }
         __syncthreads();
-        // vectorized writers: h slice (Hout + exchange), c, gate stash
+        // vectorized writers: h slice, c, gate stash
         {
             int tid = threadIdx.x;
             constexpr int H8 = UNITS / 8;        // 16-B pieces per h row
             if (tid < 64 * H8) {
                 int b = tid / H8, piece = tid % H8;
                 if (b < Bl) {
-                    bf16x8 v = *reinterpret_cast<bf16x8*>(
-                        &s_hrow[b][piece * 8]);
                     long off = ((long)(b0 + b) * (T + 1) + t + 1) * H + u0
                                + piece * 8;
-                    lstore8(Hout + off, v);
-                    lstore8(hx + ((long)((t + 1) & 1) * B + b0 + b) * H + u0
-                            + piece * 8, v);
+                    lstore8(Hout + off, *reinterpret_cast<bf16x8*>(
+                        &s_hrow[b][piece * 8]));
                 }
             } else if (tid >= 128) {
                 // c writers: threads [128, 256) cover BROWS rows x C4
@@ -784,8 +772,6 @@ std::vector<torch::Tensor> lstm_fwd(
     auto H1 = two ? torch::empty({B, T + 1, H}, bf) : torch::Tensor();
     auto C1 = two ? torch::empty({B, T + 1, H}, f32) : torch::Tensor();
     auto stash = want_stash ? torch::empty({B, T, H4}, bf) : torch::Tensor();
-    auto hx0 = torch::empty({2, B, H}, bf);
-    auto hx1 = two ? torch::empty({2, B, H}, bf) : torch::Tensor();
 
     // UNITS: hidden units per workgroup.  16 is the measured default for
     // the dual-net batch-split launch (fewer/fatter workgroups: half the
@@ -827,7 +813,7 @@ std::vector<torch::Tensor> lstm_fwd(
         two ? init1.data_ptr<float>() : nullptr,                              \
         lens.data_ptr<int>(), bp(H0), bp(H1),                                 \
         C0.data_ptr<float>(), two ? C1.data_ptr<float>() : nullptr,           \
-        want_stash ? bp(stash) : nullptr, bp(hx0), bp(hx1),                   \
+        want_stash ? bp(stash) : nullptr,                                     \
         reinterpret_cast<GridBar*>(barrier_ws.data_ptr()),                    \
         (int)B, (int)T, nblocks, nhalves)
     if (fbrows == 16) LSTMF(16, 16);
