@@ -342,14 +342,6 @@ class HipNetworkEngine:
         self._graph_key = None
         self._use_graph = (os.environ.get("R2D2_HIP_GRAPH", "0") == "1"
                            and not self.timing)
-        # side stream for the target-net encoder (concurrent with the
-        # online chain; R2D2_TGT_STREAM=0 serializes for comparison).
-        # IMPALA shares the packed-frame arena between the two nets, so it
-        # keeps the serial path.
-        self._tgt_stream = None
-        if (os.environ.get("R2D2_TGT_STREAM", "1") != "0"
-                and not self.impala):
-            self._tgt_stream = torch.cuda.Stream(device=self.device)
 
     def _mark(self, name):
         if self.timing:
@@ -574,27 +566,12 @@ class HipNetworkEngine:
         if self.impala:
             # pack frames once; online and target share the padded u8 tensor
             self._xp = impala_ops.pack_obs(m, self.online.imp, obs_hwc)
-        # the target-net encoder chain is independent of the online one
-        # until the dual-net LSTM launch — run it on a side stream so its
-        # kernels fill the online chain's tail waves
-        tgt_stream = self._tgt_stream
-        if tgt_stream is not None:
-            tgt_stream.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(tgt_stream):
-                lat_t, _ = self._encoder_fwd(self.target, obs_hwc,
-                                             want_stash=False)
-                _, X_t = self._lstm_input(self.target, lat_t, la, lr)
         lat_o, enc_stash = self._encoder_fwd(self.online, obs_hwc)
         self._mark("enc_online")
-        if tgt_stream is None:
-            lat_t, _ = self._encoder_fwd(self.target, obs_hwc,
-                                         want_stash=False)
+        lat_t, _ = self._encoder_fwd(self.target, obs_hwc, want_stash=False)
         self._mark("enc_target")
         rin_o, X_o = self._lstm_input(self.online, lat_o, la, lr)
-        if tgt_stream is None:
-            _, X_t = self._lstm_input(self.target, lat_t, la, lr)
-        else:
-            torch.cuda.current_stream().wait_stream(tgt_stream)
+        _, X_t = self._lstm_input(self.target, lat_t, la, lr)
         self._mark("lstm_input")
         Xo = X_o.view(B, T, 4 * H)
         Xt = X_t.view(B, T, 4 * H)
