@@ -254,15 +254,28 @@ __global__ __launch_bounds__(256) void conv_dgrad_kernel(
 // ((act_x > 0) at the dX address) on store, so the next layer's dgrad and
 // wgrad consume a pre-masked tensor directly.
 // ---------------------------------------------------------------------------
-template <int TAPS, int NCOL, int CO_T, bool OUT_MASK>
+// GT (geometry template): nonzero = compile-time (OH==OW, XH==XW, S) for
+// the hot Nature shapes — the per-lane row->(n,y,x) and tap divisions
+// lower to mul-shift by constants (this kernel is otherwise VALU-bound on
+// address math, r06 PMC 72:1 VALU:MFMA).  Encoded GT = OH*1000 + XH*10 + S.
+template <int TAPS, int NCOL, int CO_T, bool OUT_MASK, int GT = 0>
 __global__ __launch_bounds__(256) void conv_dgrad_dense_kernel(
     const __hip_bfloat16* __restrict__ dY,   // (N, OH, OW, COUT)
     const __hip_bfloat16* __restrict__ Wd,   // (CIN, TAPS*COUT)
     const __hip_bfloat16* __restrict__ actx, // (N, XH, XW, CIN) or null
     __hip_bfloat16* __restrict__ dX,         // (N, XH, XW, CIN)
     const int* __restrict__ taps,            // (TAPS, 2): dy, dx
-    int Mc, int YY, int XX, int y0, int x0, int S,
-    int OH, int OW, int COUT, int XH, int XW, int CIN) {
+    int Mc, int YY_, int XX_, int y0, int x0, int S_,
+    int OH_, int OW_, int COUT, int XH_, int XW_, int CIN) {
+    const int S = GT ? (GT % 10) : S_;
+    const int OH = GT ? (GT / 1000) : OH_;
+    const int OW = GT ? (GT / 1000) : OW_;
+    const int XH = GT ? ((GT / 10) % 100) : XH_;
+    const int XW = GT ? ((GT / 10) % 100) : XW_;
+    // for both hot geometries XH/S equals the class grid extent for every
+    // parity offset (20/2=10, 9/1=9), keeping YY/XX compile-time under GT
+    const int YY = GT ? (XH / S) : YY_;
+    const int XX = GT ? (XW / S) : XX_;
     const int COUTc = CO_T ? CO_T : COUT;
     const int K = TAPS * COUTc;
     int wave = threadIdx.x / WAVE;
@@ -849,16 +862,27 @@ torch::Tensor conv_dgrad_dense(torch::Tensor dY, torch::Tensor Wd,
         ? reinterpret_cast<const __hip_bfloat16*>(actx.data_ptr()) : nullptr;
     auto* dx = reinterpret_cast<__hip_bfloat16*>(dX.data_ptr());
     const int* tp = taps.data_ptr<int>();
-#define DD1(T, NC, CO, OM)                                                     \
-    hipLaunchKernelGGL((conv_dgrad_dense_kernel<T, NC, CO, OM>), grid,         \
+#define DD1(T, NC, CO, OM, GT)                                                 \
+    hipLaunchKernelGGL((conv_dgrad_dense_kernel<T, NC, CO, OM, GT>), grid,     \
                        dim3(256), 0, stream.stream(), dy, w, am, dx, tp,       \
                        (int)Mc, (int)YY, (int)XX, (int)y0, (int)x0, (int)S,    \
                        (int)OH, (int)OW, (int)COUT, (int)XH, (int)XW, (int)CIN)
-#define DDM(T, NC, CO)                                                         \
-    do { if (has_m) DD1(T, NC, CO, true); else DD1(T, NC, CO, false); } while (0)
+#define DDM(T, NC, CO, GT)                                                     \
+    do { if (has_m) DD1(T, NC, CO, true, GT);                                  \
+         else DD1(T, NC, CO, false, GT); } while (0)
+    // compile-time geometry for the hot Nature shapes (address math
+    // lowers to mul-shift; GT = OH*1000 + XH*10 + S)
+    if (TAPS == 9 && CIN == 64 && COUT == 64 && OH == 7 && XH == 9 && S == 1) {
+        DDM(9, 64, 64, 7091);
+        return dX;
+    }
+    if (TAPS == 4 && CIN == 32 && COUT == 64 && OH == 9 && XH == 20 && S == 2) {
+        DDM(4, 32, 64, 9202);
+        return dX;
+    }
 #define DDLAUNCH(T)                                                            \
-    if (CIN <= 32) { if (COUT == 64) DDM(T, 32, 64); else DDM(T, 32, 0); }     \
-    else { if (COUT == 64) DDM(T, 64, 64); else DDM(T, 64, 0); }
+    if (CIN <= 32) { if (COUT == 64) DDM(T, 32, 64, 0); else DDM(T, 32, 0, 0); } \
+    else { if (COUT == 64) DDM(T, 64, 64, 0); else DDM(T, 64, 0, 0); }
     if (TAPS == 4) { DDLAUNCH(4); }
     else if (TAPS == 9) { DDLAUNCH(9); }
     else TORCH_CHECK(false, "unsupported tap count");
