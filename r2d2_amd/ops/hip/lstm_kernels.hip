@@ -5,28 +5,33 @@
 //
 // Structure (forward):
 // - X = rin @ W_ih^T + b precomputed for all (B, T) steps by the MFMA GEMM.
-// - Persistent launch, one workgroup per 8 hidden units (x2 networks in one
-//   launch).  W_hh slice and the cell state c stay LDS-resident for all T
-//   steps; each step bulk-stages the previous h (B x H) into LDS with
-//   independent coalesced 16 B loads, runs the gate GEMM entirely out of
-//   LDS, and advances h/c with vectorized stores.
+// - Persistent launch, one workgroup per 16 hidden units x batch half
+//   (x2 networks in one launch = 128 workgroups; slice width and batch
+//   split are template parameters — 8-unit and quartered variants stay
+//   selectable for comparison).  W_hh slice and the cell state c stay
+//   LDS-resident for all T steps; each step bulk-stages the previous h
+//   into LDS with coalesced 16 B loads, runs the gate GEMM entirely out
+//   of LDS, and advances h/c with vectorized stores.
 // - NO global barrier: the time index makes the h exchange naturally
 //   double-buffered, so each step is synchronized by PRODUCER FLAGS only
 //   (HIP guide §6 G16: plain h stores -> per-wave vmcnt drain ->
 //   __syncthreads -> release fence -> relaxed per-slice flag; consumers
 //   poll 64 flags lane-parallel, one acquire fence, then plain loads).
-//   Flag words are zeroed by hipMemsetAsync before every launch; max skew
+//   Flag words are zeroed by a tiny kernel before every launch (a
+//   CAPTURED hipMemsetAsync replays a garbage fill value — see
+//   zero_gridbar_kernel); max skew
 //   between workgroups is self-limiting (each wg is both producer and
 //   consumer).  Spins are bounded; a poison word aborts the launch.
 // - Per-sample length masks replace pack_padded semantics: masked steps
 //   copy h/c through unchanged.
 //
-// Backward (online net): reverse recurrence with the same flag protocol on
-// the dgates stream; dgates_{t+1} (B x 4H) is streamed through LDS in
-// 256-column pieces (cooperative staging, then all-LDS MFMA); per-step gate
-// math reads the stashed activations via vectorized row staging.  The bulk
-// weight gradients (dW_hh, dW_ih, db) and dX are plain GEMMs outside the
-// kernel (gemm_wgrad / gemm_dgrad over B*T rows).
+// Backward (online net): reverse recurrence with the same counter protocol
+// on the dgates stream — one workgroup per 16 hidden units x batch QUARTER
+// (128 workgroups; measured faster than batch halves — the backward is
+// CU-coverage bound); the recurrent GEMM streams dgates_{t+1} through a
+// 4-deep register prefetch ring with the 4H reduction split across waves.
+// The bulk weight gradients (dW_hh, dW_ih, db) and dX are plain GEMMs
+// outside the kernel (gemm_wgrad / gemm_dgrad over B*T rows).
 //
 // Gate order matches torch.nn.LSTM: [i, f, g, o] chunks of H.
 
