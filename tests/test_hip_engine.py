@@ -212,3 +212,33 @@ def test_engine_matches_eager_on_ragged_layouts():
             if cs < 0.98 and rel > 0.12:
                 bad.append((trial, n, cs, rel))
         assert not bad, bad
+
+
+def test_engine_checkpoint_resume_roundtrip(tmp_path):
+    """Save/resume on the HIP engine path: the 4-tuple + .train.pth sidecar
+    must restore flat params AND the engine's fused-Adam moments so
+    training continues bit-compatibly."""
+    import os
+    c, eager, hip = make_learners(seed=21)
+    from bench import build_batch
+    batch = build_batch(c, torch.device("cuda"), seed=31)
+    for _ in range(3):
+        loss, prio = hip.train_step(batch)
+    hip.model_dir = str(tmp_path)
+    hip.game_name = "Resume"
+    hip.save(start_time=0.0)
+    path = os.path.join(str(tmp_path), f"Resume{hip.num_updates}.pth")
+    assert os.path.exists(path) and os.path.exists(
+        path[:-4] + ".train.pth")
+
+    _, _, hip2 = make_learners(seed=99)   # different init
+    hip2.load_checkpoint(path)
+    assert torch.equal(hip2.engine.flat_param, hip.engine.flat_param)
+    assert torch.equal(hip2.engine.exp_avg, hip.engine.exp_avg)
+    assert torch.equal(hip2.engine.exp_avg_sq, hip.engine.exp_avg_sq)
+    assert hip2.engine.adam_t == hip.engine.adam_t
+    assert hip2.num_updates == hip.num_updates
+
+    l1, _ = hip.train_step(batch)
+    l2, _ = hip2.train_step(batch)
+    assert abs(float(l1) - float(l2)) < 1e-5, (float(l1), float(l2))
