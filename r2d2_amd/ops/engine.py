@@ -426,8 +426,10 @@ class HipNetworkEngine:
         batches (the common case) skip the python loop and re-upload."""
         key = (T, burn.numpy().tobytes(), learn.numpy().tobytes(),
                fwd.numpy().tobytes())
-        if self._pos_cache is not None and self._pos_cache[0] == key:
-            return self._pos_cache[1]
+        if self._pos_cache is None:
+            self._pos_cache = {}
+        if key in self._pos_cache:
+            return self._pos_cache[key]
         n = self.cfg.forward_steps
         bu = burn.numpy().astype(np.int64)
         le = learn.numpy().astype(np.int64)
@@ -456,7 +458,9 @@ class HipNetworkEngine:
         seg = torch.zeros(len(burn) + 1, dtype=torch.int32)
         seg[1:] = torch.cumsum(learn.to(torch.int32), 0)
         out = (lp_t, tp_t, row_of_t, lens_dev, seg.to(dev))
-        self._pos_cache = (key, out)
+        if len(self._pos_cache) > 64:   # bounded (fixed layouts in practice)
+            self._pos_cache.clear()
+        self._pos_cache[key] = out
         return out
 
     # ------------------------------------------------------------------
@@ -555,9 +559,19 @@ class HipNetworkEngine:
             obs_hwc = obs_hwc.contiguous()
         la = batch.last_action.to(dev)
         lr = batch.last_reward.to(dev)
-        lp_t, tp_t, row_of, lens, seg = self._positions(
-            batch.burn_in_steps, batch.learning_steps,
-            batch.forward_steps, T)
+        md = getattr(batch, "meta_dev", None)
+        if md is not None:
+            # replay-sampled batches: position arrays computed on-device
+            # from the sample metadata (one kernel; every such batch is
+            # ragged, so the host cache below would miss every step)
+            lp_t, tp_t, row_of, lens = m.positions_meta(
+                md, batch.seg_dev, T, c.forward_steps,
+                batch.action.shape[0], int(c.learning_steps))
+            seg = batch.seg_dev
+        else:
+            lp_t, tp_t, row_of, lens, seg = self._positions(
+                batch.burn_in_steps, batch.learning_steps,
+                batch.forward_steps, T)
         R = lp_t.shape[0]
         init = batch.hidden.float().contiguous()   # (2, B, H)
 

@@ -21,6 +21,9 @@ std::vector<torch::Tensor> sumtree_sample(torch::Tensor tree, int64_t leaf_offse
                                           int64_t max_idx);
 
 // replay_gather.hip
+std::vector<torch::Tensor> positions_meta(torch::Tensor meta, torch::Tensor seg,
+                                          int64_t T, int64_t n, int64_t R,
+                                          int64_t max_learn);
 std::vector<torch::Tensor> replay_gather_meta(
     torch::Tensor idx, torch::Tensor burn_s, torch::Tensor learn_s,
     torch::Tensor fwd_s, torch::Tensor obs_start_s, torch::Tensor learn_off_s,
@@ -137,6 +140,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "GPU sum-tree stratified sample -> (idx, prio, is_weight)");
     m.def("replay_gather_meta", &replay_gather_meta,
           "Sampled-sequence metadata + segment offsets");
+    m.def("positions_meta", &positions_meta,
+          "engine gather/scatter position arrays from device-side sample "
+          "metadata (no host round trip per ragged batch)");
     m.def("replay_gather_batch", &replay_gather_batch,
           "On-device padded batch assembly from the GPU block store");
     m.def("gemm_bias_act", &gemm_bias_act, "MFMA GEMM + bias + activation");

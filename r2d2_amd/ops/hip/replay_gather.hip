@@ -136,6 +136,60 @@ __global__ void gather_flat_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// positions_meta: the engine's gather/scatter position arrays straight from
+// the device-side sample metadata — lp/tp (learning- and
+// target-position rows of the (B, T+1, H) LSTM output), the inverse map
+// feeding the dHext scatter, and per-sample lengths.  Replay-sampled
+// batches have a different ragged layout every step, so the host-side
+// numpy rebuild this replaces sat in the headline loop.
+// ---------------------------------------------------------------------------
+__global__ void positions_meta_kernel(
+    const int* __restrict__ meta,   // (5, B): burn, learn, fwd, ...
+    const int* __restrict__ seg,    // (B+1,)
+    long* __restrict__ lp,          // (R,)
+    long* __restrict__ tp,          // (R,)
+    int* __restrict__ row_of,       // (B*T,) pre-filled with -1
+    int* __restrict__ lens,         // (B,)
+    int B, int T, int n, int max_learn) {
+    int tid = blockIdx.x * blockDim.x + threadIdx.x;
+    int b = tid / max_learn, j = tid % max_learn;
+    if (b >= B) return;
+    int burn = meta[b], learn = meta[B + b], fwd = meta[2 * B + b];
+    if (j == 0) lens[b] = burn + learn + fwd;
+    if (j < learn) {
+        long r = seg[b] + j;
+        int tl = burn + j;
+        int tt = min(tl + n, burn + learn + fwd - 1);
+        lp[r] = (long)b * (T + 1) + tl + 1;
+        tp[r] = (long)b * (T + 1) + tt + 1;
+        row_of[(long)b * T + tl] = (int)r;
+    }
+}
+
+std::vector<torch::Tensor> positions_meta(torch::Tensor meta, torch::Tensor seg,
+                                          int64_t T, int64_t n, int64_t R,
+                                          int64_t max_learn) {
+    TORCH_CHECK(meta.is_cuda() && meta.dtype() == torch::kInt32);
+    int B = meta.size(1);
+    auto i64 = meta.options().dtype(torch::kInt64);
+    auto i32 = meta.options();
+    auto lp = torch::empty({R}, i64);
+    auto tp = torch::empty({R}, i64);
+    auto row_of = torch::full({B * T}, -1, i32);
+    auto lens = torch::empty({B}, i32);
+    long total = (long)B * max_learn;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(positions_meta_kernel,
+                       dim3((int)((total + 255) / 256)), dim3(256), 0,
+                       stream.stream(), meta.data_ptr<int>(),
+                       seg.data_ptr<int>(), lp.data_ptr<long>(),
+                       tp.data_ptr<long>(), row_of.data_ptr<int>(),
+                       lens.data_ptr<int>(), B, (int)T, (int)n,
+                       (int)max_learn);
+    return {lp, tp, row_of, lens};
+}
+
+// ---------------------------------------------------------------------------
 // Host wrappers
 // ---------------------------------------------------------------------------
 

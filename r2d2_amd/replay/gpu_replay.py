@@ -247,13 +247,14 @@ class GpuReplayBuffer:
                 meta_h = meta[:3].cpu()
                 seg_h = seg.cpu()
             self._ev_sampled.record(self._sample_stream)
-        return (idx, outs, meta_h, seg_h, self.block_ptr, self.env_steps)
+        return (idx, outs, meta, seg, meta_h, seg_h, self.block_ptr,
+                self.env_steps)
 
     def sample_wait(self, token):
         """Complete a sample_async(): host-syncs the side stream only."""
         from ..worker import TrainingBatch
 
-        idx, outs, meta_h, seg_h, old_ptr, env_steps = token
+        idx, outs, meta_dev, seg_dev, meta_h, seg_h, old_ptr, env_steps = token
         self._ev_sampled.synchronize()
         obs, la, lr, act, nsr, gam, w_rep, hid = outs
         B = idx.shape[0]
@@ -267,6 +268,11 @@ class GpuReplayBuffer:
             forward_steps=meta_h[2].long(),
             idxes=idx, is_weights=w_rep[:R],
             old_ptr=old_ptr, env_steps=env_steps)
+        # device-side layout metadata: the engine computes its gather/
+        # scatter position arrays from these with one kernel instead of a
+        # per-batch host numpy rebuild (every replay batch is ragged)
+        batch.meta_dev = meta_dev
+        batch.seg_dev = seg_dev
         # training-stream consumers must see the gathered tensors
         torch.cuda.current_stream().wait_event(self._ev_sampled)
         return batch
