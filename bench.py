@@ -235,9 +235,15 @@ def main():
     value = c.batch_size * args.steps * n_gpus / elapsed
     baseline = 363.0  # BASELINE.md derived learner seq-samples/sec
 
+    # BASELINE.json's headline metric name belongs to the default preset;
+    # alternate presets get an honest label
+    metric = "learner seq-samples/sec (bs=64, L=80, 84x84x4 frames)"
+    if args.preset != "mspacman":
+        metric = f"learner seq-samples/sec ({args.preset} preset)"
+
     if rank == 0:
         print(json.dumps({
-            "metric": "learner seq-samples/sec (bs=64, L=80, 84x84x4 frames)",
+            "metric": metric,
             "value": round(value, 2),
             "unit": "seq/s",
             "n_gpus": n_gpus,
