@@ -752,6 +752,10 @@ class Learner:
         # priority staleness — far inside the reference's own <=12-batch
         # staleness)
         pending = None
+        # loss accumulates ON-DEVICE; the host reads it once per log
+        # interval instead of syncing the stream every update
+        loss_accum = (torch.zeros((), device=self.device)
+                      if self.device.type == "cuda" else None)
         while self.num_updates < c.training_steps:
             with lock:
                 batch = pending if pending is not None else replay.sample()
@@ -763,7 +767,10 @@ class Learner:
                 replay.update_priorities(batch.idxes, priorities,
                                          batch.old_ptr)
                 pending = replay.sample_wait(tok)
-            stats["sum_loss"] += float(loss)
+            if loss_accum is not None:
+                loss_accum += loss.detach()
+            else:
+                stats["sum_loss"] += float(loss)
             self.env_steps = replay.env_steps
             if self.num_updates % 4 == 0:
                 self.store_weights()
@@ -777,6 +784,9 @@ class Learner:
             if now - last_log > c.log_interval:
                 interval = now - last_log
                 delta = self.num_updates - stats["last_updates"]
+                if loss_accum is not None:
+                    stats["sum_loss"] = float(loss_accum)
+                    loss_accum.zero_()
                 rec = {
                     "wall_s": round(now - start_time, 1),
                     "buffer_size": len(replay),
