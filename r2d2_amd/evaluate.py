@@ -8,9 +8,9 @@ reference test.py:18,32-33 runs 5 episodes in mp.Pool(5)), and emits the
 learning curve — reward vs env frames (= env_steps * frame_skip,
 reference test.py:28,36) and vs wall-clock hours (test.py:29).
 
-This image has no matplotlib, so curves are written as ``{game}_eval.csv``
-and ``{game}_eval.jsonl``; a ``{game}.jpg`` plot is produced only when
-matplotlib is importable (gated, reference test.py:42-58 behavior).
+Curves are written as ``{game}_eval.csv`` and ``{game}_eval.jsonl``, plus
+the reference's ``{game}.jpg`` two-panel plot (reward vs env frames and vs
+wall clock, reference test.py:42-58) when matplotlib is importable.
 """
 
 import csv
