@@ -566,9 +566,17 @@ class Learner:
 
     # -- weight publication -------------------------------------------------
 
-    def store_weights(self):
-        state = {k: v.cpu() for k, v in self.online_net.state_dict().items()}
-        self.shared_model.load_state_dict(state)
+    def store_weights(self, cpu: bool = True):
+        """Publish weights: device bus (VectorActor consumers, cheap D2D)
+        and/or the shared CPU model (reference worker.py:306-307 contract,
+        process-Actor consumers + watchdog restarts).  When the bus serves
+        the actors, the CPU publish is throttled by the caller — the 17 MB
+        GPU->CPU state_dict round trip otherwise blocks the learner thread
+        every 4 updates for consumers that rarely read it."""
+        if cpu:
+            state = {k: v.cpu()
+                     for k, v in self.online_net.state_dict().items()}
+            self.shared_model.load_state_dict(state)
         if self.weight_bus is not None and self.engine is not None:
             self.weight_bus.publish(self.engine.online)
 
@@ -773,7 +781,10 @@ class Learner:
                 stats["sum_loss"] += float(loss)
             self.env_steps = replay.env_steps
             if self.num_updates % 4 == 0:
-                self.store_weights()
+                # bus publish every 4 (reference cadence, cheap D2D); CPU
+                # copy throttled when the bus serves the actors
+                self.store_weights(cpu=(self.weight_bus is None
+                                        or self.num_updates % 100 == 0))
             if self.num_updates % self.target_net_update_interval == 0:
                 self.target_net.load_state_dict(self.online_net.state_dict())
                 if self.engine is not None:
