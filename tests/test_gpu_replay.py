@@ -186,3 +186,37 @@ def test_gpu_replay_train_integration():
     torch.cuda.synchronize()
     assert replay.total_priority != total0
     assert np.isfinite(replay.total_priority)
+
+
+def test_device_positions_match_host_positions():
+    """The on-device positions_meta path (used for every replay-sampled
+    batch) must produce the same loss/priorities/grads as the host numpy
+    positions path on the same batch."""
+    c = cfg.apply("mspacman", buffer_capacity=4000, batch_size=8)
+    from r2d2_amd.replay.gpu_replay import GpuReplayBuffer
+    from r2d2_amd.models.network import Network
+    from r2d2_amd.worker import Learner
+    from bench import build_synthetic_block
+    torch.manual_seed(3)
+    rng = np.random.default_rng(7)
+    replay = GpuReplayBuffer(device="cuda", capacity=4000)
+    for _ in range(replay.num_blocks):
+        replay.ingest(build_synthetic_block(c, rng),
+                      rng.random(replay.spb).astype(np.float32) + 0.5)
+    model = Network(c.action_dim, c.obs_shape, c.hidden_dim, encoder="nature",
+                    forward_steps=c.forward_steps)
+    learner = Learner(None, None, model)
+    learner.enable_hip_engine()
+    batch = replay.sample(8)
+    assert batch.meta_dev is not None
+
+    loss_d, prio_d = learner.engine.train_step(batch)
+    grads_d = learner.engine.flat_grad.clone()
+
+    batch.meta_dev = None   # force the host numpy positions path
+    loss_h, prio_h = learner.engine.train_step(batch)
+    grads_h = learner.engine.flat_grad.clone()
+
+    assert torch.equal(loss_d, loss_h)
+    assert torch.equal(prio_d, prio_h)
+    assert torch.equal(grads_d, grads_h)
