@@ -715,10 +715,13 @@ class HipNetworkEngine:
                 m.conv_dgrad_dense(d_a2f, ON.w2d[(py, px)],
                                    ON.taps2[(py, px)], a1,
                                    M, 9, 9, 64, 20, 20, 32, py, px, 2, d_a1)
-        # conv1 wgrad (no dgrad: input is data; d_a1 pre-masked by a1):
-        # per-image band at 2 images/wg (single dequant per input element)
-        dW1, db1 = m.conv_wgrad_band(d_a1.view(M * 400, 32), a1, obs_hwc,
-                                     1, M)
+        # conv1 wgrad (no dgrad: input is data; d_a1 pre-masked by a1).
+        # Stays on the CHUNKED kernel: measured 0.434 vs 0.474 ms for the
+        # per-image band variant (the 8x8 s4 patch field re-reads only 4x,
+        # and the band's whole-image slab costs more occupancy than the
+        # dequant re-reads save; conv2/conv3 with 9x/4x re-read DO win).
+        dW1, db1 = m.conv_wgrad(d_a1.view(M * 400, 32), a1, obs_hwc, 1,
+                                M, 84, 84, 20, 20, 32, 8 * 8 * self.C)
         self._mark("conv_bwd")
 
         enc = net.encoder

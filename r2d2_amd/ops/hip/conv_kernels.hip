@@ -936,10 +936,7 @@ std::vector<torch::Tensor> conv_wgrad_band(torch::Tensor dY, torch::Tensor act,
                                                          : 3 * 3 * 64;
     auto dWt = torch::zeros({COUT, K}, dY.options().dtype(torch::kFloat32));
     auto db = torch::zeros({COUT}, dY.options().dtype(torch::kFloat32));
-    // conv1 (13 tiles/image, 2 wgs/CU) runs best near 2 images per wg;
-    // the small convs aim for ~1024 workgroups
-    int imgs = (conv_id == 1) ? (int)std::max(1L, (N + 2719) / 2720)
-                              : (int)((N + 1023) / 1024);
+    int imgs = (int)((N + 1023) / 1024);
     int grid = (int)((N + imgs - 1) / imgs);
     auto stream = at::cuda::getCurrentCUDAStream();
     auto* dy = reinterpret_cast<const __hip_bfloat16*>(dY.data_ptr());
