@@ -453,7 +453,7 @@ __global__ __launch_bounds__(256, 1) void lstm_bwd_kernel(
     GridBar* bar, int B, int T, int nblocks, int nhalves) {
     constexpr int WGS = H / UNITS;
     constexpr int KSPLIT = (UNITS == 16 && BROWS == 32) ? 2
-                           : (UNITS == 16 && BROWS == 16) ? 4 : 1;
+                           : (UNITS == 16 && BROWS <= 16) ? 4 : 1;
     constexpr int NCWAVE = UNITS / 16;         // col fragments across waves
     const int half = blockIdx.x / WGS;
     const int wid = blockIdx.x % WGS;
@@ -862,9 +862,11 @@ torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
     }();
     int nhalves = B > 32 ? 2 : 1;
     int units = (nhalves == 2 && bunits_env == 32) ? 32 : 16;
-    int brows = (nhalves == 2 && units == 16 && brows_env == 16 && B > 48)
-                    ? 16 : (nhalves == 2 ? 32 : 64);
+    int brows = (nhalves == 2 && units == 16 && B > 48)
+                    ? brows_env : (nhalves == 2 ? 32 : 64);
+    if (brows != 8 && brows != 16 && brows != 32) brows = 16;
     if (brows == 16) nhalves = 4;
+    else if (brows == 8) nhalves = 8;
     int nblocks = (int)H / units * nhalves;
     auto stream = at::cuda::getCurrentCUDAStream();
     zero_ws(barrier_ws, stream.stream());
@@ -880,7 +882,8 @@ torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
         reinterpret_cast<__hip_bfloat16*>(dgates.data_ptr()),                 \
         reinterpret_cast<GridBar*>(barrier_ws.data_ptr()),                    \
         (int)B, (int)T, nblocks, nhalves)
-    if (brows == 16) LSTMB(16, 16);
+    if (brows == 8) LSTMB(8, 16);
+    else if (brows == 16) LSTMB(16, 16);
     else if (nhalves == 2) {
         if (units == 32) LSTMB(32, 32); else LSTMB(32, 16);
     } else {
