@@ -853,9 +853,11 @@ torch::Tensor lstm_bwd(torch::Tensor stash, torch::Tensor Cout,
         const char* e = getenv("R2D2_LSTM_BWD_UNITS");
         return e ? atoi(e) : 16;
     }();
-    // batch QUARTERED by default (128 wgs x 2 CUs' worth each, KSPLIT=4):
-    // measured 0.71 vs 0.95 ms at BROWS=32 (the backward is CU-coverage
-    // bound; R2D2_LSTM_BWD_BROWS=32 selects the half-split for comparison)
+    // batch QUARTERED by default (128 wgs, KSPLIT=4): measured 0.71 vs
+    // 0.95 ms at BROWS=32 (the backward is CU-coverage bound).  Pushing
+    // further to BROWS=8 (256 wgs) LOSES (12.03k vs 12.43k end-to-end) —
+    // 8-row batches leave the 16-row MFMA A-fragments half empty.
+    // R2D2_LSTM_BWD_BROWS selects 8/16/32 for comparison.
     static const int brows_env = [] {
         const char* e = getenv("R2D2_LSTM_BWD_BROWS");
         return e ? atoi(e) : 16;
