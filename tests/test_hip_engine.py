@@ -242,3 +242,58 @@ def test_engine_checkpoint_resume_roundtrip(tmp_path):
     l1, _ = hip.train_step(batch)
     l2, _ = hip2.train_step(batch)
     assert abs(float(l1) - float(l2)) < 1e-5, (float(l1), float(l2))
+
+
+def test_weight_bus_concurrent_publish_pull_consistency():
+    """Torn-read protection: a puller racing a publisher must always see an
+    internally consistent snapshot (every tensor from the same publish
+    generation — the version word re-check redoes overlapped pulls)."""
+    import threading
+    import types
+    from r2d2_amd.parallel.weight_bus import WeightBus, pack_tensors
+
+    def make_pack(val):
+        return types.SimpleNamespace(
+            a=torch.full((1000,), float(val), device="cuda",
+                         dtype=torch.bfloat16),
+            b=torch.full((64, 64), float(val), device="cuda",
+                         dtype=torch.bfloat16),
+            c=torch.full((333,), float(val), device="cuda",
+                         dtype=torch.float32))
+
+    src = make_pack(0)
+    dst = make_pack(-1)
+    bus = WeightBus(pack_tensors(src), "cuda")
+    stop = threading.Event()
+    errs = []
+
+    def publisher():
+        g = 0
+        while not stop.is_set():
+            g += 1
+            for t in pack_tensors(src).values():
+                t.fill_(float(g % 200))
+            bus.publish(src)
+
+    def puller():
+        ver = 0
+        try:
+            for _ in range(300):
+                ver = bus.pull_into(dst, ver)
+                vals = {float(t.reshape(-1)[0]) for t in
+                        pack_tensors(dst).values()}
+                uniform = all(
+                    bool((t == t.reshape(-1)[0]).all())
+                    for t in pack_tensors(dst).values())
+                if len(vals) > 1 or not uniform:
+                    errs.append(f"torn snapshot: {vals}")
+                    return
+        except Exception as e:  # pragma: no cover
+            errs.append(repr(e))
+
+    pub = threading.Thread(target=publisher, daemon=True)
+    pub.start()
+    puller()
+    stop.set()
+    pub.join(10)
+    assert not errs, errs[:3]
