@@ -87,30 +87,39 @@ class WeightBus:
 
     # -- learner side -------------------------------------------------------
     def publish(self, pack):
-        """D2D-copy the pack's tensors into the bus and bump the version.
-        All copies launch on the current stream; the version bump is ordered
-        after them, so a reader that observes the new version sees the
-        completed copies."""
+        """D2D-copy the pack's tensors into the bus under a SEQLOCK: the
+        version goes ODD before the copies and EVEN after, all on the
+        current stream — a reader that observes an odd version (or a
+        version change across its snapshot) knows a publish overlapped and
+        redoes the pull.  A plain post-bump alone is NOT enough: a reader
+        whose copies interleave with an in-flight publish would see a torn
+        snapshot with an unchanged version (caught by
+        test_weight_bus_concurrent_publish_pull_consistency)."""
         tensors = pack_tensors(pack)
+        self.ver.add_(1)          # odd: publish in progress
         for name, dtype, shape, ofs in self.layout:
             t = tensors[name]
             self.buf[dtype][ofs:ofs + t.numel()].copy_(
                 t.detach().view(-1), non_blocking=True)
-        self.ver.add_(1)
+        self.ver.add_(1)          # even: stable
 
     # -- actor side ---------------------------------------------------------
     def version(self) -> int:
         return int(self.ver.item())
 
     def pull_into(self, pack, last_ver: int) -> int:
-        """If the bus moved past last_ver, copy the slices straight into the
-        pack's tensors (redoing the snapshot if a publish overlapped).
-        Returns the version that was applied (== last_ver if unchanged)."""
+        """If the bus moved past last_ver, copy the slices straight into
+        the pack's tensors (seqlock read: retry while a publish is in
+        flight or overlapped the snapshot).  Returns the (even) version
+        that was applied (== last_ver if unchanged)."""
         v = self.version()
         if v == last_ver:
             return last_ver
         tensors = pack_tensors(pack)
         while True:
+            if v % 2 == 1:        # publish in progress
+                v = self.version()
+                continue
             for name, dtype, shape, ofs in self.layout:
                 t = tensors[name]
                 t.view(-1).copy_(self.buf[dtype][ofs:ofs + t.numel()],
