@@ -118,6 +118,10 @@ def test_gpu_replay_roundtrip():
         replay.ingest(blk, np.ones(replay.spb, dtype=np.float32) * (v + 1))
     torch.cuda.synchronize()
     assert len(replay) == 8000
+    # sum-tree invariant: root == sum of leaves after streamed ingest
+    leaves = replay.tree[replay.leaf_offset:].sum()
+    assert abs(float(replay.tree[0]) - float(leaves)) < 1e-6 * max(
+        1.0, float(leaves))
 
     batch = replay.sample(16)
     torch.cuda.synchronize()
