@@ -414,7 +414,11 @@ class ReplayBuffer:
                 burn[i] = blk.burn_in_steps[si]
                 learn[i] = blk.learning_steps[si]
                 fwd[i] = blk.forward_steps[si]
-                starts[i] = blk.burn_in_steps[0] + int(np.sum(blk.learning_steps[:si]))
+                # int() both operands: a uint8 SCALAR + a >255 python int
+                # raises OverflowError under NumPy 2 promotion (NEP 50) —
+                # reachable only at full block scale (learn sums > 255)
+                starts[i] = (int(blk.burn_in_steps[0])
+                             + int(np.sum(blk.learning_steps[:si])))
 
             T = int((burn + learn + fwd).max())
             obs_shape = blocks[0].obs.shape[1:]

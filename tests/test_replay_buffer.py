@@ -102,3 +102,28 @@ def test_priority_zero_blocks_never_sampled():
     for _ in range(20):
         idx, _ = rb.priority_tree.sample(8)
         assert (idx < 5).all()
+
+
+def test_sample_batch_full_scale_blocks():
+    """FULL mspacman geometry (block 400, learning 40, seq_per_block 10):
+    per-sequence learn-offsets exceed 255, which tripped NumPy-2 uint8
+    scalar promotion (OverflowError) in the assembler — the reference
+    config must assemble cleanly."""
+    c = cfg.apply("cartpole", buffer_capacity=8000, block_length=400,
+                  burn_in_steps=40, learning_steps=40, forward_steps=5,
+                  batch_size=16, learning_starts=400, hidden_dim=16)
+    rb = make_rb(seed=3)
+    for v in range(1, 8):
+        blk, prios = make_block(float(v), steps=400, burn=40, learn=40, n=5)
+        rb.add(blk, prios + v, None)
+    for _ in range(5):
+        batch = rb.sample_batch()
+        assert batch.obs.shape[0] == 16
+        # sequences deep into a block (learn offset > 255) must slice the
+        # right subsequence: obs value == block value everywhere
+        for i in range(16):
+            bi = batch.idxes[i] // rb.seq_per_block
+            v = float(bi + 1)
+            L = int(batch.burn_in_steps[i] + batch.learning_steps[i]
+                    + batch.forward_steps[i])
+            assert np.allclose(batch.obs[i, :L].numpy(), v), (i, v)
