@@ -62,8 +62,20 @@ class PriorityTree:
             u = np.where(go_right, u - left, u)
             idxes = 2 * idxes + go_right
         priorities = self.levels[-1][idxes]
-        # guard: jitter can land on a zero-priority leaf only via fp edge cases
-        min_p = priorities[priorities > 0].min() if (priorities > 0).any() else 1.0
+        # fp-edge repair: accumulated rounding in the parent sums can steer
+        # a descent into a ZERO-priority leaf (zero-padded capacity tail,
+        # or the unused slots of a partial block).  Snap such samples to
+        # the nearest preceding nonzero leaf — the stratum's true mass
+        # lives at/left of the overshoot.  (The consumer indexes blocks
+        # with these, so returning a dead slot is a crash, not just a
+        # skewed IS weight.)
+        bad = priorities <= 0.0
+        if bad.any():
+            valid = np.flatnonzero(self.levels[-1])
+            pos = np.searchsorted(valid, idxes[bad], side="right") - 1
+            idxes[bad] = valid[np.clip(pos, 0, len(valid) - 1)]
+            priorities = self.levels[-1][idxes]
+        min_p = priorities.min() if len(priorities) else 1.0
         is_weights = np.power(np.maximum(priorities, min_p * 1e-12) / min_p,
                               -self.is_exponent)
         return idxes, is_weights

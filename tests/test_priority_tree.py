@@ -100,3 +100,26 @@ def test_property_random_ops_keep_tree_consistent():
                                        rtol=1e-9)
 
     run()
+
+
+def test_sample_never_returns_dead_slots():
+    """Zero-priority leaves (partial-block slots, zero-padded capacity
+    tail) must never be returned even under fp-edge descents — the
+    consumer indexes the block ring with these (a dead slot crashed the
+    full-scale host assembler before the snap-repair)."""
+    rng = np.random.default_rng(0)
+    tree = PriorityTree(50_000, 0.9, 0.6, rng=np.random.default_rng(1))
+    # gappy occupancy like a ring of partial blocks: 10-slot groups with
+    # random tails unwritten
+    idx = []
+    for g in range(0, 49_990, 10):
+        n = int(rng.integers(1, 11))
+        idx.extend(range(g, g + n))
+    idx = np.array(idx, dtype=np.int64)
+    tree.update(idx, rng.random(len(idx)).astype(np.float64) + 1e-3)
+    occupied = np.zeros(tree.num_leaves, dtype=bool)
+    occupied[idx] = True
+    for _ in range(200):
+        sampled, w = tree.sample(64)
+        assert occupied[sampled].all()
+        assert np.isfinite(w).all() and (w > 0).all()
