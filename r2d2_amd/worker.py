@@ -79,6 +79,7 @@ class TrainingBatch:
     is_weights: torch.Tensor     # (sum_learn,)
     old_ptr: int
     env_steps: int
+    old_count: int = 0           # monotone blocks-added at sample time
 
     def to(self, device, non_blocking=True):
         self.obs = self.obs.to(device, non_blocking=non_blocking)
@@ -272,6 +273,7 @@ class ReplayBuffer:
             rng=np.random.default_rng(seed))
 
         self.block_ptr = 0
+        self.blocks_added = 0    # monotone (never wraps)
         self.size = 0
         self.env_steps = 0
         self.num_episodes = 0
@@ -389,6 +391,7 @@ class ReplayBuffer:
             self.env_steps += int(np.sum(block.learning_steps))
             self.buffer[self.block_ptr] = block
             self.block_ptr = (self.block_ptr + 1) % self.num_blocks
+            self.blocks_added += 1
             if episode_reward is not None:
                 self.episode_reward += episode_reward
                 self.num_episodes += 1
@@ -457,15 +460,29 @@ class ReplayBuffer:
                 is_weights=torch.from_numpy(is_rep),
                 old_ptr=self.block_ptr,
                 env_steps=self.env_steps,
+                old_count=self.blocks_added,
             )
         return batch
 
     def update_priorities(self, idxes: np.ndarray, td_errors: np.ndarray,
-                          old_ptr: int, loss: float):
+                          old_ptr: int, loss: float,
+                          old_count: Optional[int] = None):
         """Masks out indexes overwritten by the ring pointer since sampling
-        (reference worker.py:242-261 wraparound semantics), then tree update."""
+        (reference worker.py:242-261 wraparound semantics), then tree update.
+
+        The pointer comparison alone CANNOT see a full ring lap: if the
+        ring advanced >= num_blocks blocks between sampling and this
+        (delayed) update, cur == old looks like "nothing overwritten" and
+        the stale priorities resurrect slots of newer, possibly SHORTER
+        blocks — dead sequence slots with nonzero priority then crash the
+        assembler.  ``old_count`` (monotone blocks-added at sample time)
+        closes the hole; the reference shares this flaw."""
         with self.lock:
-            if self.block_ptr > old_ptr:
+            if (old_count is not None
+                    and self.blocks_added - old_count >= self.num_blocks):
+                idxes = idxes[:0]
+                td_errors = td_errors[:0]
+            elif self.block_ptr > old_ptr:
                 mask = ((idxes < old_ptr * self.seq_per_block)
                         | (idxes >= self.block_ptr * self.seq_per_block))
                 idxes, td_errors = idxes[mask], td_errors[mask]
@@ -700,7 +717,8 @@ class Learner:
             loss = float(loss)
             if torch.is_tensor(priorities):
                 priorities = priorities.cpu().numpy()
-            self.priority_queue.put((batch.idxes, priorities, batch.old_ptr, loss))
+            self.priority_queue.put((batch.idxes, priorities, batch.old_ptr, loss,
+                                     batch.old_count))
             self.env_steps = batch.env_steps
             if self.num_updates % 4 == 0:
                 self.store_weights()

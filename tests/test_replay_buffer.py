@@ -127,3 +127,45 @@ def test_sample_batch_full_scale_blocks():
             L = int(batch.burn_in_steps[i] + batch.learning_steps[i]
                     + batch.forward_steps[i])
             assert np.allclose(batch.obs[i, :L].numpy(), v), (i, v)
+
+
+def test_delayed_priority_update_cannot_resurrect_dead_slots():
+    """A priority update delayed by MORE than a full ring lap must be
+    dropped entirely: the pointer-interval mask alone sees cur == old
+    after exactly num_blocks additions and would write stale priorities
+    onto slots now holding shorter blocks — nonzero priority on a dead
+    sequence slot crashes the assembler (reference worker.py:247-256
+    shares this flaw; the monotone blocks-added counter closes it)."""
+    c = cfg.apply("cartpole", buffer_capacity=320, block_length=40,
+                  burn_in_steps=8, learning_steps=8, forward_steps=3,
+                  batch_size=4, learning_starts=40, hidden_dim=16)
+    rb = make_rb(seed=5)
+    # fill all 8 slots with FULL blocks (5 sequences each)
+    for v in range(8):
+        blk, prios = make_block(float(v + 1))
+        rb.add(blk, prios + 1.0, None)
+    batch = rb.sample_batch()
+
+    # ring laps EXACTLY once, refilled with PARTIAL blocks (1 sequence ->
+    # slots 1..4 of every block are dead with zero priority)
+    for v in range(8):
+        blk, prios = make_block(float(v + 100), steps=8)
+        assert blk.num_sequences == 1
+        padded = np.zeros(rb.seq_per_block, dtype=np.float32)
+        padded[:1] = 1.0
+        rb.add(blk, padded, None)
+    assert rb.block_ptr == batch.old_ptr    # the lap the ptr mask can't see
+
+    rb.update_priorities(batch.idxes,
+                         np.ones(len(batch.idxes), dtype=np.float32) * 5.0,
+                         batch.old_ptr, 0.1, batch.old_count)
+    # every dead slot must still have zero priority
+    leaves = rb.priority_tree.levels[-1]
+    for b in range(8):
+        for s in range(1, rb.seq_per_block):
+            assert leaves[b * rb.seq_per_block + s] == 0.0, (b, s)
+    # and sampling stays within live slots
+    for _ in range(50):
+        bt = rb.sample_batch()
+        for idx in bt.idxes:
+            assert idx % rb.seq_per_block == 0
