@@ -301,7 +301,7 @@ class ReplayBuffer:
     def run(self):
         threads = [threading.Thread(target=f, daemon=True)
                    for f in (self._ingest_loop, self._assemble_loop,
-                             self._priority_loop)]
+                             self._assemble_loop, self._priority_loop)]
         for t in threads:
             t.start()
         log_interval = self.cfg.log_interval
@@ -398,31 +398,44 @@ class ReplayBuffer:
 
     def sample_batch(self) -> TrainingBatch:
         """Vectorized batch assembly (replaces the reference's per-sample
-        Python slice loop, worker.py:176-210)."""
+        Python slice loop, worker.py:176-210).
+
+        The LOCK covers only the tree sample and capturing Block
+        REFERENCES (a Block is immutable once stored; a ring overwrite
+        replaces the list slot but our reference keeps the sampled data
+        alive and coherent).  The heavy slice copies run unlocked, so
+        several assemble threads overlap (the copies are memcpys that
+        release the GIL) and ingest/priority threads never stall behind
+        an assembly."""
         with self.lock:
             idxes, is_weights = self.priority_tree.sample(self.batch_size)
             block_idxes = idxes // self.seq_per_block
             seq_idxes = idxes % self.seq_per_block
-
-            B = self.batch_size
-            burn = np.empty(B, dtype=np.int64)
-            learn = np.empty(B, dtype=np.int64)
-            fwd = np.empty(B, dtype=np.int64)
-            starts = np.empty(B, dtype=np.int64)
             blocks = []
-            for i, (bi, si) in enumerate(zip(block_idxes, seq_idxes)):
+            for bi, si in zip(block_idxes, seq_idxes):
                 blk = self.buffer[bi]
                 assert blk is not None and si < blk.num_sequences
                 blocks.append(blk)
-                burn[i] = blk.burn_in_steps[si]
-                learn[i] = blk.learning_steps[si]
-                fwd[i] = blk.forward_steps[si]
-                # int() both operands: a uint8 SCALAR + a >255 python int
-                # raises OverflowError under NumPy 2 promotion (NEP 50) —
-                # reachable only at full block scale (learn sums > 255)
-                starts[i] = (int(blk.burn_in_steps[0])
-                             + int(np.sum(blk.learning_steps[:si])))
+            old_ptr = self.block_ptr
+            old_count = self.blocks_added
+            env_steps = self.env_steps
 
+        B = self.batch_size
+        burn = np.empty(B, dtype=np.int64)
+        learn = np.empty(B, dtype=np.int64)
+        fwd = np.empty(B, dtype=np.int64)
+        starts = np.empty(B, dtype=np.int64)
+        for i, (blk, si) in enumerate(zip(blocks, seq_idxes)):
+            burn[i] = blk.burn_in_steps[si]
+            learn[i] = blk.learning_steps[si]
+            fwd[i] = blk.forward_steps[si]
+            # int() both operands: a uint8 SCALAR + a >255 python int
+            # raises OverflowError under NumPy 2 promotion (NEP 50) —
+            # reachable only at full block scale (learn sums > 255)
+            starts[i] = (int(blk.burn_in_steps[0])
+                         + int(np.sum(blk.learning_steps[:si])))
+
+        if True:
             T = int((burn + learn + fwd).max())
             obs_shape = blocks[0].obs.shape[1:]
             A = blocks[0].last_action.shape[1]
@@ -458,9 +471,9 @@ class ReplayBuffer:
                 forward_steps=torch.from_numpy(fwd),
                 idxes=idxes,
                 is_weights=torch.from_numpy(is_rep),
-                old_ptr=self.block_ptr,
-                env_steps=self.env_steps,
-                old_count=self.blocks_added,
+                old_ptr=old_ptr,
+                env_steps=env_steps,
+                old_count=old_count,
             )
         return batch
 
