@@ -169,3 +169,52 @@ def test_delayed_priority_update_cannot_resurrect_dead_slots():
         bt = rb.sample_batch()
         for idx in bt.idxes:
             assert idx % rb.seq_per_block == 0
+
+
+def test_sample_batch_races_ring_overwrite_without_tearing():
+    """The assembler's heavy copies now run OUTSIDE the lock, reading from
+    Block references captured under it.  A concurrent ring overwrite
+    replaces the buffer slot but must not affect an in-flight assembly:
+    Blocks are immutable, so every sampled row must be internally
+    consistent (all obs values in the row equal) even while the writer
+    laps the ring continuously."""
+    import threading
+
+    small_cfg()
+    rb = make_rb(seed=7)
+    for v in range(1, 9):
+        blk, prios = make_block(float(v))
+        rb.add(blk, prios + 1.0, None)
+
+    stop = threading.Event()
+    err = []
+
+    def writer():
+        v = 9
+        while not stop.is_set():
+            blk, prios = make_block(float(v % 100 + 1))
+            rb.add(blk, prios + 1.0, None)
+            v += 1
+
+    def sampler():
+        try:
+            for _ in range(60):
+                batch = rb.sample_batch()
+                for i in range(batch.obs.shape[0]):
+                    L = int(batch.burn_in_steps[i] + batch.learning_steps[i]
+                            + batch.forward_steps[i])
+                    row = batch.obs[i, :L].numpy()
+                    assert (row == row.flat[0]).all(), "torn row"
+        except Exception as e:  # surface into the main thread
+            err.append(e)
+
+    w = threading.Thread(target=writer, daemon=True)
+    samplers = [threading.Thread(target=sampler) for _ in range(2)]
+    w.start()
+    for s in samplers:
+        s.start()
+    for s in samplers:
+        s.join()
+    stop.set()
+    w.join(timeout=5)
+    assert not err, err
