@@ -58,6 +58,13 @@ def _worker(rank, world_size, port, fail_q):
         raise
 
 
+def _free_port() -> int:
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 @pytest.mark.timeout(300)
 def test_torchrun_multirank_bench_path():
     """The EXACT multi-rank entry the driver uses — torch.distributed.run
@@ -73,7 +80,7 @@ def test_torchrun_multirank_bench_path():
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29517", "bench.py", "--gpus", "2",
+         "--master-port", str(_free_port()), "bench.py", "--gpus", "2",
          "--steps", "2", "--warmup", "1", "--batches", "1",
          "--preset", "cartpole"],
         cwd=repo, capture_output=True, text=True, timeout=240)
@@ -89,7 +96,7 @@ def test_torchrun_multirank_bench_path():
 def test_bucketed_allreduce_matches_mean_grad():
     ctx = mp.get_context("spawn")
     fail_q = ctx.Queue()
-    port = 29511
+    port = _free_port()
     procs = [ctx.Process(target=_worker, args=(r, 2, port, fail_q))
              for r in range(2)]
     for p in procs:
