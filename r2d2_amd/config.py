@@ -159,6 +159,12 @@ def apply(preset: Optional[str] = None, **overrides) -> Config:
     """Apply a named preset and/or field overrides, updating both the Config
     object and this module's flat attribute namespace."""
     global _current
+    unknown = [k for k in overrides
+               if k not in Config.__dataclass_fields__
+               and k not in ("seq_len", "seq_per_block", "num_blocks")]
+    if unknown:
+        raise TypeError(f"unknown config field(s): {unknown} — a silently "
+                        f"dropped override is a misconfigured run")
     base = dict(PRESETS[preset]) if preset else asdict(_current)
     base.update(overrides)
     # drop derived keys if present

@@ -43,3 +43,17 @@ def test_override_without_preset_keeps_state():
     assert c.game_name == "CartPole"       # preset retained
     assert c.num_blocks == 1280 // 16
     cfg.apply("mspacman")
+
+
+def test_apply_rejects_unknown_override():
+    """A typo'd override must raise, not silently configure a different
+    run (apply() used to drop unknown keys on the floor)."""
+    import pytest
+
+    from r2d2_amd import config as cfg
+
+    with pytest.raises(TypeError, match="unknown config field"):
+        cfg.apply("cartpole", num_envz=32)
+    # derived keys remain accepted (and recomputed)
+    c = cfg.apply("cartpole", seq_len=999)
+    assert c.seq_len == c.burn_in_steps + c.learning_steps
