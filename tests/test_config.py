@@ -11,7 +11,8 @@ def test_presets_cover_baseline_configs():
     for name in ("cartpole", "mspacman", "mspacman_gpu_replay",
                  "mspacman_dp", "seaquest_impala"):
         c = cfg.apply(name)
-        assert c.seq_len == c.burn_in_steps + c.learning_steps + c.forward_steps
+        assert c.seq_len == (c.burn_in_steps + c.learning_steps
+                         + c.forward_steps) + c.forward_steps
         assert c.block_length % c.learning_steps == 0
     cfg.apply("mspacman")
 
@@ -56,4 +57,5 @@ def test_apply_rejects_unknown_override():
         cfg.apply("cartpole", num_envz=32)
     # derived keys remain accepted (and recomputed)
     c = cfg.apply("cartpole", seq_len=999)
-    assert c.seq_len == c.burn_in_steps + c.learning_steps
+    assert c.seq_len == (c.burn_in_steps + c.learning_steps
+                         + c.forward_steps)
