@@ -12,7 +12,7 @@ def test_presets_cover_baseline_configs():
                  "mspacman_dp", "seaquest_impala"):
         c = cfg.apply(name)
         assert c.seq_len == (c.burn_in_steps + c.learning_steps
-                         + c.forward_steps) + c.forward_steps
+                         + c.forward_steps)
         assert c.block_length % c.learning_steps == 0
     cfg.apply("mspacman")
 
