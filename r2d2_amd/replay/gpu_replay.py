@@ -285,7 +285,15 @@ class GpuReplayBuffer:
     def update_priorities(self, idxes: torch.Tensor, priorities: torch.Tensor,
                           old_ptr: int):
         """idxes/priorities stay on device; ring-stale mask applied in-kernel.
-        Waits on any in-flight sample's tree reads (side stream)."""
+        Waits on any in-flight sample's tree reads (side stream).
+
+        The host ReplayBuffer's full-ring-lap hole (see worker.py
+        update_priorities: a pointer-only mask cannot distinguish "nothing
+        overwritten" from "everything overwritten") does not arise here:
+        priority staleness is bounded at ONE batch by construction — the
+        learner updates synchronously after each step and prefetch depth
+        is 1 — so the ring can never advance num_blocks slots between a
+        sample and its update."""
         torch.cuda.current_stream().wait_event(self._ev_sampled)
         self._ext.sumtree_update(self.tree, self.leaf_offset,
                                  idxes.to(torch.int64),
