@@ -66,14 +66,25 @@ def test_one_case(args):
 
 
 def _checkpoint_paths(model_dir: str, game_name: str, save_interval: int):
-    """Yield (num_updates, path) for consecutive checkpoints until a gap."""
-    k = save_interval
-    while True:
-        path = os.path.join(model_dir, f"{game_name}{k}.pth")
-        if not os.path.exists(path):
-            return
-        yield k, path
-        k += save_interval
+    """Yield (num_updates, path) in update order.
+
+    The reference (test.py:18-20) steps k=500,1000,... and stops at the
+    first missing file — which silently skips EVERYTHING when the run
+    used a different save_interval, or stops early at any gap.  Glob the
+    actual ``{game}{N}.pth`` files and sort numerically instead (the
+    ``.train.pth`` optimizer sidecars are not checkpoints)."""
+    import glob
+    import re
+
+    found = []
+    for path in glob.glob(os.path.join(model_dir, f"{game_name}*.pth")):
+        if path.endswith(".train.pth"):
+            continue
+        m = re.fullmatch(re.escape(game_name) + r"(\d+)\.pth",
+                         os.path.basename(path))
+        if m:
+            found.append((int(m.group(1)), path))
+    yield from sorted(found)
 
 
 def test(game_name: Optional[str] = None, model_dir: str = "models",

@@ -100,3 +100,19 @@ def test_eval_harness(tmp_path):
         assert np.isfinite(r["mean_reward"])
     assert (tmp_path / f"{c.game_name}_eval.csv").exists()
     assert (tmp_path / f"{c.game_name}_eval.jsonl").exists()
+
+
+def test_checkpoint_discovery_mixed_intervals(tmp_path):
+    """Discovery must find every {game}{N}.pth regardless of the configured
+    save_interval and survive gaps (the reference's k*interval walk stops
+    at the first hole and finds NOTHING when the run used a different
+    interval); .train.pth optimizer sidecars are not checkpoints."""
+    from r2d2_amd.evaluate import _checkpoint_paths
+
+    for n in (500, 2000, 10000):        # mixed intervals + a gap
+        (tmp_path / f"Pong{n}.pth").touch()
+        (tmp_path / f"Pong{n}.train.pth").touch()
+    (tmp_path / "PongBest.pth").touch()  # non-numeric: ignored
+    got = list(_checkpoint_paths(str(tmp_path), "Pong", save_interval=777))
+    assert [n for n, _ in got] == [500, 2000, 10000]
+    assert all(p.endswith(f"Pong{n}.pth") for n, p in got)
