@@ -280,3 +280,50 @@ def test_assembler_rows_match_origin_blocks(block_steps, seed):
         # IS weights repeated once per learning step
         assert batch.is_weights.shape[0] == int(batch.learning_steps.sum())
         assert batch.action.shape[0] == int(batch.learning_steps.sum())
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.integers(min_value=0, max_value=30),   # adds before sampling
+       st.integers(min_value=0, max_value=20),   # adds between sample/update
+       st.integers(min_value=0, max_value=2 ** 31 - 1))
+def test_stale_mask_matches_generation_model(pre_adds, mid_adds, seed):
+    """update_priorities' pointer-interval + lap-counter mask must agree
+    with the ground truth: a sampled index's update lands iff its block
+    SLOT was not overwritten between sample and update (tracked here by
+    per-slot generation numbers — the model the mask approximates)."""
+    small_cfg()
+    rb = make_rb(seed=seed % 10_000)
+    gen = {}
+
+    def add_one(g):
+        blk, prios = make_block(float(g % 7 + 1))
+        rb.add(blk, prios + 1.0, None)
+        gen[(rb.block_ptr - 1) % rb.num_blocks] = g
+
+    g = 0
+    for _ in range(max(pre_adds, 2)):
+        add_one(g); g += 1
+
+    batch = rb.sample_batch()
+    gen_at_sample = dict(gen)
+    for _ in range(mid_adds):
+        add_one(g); g += 1
+
+    new_td = np.full(len(batch.idxes), 3.0, dtype=np.float32)
+    before = rb.priority_tree.levels[-1][batch.idxes].copy()
+    rb.update_priorities(batch.idxes, new_td, batch.old_ptr, 0.0,
+                         batch.old_count)
+    after = rb.priority_tree.levels[-1][batch.idxes]
+
+    alpha = rb.priority_tree.prio_exponent
+    for j, idx in enumerate(batch.idxes):
+        slot = idx // rb.seq_per_block
+        fresh = gen.get(slot) == gen_at_sample.get(slot)
+        if fresh:
+            assert after[j] == pytest.approx(3.0 ** alpha), (j, idx)
+        else:
+            # overwritten since sampling: priority must be whatever the
+            # overwrite wrote, NOT the stale update
+            assert after[j] == before[j], (j, idx)
+            assert after[j] != pytest.approx(3.0 ** alpha) or \
+                before[j] == pytest.approx(3.0 ** alpha)
