@@ -11,6 +11,25 @@ from functools import lru_cache
 
 import numpy as np
 
+_blas_limiter = None
+
+
+def limit_blas_threads():
+    """Pin BLAS pools to ONE thread in this process (idempotent, kept for
+    the process lifetime).  Actor processes are single-threaded by design
+    (the reference sets torch.set_num_threads(1), train.py:13) but numpy's
+    OpenBLAS pool is separate: on a saturated machine the multithreaded
+    pool collapses on the tiny per-frame resize matmuls (measured 28
+    frames/s loaded vs 6,078 single-threaded on 8 contended cores —
+    thread-pool handoff costs 200x the 3 MFLOP of work)."""
+    global _blas_limiter
+    if _blas_limiter is None:
+        try:
+            from threadpoolctl import threadpool_limits
+            _blas_limiter = threadpool_limits(limits=1)
+        except Exception:      # threadpoolctl absent: leave pools alone
+            _blas_limiter = False
+
 
 @lru_cache(maxsize=8)
 def _area_weights(n_in: int, n_out: int) -> np.ndarray:
@@ -42,6 +61,7 @@ class AtariEnv:
     def __init__(self, game_name: str, obs_shape=(1, 84, 84), noop_max: int = 30,
                  seed=None):
         import gymnasium as gym
+        limit_blas_threads()   # this process does per-frame resize matmuls
         self.env = gym.make(f"ALE/{game_name}-v5", obs_type="grayscale",
                             frameskip=4, repeat_action_probability=0)
         self.action_dim = self.env.action_space.n

@@ -75,6 +75,9 @@ def test_resize_throughput():
     regression this guards (the original per-pixel Python loop) was 60x
     slower, far outside any contention noise."""
     import time
+
+    from r2d2_amd.envs.atari import limit_blas_threads
+    limit_blas_threads()   # the production (actor-process) configuration
     rng = np.random.default_rng(3)
     img = rng.integers(0, 256, size=(210, 160)).astype(np.uint8)
     _area_resize_84(img)  # warm the weight cache
