@@ -299,6 +299,14 @@ class ReplayBuffer:
     # -- threads -----------------------------------------------------------
 
     def run(self):
+        # Complete torch's lazy parallel-backend init on THIS thread before
+        # the assemble threads start: in a forked child (this process) two
+        # threads racing the first parallel-sized torch op can hit a dead
+        # inherited OpenMP pool ("Invalid thread pool!", ParallelOpenMP.cpp
+        # lazy-init race) and kill an assembler at startup.  Only reachable
+        # at full shapes — the op must cross the parallel grain size.
+        torch.set_num_threads(1)
+        torch.zeros((64, 2, 512)).transpose(0, 1).contiguous()
         threads = [threading.Thread(target=f, daemon=True)
                    for f in (self._ingest_loop, self._assemble_loop,
                              self._assemble_loop, self._priority_loop)]
