@@ -150,7 +150,7 @@ def test(game_name: Optional[str] = None, model_dir: str = "models",
 
 def _maybe_plot(results, game_name, out_dir):
     """Reward vs env-frames and vs wall-clock plot (reference test.py:42-58);
-    silently skipped when matplotlib is absent (it is not in this image)."""
+    silently skipped when matplotlib is absent."""
     try:
         import matplotlib
         matplotlib.use("Agg")
