@@ -164,8 +164,14 @@ def main():
         replay = GpuReplayBuffer(device=device,
                                  capacity=args.replay_transitions)
         rng = np.random.default_rng(42 + rank)
-        for _ in range(replay.num_blocks):
-            replay.ingest(build_synthetic_block(c, rng),
+        # prefill (OUTSIDE the timed region): rotate a small pool of
+        # prebuilt random blocks with fresh per-ingest priorities — frame
+        # CONTENT doesn't affect learner throughput, and building 400
+        # distinct 12 MB random blocks costs ~10 s of host RNG per rank
+        # before the clock starts (noticeable at N=8)
+        pool = [build_synthetic_block(c, rng) for _ in range(8)]
+        for i in range(replay.num_blocks):
+            replay.ingest(pool[i % len(pool)],
                           rng.random(replay.spb).astype(np.float32) + 0.1)
         torch.cuda.synchronize()
         batches = None
