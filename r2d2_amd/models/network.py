@@ -166,3 +166,45 @@ class Network(nn.Module):
         h_learn = self._gather_positions(out, pos_learn)
         h_tgt = self._gather_positions(out, pos_tgt)
         return self._dueling_q(h_learn), self._dueling_q(h_tgt)
+
+
+# ---------------------------------------------------------------------------
+# Reference-checkpoint interop
+# ---------------------------------------------------------------------------
+
+_REFERENCE_ENCODER_KEYS = {
+    # reference model.py:39-49 nn.Sequential indices -> our named modules
+    "feature.0": "encoder.conv1",   # Conv2d(1, 32, 8, 4)
+    "feature.2": "encoder.conv2",   # Conv2d(32, 64, 4, 2)
+    "feature.4": "encoder.conv3",   # Conv2d(64, 64, 3, 1)
+    "feature.7": "encoder.fc",      # Linear(3136, 512)
+}
+
+
+def reference_state_dict_to_native(state_dict):
+    """Rename a ZiyuanMa/R2D2 reference ``Network`` state_dict to this
+    framework's module names.  Only the encoder differs (the reference's
+    anonymous ``feature`` Sequential, model.py:39-49); ``recurrent``,
+    ``advantage`` and ``value`` already share names and structure.  The
+    result loads into ``Network(action_dim, obs_shape=(1, 84, 84))`` —
+    the reference's single-grayscale-frame architecture."""
+    out = {}
+    for k, v in state_dict.items():
+        head, _, tail = k.rpartition(".")   # e.g. "feature.0", "weight"
+        out[f"{_REFERENCE_ENCODER_KEYS.get(head, head)}.{tail}"
+            if head else k] = v
+    return out
+
+
+def load_reference_checkpoint(path, map_location="cpu"):
+    """Load a reference 4-tuple checkpoint (test.py:27 contract) into a
+    native ``Network``.  Returns (network, num_updates, env_steps, minutes)."""
+    state, num_updates, env_steps, minutes = torch.load(
+        path, map_location=map_location, weights_only=False)
+    state = reference_state_dict_to_native(state)
+    action_dim = state["advantage.2.weight"].shape[0]
+    hidden_dim = state["value.0.weight"].shape[0]
+    net = Network(action_dim, obs_shape=(1, 84, 84), hidden_dim=hidden_dim,
+                  encoder="nature")
+    net.load_state_dict(state)
+    return net, int(num_updates), int(env_steps), float(minutes)

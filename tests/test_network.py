@@ -118,3 +118,37 @@ def test_agent_state_no_shared_default():
     s2 = AgentState(torch.zeros(1, 4), 2)
     s1.last_reward[0, 0] = 99.0
     assert s2.last_reward[0, 0] == 0.0
+
+
+def test_reference_checkpoint_key_mapping_roundtrip(tmp_path):
+    """A reference-named state_dict (anonymous `feature` Sequential,
+    reference model.py:39-49) must load into our Network bit-exactly via
+    reference_state_dict_to_native / load_reference_checkpoint."""
+    import torch
+
+    from r2d2_amd.models.network import (Network, load_reference_checkpoint,
+                                         reference_state_dict_to_native)
+
+    torch.manual_seed(11)
+    src = Network(9, obs_shape=(1, 84, 84), hidden_dim=512, encoder="nature")
+    inv = {"encoder.conv1": "feature.0", "encoder.conv2": "feature.2",
+           "encoder.conv3": "feature.4", "encoder.fc": "feature.7"}
+    ref_sd = {}
+    for k, v in src.state_dict().items():
+        head, _, tail = k.rpartition(".")
+        ref_sd[f"{inv.get(head, head)}.{tail}" if head else k] = v
+    assert any(k.startswith("feature.") for k in ref_sd)
+
+    native = reference_state_dict_to_native(ref_sd)
+    dst = Network(9, obs_shape=(1, 84, 84), hidden_dim=512, encoder="nature")
+    dst.load_state_dict(native)
+    for k, v in src.state_dict().items():
+        assert torch.equal(v, dst.state_dict()[k]), k
+
+    # full 4-tuple loader (reference test.py:27 contract)
+    p = tmp_path / "MsPacman500.pth"
+    torch.save((ref_sd, 500, 123456, 7.5), p)
+    net, nu, es, mins = load_reference_checkpoint(str(p))
+    assert (nu, es, mins) == (500, 123456, 7.5)
+    assert torch.equal(net.state_dict()["encoder.conv1.weight"],
+                       src.state_dict()["encoder.conv1.weight"])
