@@ -76,6 +76,11 @@ class Config:
     gpu_replay: bool = True                      # GPU-resident block store + sum-tree
     log_interval: int = 10                       # seconds; reference: config.py (log_interval)
     metrics_path: Optional[str] = None           # JSONL metrics emit (None = console only)
+    # host-replay persistence (elastic resume; the reference never saves
+    # replay contents — SURVEY §5).  Set a path to snapshot the ReplayBuffer
+    # every replay_snapshot_interval seconds; train(resume=...) restores it.
+    replay_snapshot_path: Optional[str] = None
+    replay_snapshot_interval: float = 300.0      # seconds between snapshots
     batch_queue_size: int = 8
     amp: bool = True
 

@@ -49,6 +49,23 @@ class PriorityTree:
             self.levels[lvl][nodes] = child[2 * nodes] + child[2 * nodes + 1]
             child = self.levels[lvl]
 
+    def leaf_values(self) -> np.ndarray:
+        """Copy of the raw stored leaf priorities (already ^alpha) — the
+        snapshot format for replay persistence."""
+        return self.levels[-1].copy()
+
+    def set_leaf_values(self, leaves: np.ndarray) -> None:
+        """Restore raw leaf priorities (as returned by leaf_values — NOT
+        td errors; no ^alpha is applied) and rebuild every parent level."""
+        if leaves.shape != self.levels[-1].shape:
+            raise ValueError(f"leaf snapshot shape {leaves.shape} != tree "
+                             f"leaves {self.levels[-1].shape}")
+        self.levels[-1][:] = leaves
+        child = self.levels[-1]
+        for lvl in range(len(self.levels) - 2, -1, -1):
+            self.levels[lvl][:] = child[0::2] + child[1::2]
+            child = self.levels[lvl]
+
     def sample(self, num_samples: int) -> Tuple[np.ndarray, np.ndarray]:
         total = self.levels[0][0]
         assert total > 0, "sampling from an empty tree"

@@ -44,18 +44,37 @@ def _run_vector_actor(epsilons, model, sample_queues, device, seed,
     va.run()
 
 
-def _run_buffer(buffer: ReplayBuffer = None, queues=None, config_dict=None):
+def _run_buffer(buffer: ReplayBuffer = None, queues=None, config_dict=None,
+                restore_path=None, initial_training_steps=0):
     if buffer is None:
         # spawn context: a ReplayBuffer (threading.Lock inside) cannot be
         # pickled — build it in the child from the handed-over config
         cfg.apply(**config_dict)
         sq, bq, pq = queues
-        buffer = ReplayBuffer(sq, bq, pq)
+        buffer = ReplayBuffer(sq, bq, pq, restore_path=restore_path,
+                              initial_training_steps=initial_training_steps)
     buffer.run()
 
 
+def _latest_checkpoint(model_dir: str, game_name: str):
+    """Newest ``{game}{N}.pth`` under model_dir, or None."""
+    from .evaluate import _checkpoint_paths
+    found = list(_checkpoint_paths(model_dir, game_name, 0))
+    return found[-1][1] if found else None
+
+
 def train(seed: int = 0, restart_dead_actors: bool = True,
-          _force_spawn: bool = False):
+          _force_spawn: bool = False, resume: str = None,
+          model_dir: str = "models"):
+    """Run the full training topology (reference train.py:20-44).
+
+    ``resume``: a checkpoint path, or ``"auto"`` for the newest
+    ``{game}{N}.pth`` in ``model_dir``.  Restores the learner (weights,
+    update/env-step counters, and — via the ``.train.pth`` sidecar —
+    optimizer moments and target net), and, when
+    ``config.replay_snapshot_path`` points at an existing snapshot, the
+    host replay contents too; the reference has no resume path at all
+    (SURVEY §5)."""
     torch.manual_seed(seed)
     np.random.seed(seed)
     random.seed(seed)
@@ -80,11 +99,30 @@ def train(seed: int = 0, restart_dead_actors: bool = True,
     priority_queue = ctx.Queue(c.batch_queue_size)
 
     use_gpu_replay = c.gpu_replay and torch.cuda.is_available()
-    buffer = None if use_gpu_replay else ReplayBuffer(
-        sample_queues, batch_queue, priority_queue)
-    learner = Learner(batch_queue, priority_queue, model)
+    learner = Learner(batch_queue, priority_queue, model,
+                      model_dir=model_dir)
     if use_gpu_replay and c.use_hip_kernels:
         learner.enable_hip_engine()
+    if resume:
+        path = (_latest_checkpoint(model_dir, c.game_name)
+                if resume == "auto" else resume)
+        if path is None:
+            print(f"[train] resume='auto': no checkpoint under {model_dir}/; "
+                  f"starting fresh")
+        else:
+            print(f"[train] resuming from {path}")
+            learner.load_checkpoint(path)
+    import os
+    snap = c.replay_snapshot_path
+    restore = snap if (resume and snap and os.path.exists(snap)) else None
+    if restore:
+        print(f"[train] restoring replay snapshot {restore}")
+    # spawn mode rebuilds the buffer in the child (_run_buffer), which does
+    # the restore there — don't load the snapshot twice
+    buffer = None if use_gpu_replay else ReplayBuffer(
+        sample_queues, batch_queue, priority_queue,
+        restore_path=None if use_spawn else restore,
+        initial_training_steps=learner.num_updates)
 
     if c.vector_actors:
         # one driver process, all envs in lockstep, batched inference.
@@ -111,7 +149,8 @@ def train(seed: int = 0, restart_dead_actors: bool = True,
         if use_spawn:
             buffer_proc = ctx.Process(target=_run_buffer, kwargs=dict(
                 queues=(sample_queues, batch_queue, priority_queue),
-                config_dict=config_dict))
+                config_dict=config_dict, restore_path=restore,
+                initial_training_steps=learner.num_updates))
         else:
             buffer_proc = ctx.Process(target=_run_buffer, args=(buffer,))
         buffer_proc.start()
@@ -148,5 +187,21 @@ def train(seed: int = 0, restart_dead_actors: bool = True,
         p.terminate()
 
 
+def main():
+    import argparse
+    ap = argparse.ArgumentParser(description="R2D2 training (reference "
+                                 "train.py topology)")
+    ap.add_argument("--preset", type=str, default=None,
+                    help="config preset (see r2d2_amd.config.PRESETS)")
+    ap.add_argument("--resume", type=str, default=None,
+                    help="checkpoint path, or 'auto' = newest in --model-dir")
+    ap.add_argument("--model-dir", type=str, default="models")
+    ap.add_argument("--seed", type=int, default=0)
+    args = ap.parse_args()
+    if args.preset:
+        cfg.apply(args.preset)
+    train(seed=args.seed, resume=args.resume, model_dir=args.model_dir)
+
+
 if __name__ == "__main__":
-    train()
+    main()

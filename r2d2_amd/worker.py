@@ -255,7 +255,9 @@ class ReplayBuffer:
                  buffer_capacity: Optional[int] = None,
                  alpha: Optional[float] = None, beta: Optional[float] = None,
                  batch_size: Optional[int] = None, seed: Optional[int] = None,
-                 metrics_path: Optional[str] = None):
+                 metrics_path: Optional[str] = None,
+                 restore_path: Optional[str] = None,
+                 initial_training_steps: int = 0):
         c = cfg.get()
         self.cfg = c
         self.block_len = c.block_length
@@ -278,8 +280,10 @@ class ReplayBuffer:
         self.env_steps = 0
         self.num_episodes = 0
         self.episode_reward = 0.0
-        self.training_steps = 0
-        self.last_training_steps = 0
+        # on resume the learner continues from a restored update count; the
+        # buffer's run-termination condition must count from the same origin
+        self.training_steps = initial_training_steps
+        self.last_training_steps = initial_training_steps
         self.sum_loss = 0.0
         self.last_size = 0
         self.lock = threading.Lock()
@@ -292,6 +296,12 @@ class ReplayBuffer:
         # (SURVEY §5 — the reference only print()s, worker.py:89-111)
         self.metrics_path = metrics_path or getattr(c, "metrics_path", None)
         self._t0 = time.time()
+        # replay persistence (elastic resume): snapshot cadence handled in
+        # run(); restore happens before the worker threads start
+        self.snapshot_path = getattr(c, "replay_snapshot_path", None)
+        self.snapshot_interval = getattr(c, "replay_snapshot_interval", 300.0)
+        if restore_path:
+            self.load_state(restore_path)
 
     def __len__(self):
         return self.size
@@ -313,8 +323,16 @@ class ReplayBuffer:
         for t in threads:
             t.start()
         log_interval = self.cfg.log_interval
+        last_snapshot = time.time()
         while True:
             self._log(log_interval)
+            if (self.snapshot_path and self.size > 0
+                    and time.time() - last_snapshot >= self.snapshot_interval):
+                try:
+                    self.save_state(self.snapshot_path)
+                except OSError as e:
+                    print(f"[buffer] replay snapshot failed: {e!r}")
+                last_snapshot = time.time()
             if self.training_steps >= self.cfg.training_steps:
                 self.stop_flag = True
                 break
@@ -515,6 +533,54 @@ class ReplayBuffer:
                 self.priority_tree.update(idxes, td_errors)
         self.training_steps += 1
         self.sum_loss += loss
+
+    # -- persistence (elastic resume) ---------------------------------------
+
+    def save_state(self, path: str):
+        """Snapshot the replay contents for elastic resume — the reference
+        never persists replay (a restart refills 2M transitions
+        from scratch).  Blocks are immutable once stored, so the lock only
+        covers capturing the slot references and counters; serialization
+        runs unlocked, and the write is atomic (tmp + rename) so a crash
+        mid-snapshot leaves the previous snapshot intact."""
+        import pickle
+
+        with self.lock:
+            state = {
+                "geometry": (self.num_blocks, self.seq_per_block,
+                             self.block_len, self.seq_len),
+                "blocks": list(self.buffer),
+                "leaves": self.priority_tree.leaf_values(),
+                "block_ptr": self.block_ptr,
+                "blocks_added": self.blocks_added,
+                "size": self.size,
+                "env_steps": self.env_steps,
+            }
+        tmp = path + ".tmp"
+        with open(tmp, "wb") as f:
+            pickle.dump(state, f, protocol=pickle.HIGHEST_PROTOCOL)
+        os.replace(tmp, path)
+
+    def load_state(self, path: str):
+        """Restore a save_state snapshot (geometry must match the live
+        config).  Call before run() — not thread-safe against the worker
+        threads."""
+        import pickle
+
+        with open(path, "rb") as f:
+            state = pickle.load(f)
+        geo = (self.num_blocks, self.seq_per_block, self.block_len,
+               self.seq_len)
+        if tuple(state["geometry"]) != geo:
+            raise ValueError(f"replay snapshot geometry {state['geometry']} "
+                             f"!= configured {geo}")
+        with self.lock:
+            self.buffer = list(state["blocks"])
+            self.priority_tree.set_leaf_values(state["leaves"])
+            self.block_ptr = int(state["block_ptr"])
+            self.blocks_added = int(state["blocks_added"])
+            self.size = int(state["size"])
+            self.env_steps = int(state["env_steps"])
 
 
 ############################## Learner ##############################

@@ -33,6 +33,34 @@ def test_train_to_completion(tmp_path, vector, monkeypatch):
     assert (tmp_path / "metrics.jsonl").exists()
 
 
+@pytest.mark.timeout(600)
+def test_train_resume_continues(tmp_path, monkeypatch):
+    """Elastic resume through the train() entry: a finished run's newest
+    checkpoint (resume='auto') restores the learner counters; the replay
+    snapshot (config.replay_snapshot_path) restores the buffer contents,
+    and the buffer's termination condition counts from the restored update
+    number (a resumed learner doing N more updates must not leave the
+    buffer process waiting for 2N)."""
+    monkeypatch.chdir(tmp_path)
+    snap = str(tmp_path / "replay.snap")
+    tiny_cfg(tmp_path, training_steps=15, save_interval=15,
+             replay_snapshot_path=snap, replay_snapshot_interval=0.0)
+    from r2d2_amd.train import train
+    train(seed=0)
+    assert (tmp_path / "models" / "CartPole15.pth").exists()
+
+    # second leg: 15 more updates on top of the restored 15
+    tiny_cfg(tmp_path, training_steps=30, save_interval=15,
+             replay_snapshot_path=snap, replay_snapshot_interval=0.0)
+    train(seed=1, resume="auto")
+    assert (tmp_path / "models" / "CartPole30.pth").exists()
+    import torch
+    _, num_updates, env_steps, _ = torch.load(
+        str(tmp_path / "models" / "CartPole30.pth"), map_location="cpu",
+        weights_only=False)
+    assert num_updates == 30 and env_steps > 0
+
+
 @pytest.mark.timeout(300)
 def test_train_spawn_context(tmp_path, monkeypatch):
     """The spawn-context actor path (used when vector actors run on cuda):
