@@ -81,6 +81,9 @@ class Config:
     # every replay_snapshot_interval seconds; train(resume=...) restores it.
     replay_snapshot_path: Optional[str] = None
     replay_snapshot_interval: float = 300.0      # seconds between snapshots
+    # host-replay batch assembler threads (the slice copies release the
+    # GIL; 2 saturates the pipeline when CPU actors share the box)
+    assemble_threads: int = 2
     batch_queue_size: int = 8
     amp: bool = True
 

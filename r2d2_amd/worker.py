@@ -99,6 +99,21 @@ class TrainingBatch:
         return self
 
 
+def _shared_tensor(shape, dtype) -> torch.Tensor:
+    """CPU tensor allocated DIRECTLY in shared memory (no intermediate
+    copy).  A torch.multiprocessing queue ships an already-shared tensor
+    as a handle; a private tensor gets copied into shared memory by the
+    queue's single feeder thread.  At the full MsPacman batch shape the
+    obs tensor is ~150 MB, and that serialized feeder copy — not the
+    assembly itself — bounded the host-replay pipeline (measured on this
+    container: assembler alone 33-60 batches/s at 2-4 threads, but only
+    7-10.5 batches/s through the queue; handle-passing removes the wall)."""
+    t = torch.empty(0, dtype=dtype)
+    numel = int(np.prod(shape))
+    storage = torch.UntypedStorage._new_shared(numel * t.element_size())
+    return t.set_(storage, 0, tuple(shape))
+
+
 def calculate_mixed_td_errors(td_error: np.ndarray, learning_steps: np.ndarray,
                               eta: float = 0.9) -> np.ndarray:
     """Per-sequence priority = eta*max + (1-eta)*mean of |TD|
@@ -296,6 +311,13 @@ class ReplayBuffer:
         # (SURVEY §5 — the reference only print()s, worker.py:89-111)
         self.metrics_path = metrics_path or getattr(c, "metrics_path", None)
         self._t0 = time.time()
+        # rotating pool of shared-memory obs tensors for outgoing batches
+        # (see _shared_tensor / sample_batch).  In-flight batches are hard
+        # bounded by batch_queue_size (<=8) + the learner's prefetch
+        # staging (<=5) + one per assemble thread (<=2) = 15 < pool depth,
+        # so a slot is never overwritten while a consumer can still see it.
+        self._obs_pool: List[Optional[torch.Tensor]] = [None] * 32
+        self._obs_pool_i = 0
         # replay persistence (elastic resume): snapshot cadence handled in
         # run(); restore happens before the worker threads start
         self.snapshot_path = getattr(c, "replay_snapshot_path", None)
@@ -317,9 +339,10 @@ class ReplayBuffer:
         # at full shapes — the op must cross the parallel grain size.
         torch.set_num_threads(1)
         torch.zeros((64, 2, 512)).transpose(0, 1).contiguous()
+        n_asm = max(1, int(getattr(self.cfg, "assemble_threads", 2)))
         threads = [threading.Thread(target=f, daemon=True)
-                   for f in (self._ingest_loop, self._assemble_loop,
-                             self._assemble_loop, self._priority_loop)]
+                   for f in ((self._ingest_loop, self._priority_loop)
+                             + (self._assemble_loop,) * n_asm)]
         for t in threads:
             t.start()
         log_interval = self.cfg.log_interval
@@ -445,6 +468,9 @@ class ReplayBuffer:
             old_ptr = self.block_ptr
             old_count = self.blocks_added
             env_steps = self.env_steps
+            # claim the next shared-memory obs slot while the lock is held
+            pool_i = self._obs_pool_i
+            self._obs_pool_i = (pool_i + 1) % len(self._obs_pool)
 
         B = self.batch_size
         burn = np.empty(B, dtype=np.int64)
@@ -465,7 +491,24 @@ class ReplayBuffer:
             T = int((burn + learn + fwd).max())
             obs_shape = blocks[0].obs.shape[1:]
             A = blocks[0].last_action.shape[1]
-            obs = np.zeros((B, T) + obs_shape, dtype=blocks[0].obs.dtype)
+            # the big obs tensor is built in shared memory so the
+            # batch_queue passes a handle instead of feeder-copying
+            # ~150 MB per batch (see _shared_tensor).  Slots are pooled:
+            # a FRESH shm segment would cold-page-fault its whole extent
+            # on every fill (measured: halves the assembler and caps it
+            # ~17 batches/s regardless of threads); reused slots stay
+            # warm.  Allocated at the max sequence length, sliced to this
+            # batch's T.
+            dt = torch.from_numpy(blocks[0].obs[:1]).dtype
+            Tmax = max(T, self.cfg.seq_len)
+            slot = self._obs_pool[pool_i]
+            if (slot is None or slot.dtype != dt
+                    or slot.shape[1] < Tmax or slot.shape[2:] != obs_shape):
+                slot = _shared_tensor((B, Tmax) + obs_shape, dt)
+                self._obs_pool[pool_i] = slot
+            obs_t = slot[:, :T]
+            obs = obs_t.numpy()   # rows are fully written below; only the
+                                  # ragged padding tail needs zeroing
             last_action = np.zeros((B, T, A), dtype=np.float32)
             last_reward = np.zeros((B, T), dtype=np.float32)
             hidden = np.empty((B, 2, blocks[0].hidden.shape[-1]), dtype=np.float32)
@@ -474,6 +517,7 @@ class ReplayBuffer:
                 si = seq_idxes[i]
                 s, L = starts[i], int(burn[i] + learn[i] + fwd[i])
                 obs[i, :L] = blk.obs[s - burn[i]: s + learn[i] + fwd[i]]
+                obs[i, L:] = 0   # padding tail (slot is reused, not fresh)
                 last_action[i, :L] = blk.last_action[s - burn[i]: s + learn[i] + fwd[i]]
                 last_reward[i, :L] = blk.last_reward[s - burn[i]: s + learn[i] + fwd[i]]
                 hidden[i] = blk.hidden[si]
@@ -485,7 +529,7 @@ class ReplayBuffer:
 
             is_rep = np.repeat(is_weights, learn).astype(np.float32)
             batch = TrainingBatch(
-                obs=torch.from_numpy(obs),
+                obs=obs_t,
                 last_action=torch.from_numpy(last_action),
                 last_reward=torch.from_numpy(last_reward),
                 hidden=torch.from_numpy(hidden).transpose(0, 1).contiguous(),
