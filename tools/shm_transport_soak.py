@@ -35,6 +35,9 @@ def consumer(bq, pq, stop_ev, out):
         except Exception:
             continue
         staged.append(b)
+        if n and n % 500 == 0:
+            print(f"[consumer] {n} batches, {errs} errors, "
+                  f"{n / (time.time() - t0):.1f}/s", flush=True)
         if len(staged) > 5:
             old = staged.popleft()
             # tearing check: every row of a constant-filled block is uniform
