@@ -312,11 +312,16 @@ class ReplayBuffer:
         self.metrics_path = metrics_path or getattr(c, "metrics_path", None)
         self._t0 = time.time()
         # rotating pool of shared-memory obs tensors for outgoing batches
-        # (see _shared_tensor / sample_batch).  In-flight batches are hard
-        # bounded by batch_queue_size (<=8) + the learner's prefetch
-        # staging (<=5) + one per assemble thread (<=2) = 15 < pool depth,
-        # so a slot is never overwritten while a consumer can still see it.
-        self._obs_pool: List[Optional[torch.Tensor]] = [None] * 32
+        # (see _shared_tensor / sample_batch).  Safety contract: a slot is
+        # reused after pool-depth claims, so no consumer may still hold a
+        # batch once pool-depth NEWER batches were claimed.  The topology
+        # enforces that by throttling: assemblers only produce while
+        # batch_queue has room, so claims past the oldest live batch are
+        # bounded by queue depth + learner staging (<=5) + one per
+        # assemble thread — the pool is sized at 2x that bound.
+        n_slots = max(32, 2 * (int(getattr(c, "batch_queue_size", 8))
+                               + int(getattr(c, "assemble_threads", 2)) + 6))
+        self._obs_pool: List[Optional[torch.Tensor]] = [None] * n_slots
         self._obs_pool_i = 0
         # replay persistence (elastic resume): snapshot cadence handled in
         # run(); restore happens before the worker threads start

@@ -183,6 +183,11 @@ def test_sample_batch_races_ring_overwrite_without_tearing():
 
     small_cfg()
     rb = make_rb(seed=7)
+    # the two unthrottled samplers here bypass the batch_queue throttling
+    # that bounds claim distance in the real topology (see the obs-pool
+    # contract in worker.py) — give the pool enough slots that 2x60
+    # back-to-back claims cannot wrap onto a batch still being checked
+    rb._obs_pool = [None] * 256
     for v in range(1, 9):
         blk, prios = make_block(float(v))
         rb.add(blk, prios + 1.0, None)
@@ -411,6 +416,58 @@ def test_assembler_rows_match_origin_blocks(block_steps, seed):
         # IS weights repeated once per learning step
         assert batch.is_weights.shape[0] == int(batch.learning_steps.sum())
         assert batch.action.shape[0] == int(batch.learning_steps.sum())
+
+
+@settings(max_examples=15, deadline=None)
+@given(st.lists(st.sampled_from(["add", "addp", "sample", "update",
+                                 "snap"]),
+                min_size=5, max_size=30),
+       st.integers(min_value=0, max_value=2 ** 31 - 1))
+def test_snapshot_under_random_op_sequences(tmp_path_factory, ops, seed):
+    """Random interleavings of add / partial-add / sample / priority-update
+    / save+load round trips: after every snapshot restore the tree root
+    equals the sum of its leaves, restored leaves equal the original's,
+    and sampling still returns only live slots."""
+    tmp_path = tmp_path_factory.mktemp("snapshot_ops")
+    small_cfg()
+    rb = make_rb(seed=seed % 10_000)
+    blk, prios = make_block(1.0)
+    rb.add(blk, prios + 1.0, None)          # never-empty tree
+    last_batch = None
+    v = 2
+    for k, op in enumerate(ops):
+        if op == "add":
+            blk, prios = make_block(float(v % 30 + 1))
+            rb.add(blk, prios + 1.0, None)
+            v += 1
+        elif op == "addp":
+            blk, prios = _make_partial_block(float(v % 30 + 1),
+                                             steps=(v % 39) + 1)
+            rb.add(blk, prios, None)
+            v += 1
+        elif op == "sample":
+            last_batch = rb.sample_batch()
+        elif op == "update" and last_batch is not None:
+            rb.update_priorities(
+                last_batch.idxes,
+                np.full(len(last_batch.idxes), 2.0, dtype=np.float32),
+                last_batch.old_ptr, 0.0, last_batch.old_count)
+        elif op == "snap":
+            p = str(tmp_path / f"s{k}")
+            rb.save_state(p)
+            rb2 = make_rb(seed=1)
+            rb2.load_state(p)
+            tree = rb2.priority_tree
+            assert tree.total == pytest.approx(
+                float(tree.levels[-1].sum()), rel=1e-9)
+            assert np.array_equal(tree.leaf_values(),
+                                  rb.priority_tree.leaf_values())
+            idx, w = tree.sample(8)
+            for i in idx:
+                b = rb2.buffer[i // rb2.seq_per_block]
+                assert b is not None
+                assert i % rb2.seq_per_block < b.num_sequences
+            assert np.all(w > 0)
 
 
 @settings(max_examples=25, deadline=None)
