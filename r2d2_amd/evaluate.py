@@ -43,7 +43,13 @@ def test_one_case(args):
     network = Network(env.action_dim, c.obs_shape, c.hidden_dim,
                       encoder=c.encoder, forward_steps=c.forward_steps,
                       mlp_hidden=c.mlp_hidden)
-    network.load_state_dict(state_dict)
+    try:
+        network.load_state_dict(state_dict)
+    except RuntimeError:
+        # checkpoint trained by the ORIGINAL reference: anonymous `feature`
+        # Sequential encoder keys (reference model.py:39-49) — remap
+        from .models.network import reference_state_dict_to_native
+        network.load_state_dict(reference_state_dict_to_native(state_dict))
     network.eval()
     rng = np.random.default_rng(seed)
 

@@ -102,6 +102,33 @@ def test_eval_harness(tmp_path):
     assert (tmp_path / f"{c.game_name}_eval.jsonl").exists()
 
 
+def test_eval_harness_accepts_reference_checkpoints(tmp_path):
+    """A checkpoint trained by the ORIGINAL reference (anonymous `feature`
+    Sequential encoder keys) must evaluate through evaluate.test() via the
+    key-mapping fallback — a reference user can point the harness at their
+    existing models/ directory."""
+    import time
+
+    c = cfg.apply("mspacman", env_type="synthetic", obs_shape=(1, 84, 84),
+                  action_dim=4, hidden_dim=32, max_episode_steps=8,
+                  device="cpu", dtype="fp32", use_hip_kernels=False,
+                  gpu_replay=False)
+    torch.manual_seed(3)
+    src = Network(4, obs_shape=(1, 84, 84), hidden_dim=32, encoder="nature")
+    inv = {"encoder.conv1": "feature.0", "encoder.conv2": "feature.2",
+           "encoder.conv3": "feature.4", "encoder.fc": "feature.7"}
+    ref_sd = {}
+    for k, v in src.state_dict().items():
+        head, _, tail = k.rpartition(".")
+        ref_sd[f"{inv.get(head, head)}.{tail}" if head else k] = v
+    torch.save((ref_sd, 500, 2000, 1.5), tmp_path / f"{c.game_name}500.pth")
+
+    results = evaluate.test(model_dir=str(tmp_path), num_episodes=1,
+                            pool_size=1, out_dir=str(tmp_path))
+    assert len(results) == 1 and results[0]["num_updates"] == 500
+    assert np.isfinite(results[0]["mean_reward"])
+
+
 def test_checkpoint_discovery_mixed_intervals(tmp_path):
     """Discovery must find every {game}{N}.pth regardless of the configured
     save_interval and survive gaps (the reference's k*interval walk stops
