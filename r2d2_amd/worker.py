@@ -1027,7 +1027,14 @@ class Learner:
         fresh optimizer)."""
         state, num_updates, env_steps, minutes = torch.load(
             path, map_location="cpu", weights_only=False)
-        self.online_net.load_state_dict(state)
+        try:
+            self.online_net.load_state_dict(state)
+        except RuntimeError:
+            # checkpoint trained by the ORIGINAL reference (anonymous
+            # `feature` Sequential encoder keys, reference model.py:39-49)
+            from .models.network import reference_state_dict_to_native
+            state = reference_state_dict_to_native(state)
+            self.online_net.load_state_dict(state)
         self.num_updates = int(num_updates)
         self.env_steps = int(env_steps)
         self._resumed_minutes = float(minutes)

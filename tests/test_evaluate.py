@@ -129,6 +129,36 @@ def test_eval_harness_accepts_reference_checkpoints(tmp_path):
     assert np.isfinite(results[0]["mean_reward"])
 
 
+def test_learner_resumes_from_reference_checkpoint(tmp_path):
+    """Learner.load_checkpoint applies the same key-mapping fallback, so
+    training can RESUME from a reference-trained checkpoint."""
+    c = cfg.apply("mspacman", env_type="synthetic", obs_shape=(1, 84, 84),
+                  action_dim=4, hidden_dim=32, device="cpu", dtype="fp32",
+                  use_hip_kernels=False, gpu_replay=False, amp=False)
+    torch.manual_seed(5)
+    src = Network(4, obs_shape=(1, 84, 84), hidden_dim=32, encoder="nature")
+    inv = {"encoder.conv1": "feature.0", "encoder.conv2": "feature.2",
+           "encoder.conv3": "feature.4", "encoder.fc": "feature.7"}
+    ref_sd = {}
+    for k, v in src.state_dict().items():
+        head, _, tail = k.rpartition(".")
+        ref_sd[f"{inv.get(head, head)}.{tail}" if head else k] = v
+    p = tmp_path / f"{c.game_name}700.pth"
+    torch.save((ref_sd, 700, 4000, 2.0), p)
+
+    model = Network(c.action_dim, c.obs_shape, c.hidden_dim,
+                    encoder="nature")
+    learner = Learner(queue.Queue(), queue.Queue(), model,
+                      model_dir=str(tmp_path))
+    learner.load_checkpoint(str(p))
+    assert learner.num_updates == 700 and learner.env_steps == 4000
+    assert torch.equal(learner.online_net.state_dict()["encoder.conv1.weight"],
+                       src.state_dict()["encoder.conv1.weight"])
+    # no sidecar: target net must equal the (remapped) online weights
+    assert torch.equal(learner.target_net.state_dict()["encoder.conv1.weight"],
+                       src.state_dict()["encoder.conv1.weight"])
+
+
 def test_checkpoint_discovery_mixed_intervals(tmp_path):
     """Discovery must find every {game}{N}.pth regardless of the configured
     save_interval and survive gaps (the reference's k*interval walk stops
