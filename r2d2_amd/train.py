@@ -159,12 +159,23 @@ def train(seed: int = 0, restart_dead_actors: bool = True,
     # (SURVEY §5 — throughput degrades with no signal); restart them.
     stop = threading.Event()
 
+    restarts = [0] * len(actor_procs)
+    max_restarts = 10
+
     def _watchdog():
         while not stop.wait(5.0):
             for i, p in enumerate(actor_procs):
-                if not p.is_alive():
+                if not p.is_alive() and restarts[i] <= max_restarts:
+                    restarts[i] += 1
+                    if restarts[i] > max_restarts:
+                        # persistent crash (bad env/config): stop the
+                        # restart storm and surface it instead
+                        print(f"[train] actor {i} died {max_restarts} "
+                              f"times; giving up on it")
+                        continue
                     print(f"[train] actor process {i} died "
-                          f"(exitcode {p.exitcode}); restarting")
+                          f"(exitcode {p.exitcode}); restarting "
+                          f"({restarts[i]}/{max_restarts})")
                     actor_procs[i] = spawners[i]()
                     actor_procs[i].start()
 
